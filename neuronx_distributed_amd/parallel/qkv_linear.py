@@ -1,0 +1,235 @@
+"""GQA-aware fused QKV column-parallel linear with KV-head replication.
+
+Parity with reference ``modules/qkv_linear.py`` (713 LoC):
+``GQAQKVColumnParallelLinear`` (:371) — when ``num_key_value_heads < tp``
+the K/V heads are replicated ``kv_size_multiplier``× so heads divide TP
+(:80-88, "adjacent" replication layout; the trn1 interleave is
+hardware-specific and not carried over).  Ranks holding the same KV-head
+replica form the kv-shared group (parallel_state kv groups); their weight
+grads are summed after backward (see ``allreduce_kv_shared_gradients``).
+
+One input all-gather (SP) feeds all three GEMMs; backward sums the three
+grad-input contributions and issues a single TP all-reduce/reduce-scatter,
+overlapped with the weight-grad GEMMs (the reference gets this from its
+fused single-GEMM; we get it from a dedicated autograd.Function).
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+import torch.distributed as dist
+
+from . import comm
+from . import parallel_state as ps
+from .layers import BaseParallelLinear, default_init_method
+from .random import get_rng_state_tracker
+from .utils import divide, set_tensor_model_parallel_attributes
+
+
+class _QKVLinearWithAsyncCommunication(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, input_, wq, wk, wv, bq, bk, bv, sequence_parallel, kv_mult):
+        ctx.sequence_parallel = sequence_parallel
+        ctx.kv_mult = kv_mult
+        ctx.use_bias = bq is not None
+        if sequence_parallel:
+            total_input = comm.all_gather(input_, dim=0, group=ps.get_group_info("tp"))
+        else:
+            total_input = input_
+        q = F.linear(total_input, wq, bq)
+        k = F.linear(total_input, wk, bk)
+        v = F.linear(total_input, wv, bv)
+        ctx.save_for_backward(input_, wq, wk, wv)
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, gq, gk, gv):
+        input_, wq, wk, wv = ctx.saved_tensors
+        tp_info = ps.get_group_info("tp")
+        world = tp_info.size
+
+        gq = gq.contiguous()
+        gk = gk.contiguous()
+        gv = gv.contiguous()
+        grad_input = gq.matmul(wq)
+        grad_input += gk.matmul(wk)
+        grad_input += gv.matmul(wv)
+
+        handle = None
+        if ctx.sequence_parallel:
+            total_input = comm.all_gather(input_, dim=0, group=tp_info)
+            if world > 1 and not ps.is_aot_mode():
+                sub = (grad_input.shape[0] // world,) + tuple(grad_input.shape[1:])
+                gi_out = torch.empty(sub, dtype=grad_input.dtype,
+                                     device=grad_input.device)
+                if comm._backend_is_gloo(tp_info.group):
+                    dist.all_reduce(grad_input, group=tp_info.group)
+                    r = tp_info.rank_in_group(dist.get_rank())
+                    gi_out.copy_(grad_input[r * sub[0]:(r + 1) * sub[0]])
+                else:
+                    handle = dist.reduce_scatter_tensor(
+                        gi_out, grad_input, group=tp_info.group, async_op=True)
+                grad_input = gi_out
+        else:
+            total_input = input_
+            if world > 1:
+                handle = comm.all_reduce(grad_input, group=tp_info, async_op=True)
+
+        in2d = total_input.reshape(-1, total_input.shape[-1])
+        gq2 = gq.reshape(-1, gq.shape[-1])
+        gk2 = gk.reshape(-1, gk.shape[-1])
+        gv2 = gv.reshape(-1, gv.shape[-1])
+        gwq = gq2.t().matmul(in2d)
+        gwk = gk2.t().matmul(in2d)
+        gwv = gv2.t().matmul(in2d)
+        gbq = gq2.sum(0) if ctx.use_bias else None
+        gbk = gk2.sum(0) if ctx.use_bias else None
+        gbv = gv2.sum(0) if ctx.use_bias else None
+
+        if handle is not None:
+            handle.wait()
+        return grad_input, gwq, gwk, gwv, gbq, gbk, gbv, None, None
+
+
+class GQAQKVColumnParallelLinear(BaseParallelLinear):
+    def __init__(self, input_size, output_sizes, bias=False, gather_output=False,
+                 dtype=None, device=None, init_method=None,
+                 sequence_parallel_enabled=False, kv_size_multiplier=1,
+                 num_attention_heads: Optional[int] = None,
+                 num_key_value_heads: Optional[int] = None,
+                 head_dim: Optional[int] = None):
+        """``output_sizes = [q_out, kv_out]`` like the reference, or pass
+        heads explicitly.  K/V replication factor ``kv_size_multiplier``
+        must make ``num_key_value_heads * kv_size_multiplier`` divisible by
+        tp."""
+        super().__init__()
+        world = ps.get_tensor_model_parallel_size()
+        dtype = dtype or torch.get_default_dtype()
+        self.dtype = dtype
+        init_method = init_method or default_init_method
+        self.sequence_parallel_enabled = sequence_parallel_enabled
+        self.gather_output = gather_output
+        assert not gather_output, "GQA QKV is used with gather_output=False"
+
+        if num_attention_heads is not None:
+            assert head_dim is not None and num_key_value_heads is not None
+            q_out = num_attention_heads * head_dim
+            kv_out = num_key_value_heads * head_dim
+        else:
+            q_out, kv_out = output_sizes
+            assert head_dim is not None, "head_dim required"
+            num_attention_heads = q_out // head_dim
+            num_key_value_heads = kv_out // head_dim
+
+        self.head_dim = head_dim
+        self.num_attention_heads = num_attention_heads
+        self.num_key_value_heads = num_key_value_heads
+        self.kv_size_multiplier = kv_size_multiplier
+        kvh_rep = num_key_value_heads * kv_size_multiplier
+        assert kvh_rep % world == 0, (
+            f"replicated kv heads {kvh_rep} not divisible by tp {world}")
+        assert num_attention_heads % world == 0
+
+        self.q_output_size_per_partition = divide(q_out, world)
+        self.kv_output_size_per_partition = divide(kvh_rep * head_dim, world)
+
+        self.weight_q = nn.Parameter(torch.empty(
+            self.q_output_size_per_partition, input_size, dtype=dtype, device=device))
+        self.weight_k = nn.Parameter(torch.empty(
+            self.kv_output_size_per_partition, input_size, dtype=dtype, device=device))
+        self.weight_v = nn.Parameter(torch.empty(
+            self.kv_output_size_per_partition, input_size, dtype=dtype, device=device))
+
+        set_tensor_model_parallel_attributes(self.weight_q, world > 1, 0, 1, world)
+        for w in (self.weight_k, self.weight_v):
+            set_tensor_model_parallel_attributes(w, world > 1, 0, 1, world)
+            w.kv_shared = kv_size_multiplier > 1
+
+        if not ps.is_aot_mode():
+            self._deterministic_init(input_size, q_out, kv_out, init_method)
+
+        if bias:
+            self.bias_q = nn.Parameter(torch.zeros(
+                self.q_output_size_per_partition, dtype=dtype, device=device))
+            self.bias_k = nn.Parameter(torch.zeros(
+                self.kv_output_size_per_partition, dtype=dtype, device=device))
+            self.bias_v = nn.Parameter(torch.zeros(
+                self.kv_output_size_per_partition, dtype=dtype, device=device))
+            for b in (self.bias_k, self.bias_v):
+                b.kv_shared = kv_size_multiplier > 1
+        else:
+            self.register_parameter("bias_q", None)
+            self.register_parameter("bias_k", None)
+            self.register_parameter("bias_v", None)
+
+    def _kv_local_rows(self, master: torch.Tensor) -> torch.Tensor:
+        """Slice this rank's replicated-KV rows out of the full
+        (kv_heads*head_dim, H) master using the adjacent replication layout:
+        replicated head slot s holds original head s // kv_size_multiplier."""
+        world = ps.get_tensor_model_parallel_size()
+        rank = ps.get_tensor_model_parallel_rank()
+        kvh_rep = self.num_key_value_heads * self.kv_size_multiplier
+        per_rank = kvh_rep // world
+        rows = []
+        for s in range(rank * per_rank, (rank + 1) * per_rank):
+            h = s // self.kv_size_multiplier
+            rows.append(master[h * self.head_dim:(h + 1) * self.head_dim])
+        return torch.cat(rows, dim=0)
+
+    def _deterministic_init(self, input_size, q_out, kv_out, init_method):
+        if self.weight_q.device.type == "meta":
+            return
+        world = ps.get_tensor_model_parallel_size()
+        rank = ps.get_tensor_model_parallel_rank()
+        with torch.no_grad():
+            mq = torch.empty(q_out, input_size, dtype=torch.float32, device="cpu")
+            init_method(mq)
+            rows = self.q_output_size_per_partition
+            self.weight_q.data.copy_(mq[rank * rows:(rank + 1) * rows].to(self.dtype))
+            for name, w in (("k", self.weight_k), ("v", self.weight_v)):
+                m = torch.empty(kv_out, input_size, dtype=torch.float32, device="cpu")
+                init_method(m)
+                w.data.copy_(self._kv_local_rows(m).to(self.dtype))
+
+    def preshard_hook(self, model_state_dict: dict, prefix: str) -> None:
+        """Shard a full checkpoint's q/k/v weights, replicating KV heads
+        (reference trace/trace.py:646-787 create_local_weight_qkv)."""
+        base = prefix.rsplit(".", 1)[0]
+        world = ps.get_tensor_model_parallel_size()
+        rank = ps.get_tensor_model_parallel_rank()
+        for pname in ("weight_q", "bias_q"):
+            key = f"{base}.{pname}"
+            if key in model_state_dict:
+                full = model_state_dict[key]
+                rows = full.shape[0] // world
+                model_state_dict[key] = full[rank * rows:(rank + 1) * rows]
+        for pname in ("weight_k", "weight_v", "bias_k", "bias_v"):
+            key = f"{base}.{pname}"
+            if key in model_state_dict:
+                model_state_dict[key] = self._kv_local_rows(model_state_dict[key])
+
+    def forward(self, input_):
+        q, k, v = _QKVLinearWithAsyncCommunication.apply(
+            input_, self.weight_q, self.weight_k, self.weight_v,
+            self.bias_q, self.bias_k, self.bias_v,
+            self.sequence_parallel_enabled, self.kv_size_multiplier)
+        return q, k, v
+
+
+def allreduce_kv_shared_gradients(parameters):
+    """Sum grads of KV-replicated weights over the kv-shared group so the
+    replicas stay identical (reference qkv_linear.py:184-198 grad scaling;
+    here the true gradient of a replicated head is the SUM of its replicas'
+    grads, each covering distinct Q heads)."""
+    if "kv" not in ps._GROUPS:
+        return
+    grads = [p.grad for p in parameters
+             if p.grad is not None and getattr(p, "kv_shared", False)]
+    if not grads:
+        return
+    flat = torch._utils._flatten_dense_tensors(grads)
+    comm.all_reduce(flat, group=ps.get_group_info("kv"))
+    for g, synced in zip(grads, torch._utils._unflatten_dense_tensors(flat, grads)):
+        g.copy_(synced)
