@@ -97,6 +97,7 @@ class ParallelGroups:
     ep_data_groups: List[List[int]]
     kv_groups: Optional[List[List[int]]] = None
     token_shuffle_groups: Optional[List[List[int]]] = None
+    zero1_groups: Optional[List[List[int]]] = None
 
 
 def _build_mesh(world_size, tp, pp, cp, ep):
@@ -130,14 +131,19 @@ def _build_mesh(world_size, tp, pp, cp, ep):
     ep_model_groups = emesh.permute(0, 1, 3, 2).reshape(-1, ep).tolist()
     ep_data_groups = emesh.permute(0, 2, 3, 1).reshape(-1, dp_exp).tolist()
 
-    return ParallelGroups(
+    # ZeRO-1 sharding groups: merged DPxCP (reference parallel_state.py:1684-1706)
+    zero1_groups = mesh.permute(0, 3, 1, 2).reshape(-1, dp * cp).tolist()
+
+    pg = ParallelGroups(
         tp_groups=tp_groups,
         dp_groups=dp_groups,
         pp_groups=pp_groups,
         cp_groups=cp_groups,
         ep_model_groups=ep_model_groups,
         ep_data_groups=ep_data_groups,
+        zero1_groups=zero1_groups,
     )
+    return pg
 
 
 def _new_group(mesh: List[List[int]], name: str, backend=None) -> GroupInfo:
@@ -214,6 +220,11 @@ def initialize_model_parallel(
 
     _TENSOR_MODEL_PARALLEL_GROUP = _new_group(groups.tp_groups, "tp")
     _DATA_PARALLEL_GROUP = _new_group(groups.dp_groups, "dp")
+    if cp > 1:
+        zero1_info = _new_group(groups.zero1_groups, "zero1")
+    else:
+        zero1_info = GroupInfo("zero1", groups.zero1_groups,
+                               _DATA_PARALLEL_GROUP.group)
     _PIPELINE_MODEL_PARALLEL_GROUP = _new_group(groups.pp_groups, "pp")
     _CONTEXT_MODEL_PARALLEL_GROUP = _new_group(groups.cp_groups, "cp")
     _EXPERT_MODEL_PARALLEL_GROUP = _new_group(groups.ep_model_groups, "ep")
@@ -242,6 +253,7 @@ def initialize_model_parallel(
         "cp": _CONTEXT_MODEL_PARALLEL_GROUP,
         "ep": _EXPERT_MODEL_PARALLEL_GROUP,
         "edp": _EXPERT_DATA_PARALLEL_GROUP,
+        "zero1": zero1_info,
     }
     if _KV_SHARED_GROUP is not None:
         _GROUPS["kv"] = _KV_SHARED_GROUP
@@ -486,6 +498,13 @@ def get_token_shuffle_group(as_list: bool = False):
 
 def get_token_shuffle_group_size() -> int:
     return _info("token_shuffle").size
+
+
+def get_zero1_sharding_group(as_list: bool = False):
+    """Merged DPxCP group that ZeRO-1 shards over (reference
+    parallel_state.py:1684-1706)."""
+    g = _info("zero1")
+    return g.mesh if as_list else g.group
 
 
 def get_world_group():
