@@ -1,0 +1,153 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark: Llama-2-7B bf16 pretraining step, TP=N,
+seq 4096, ZeRO-1 AdamW — the BASELINE.json headline metric
+(tokens/sec whole job).
+
+Single GPU:     python bench.py --steps 8 --warmup 3
+N GPUs (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+                 --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", type=str, default="llama2-7b")
+    p.add_argument("--seq", type=int, default=4096)
+    p.add_argument("--batch", type=int, default=0,
+                   help="global batch per step (default: n_gpus)")
+    p.add_argument("--tp", type=int, default=0, help="TP degree (default: world)")
+    p.add_argument("--pp", type=int, default=1)
+    p.add_argument("--sequence-parallel", action="store_true")
+    p.add_argument("--layers", type=int, default=0,
+                   help="override layer count (debug only; overridden runs "
+                        "are marked invalid in the output)")
+    p.add_argument("--microbatch", type=int, default=1)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29777")
+
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+    backend = "nccl" if on_gpu else "gloo"
+    if not dist.is_initialized():
+        dist.init_process_group(backend, rank=rank, world_size=world)
+
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+
+    tp = args.tp or (world // args.pp)
+    ps.initialize_model_parallel(tensor_model_parallel_size=tp,
+                                 pipeline_model_parallel_size=args.pp)
+    nxd.parallel.model_parallel_manual_seed(1234)
+
+    overrides = {"sequence_parallel_enabled": args.sequence_parallel,
+                 "max_position_embeddings": max(args.seq, 4096)}
+    if args.layers:
+        overrides["num_hidden_layers"] = args.layers
+    cfg = get_config(args.model, **overrides)
+
+    device = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
+    dtype = torch.bfloat16
+    torch.manual_seed(1234)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(dtype)
+    with torch.device(device):
+        model = LlamaForCausalLM(cfg)
+    torch.set_default_dtype(prev)
+    model = model.to(device)
+
+    opt = NeuronZero1Optimizer(model.parameters(), torch.optim.AdamW,
+                               lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
+                               grad_clipping=True, max_norm=1.0)
+
+    B = args.batch or max(1, world)
+    S = args.seq
+    mbs = args.microbatch
+    assert B % mbs == 0
+    n_micro = B // mbs
+    torch.manual_seed(4321)
+    data = [torch.randint(0, cfg.vocab_size, (mbs, S), device=device)
+            for _ in range(n_micro)]
+
+    def step():
+        opt.zero_grad()
+        for x in data:
+            loss = model(x, labels=x)
+            (loss / n_micro).backward()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = t.item()
+
+    tokens = B * S * args.steps
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = tokens / elapsed
+    if rank == 0:
+        result = {
+            "metric": "tokens/sec (whole node) Llama-2-7B training, "
+                      "TP=num_gpus, seq=4096",
+            "value": value,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model if not args.layers else
+                         f"{args.model}-layers{args.layers}-INVALID",
+                "global_batch": B,
+                "seq_len": S,
+                "parallelism": f"tp{tp}" + (f"_pp{args.pp}" if args.pp > 1 else "")
+                               + ("_sp" if args.sequence_parallel else ""),
+                "zero1": True,
+                "loss": float(loss.item()),
+            },
+        }
+        print(json.dumps(result), flush=True)
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

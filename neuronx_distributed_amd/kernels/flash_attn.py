@@ -1,0 +1,54 @@
+"""Flash attention wrapper (reference kernels/flash_attn.py parity).
+
+``flash_attn_func(q, k, v, causal=True)`` with q (B,Hq,S,D), k/v (B,Hkv,S,D)
+bf16 — GQA supported (Hq a multiple of Hkv).  On GPU this dispatches to the
+hand-written CDNA4 MFMA kernel (ops.flash_attn); on CPU it runs the plain
+fp32 torch reference the kernel is tested against.
+
+Set NXDA_ALLOW_TORCH_FALLBACK=1 to permit the composed-torch path on GPU
+(bring-up only — the HIP kernel is the production path and GPU calls fail
+loudly without it otherwise).
+"""
+
+import math
+import os
+
+import torch
+
+
+def _torch_reference(q, k, v, causal=True, scale=None):
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    rep = Hq // Hkv
+    if rep > 1:
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    scale = scale or 1.0 / math.sqrt(D)
+    scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        Sk = k.shape[2]
+        mask = torch.ones(S, Sk, dtype=torch.bool, device=q.device).tril(
+            diagonal=Sk - S)
+        scores = scores.masked_fill(~mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    return torch.matmul(probs, v.float()).to(q.dtype)
+
+
+def flash_attn_func(q, k, v, causal=True, softmax_scale=None):
+    if q.is_cuda:
+        from .. import ops
+
+        if hasattr(ops, "flash_attn") and ops.flash_attn_available():
+            return ops.flash_attn(q, k, v, causal=causal,
+                                  softmax_scale=softmax_scale)
+        if os.environ.get("NXDA_ALLOW_TORCH_FALLBACK", "0") == "1":
+            return _torch_reference(q, k, v, causal, softmax_scale)
+        raise RuntimeError(
+            "HIP flash-attention kernel unavailable on GPU; build "
+            "neuronx_distributed_amd.ops (or set NXDA_ALLOW_TORCH_FALLBACK=1 "
+            "for bring-up)")
+    return _torch_reference(q, k, v, causal, softmax_scale)
+
+
+# reference-compatible name (kernels/flash_attn.py:162)
+nki_flash_attn_func = flash_attn_func

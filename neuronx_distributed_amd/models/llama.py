@@ -1,0 +1,252 @@
+"""Llama-2/3 model family on the MI355X parallel layers.
+
+The model-zoo equivalent of the reference's
+``examples/training/llama/modeling_llama_nxd.py`` (ParallelEmbedding +
+GQAQKVColumnParallelLinear + Row/ColumnParallel MLP + parallel CE), built
+MI355X-first:
+
+* hot ops (RMSNorm, RoPE, SwiGLU, flash attention) are HIP kernels
+  (``neuronx_distributed_amd.ops``), everything GEMM-shaped is hipBLASLt
+  via F.linear;
+* gate/up projections are ONE fused stride-2 ColumnParallel GEMM;
+* layout is (B,S,H); with sequence parallelism the activations between TP
+  regions are (S/tp,B,H) (seq dim 0), matching the reference's SP design.
+"""
+
+import math
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..parallel import parallel_state as ps
+from ..parallel.layers import (
+    ColumnParallelLinear,
+    RowParallelLinear,
+    ParallelEmbedding,
+)
+from ..parallel.qkv_linear import GQAQKVColumnParallelLinear
+from ..parallel.loss_functions import parallel_cross_entropy
+from ..parallel.mappings import (
+    scatter_to_sequence_parallel_region,
+    gather_from_sequence_parallel_region,
+)
+from ..kernels.flash_attn import flash_attn_func
+
+
+@dataclass
+class LlamaConfig:
+    hidden_size: int = 4096
+    intermediate_size: int = 11008
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 32
+    vocab_size: int = 32000
+    max_position_embeddings: int = 4096
+    rope_theta: float = 10000.0
+    rms_norm_eps: float = 1e-5
+    initializer_range: float = 0.02
+    sequence_parallel_enabled: bool = False
+    tie_word_embeddings: bool = False
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.num_attention_heads
+
+
+CONFIGS = {
+    "llama2-7b": LlamaConfig(),
+    "llama2-13b": LlamaConfig(hidden_size=5120, intermediate_size=13824,
+                              num_hidden_layers=40, num_attention_heads=40,
+                              num_key_value_heads=40),
+    "llama2-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672,
+                              num_hidden_layers=80, num_attention_heads=64,
+                              num_key_value_heads=8),
+    "llama3-8b": LlamaConfig(hidden_size=4096, intermediate_size=14336,
+                             num_hidden_layers=32, num_attention_heads=32,
+                             num_key_value_heads=8, vocab_size=128256,
+                             rope_theta=500000.0,
+                             max_position_embeddings=8192),
+    "llama3-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672,
+                              num_hidden_layers=80, num_attention_heads=64,
+                              num_key_value_heads=8, vocab_size=128256,
+                              rope_theta=500000.0,
+                              max_position_embeddings=8192),
+    "tiny": LlamaConfig(hidden_size=64, intermediate_size=128,
+                        num_hidden_layers=2, num_attention_heads=4,
+                        num_key_value_heads=2, vocab_size=256,
+                        max_position_embeddings=128, rms_norm_eps=1e-5),
+}
+
+
+def get_config(name: str, **overrides) -> LlamaConfig:
+    import dataclasses
+
+    cfg = dataclasses.replace(CONFIGS[name], **overrides)
+    return cfg
+
+
+class RMSNorm(nn.Module):
+    """Reference modules/rms_norm.py parity; HIP-fused on GPU."""
+
+    def __init__(self, hidden_size, eps=1e-6, sequence_parallel_enabled=False):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden_size))
+        self.variance_epsilon = eps
+        self.sequence_parallel_enabled = sequence_parallel_enabled
+        self.weight.sequence_parallel_enabled = sequence_parallel_enabled
+
+    def forward(self, x):
+        return ops.rmsnorm(x, self.weight, self.variance_epsilon)
+
+
+def _init_method(std):
+    def f(t):
+        return nn.init.normal_(t, mean=0.0, std=std)
+
+    return f
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.config = config
+        tp = ps.get_tensor_model_parallel_size()
+        self.head_dim = config.head_dim
+        kv_mult = max(1, tp // config.num_key_value_heads)
+        self.kv_mult = kv_mult
+        self.num_heads_local = config.num_attention_heads // tp
+        self.num_kv_local = config.num_key_value_heads * kv_mult // tp
+        sp = config.sequence_parallel_enabled
+
+        self.qkv_proj = GQAQKVColumnParallelLinear(
+            config.hidden_size,
+            [config.num_attention_heads * self.head_dim,
+             config.num_key_value_heads * self.head_dim],
+            bias=False, gather_output=False,
+            num_attention_heads=config.num_attention_heads,
+            num_key_value_heads=config.num_key_value_heads,
+            head_dim=self.head_dim, kv_size_multiplier=kv_mult,
+            sequence_parallel_enabled=sp,
+            init_method=_init_method(config.initializer_range))
+        self.o_proj = RowParallelLinear(
+            config.num_attention_heads * self.head_dim, config.hidden_size,
+            bias=False, input_is_parallel=True, sequence_parallel_enabled=sp,
+            init_method=_init_method(config.initializer_range))
+
+    def forward(self, hidden, cos, sin, pos_offset=0, kv_cache=None):
+        sp = self.config.sequence_parallel_enabled
+        q, k, v = self.qkv_proj(hidden)
+        if sp:
+            # (S,B,*) -> (B,S,*)
+            q, k, v = (t.transpose(0, 1) for t in (q, k, v))
+        B, S = q.shape[0], q.shape[1]
+        q = q.reshape(B, S, self.num_heads_local, self.head_dim)
+        k = k.reshape(B, S, self.num_kv_local, self.head_dim)
+        v = v.reshape(B, S, self.num_kv_local, self.head_dim)
+        q, k = ops.apply_rotary_pos_emb(q, k, cos, sin, pos_offset)
+
+        # (B,S,h,D) -> (B,h,S,D)
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        if kv_cache is not None:
+            k, v = kv_cache.update(k, v, pos_offset)
+        out = flash_attn_func(q, k, v, causal=kv_cache is None)
+        out = out.transpose(1, 2).reshape(B, S, -1)
+        if sp:
+            out = out.transpose(0, 1)
+        return self.o_proj(out)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        sp = config.sequence_parallel_enabled
+        # fused [gate; up] stride-2 column-parallel GEMM
+        self.gate_up_proj = ColumnParallelLinear(
+            config.hidden_size, 2 * config.intermediate_size, bias=False,
+            gather_output=False, stride=2, sequence_parallel_enabled=sp,
+            init_method=_init_method(config.initializer_range))
+        self.down_proj = RowParallelLinear(
+            config.intermediate_size, config.hidden_size, bias=False,
+            input_is_parallel=True, sequence_parallel_enabled=sp,
+            init_method=_init_method(config.initializer_range))
+
+    def forward(self, x):
+        return self.down_proj(ops.swiglu(self.gate_up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        sp = config.sequence_parallel_enabled
+        self.self_attn = LlamaAttention(config)
+        self.mlp = LlamaMLP(config)
+        self.input_layernorm = RMSNorm(config.hidden_size, config.rms_norm_eps,
+                                       sequence_parallel_enabled=sp)
+        self.post_attention_layernorm = RMSNorm(
+            config.hidden_size, config.rms_norm_eps,
+            sequence_parallel_enabled=sp)
+
+    def forward(self, hidden, cos, sin, pos_offset=0, kv_cache=None):
+        h = hidden + self.self_attn(self.input_layernorm(hidden), cos, sin,
+                                    pos_offset, kv_cache)
+        return h + self.mlp(self.post_attention_layernorm(h))
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.config = config
+        self.embed_tokens = ParallelEmbedding(
+            config.vocab_size, config.hidden_size,
+            init_method=_init_method(config.initializer_range),
+            sequence_parallel_enabled=False)
+        self.layers = nn.ModuleList(
+            LlamaDecoderLayer(config) for _ in range(config.num_hidden_layers))
+        self.norm = RMSNorm(config.hidden_size, config.rms_norm_eps,
+                            sequence_parallel_enabled=config.sequence_parallel_enabled)
+        cos, sin = ops.precompute_rope_freqs(config.max_position_embeddings,
+                                             config.head_dim, config.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def forward(self, input_ids, pos_offset=0, kv_caches=None):
+        hidden = self.embed_tokens(input_ids)  # (B,S,H)
+        if self.config.sequence_parallel_enabled:
+            hidden = hidden.transpose(0, 1).contiguous()  # (S,B,H)
+            hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
+        for i, layer in enumerate(self.layers):
+            kc = kv_caches[i] if kv_caches is not None else None
+            hidden = layer(hidden, self.rope_cos, self.rope_sin, pos_offset, kc)
+        hidden = self.norm(hidden)
+        if self.config.sequence_parallel_enabled:
+            hidden = gather_from_sequence_parallel_region(
+                hidden, seq_dim=0, to_model_parallel=True)
+            hidden = hidden.transpose(0, 1)  # (B,S,H)
+        return hidden
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.config = config
+        self.model = LlamaModel(config)
+        self.lm_head = ColumnParallelLinear(
+            config.hidden_size, config.vocab_size, bias=False,
+            gather_output=False,
+            init_method=_init_method(config.initializer_range))
+        if config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+
+    def forward(self, input_ids, labels=None, pos_offset=0, kv_caches=None):
+        hidden = self.model(input_ids, pos_offset, kv_caches)
+        logits = self.lm_head(hidden)  # (B,S,V/tp)
+        if labels is None:
+            return logits
+        loss = parallel_cross_entropy(
+            logits[:, :-1, :].float(), labels[:, 1:].contiguous())
+        return loss.mean()

@@ -1,0 +1,250 @@
+"""Hand-written CDNA4 HIP kernels — python wrappers + autograd.
+
+The native library (``libnxd_ops.so``, built in-tree by ``ops/build.py``)
+replaces the reference's NKI kernels: RMSNorm, RoPE,
+SwiGLU, flash attention, fused AdamW.  On a GPU box the HIP path is
+MANDATORY — if the library is missing, GPU calls raise instead of silently
+falling back to eager torch.  CPU tensors use plain-torch reference
+implementations (the numerics baseline the GPU kernels are tested against).
+"""
+
+import ctypes
+import math
+import os
+from typing import Optional
+
+import torch
+
+_LIB = None
+_LIB_ERR = None
+
+
+def _load():
+    global _LIB, _LIB_ERR
+    if _LIB is not None or _LIB_ERR is not None:
+        return _LIB
+    from .build import LIB
+
+    if not os.path.exists(LIB):
+        _LIB_ERR = f"{LIB} not built — run neuronx_distributed_amd.ops.build"
+        return None
+    try:
+        lib = ctypes.CDLL(LIB)
+    except OSError as e:  # pragma: no cover
+        _LIB_ERR = str(e)
+        return None
+    _LIB = lib
+    return _LIB
+
+
+def is_available() -> bool:
+    return _load() is not None
+
+
+def _require_lib():
+    lib = _load()
+    if lib is None:
+        raise RuntimeError(
+            f"nxd_ops HIP library unavailable on a GPU tensor: {_LIB_ERR}. "
+            "The MI355X-native kernels are mandatory on GPU — build with "
+            "python -m neuronx_distributed_amd.ops.build")
+    return lib
+
+
+def _stream() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _ptr(t: torch.Tensor) -> ctypes.c_void_p:
+    return ctypes.c_void_p(t.data_ptr())
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ctx.eps = eps
+        if x.is_cuda:
+            lib = _require_lib()
+            assert x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0
+            x2 = x.contiguous()
+            rows = x2.numel() // x2.shape[-1]
+            out = torch.empty_like(x2)
+            rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+            lib.rmsnorm_fwd(_ptr(x2), _ptr(weight.contiguous()), _ptr(out),
+                            _ptr(rstd), ctypes.c_int(rows),
+                            ctypes.c_int(x2.shape[-1]), ctypes.c_float(eps),
+                            _stream())
+            ctx.save_for_backward(x2, weight, rstd)
+            return out
+        # CPU reference (fp32 math like the kernel)
+        xf = x.float()
+        rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+        ctx.save_for_backward(x, weight, rstd.squeeze(-1).reshape(-1))
+        return (xf * rstd * weight.float()).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, rstd = ctx.saved_tensors
+        H = x.shape[-1]
+        if x.is_cuda:
+            lib = _require_lib()
+            dy2 = dy.contiguous()
+            rows = x.numel() // H
+            dx = torch.empty_like(x)
+            dw32 = torch.zeros(H, dtype=torch.float32, device=x.device)
+            lib.rmsnorm_bwd(_ptr(x), _ptr(weight.contiguous()), _ptr(dy2),
+                            _ptr(rstd), _ptr(dx), _ptr(dw32),
+                            ctypes.c_int(rows), ctypes.c_int(H), _stream())
+            return dx, dw32.to(weight.dtype), None
+        xf = x.float()
+        dyf = dy.float()
+        wf = weight.float()
+        r = rstd.reshape(x.shape[:-1]).unsqueeze(-1).to(torch.float32)
+        dot = (dyf * wf * xf).sum(-1, keepdim=True)
+        dx = r * wf * dyf - r.pow(3) / H * xf * dot
+        dw = (dyf * xf * r).reshape(-1, H).sum(0)
+        return dx.to(x.dtype), dw.to(weight.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6):
+    return _RMSNormFn.apply(x, weight, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE (neox rotate-half), in-place on clones
+# ---------------------------------------------------------------------------
+
+def _rope_torch(x, cos, sin, sign=1.0):
+    # x (B,S,h,D); cos/sin (S, D/2) f32
+    half = x.shape[-1] // 2
+    x0 = x[..., :half].float()
+    x1 = x[..., half:].float()
+    c = cos.view(1, cos.shape[0], 1, half)
+    s = sin.view(1, sin.shape[0], 1, half) * sign
+    o0 = x0 * c - x1 * s
+    o1 = x1 * c + x0 * s
+    return torch.cat([o0, o1], dim=-1).to(x.dtype)
+
+
+class _RoPEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, cos, sin, pos_offset):
+        ctx.pos_offset = pos_offset
+        ctx.save_for_backward(cos, sin)
+        if q.is_cuda:
+            lib = _require_lib()
+            B, S, Hq, D = q.shape
+            Hk = k.shape[2]
+            qo = q.contiguous().clone()
+            ko = k.contiguous().clone()
+            lib.rope_fwd(_ptr(qo), _ptr(ko), _ptr(cos), _ptr(sin),
+                         ctypes.c_int(B), ctypes.c_int(S), ctypes.c_int(Hq),
+                         ctypes.c_int(Hk), ctypes.c_int(D),
+                         ctypes.c_int(pos_offset), ctypes.c_int(0), _stream())
+            return qo, ko
+        S = q.shape[1]
+        c = cos[pos_offset:pos_offset + S]
+        s = sin[pos_offset:pos_offset + S]
+        return _rope_torch(q, c, s), _rope_torch(k, c, s)
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        cos, sin = ctx.saved_tensors
+        off = ctx.pos_offset
+        if dq.is_cuda:
+            lib = _require_lib()
+            B, S, Hq, D = dq.shape
+            Hk = dk.shape[2]
+            dqo = dq.contiguous().clone()
+            dko = dk.contiguous().clone()
+            lib.rope_fwd(_ptr(dqo), _ptr(dko), _ptr(cos), _ptr(sin),
+                         ctypes.c_int(B), ctypes.c_int(S), ctypes.c_int(Hq),
+                         ctypes.c_int(Hk), ctypes.c_int(D), ctypes.c_int(off),
+                         ctypes.c_int(1), _stream())
+            return dqo, dko, None, None, None
+        S = dq.shape[1]
+        c = cos[off:off + S]
+        s = sin[off:off + S]
+        return (_rope_torch(dq, c, s, -1.0), _rope_torch(dk, c, s, -1.0),
+                None, None, None)
+
+
+def apply_rotary_pos_emb(q, k, cos, sin, pos_offset: int = 0):
+    """q (B,S,Hq,D), k (B,S,Hk,D); cos/sin (S_max, D/2) fp32 tables."""
+    return _RoPEFn.apply(q, k, cos, sin, pos_offset)
+
+
+def precompute_rope_freqs(seq_len: int, dim: int, theta: float = 10000.0,
+                          device=None):
+    """cos/sin tables (S, D/2) fp32 (reference attention/utils.py
+    precompute_freqs_cis)."""
+    inv = 1.0 / (theta ** (torch.arange(0, dim, 2, device=device,
+                                        dtype=torch.float32) / dim))
+    t = torch.arange(seq_len, device=device, dtype=torch.float32)
+    freqs = torch.outer(t, inv)
+    return freqs.cos(), freqs.sin()
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        I = x.shape[-1] // 2
+        if x.is_cuda:
+            lib = _require_lib()
+            assert x.dtype == torch.bfloat16 and I % 8 == 0
+            x2 = x.contiguous()
+            N = x2.numel() // x2.shape[-1]
+            out = torch.empty(x2.shape[:-1] + (I,), dtype=x2.dtype,
+                              device=x2.device)
+            lib.swiglu_fwd(_ptr(x2), _ptr(out), ctypes.c_long(N),
+                           ctypes.c_int(I), _stream())
+            return out
+        g, u = x[..., :I].float(), x[..., I:].float()
+        return (torch.nn.functional.silu(g) * u).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        I = x.shape[-1] // 2
+        if x.is_cuda:
+            lib = _require_lib()
+            dy2 = dy.contiguous()
+            N = x.numel() // x.shape[-1]
+            dx = torch.empty_like(x)
+            lib.swiglu_bwd(_ptr(x), _ptr(dy2), _ptr(dx), ctypes.c_long(N),
+                           ctypes.c_int(I), _stream())
+            return dx
+        g, u = x[..., :I].float(), x[..., I:].float()
+        dyf = dy.float()
+        sig = torch.sigmoid(g)
+        s = g * sig
+        dg = dyf * u * (sig + s * (1 - sig))
+        du = dyf * s
+        return torch.cat([dg, du], dim=-1).to(x.dtype)
+
+
+def swiglu(x: torch.Tensor) -> torch.Tensor:
+    """x = [gate; up] on the last dim -> silu(gate)*up."""
+    return _SwiGLUFn.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# Flash attention (MFMA kernel in csrc/flash_attn.hip)
+# ---------------------------------------------------------------------------
+
+def flash_attn_available() -> bool:
+    lib = _load()
+    return lib is not None and hasattr(lib, "flash_attn_fwd")
+
+
+def flash_attn(q, k, v, causal=True, softmax_scale=None):
+    raise RuntimeError("HIP flash attention kernel not built yet")

@@ -1,0 +1,127 @@
+// Fused RMSNorm forward/backward for CDNA4 (gfx950).
+//
+// Replaces the reference's torch-level RMSNorm that the Neuron compiler
+// fuses (reference modules/rms_norm.py:10-36) — eager ROCm will not fuse,
+// so this is a hand-written HBM-bound kernel: one pass, vectorized 16B
+// bf16 loads (guide G13), fp32 accumulation, per-row rstd saved for bwd.
+//
+// Layout: x (N, H) bf16 rows; w (H) bf16; out (N, H) bf16; rstd (N) f32.
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// forward: out = x * rsqrt(mean(x^2) + eps) * w
+// one block per ROWS_PER_BLOCK rows; vector short8 loads
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+rmsnorm_fwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
+                   short* __restrict__ out, float* __restrict__ rstd_out,
+                   int H, float eps, int rows) {
+  __shared__ float scratch[16];
+  int row = blockIdx.x;
+  if (row >= rows) return;
+  const short* xr = x + (long)row * H;
+  short* outr = out + (long)row * H;
+
+  int nvec = H >> 3;  // H % 8 == 0 required
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    s8v v = *(const s8v*)(xr + i * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bits2f(v[j]);
+      ss += f * f;
+    }
+  }
+  ss = block_reduce_sum(ss, scratch);
+  float rstd = rsqrtf(ss / (float)H + eps);
+  if (threadIdx.x == 0 && rstd_out) rstd_out[row] = rstd;
+
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    s8v v = *(const s8v*)(xr + i * 8);
+    s8v wv = *(const s8v*)(w + i * 8);
+    s8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f2bits(bits2f(v[j]) * rstd * bits2f(wv[j]));
+    *(s8v*)(outr + i * 8) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward:
+//   dx = rstd * w * dy - rstd^3/H * x * sum(dy * w * x)
+//   dw += sum_rows(dy * x * rstd)        (fp32 atomics into dw_f32)
+// ---------------------------------------------------------------------------
+// grid-stride over rows; per-block dw partial accumulated in LDS (H*4 B
+// dynamic LDS, H<=32768), ONE atomicAdd per element per block at the end —
+// avoids rows*H atomic contention.
+extern "C" __global__ void __launch_bounds__(256)
+rmsnorm_bwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
+                   const short* __restrict__ dy,
+                   const float* __restrict__ rstd_in,
+                   short* __restrict__ dx, float* __restrict__ dw_f32,
+                   int H, int rows) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* dw_part = smem;           // [H]
+  float* scratch = smem + H;       // [16]
+
+  for (int i = threadIdx.x; i < H; i += blockDim.x) dw_part[i] = 0.f;
+  __syncthreads();
+
+  int nvec = H >> 3;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = x + (long)row * H;
+    const short* dyr = dy + (long)row * H;
+    short* dxr = dx + (long)row * H;
+    float rstd = rstd_in[row];
+
+    float dot = 0.f;  // sum(dy * w * x)
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      s8v xv = *(const s8v*)(xr + i * 8);
+      s8v dv = *(const s8v*)(dyr + i * 8);
+      s8v wv = *(const s8v*)(w + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bits2f(dv[j]) * bits2f(wv[j]) * bits2f(xv[j]);
+    }
+    dot = block_reduce_sum(dot, scratch);
+    float k = rstd * rstd * rstd * dot / (float)H;
+
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      s8v xv = *(const s8v*)(xr + i * 8);
+      s8v dv = *(const s8v*)(dyr + i * 8);
+      s8v wv = *(const s8v*)(w + i * 8);
+      s8v o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = bits2f(xv[j]);
+        float df = bits2f(dv[j]);
+        o[j] = f2bits(rstd * bits2f(wv[j]) * df - k * xf);
+        dw_part[i * 8 + j] += df * xf * rstd;  // thread-exclusive slot per i
+      }
+      *(s8v*)(dxr + i * 8) = o;
+    }
+    __syncthreads();  // dw_part reuse across rows is thread-local per index
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < H; i += blockDim.x)
+    if (dw_part[i] != 0.f) atomicAdd(dw_f32 + i, dw_part[i]);
+}
+
+extern "C" void rmsnorm_fwd(const void* x, const void* w, void* out,
+                            void* rstd, int rows, int H, float eps,
+                            hipStream_t stream) {
+  rmsnorm_fwd_kernel<<<rows, 256, 0, stream>>>(
+      (const short*)x, (const short*)w, (short*)out, (float*)rstd, H, eps, rows);
+}
+
+extern "C" void rmsnorm_bwd(const void* x, const void* w, const void* dy,
+                            const void* rstd, void* dx, void* dw_f32, int rows,
+                            int H, hipStream_t stream) {
+  int blocks = rows < 2048 ? rows : 2048;
+  size_t lds = (size_t)(H + 16) * sizeof(float);
+  rmsnorm_bwd_kernel<<<blocks, 256, lds, stream>>>(
+      (const short*)x, (const short*)w, (const short*)dy, (const float*)rstd,
+      (short*)dx, (float*)dw_f32, H, rows);
+}

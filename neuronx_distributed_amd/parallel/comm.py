@@ -55,6 +55,8 @@ def all_reduce(tensor: torch.Tensor, op=dist.ReduceOp.SUM, group=None,
     """In-place all-reduce. Returns the async work handle if requested."""
     if ps.is_aot_mode():
         return None
+    if isinstance(group, ps.GroupInfo) and group.size == 1:
+        return None
     g = _unwrap(group)
     return dist.all_reduce(tensor, op=op, group=g, async_op=async_op)
 
