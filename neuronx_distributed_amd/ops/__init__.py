@@ -246,5 +246,57 @@ def flash_attn_available() -> bool:
     return lib is not None and hasattr(lib, "flash_attn_fwd")
 
 
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        lib = _require_lib()
+        B, Hq, S, D = q.shape
+        Hkv = k.shape[1]
+        assert D == 128, "flash kernel supports D=128"
+        assert q.dtype == torch.bfloat16
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        out = torch.empty_like(q)
+        lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
+        lib.flash_attn_fwd(_ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(lse),
+                           ctypes.c_int(B), ctypes.c_int(Hq),
+                           ctypes.c_int(Hkv), ctypes.c_int(S),
+                           ctypes.c_float(scale), ctypes.c_int(1 if causal else 0),
+                           _stream())
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        lib = _require_lib()
+        if not hasattr(lib, "flash_attn_bwd"):
+            raise RuntimeError("flash_attn_bwd kernel not built")
+        B, Hq, S, D = q.shape
+        Hkv = k.shape[1]
+        dout = dout.contiguous()
+        delta = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
+        dq = torch.empty_like(q)
+        dk32 = torch.zeros(B, Hq, S, D, dtype=torch.float32, device=q.device)
+        dv32 = torch.zeros(B, Hq, S, D, dtype=torch.float32, device=q.device)
+        lib.flash_attn_bwd(_ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(dout),
+                           _ptr(lse), _ptr(delta), _ptr(dq), _ptr(dk32),
+                           _ptr(dv32), ctypes.c_int(B), ctypes.c_int(Hq),
+                           ctypes.c_int(Hkv), ctypes.c_int(S),
+                           ctypes.c_float(ctx.scale),
+                           ctypes.c_int(1 if ctx.causal else 0), _stream())
+        rep = Hq // Hkv
+        if rep > 1:
+            dk = dk32.view(B, Hkv, rep, S, D).sum(2).to(q.dtype)
+            dv = dv32.view(B, Hkv, rep, S, D).sum(2).to(q.dtype)
+        else:
+            dk = dk32.to(q.dtype)
+            dv = dv32.to(q.dtype)
+        return dq, dk, dv, None, None
+
+
 def flash_attn(q, k, v, causal=True, softmax_scale=None):
-    raise RuntimeError("HIP flash attention kernel not built yet")
+    """q (B,Hq,S,D=128), k/v (B,Hkv,S,D) bf16 on GPU."""
+    scale = softmax_scale or 1.0 / math.sqrt(q.shape[-1])
+    return _FlashAttnFn.apply(q, k, v, causal, scale)

@@ -1,0 +1,273 @@
+// FlashAttention forward for CDNA4 (gfx950) — hand-written MFMA kernel.
+//
+// Replaces the reference's NKI flash kernels K1 (SURVEY.md §2.3,
+// reference kernels/flash_attn.py:18,51-63).  MI355X-first design per
+// cdna_hip_programming.md §B:
+//   * 8 waves/WG, each wave owns 32 q rows (WG = 256 q rows); KV tiles of
+//     64 double-buffered in LDS with the (row&7)<<4 XOR swizzle (G4).
+//   * swapped QK^T — mfma(A=K, B=Q^T) — so each lane's softmax stats are
+//     for ONE q row (l&31): row max / rescale / denom are lane-local
+//     (+ one shfl_xor with the partner half-wave).
+//   * P is repacked to bf16 PV B-fragments with pack_bf16x2 +
+//     v_permlane32_swap (T12/T21 primitive).
+//   * O is accumulated TRANSPOSED (O^T[d][q], q = lane) so the online
+//     rescale is a per-lane scalar multiply; V is staged transposed (Vt).
+//   * GQA: kv head = q head / (Hq/Hkv); causal masking per element on
+//     diagonal tiles, whole-tile skip below the diagonal.
+//
+// Layouts: q,k,v,out (B, H, S, D) bf16 contiguous, D = 128; lse (B,Hq,S)
+// f32 (natural-log row logsumexp, saved for backward).
+
+#include "common.h"
+#include "mfma.h"
+
+#define FA_D 128
+#define FA_QW 32       // q rows per wave
+#define FA_WAVES 8
+#define FA_QBLK (FA_QW * FA_WAVES)  // 256 q rows per workgroup
+#define FA_KV 64       // kv tile
+#define LOG2E 1.4426950408889634f
+#define NEG_INF (-1e30f)
+
+// LDS: double-buffered K [64][128] and Vt [128][64], both bf16, XOR-swizzled
+#define K_TILE_B (FA_KV * FA_D * 2)    // 16 KB
+#define VT_TILE_B (FA_D * FA_KV * 2)   // 16 KB
+
+extern "C" __global__ void __launch_bounds__(512, 2)
+flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
+                 const short* __restrict__ vp, short* __restrict__ op,
+                 float* __restrict__ lsep, int B, int Hq, int Hkv, int S,
+                 float scale, int causal) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto kbuf = [&](int i) { return smem + (size_t)i * K_TILE_B; };
+  auto vbuf = [&](int i) {
+    return smem + 2 * K_TILE_B + (size_t)i * VT_TILE_B;
+  };
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int col = lane & 31;       // q column owned by this lane
+  const int hi = lane >> 5;
+
+  const int qblk = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / (Hq / Hkv);
+
+  const long q_base = ((long)(b * Hq + h) * S) * FA_D;
+  const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
+
+  const int q0 = qblk * FA_QBLK;
+  const int qw0 = q0 + wid * FA_QW;      // this wave's first q row
+  const int my_q = qw0 + col;            // this lane's q row
+  const int q_row_ld = my_q < S ? my_q : S - 1;
+
+  // ---- Q fragments: 8 chunks of (16 d), each lane 8 bf16 --------------
+  frag_u qf[8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    const short* src = qp + q_base + (long)q_row_ld * FA_D + c * 16 + hi * 8;
+    qf[c].u4 = *(const uint4v*)src;
+  }
+
+  // ---- accumulators ----------------------------------------------------
+  f32x16 ot[4] = {};          // O^T: d-block nb, rows d_local, col q
+  float m_run = NEG_INF;      // running max (exp2 domain)
+  float l_run = 0.f;
+
+  const int kv_end = causal ? min(S, q0 + FA_QBLK) : S;
+  const int ntiles = (kv_end + FA_KV - 1) / FA_KV;
+  // this wave can skip tiles fully above its causal row range
+  const int my_kv_end = causal ? min(S, qw0 + FA_QW) : S;
+
+  const float s2 = scale * LOG2E;
+
+  // ---- staging: thread t loads pieces t, t+512 of each tile -----------
+  // K: piece p = (row p>>3, 16B col p&7) -> ds_write_b128 swizzled
+  // V: same global piece, scatter-written transposed into Vt
+  const int tid = threadIdx.x;
+  int st_row[2], st_col[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int p = tid + i * 512;
+    st_row[i] = p >> 3;
+    st_col[i] = p & 7;
+  }
+
+  uint4v kreg[2], vreg[2];
+  auto issue_loads = [&](int t) {
+    int kv0 = t * FA_KV;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int row = kv0 + st_row[i];
+      int rr = row < S ? row : S - 1;  // clamp; masked later
+      kreg[i] = *(const uint4v*)(kp + kv_base + (long)rr * FA_D + st_col[i] * 8);
+      vreg[i] = *(const uint4v*)(vp + kv_base + (long)rr * FA_D + st_col[i] * 8);
+    }
+  };
+
+  auto write_tile = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      // K row-major [64][128], swizzled 16B slots
+      *(uint4v*)(kbuf(buf) + swz(st_row[i], st_col[i] * 16)
+                 + st_row[i] * (FA_D * 2)) = kreg[i];
+      // Vt transposed [128][64]: element j -> row d = col*8+j, col k = row
+      union { uint4v u; short s[8]; } vv;
+      vv.u = vreg[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = st_col[i] * 8 + j;
+        *(short*)(vbuf(buf) + d * (FA_KV * 2) + swz(d, st_row[i] * 2)) = vv.s[j];
+      }
+    }
+  };
+
+  issue_loads(0);
+  write_tile(0);
+  __syncthreads();
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * FA_KV;
+    const int cur = t & 1;
+    if (t + 1 < ntiles) issue_loads(t + 1);
+
+    const bool wave_active = kv0 < my_kv_end;
+    if (wave_active) {
+      // ---- QK^T: S^T[k][q] = sum_d K[k][d] Q^T[d][q] ------------------
+      f32x16 acc[2] = {};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+#pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+          int row = col + 32 * kb;
+          frag_u kf;
+          kf.u4 = *(const uint4v*)(kbuf(cur) + row * (FA_D * 2)
+                                   + swz(row, (c * 16 + hi * 8) * 2));
+          acc[kb] = mfma_bf16(kf.bf, qf[c].bf, acc[kb]);
+        }
+      }
+
+      // ---- online softmax (exp2 domain), lane-local per q row ---------
+      float sc[2][16];
+      const bool need_mask =
+          (causal && kv0 + FA_KV > qw0) || (kv0 + FA_KV > S);
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float v = acc[kb][r] * s2;
+          if (need_mask) {
+            int kg = kv0 + 32 * kb + acc_row(r, hi);
+            if ((causal && kg > my_q) || kg >= S) v = NEG_INF;
+          }
+          sc[kb][r] = v;
+        }
+
+      float mt = NEG_INF;
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) mt = fmaxf(mt, sc[kb][r]);
+      mt = fmaxf(mt, __shfl_xor(mt, 32, 64));  // combine partner half
+
+      float m_new = fmaxf(m_run, mt);
+      // m_eff floor: when a lane's rows are ALL masked so far, m_new is
+      // -1e30 and exp2(sc - m_new) would be exp2(0)=1 for masked scores;
+      // the floor keeps those at exp2(-9.9e29) = 0.
+      float m_eff = fmaxf(m_new, -1e28f);
+      float alpha = __builtin_exp2f(m_run - m_eff);  // 0 when m_run=-1e30
+      m_run = m_new;
+
+      float psum = 0.f;
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float p = __builtin_exp2f(sc[kb][r] - m_eff);
+          sc[kb][r] = p;
+          psum += p;
+        }
+      psum += __shfl_xor(psum, 32, 64);
+      l_run = l_run * alpha + psum;
+
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb) ot[nb] *= alpha;
+
+      // ---- pack P -> PV B-fragments (4 chunks of 16 k) ----------------
+      frag_u pf[4];
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+        for (int cc = 0; cc < 2; ++cc) {
+          uint b0 = pack_bf16x2(sc[kb][8 * cc + 0], sc[kb][8 * cc + 1]);
+          uint b1 = pack_bf16x2(sc[kb][8 * cc + 2], sc[kb][8 * cc + 3]);
+          uint b2 = pack_bf16x2(sc[kb][8 * cc + 4], sc[kb][8 * cc + 5]);
+          uint b3 = pack_bf16x2(sc[kb][8 * cc + 6], sc[kb][8 * cc + 7]);
+          {
+            auto r01 = __builtin_amdgcn_permlane32_swap(b0, b2, false, false);
+            b0 = r01[0]; b2 = r01[1];
+          }
+          {
+            auto r23 = __builtin_amdgcn_permlane32_swap(b1, b3, false, false);
+            b1 = r23[0]; b3 = r23[1];
+          }
+          frag_u& f = pf[2 * kb + cc];
+          f.u[0] = b0; f.u[1] = b1; f.u[2] = b2; f.u[3] = b3;
+        }
+
+      // ---- PV: O^T[d][q] += V^T[d][k] P^T[k][q] -----------------------
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb) {
+#pragma unroll
+        for (int c16 = 0; c16 < 4; ++c16) {
+          int d = 32 * nb + col;
+          frag_u vf;
+          vf.u4 = *(const uint4v*)(vbuf(cur) + d * (FA_KV * 2)
+                                   + swz(d, (c16 * 16 + hi * 8) * 2));
+          ot[nb] = mfma_bf16(vf.bf, pf[c16].bf, ot[nb]);
+        }
+      }
+    }
+
+    __syncthreads();
+    if (t + 1 < ntiles) {
+      write_tile((t + 1) & 1);
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: normalize, store O (transposed back) and LSE ---------
+  if (my_q >= S) return;
+  float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+  long o_row = q_base + (long)my_q * FA_D;
+#pragma unroll
+  for (int nb = 0; nb < 4; ++nb) {
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg) {
+      int d = 32 * nb + 8 * rg + 4 * hi;
+      uint w0 = pack_bf16x2(ot[nb][4 * rg + 0] * inv_l,
+                            ot[nb][4 * rg + 1] * inv_l);
+      uint w1 = pack_bf16x2(ot[nb][4 * rg + 2] * inv_l,
+                            ot[nb][4 * rg + 3] * inv_l);
+      uint2 st = {w0, w1};
+      *(uint2*)(op + o_row + d) = st;
+    }
+  }
+  if (hi == 0 && lsep) {
+    // natural-log LSE: scores were in exp2 domain
+    lsep[(long)(b * Hq + h) * S + my_q] =
+        m_run * 0.6931471805599453f + __logf(l_run);
+  }
+}
+
+extern "C" void flash_attn_fwd(const void* q, const void* k, const void* v,
+                               void* out, void* lse, int B, int Hq, int Hkv,
+                               int S, float scale, int causal,
+                               hipStream_t stream) {
+  dim3 grid((S + FA_QBLK - 1) / FA_QBLK, Hq, B);
+  size_t lds = 2 * (K_TILE_B + VT_TILE_B);
+  flash_fwd_kernel<<<grid, 512, lds, stream>>>(
+      (const short*)q, (const short*)k, (const short*)v, (short*)out,
+      (float*)lse, B, Hq, Hkv, S, scale, causal);
+}
