@@ -83,15 +83,16 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const float s2 = scale * LOG2E;
 
   // ---- staging: thread t loads pieces t, t+512 of each tile -----------
-  // K: piece p = (row p>>3, 16B col p&7) -> ds_write_b128 swizzled
-  // V: same global piece, scatter-written transposed into Vt
+  // tile = 64 rows x 128 cols bf16 = 1024 16B pieces, 16 pieces per row:
+  // piece p -> (row p>>4, 16B slot p&15).  K: ds_write_b128 swizzled;
+  // V: same global piece, scatter-written transposed into Vt.
   const int tid = threadIdx.x;
   int st_row[2], st_col[2];
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     int p = tid + i * 512;
-    st_row[i] = p >> 3;
-    st_col[i] = p & 7;
+    st_row[i] = p >> 4;
+    st_col[i] = p & 15;
   }
 
   uint4v kreg[2], vreg[2];
