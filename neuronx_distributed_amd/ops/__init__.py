@@ -278,21 +278,23 @@ class _FlashAttnFn(torch.autograd.Function):
         dout = dout.contiguous()
         delta = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
         dq = torch.empty_like(q)
-        dk32 = torch.zeros(B, Hq, S, D, dtype=torch.float32, device=q.device)
-        dv32 = torch.zeros(B, Hq, S, D, dtype=torch.float32, device=q.device)
+        # dk/dv are written bf16 PER Q-HEAD (B,Hq,S,D); GQA replicas are
+        # reduced here (each replica covers distinct Q heads -> SUM)
+        dk_pq = torch.empty(B, Hq, S, D, dtype=q.dtype, device=q.device)
+        dv_pq = torch.empty(B, Hq, S, D, dtype=q.dtype, device=q.device)
         lib.flash_attn_bwd(_ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(dout),
-                           _ptr(lse), _ptr(delta), _ptr(dq), _ptr(dk32),
-                           _ptr(dv32), ctypes.c_int(B), ctypes.c_int(Hq),
+                           _ptr(lse), _ptr(delta), _ptr(dq), _ptr(dk_pq),
+                           _ptr(dv_pq), ctypes.c_int(B), ctypes.c_int(Hq),
                            ctypes.c_int(Hkv), ctypes.c_int(S),
                            ctypes.c_float(ctx.scale),
                            ctypes.c_int(1 if ctx.causal else 0), _stream())
         rep = Hq // Hkv
         if rep > 1:
-            dk = dk32.view(B, Hkv, rep, S, D).sum(2).to(q.dtype)
-            dv = dv32.view(B, Hkv, rep, S, D).sum(2).to(q.dtype)
+            dk = dk_pq.view(B, Hkv, rep, S, D).float().sum(2).to(q.dtype)
+            dv = dv_pq.view(B, Hkv, rep, S, D).float().sum(2).to(q.dtype)
         else:
-            dk = dk32.to(q.dtype)
-            dv = dv32.to(q.dtype)
+            dk = dk_pq
+            dv = dv_pq
         return dq, dk, dv, None, None
 
 
