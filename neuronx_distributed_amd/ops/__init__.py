@@ -302,3 +302,27 @@ def flash_attn(q, k, v, causal=True, softmax_scale=None):
     """q (B,Hq,S,D=128), k/v (B,Hkv,S,D) bf16 on GPU."""
     scale = softmax_scale or 1.0 / math.sqrt(q.shape[-1])
     return _FlashAttnFn.apply(q, k, v, causal, scale)
+
+
+# ---------------------------------------------------------------------------
+# Fused AdamW (ZeRO-1 master-shard update; csrc/adamw.hip)
+# ---------------------------------------------------------------------------
+
+def adamw_step(master: torch.Tensor, m: torch.Tensor, v: torch.Tensor,
+               grad_bf16: torch.Tensor, param_bf16: torch.Tensor,
+               clip: Optional[torch.Tensor], lr: float, beta1: float,
+               beta2: float, eps: float, weight_decay: float, step: int):
+    """One fused pass: bf16 grad -> fp32 m/v/master update -> bf16 param."""
+    lib = _require_lib()
+    n = master.numel()
+    assert m.numel() == n and v.numel() == n and grad_bf16.numel() == n
+    assert grad_bf16.dtype == torch.bfloat16 and param_bf16.dtype == torch.bfloat16
+    inv_bc1 = 1.0 / (1.0 - beta1 ** step)
+    inv_bc2 = 1.0 / (1.0 - beta2 ** step)
+    lib.adamw_step(ctypes.c_long(n), _ptr(master), _ptr(m), _ptr(v),
+                   _ptr(grad_bf16), _ptr(param_bf16),
+                   _ptr(clip) if clip is not None else None,
+                   ctypes.c_float(lr), ctypes.c_float(beta1),
+                   ctypes.c_float(beta2), ctypes.c_float(eps),
+                   ctypes.c_float(weight_decay), ctypes.c_float(inv_bc1),
+                   ctypes.c_float(inv_bc2), _stream())

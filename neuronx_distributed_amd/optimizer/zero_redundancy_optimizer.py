@@ -216,7 +216,7 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
             w.wait()
 
     @torch.no_grad()
-    def _clip_grads(self):
+    def _clip_grads(self, apply: bool = True):
         device = self.buckets[0].master.device
         sq = torch.zeros(1, dtype=torch.float32, device=device)
         for b in self.buckets:
@@ -229,15 +229,18 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
         total_norm = sq.sqrt()
         self._grad_norm = total_norm.squeeze()
         clip = torch.clamp(self.max_norm / (total_norm + 1e-6), max=1.0)
-        for b in self.buckets:
-            b.grad_shard.mul_(clip)
+        if apply:
+            for b in self.buckets:
+                b.grad_shard.mul_(clip)
+        return clip
 
     @torch.no_grad()
-    def _all_gather_params(self):
+    def _all_gather_params(self, copy_master: bool = True):
         works = []
         for b in self.buckets:
             shard = b.flat_param[b.shard_lo:b.shard_hi]
-            shard.copy_(b.master.to(b.flat_param.dtype))
+            if copy_master:
+                shard.copy_(b.master.to(b.flat_param.dtype))
             world = b.group_info.size
             if world == 1:
                 continue
@@ -252,6 +255,34 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
         for w in works:
             w.wait()
 
+    def _use_fused(self) -> bool:
+        from .. import ops
+
+        return (self.use_fused_kernel
+                and self.optimizer_class is torch.optim.AdamW
+                and self.optimizer_dtype == torch.float32
+                and self.buckets and self.buckets[0].master.is_cuda
+                and ops.is_available())
+
+    @torch.no_grad()
+    def _fused_step(self, clip):
+        """One HIP kernel per bucket shard: bf16 grad -> fp32 m/v/master ->
+        bf16 param written straight into the all-gather source slice."""
+        from .. import ops
+
+        self._step_count = getattr(self, "_step_count", 0) + 1
+        for b in self.buckets:
+            group = self.param_groups[b.group_index]
+            if not hasattr(b, "fused_m"):
+                b.fused_m = torch.zeros_like(b.master)
+                b.fused_v = torch.zeros_like(b.master)
+            beta1, beta2 = group.get("betas", (0.9, 0.999))
+            ops.adamw_step(b.master, b.fused_m, b.fused_v, b.grad_shard,
+                           b.flat_param[b.shard_lo:b.shard_hi], clip,
+                           group.get("lr", 1e-3), beta1, beta2,
+                           group.get("eps", 1e-8),
+                           group.get("weight_decay", 0.0), self._step_count)
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None
@@ -262,6 +293,15 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
         for b in self.buckets:
             b.relink_grads()
         self._reduce_scatter_grads()
+
+        if self._use_fused():
+            clip = None
+            if self.grad_clipping:
+                clip = self._clip_grads(apply=False)
+            self._fused_step(clip)
+            self._all_gather_params(copy_master=False)
+            return loss
+
         if self.grad_clipping:
             self._clip_grads()
         for b in self.buckets:
@@ -288,6 +328,11 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
     def state_dict(self):
         return {
             "base_optimizer": self.base_optimizer.state_dict(),
+            "step_count": getattr(self, "_step_count", 0),
+            "fused_state": [
+                {"m": b.fused_m.cpu(), "v": b.fused_v.cpu()}
+                if hasattr(b, "fused_m") else None for b in self.buckets
+            ],
             "masters": [b.master.detach().cpu() for b in self.buckets],
             "shard_meta": [
                 {"padded": b.padded, "rank": b.rank, "world": b.group_info.size}
@@ -297,6 +342,11 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
 
     def load_state_dict(self, state_dict):
         self.base_optimizer.load_state_dict(state_dict["base_optimizer"])
+        self._step_count = state_dict.get("step_count", 0)
+        for b, fs in zip(self.buckets, state_dict.get("fused_state", [])):
+            if fs is not None:
+                b.fused_m = fs["m"].to(b.master.device)
+                b.fused_v = fs["v"].to(b.master.device)
         for b, m in zip(self.buckets, state_dict["masters"]):
             b.master.data.copy_(m.to(b.master.device))
         self._all_gather_params()

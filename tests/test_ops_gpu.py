@@ -165,3 +165,29 @@ def test_flash_attn_bwd(B, Hq, Hkv, S, causal):
     _cmp(q.grad, qr.grad * 0.5, atol=5e-2, rtol=5e-2, name="dq")
     _cmp(k.grad, kr.grad * 0.5, atol=5e-2, rtol=5e-2, name="dk")
     _cmp(v.grad, vr.grad * 0.5, atol=5e-2, rtol=5e-2, name="dv")
+
+
+def test_fused_adamw_matches_torch():
+    torch.manual_seed(7)
+    n = 10007
+    master = torch.randn(n, dtype=torch.float32, device="cuda")
+    m = torch.randn(n, device="cuda").abs() * 0.01
+    v = torch.randn(n, device="cuda").abs() * 0.001
+    grad = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+    param = torch.zeros(n, dtype=torch.bfloat16, device="cuda")
+    clip = torch.tensor([0.7], device="cuda")
+
+    mm, vv, ww = m.clone(), v.clone(), master.clone()
+    lr, b1, b2, eps, wd, t = 1e-3, 0.9, 0.95, 1e-8, 0.1, 3
+    g = grad.float() * clip
+    mm = b1 * mm + (1 - b1) * g
+    vv = b2 * vv + (1 - b2) * g * g
+    ww = ww * (1 - lr * wd)
+    ww = ww - lr * (mm / (1 - b1 ** t)) / ((vv / (1 - b2 ** t)).sqrt() + eps)
+
+    ops.adamw_step(master, m, v, grad, param, clip, lr, b1, b2, eps, wd, t)
+    torch.cuda.synchronize()
+    _cmp(master, ww, atol=1e-5, rtol=1e-4, name="fused adamw master")
+    _cmp(m, mm, atol=1e-6, name="fused adamw m")
+    _cmp(v, vv, atol=1e-7, name="fused adamw v")
+    _cmp(param, ww.to(torch.bfloat16), atol=1e-6, name="fused adamw param")
