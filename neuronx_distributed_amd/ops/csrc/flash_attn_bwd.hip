@@ -60,26 +60,79 @@ fa_bwd_delta_kernel(const short* __restrict__ dout,
 // stage a 32x128 bf16 tile row-major (XOR-swizzled) + transposed [128][32]
 // copy, using all 256 threads (2x16B pieces each for row-major; scatter b16
 // for the transpose).  row stride row-major: 256 B; transposed: 64 B.
+
+struct StageRegs { uint4v v0, v1; };
+
+__device__ __forceinline__ StageRegs load_tile32(const short* __restrict__ src,
+                                                 long src_row0,
+                                                 long src_stride,
+                                                 int rows_valid) {
+  int rp = (threadIdx.x & 255) >> 4;
+  int c16 = threadIdx.x & 15;
+  int r0 = 2 * rp, r1 = 2 * rp + 1;
+  int rr0 = r0 < rows_valid ? r0 : (rows_valid > 0 ? rows_valid - 1 : 0);
+  int rr1 = r1 < rows_valid ? r1 : (rows_valid > 0 ? rows_valid - 1 : 0);
+  StageRegs r;
+  r.v0 = *(const uint4v*)(src + (src_row0 + rr0) * src_stride + c16 * 8);
+  r.v1 = *(const uint4v*)(src + (src_row0 + rr1) * src_stride + c16 * 8);
+  return r;
+}
+
+__device__ __forceinline__ void write_tile32(StageRegs r, char* lds_rm,
+                                             char* lds_tr) {
+  if (blockDim.x > 256 && threadIdx.x >= 256) return;
+  int rp = threadIdx.x >> 4;
+  int c16 = threadIdx.x & 15;
+  int r0 = 2 * rp, r1 = 2 * rp + 1;
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz(r0, c16 * 16)) = r.v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz(r1, c16 * 16)) = r.v1;
+  union { uint4v u; short s[8]; } a, b;
+  a.u = r.v0; b.u = r.v1;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int d = c16 * 8 + j;
+    uint pair = ((uint)(unsigned short)a.s[j]) |
+                (((uint)(unsigned short)b.s[j]) << 16);
+    *(uint*)(lds_tr + d * 64 + swz32(d, r0 * 2)) = pair;
+  }
+}
+
+__device__ __forceinline__ void write_tile32_rm(StageRegs r, char* lds_rm) {
+  if (blockDim.x > 256 && threadIdx.x >= 256) return;
+  int rp = threadIdx.x >> 4;
+  int c16 = threadIdx.x & 15;
+  int r0 = 2 * rp, r1 = 2 * rp + 1;
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz(r0, c16 * 16)) = r.v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz(r1, c16 * 16)) = r.v1;
+}
+
 __device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
                                              long src_row0, long src_stride,
                                              int rows_valid, char* lds_rm,
                                              char* lds_tr) {
-  int tid = threadIdx.x;
+  if (blockDim.x > 256 && threadIdx.x >= 256) return;
+  // thread t owns rows {2rp, 2rp+1} at 16B slot c16 (rp = t>>4, c16 = t&15):
+  // row-major: 2 x b128 writes; transposed: 8 x b32 (two k cols per write).
+  // NOTE: swz32 flips byte bits 4-5; a b32 write at (k=2rp)*2 has the pair
+  // within one 4-byte word only when the two k's share the swizzled word,
+  // which holds because swz32 only permutes 16-byte groups.
+  int rp = threadIdx.x >> 4;
+  int c16 = threadIdx.x & 15;
+  int r0 = 2 * rp, r1 = 2 * rp + 1;
+  int rr0 = r0 < rows_valid ? r0 : (rows_valid > 0 ? rows_valid - 1 : 0);
+  int rr1 = r1 < rows_valid ? r1 : (rows_valid > 0 ? rows_valid - 1 : 0);
+  uint4v v0 = *(const uint4v*)(src + (src_row0 + rr0) * src_stride + c16 * 8);
+  uint4v v1 = *(const uint4v*)(src + (src_row0 + rr1) * src_stride + c16 * 8);
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz(r0, c16 * 16)) = v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz(r1, c16 * 16)) = v1;
+  union { uint4v u; short s[8]; } a, b;
+  a.u = v0; b.u = v1;
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    int p = tid + i * 256;          // 512 pieces: row p>>4, slot p&15
-    int row = p >> 4;
-    int c16 = p & 15;
-    int rr = row < rows_valid ? row : (rows_valid > 0 ? rows_valid - 1 : 0);
-    uint4v vv = *(const uint4v*)(src + (src_row0 + rr) * src_stride + c16 * 8);
-    *(uint4v*)(lds_rm + row * (FA_D * 2) + swz(row, c16 * 16)) = vv;
-    union { uint4v u; short s[8]; } u;
-    u.u = vv;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int d = c16 * 8 + j;
-      *(short*)(lds_tr + d * 64 + swz32(d, row * 2)) = u.s[j];
-    }
+  for (int j = 0; j < 8; ++j) {
+    int d = c16 * 8 + j;
+    uint pair = ((uint)(unsigned short)a.s[j]) |
+                (((uint)(unsigned short)b.s[j]) << 16);
+    *(uint*)(lds_tr + d * 64 + swz32(d, r0 * 2)) = pair;
   }
 }
 
@@ -102,10 +155,9 @@ __device__ __forceinline__ void write_acc_tile(char* tile, const f32x16& a,
 // contiguous) from a PW_PITCH tile
 __device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
                                               int col0) {
+  // (row*40 + col0)*2 with col0 in {0,8,16,24} is 16-byte aligned
   frag_u f;
-#pragma unroll
-  for (int j = 0; j < 8; ++j)
-    f.bf[j] = *(const __bf16*)(tile + (row * PW_PITCH + col0 + j) * 2);
+  f.u4 = *(const uint4v*)(tile + (row * PW_PITCH + col0) * 2);
   return f.bf;
 }
 
@@ -121,7 +173,7 @@ __device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
 #define BW_LDS_PW (4 * 32 * FA_D * 2)
 #define PW_BYTES (32 * PW_PITCH * 2)
 
-extern "C" __global__ void __launch_bounds__(256, 1)
+extern "C" __global__ void __launch_bounds__(512, 2)
 fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const short* __restrict__ vp,
                    const short* __restrict__ dop,
@@ -129,11 +181,17 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const float* __restrict__ deltap,
                    short* __restrict__ dkp, short* __restrict__ dvp,
                    int B, int Hq, int Hkv, int S, float scale, int causal) {
+  // 8 waves: wave w -> kv-row group (w>>1, 32 rows), d-half (w&1, 64 cols).
+  // The two waves of a pair recompute the same S^T/dP^T (16 extra mfmas)
+  // but halve the dK/dV accumulators, reaching 2 waves/SIMD instead of 1 -
+  // a net win since the kernel is latency-bound, not mfma-bound.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int col = lane & 31;
   const int hi = lane >> 5;
+  const int kvg = wid >> 1;        // kv row group 0..3
+  const int dhalf = wid & 1;       // d half 0..1
 
   const int kvblk = blockIdx.x;
   const int h = blockIdx.y;
@@ -143,10 +201,9 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
   const long lse_base = (long)(b * Hq + h) * S;
 
-  const int kv0 = kvblk * 128 + wid * 32;   // this wave's kv rows
-  const int my_k = kv0 + col;               // lane's kv row (for masks only)
+  const int kv0 = kvblk * 128 + kvg * 32;   // this wave's kv rows
+  const int my_k = kv0 + col;
 
-  // K,V fragments in registers: lane holds row (kv0 + col), d chunks
   frag_u kf[8], vf[8];
   {
     int row = my_k < S ? my_k : S - 1;
@@ -157,74 +214,67 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     }
   }
 
-  f32x16 dv_acc[4] = {};
-  f32x16 dk_acc[4] = {};
+  f32x16 dv_acc[2] = {};
+  f32x16 dk_acc[2] = {};
 
-  char* pw_p = smem + BW_LDS_PW + wid * 2 * PW_BYTES;        // P^T tile
-  char* pw_ds = pw_p + PW_BYTES;                             // dS^T tile
+  char* pw_p = smem + BW_LDS_PW + wid * 2 * PW_BYTES;
+  char* pw_ds = pw_p + PW_BYTES;
 
   const float s2 = scale * LOG2E;
-  int q_start = causal ? (kvblk * 128) / 32 * 32 : 0;
-  // NOTE: q_start aligned to the WG's first kv row (not per-wave) so all
-  // waves stay in the same barrier schedule.
+  int q_start = causal ? kvblk * 128 : 0;
 
   stage_tile32(qp + q_base, q_start, FA_D, S - q_start, smem + BW_LDS_Q,
                smem + BW_LDS_QT);
   stage_tile32(dop + q_base, q_start, FA_D, S - q_start, smem + BW_LDS_DO,
                smem + BW_LDS_DOT);
 
+  StageRegs nq, ndo;
   for (int q0 = q_start; q0 < S; q0 += 32) {
     __syncthreads();
-    const int my_q = q0 + col;             // lane's q (col axis)
+    if (q0 + 32 < S) {
+      nq = load_tile32(qp + q_base, q0 + 32, FA_D, S - q0 - 32);
+      ndo = load_tile32(dop + q_base, q0 + 32, FA_D, S - q0 - 32);
+    }
+    const int my_q = q0 + col;
     const bool wave_active = !causal || (q0 + 31 >= kv0);
 
     if (wave_active) {
       const float lse2 = lsep[lse_base + (my_q < S ? my_q : S - 1)] * LOG2E;
       const float dlt = deltap[lse_base + (my_q < S ? my_q : S - 1)];
 
-      // S^T[k][q] = sum_d K[k][d] Q^T[d][q]
       f32x16 st = {};
-#pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        frag_u qfr;
-        qfr.u4 = *(const uint4v*)(smem + BW_LDS_Q + col * (FA_D * 2)
-                                  + swz(col, (c * 16 + hi * 8) * 2));
-        st = mfma_bf16(kf[c].bf, qfr.bf, st);
-      }
-      // dP^T[k][q] = sum_d V[k][d] dO^T[d][q]
       f32x16 dpt = {};
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
-        frag_u dofr;
+        frag_u qfr, dofr;
+        qfr.u4 = *(const uint4v*)(smem + BW_LDS_Q + col * (FA_D * 2)
+                                  + swz(col, (c * 16 + hi * 8) * 2));
+        st = mfma_bf16(kf[c].bf, qfr.bf, st);
         dofr.u4 = *(const uint4v*)(smem + BW_LDS_DO + col * (FA_D * 2)
                                    + swz(col, (c * 16 + hi * 8) * 2));
         dpt = mfma_bf16(vf[c].bf, dofr.bf, dpt);
       }
 
-      // P^T = exp2(s*s2 - lse2); dS^T = P^T * (dP^T - delta)
-      f32x16 pt, dst;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int kg = kv0 + acc_row(r, hi);
         bool masked = (causal && kg > my_q) || kg >= S || my_q >= S;
         float p = masked ? 0.f : __builtin_exp2f(st[r] * s2 - lse2);
-        pt[r] = p;
-        dst[r] = p * (dpt[r] - dlt);
+        st[r] = p;
+        dpt[r] = p * (dpt[r] - dlt);
       }
-      write_acc_tile(pw_p, pt, lane);
-      write_acc_tile(pw_ds, dst, lane);
+      write_acc_tile(pw_p, st, lane);
+      write_acc_tile(pw_ds, dpt, lane);
 
-      // dV[k][d] += sum_q P^T[k][q] dO[q][d]
-      //   A = P^T rows k=col (b128 from pw_p), B = dO[q][d] via dOt rows d
+      // dV/dK over this wave's 64-column d-half
 #pragma unroll
       for (int cq = 0; cq < 2; ++cq) {
-        frag_u pa;
+        frag_u pa, da;
         pa.bf = read_pw_row(pw_p, col, cq * 16 + hi * 8);
-        frag_u da;
         da.bf = read_pw_row(pw_ds, col, cq * 16 + hi * 8);
 #pragma unroll
-        for (int nb = 0; nb < 4; ++nb) {
-          int d = nb * 32 + col;
+        for (int nb = 0; nb < 2; ++nb) {
+          int d = dhalf * 64 + nb * 32 + col;
           frag_u dofr, qfr;
           dofr.u4 = *(const uint4v*)(smem + BW_LDS_DOT + d * 64
                                      + swz32(d, (cq * 16 + hi * 8) * 2));
@@ -238,56 +288,53 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
     __syncthreads();
     if (q0 + 32 < S) {
-      stage_tile32(qp + q_base, q0 + 32, FA_D, S - q0 - 32, smem + BW_LDS_Q,
-                   smem + BW_LDS_QT);
-      stage_tile32(dop + q_base, q0 + 32, FA_D, S - q0 - 32,
-                   smem + BW_LDS_DO, smem + BW_LDS_DOT);
+      write_tile32(nq, smem + BW_LDS_Q, smem + BW_LDS_QT);
+      write_tile32(ndo, smem + BW_LDS_DO, smem + BW_LDS_DOT);
     }
   }
 
-  // ---- epilogue: dK/dV out (B,Hq,S,D) bf16; transpose via LDS ----------
-  // accumulators: col = d_local (n), rows k per reg.  Reuse smem tile:
-  // write [k][d] rows then vector-store.
+  // ---- epilogue: per-wave 32x64 halves via LDS transpose ---------------
   __syncthreads();
-  char* otile = smem + wid * (32 * FA_D * 2);  // per-wave 8KB scratch
+  char* otile = smem + wid * (32 * 64 * 2);  // 4 KB per wave
 #pragma unroll
-  for (int nb = 0; nb < 4; ++nb)
+  for (int nb = 0; nb < 2; ++nb)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int k = acc_row(r, hi);
-      int d = nb * 32 + col;
-      *(short*)(otile + (k * FA_D + d) * 2) = f2bits(dv_acc[nb][r] );
+      int dd = nb * 32 + col;  // within the 64-col half
+      *(short*)(otile + (k * 64 + dd) * 2) = f2bits(dv_acc[nb][r]);
     }
-  // wave-internal write->read; compiler orders via lgkmcnt on aliasing LDS
-  __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0) conservative
+  __builtin_amdgcn_s_waitcnt(0);
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {   // 64 lanes * 8 = 512 pieces of 16B
+  for (int i = 0; i < 4; ++i) {   // 64 lanes * 4 = 256 pieces of 16B
     int p = lane + i * 64;
-    int row = p >> 4, c16 = p & 15;
+    int row = p >> 3, c16 = p & 7;
     int kg = kv0 + row;
     if (kg < S) {
-      uint4v vv = *(const uint4v*)(otile + (row * FA_D + c16 * 8) * 2);
-      *(uint4v*)(dvp + ((long)(b * Hq + h) * S + kg) * FA_D + c16 * 8) = vv;
+      uint4v vv = *(const uint4v*)(otile + (row * 64 + c16 * 8) * 2);
+      *(uint4v*)(dvp + ((long)(b * Hq + h) * S + kg) * FA_D + dhalf * 64
+                 + c16 * 8) = vv;
     }
   }
   __builtin_amdgcn_s_waitcnt(0);
 #pragma unroll
-  for (int nb = 0; nb < 4; ++nb)
+  for (int nb = 0; nb < 2; ++nb)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int k = acc_row(r, hi);
-      int d = nb * 32 + col;
-      *(short*)(otile + (k * FA_D + d) * 2) = f2bits(dk_acc[nb][r] * scale);
+      int dd = nb * 32 + col;
+      *(short*)(otile + (k * 64 + dd) * 2) = f2bits(dk_acc[nb][r] * scale);
     }
   __builtin_amdgcn_s_waitcnt(0);
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < 4; ++i) {
     int p = lane + i * 64;
-    int row = p >> 4, c16 = p & 15;
+    int row = p >> 3, c16 = p & 7;
     int kg = kv0 + row;
     if (kg < S) {
-      uint4v vv = *(const uint4v*)(otile + (row * FA_D + c16 * 8) * 2);
-      *(uint4v*)(dkp + ((long)(b * Hq + h) * S + kg) * FA_D + c16 * 8) = vv;
+      uint4v vv = *(const uint4v*)(otile + (row * 64 + c16 * 8) * 2);
+      *(uint4v*)(dkp + ((long)(b * Hq + h) * S + kg) * FA_D + dhalf * 64
+                 + c16 * 8) = vv;
     }
   }
 }
@@ -343,21 +390,15 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const float s2 = scale * LOG2E;
 
   stage_tile32(kp + kv_base, 0, FA_D, S, smem + DQ_LDS_K, smem + DQ_LDS_KT);
-  {  // V row-major only: reuse stage but transposed target unused ->
-     // cheap variant: stage V rm with same piece mapping
-    int tid = threadIdx.x;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int p = tid + i * 256;
-      int row = p >> 4, c16 = p & 15;
-      int rr = row < S ? row : S - 1;
-      *(uint4v*)(smem + DQ_LDS_V + row * (FA_D * 2) + swz(row, c16 * 16)) =
-          *(const uint4v*)(vp + kv_base + (long)rr * FA_D + c16 * 8);
-    }
-  }
+  write_tile32_rm(load_tile32(vp + kv_base, 0, FA_D, S), smem + DQ_LDS_V);
 
+  StageRegs nk, nv;
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     __syncthreads();
+    if (kv0 + 32 < kv_end) {
+      nk = load_tile32(kp + kv_base, kv0 + 32, FA_D, S - kv0 - 32);
+      nv = load_tile32(vp + kv_base, kv0 + 32, FA_D, S - kv0 - 32);
+    }
     const bool wave_active = !causal || (kv0 <= q0 + 31);
 
     if (wave_active) {
@@ -413,17 +454,8 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
     __syncthreads();
     if (kv0 + 32 < kv_end) {
-      stage_tile32(kp + kv_base, kv0 + 32, FA_D, S - kv0 - 32,
-                   smem + DQ_LDS_K, smem + DQ_LDS_KT);
-      int tid = threadIdx.x;
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        int p = tid + i * 256;
-        int row = p >> 4, c16 = p & 15;
-        int rr = (kv0 + 32 + row) < S ? kv0 + 32 + row : S - 1;
-        *(uint4v*)(smem + DQ_LDS_V + row * (FA_D * 2) + swz(row, c16 * 16)) =
-            *(const uint4v*)(vp + kv_base + (long)rr * FA_D + c16 * 8);
-      }
+      write_tile32(nk, smem + DQ_LDS_K, smem + DQ_LDS_KT);
+      write_tile32_rm(nv, smem + DQ_LDS_V);
     }
   }
 
@@ -464,8 +496,8 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
                                               (const short*)out,
                                               (float*)delta, rows);
   dim3 gkv((S + 127) / 128, Hq, B);
-  size_t lds1 = 4 * 32 * FA_D * 2 + 4 * 2 * PW_BYTES;
-  fa_bwd_dkdv_kernel<<<gkv, 256, lds1, stream>>>(
+  size_t lds1 = 4 * 32 * FA_D * 2 + 8 * 2 * PW_BYTES;
+  fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
       Hkv, S, scale, causal);
