@@ -84,6 +84,10 @@ def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0):
                                        label_smoothing)
 
 
+# keep the collective-bearing loss opaque to the pipeline FX tracer
+torch.fx.wrap("parallel_cross_entropy")
+
+
 def from_parallel_logits_to_logprobs(vocab_parallel_logits, target,
                                      inference=False):
     """log p(target_{t+1} | context_t): shifts target left by one like the

@@ -1,0 +1,138 @@
+"""Pipeline P2P communication over RCCL/gloo.
+
+The reference implements PP sends as 2-rank all-gathers because XLA lacked
+real P2P (reference pipeline/comm.py:30-72); on MI355X we use genuine
+``torch.distributed`` send/recv over xGMI, batched with
+``batch_isend_irecv`` where both directions fly together (deadlock-free by
+construction; the reference's ordering rules scheduler.py:226-233 become
+unnecessary).
+
+Shape metadata travels as a small fixed-size header tensor before each
+payload (replacing the reference's TCPStore handshake, pipeline/comm.py:
+114-211) — eager P2P makes the out-of-band store unnecessary.
+"""
+
+import pickle
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..parallel import parallel_state as ps
+
+_MAX_DIMS = 8
+_DTYPES = [torch.float32, torch.float16, torch.bfloat16, torch.int64,
+           torch.int32, torch.bool, torch.float64, torch.uint8]
+
+
+def _header_from(tensor: torch.Tensor) -> torch.Tensor:
+    h = torch.zeros(2 + _MAX_DIMS, dtype=torch.int64)
+    h[0] = _DTYPES.index(tensor.dtype)
+    h[1] = tensor.dim()
+    for i, d in enumerate(tensor.shape):
+        h[2 + i] = d
+    return h
+
+
+def _tensor_from_header(h: torch.Tensor, device) -> torch.Tensor:
+    dtype = _DTYPES[int(h[0])]
+    dims = [int(x) for x in h[2:2 + int(h[1])]]
+    return torch.empty(dims, dtype=dtype, device=device)
+
+
+def _p2p_device():
+    if torch.cuda.is_available() and dist.get_backend() == "nccl":
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def _header_device():
+    # nccl needs device tensors even for the header
+    return _p2p_device()
+
+
+def send(tensors: List[torch.Tensor], dst: int):
+    """Send a list of tensors (shapes inline)."""
+    dev = _p2p_device()
+    n = torch.tensor([len(tensors)], dtype=torch.int64, device=_header_device())
+    dist.send(n, dst=dst)
+    for t in tensors:
+        h = _header_from(t).to(_header_device())
+        dist.send(h, dst=dst)
+        dist.send(t.detach().contiguous().to(dev), dst=dst)
+
+
+def recv_from(src: int) -> List[torch.Tensor]:
+    dev = _p2p_device()
+    n = torch.empty(1, dtype=torch.int64, device=_header_device())
+    dist.recv(n, src=src)
+    out = []
+    for _ in range(int(n.item())):
+        h = torch.empty(2 + _MAX_DIMS, dtype=torch.int64, device=_header_device())
+        dist.recv(h, src=src)
+        t = _tensor_from_header(h.cpu(), dev)
+        dist.recv(t, src=src)
+        out.append(t)
+    return out
+
+
+def send_recv(send_tensors: Optional[List[torch.Tensor]], dst: Optional[int],
+              recv_src: Optional[int]):
+    """Bidirectional exchange used by 1F1B steady state: both directions
+    batched so neither side blocks (send_forward_recv_backward etc.)."""
+    recv_out = None
+    if send_tensors is not None and recv_src is not None:
+        # Post sends async and NEVER wait on them before the matching
+        # peer's receives can be posted (waiting an isend whose completion
+        # needs the peer to progress past ITS sends deadlocks on ordered
+        # transports).  Only irecvs gate each phase; isends drain at the end.
+        dev = _p2p_device()
+        pending_sends = []
+        n = torch.tensor([len(send_tensors)], dtype=torch.int64,
+                         device=_header_device())
+        nr = torch.empty(1, dtype=torch.int64, device=_header_device())
+        pending_sends.append(dist.isend(n, dst))
+        dist.irecv(nr, recv_src).wait()
+
+        hs = [_header_from(t).to(_header_device()) for t in send_tensors]
+        send_payloads = [t.detach().contiguous().to(dev) for t in send_tensors]
+        for h in hs:
+            pending_sends.append(dist.isend(h, dst))
+        for t in send_payloads:
+            pending_sends.append(dist.isend(t, dst))
+
+        n_recv = int(nr.item())
+        hr = [torch.empty(2 + _MAX_DIMS, dtype=torch.int64,
+                          device=_header_device()) for _ in range(n_recv)]
+        for h in hr:
+            dist.irecv(h, recv_src).wait()
+        payload_recv = [_tensor_from_header(h.cpu(), dev) for h in hr]
+        for t in payload_recv:
+            dist.irecv(t, recv_src).wait()
+        for r in pending_sends:
+            r.wait()
+        recv_out = payload_recv
+    elif send_tensors is not None:
+        send(send_tensors, dst)
+    elif recv_src is not None:
+        recv_out = recv_from(recv_src)
+    return recv_out
+
+
+def send_python_object(obj, dst: int):
+    """Arbitrary python objects (reference pipeline/comm.py:114-211 metadata
+    channel)."""
+    data = pickle.dumps(obj)
+    buf = torch.frombuffer(bytearray(data), dtype=torch.uint8).clone()
+    n = torch.tensor([buf.numel()], dtype=torch.int64, device=_header_device())
+    dist.send(n, dst=dst)
+    dist.send(buf.to(_header_device()), dst=dst)
+
+
+def recv_python_object(src: int):
+    n = torch.empty(1, dtype=torch.int64, device=_header_device())
+    dist.recv(n, src=src)
+    buf = torch.empty(int(n.item()), dtype=torch.uint8,
+                      device=_header_device())
+    dist.recv(buf, src=src)
+    return pickle.loads(bytes(buf.cpu().numpy().tobytes()))

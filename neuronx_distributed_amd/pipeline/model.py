@@ -1,0 +1,322 @@
+"""NxDPPModel — the pipeline-parallel engine.
+
+Parity with reference ``pipeline/model.py`` (2,038 LoC): delayed FX
+trace/cut/partition (:549-670,946-966), task executor over the schedule
+streams (:1716-1743), microbatch handling (:1059-1091), loss processing
+(:1974-2028), ``local_*`` parameter namespaces (:1793-1922).
+
+MI355X-native differences: P2P is real RCCL send/recv with inline shape
+headers (pipeline/comm.py here) instead of the reference's 2-rank
+all-gather trick + TCPStore metadata; each task runs eagerly (no
+mark_step graph breaks)."""
+
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from ..parallel import comm as pcomm, parallel_state as ps
+from ..utils.logger import get_logger
+from . import comm as ppcomm
+from .manual_pipe_stage import PipelineStageModule
+from .partition import partition_model
+from .scheduler import (
+    BackwardStep,
+    ForwardStep,
+    InferenceSchedule,
+    RecvBackward,
+    RecvForward,
+    ReduceGrads,
+    SendBackward,
+    SendForward,
+    SendForwardRecvBackward,
+    Train1F1BSchedule,
+    TrainInterleavedSchedule,
+)
+
+logger = get_logger(__name__)
+
+
+class NxDPPModel(nn.Module):
+    def __init__(self, module: nn.Module,
+                 transformer_layer_cls=None,
+                 num_microbatches: int = 1,
+                 virtual_pipeline_size: int = 1,
+                 output_loss_value_spec: bool = True,
+                 pipeline_cuts: Optional[List[str]] = None,
+                 input_names: Optional[List[str]] = None,
+                 leaf_module_cls=(),
+                 manual_pp_partition: bool = False,
+                 manual_pp_stage_partition_fn=None,
+                 manual_pp_loss_fn=None,
+                 broadcast_and_average_loss: bool = True,
+                 auto_partition: bool = True,
+                 _debug_pre_partitioned=False):
+        super().__init__()
+        self.original_torch_module = module
+        self.transformer_layer_cls = transformer_layer_cls
+        self.num_microbatches = num_microbatches
+        self.virtual_pipeline_size = virtual_pipeline_size
+        self.output_loss_value_spec = output_loss_value_spec
+        self.pipeline_cuts = pipeline_cuts
+        self.input_names = input_names
+        self.leaf_module_cls = leaf_module_cls
+        self.broadcast_and_average_loss = broadcast_and_average_loss
+        self.manual_pp_loss_fn = manual_pp_loss_fn
+
+        self.pp_rank = ps.get_pipeline_model_parallel_rank()
+        self.pp_size = ps.get_pipeline_model_parallel_size()
+        self.next_rank = ps.get_pipeline_model_parallel_next_rank()
+        self.prev_rank = ps.get_pipeline_model_parallel_prev_rank()
+
+        self.partitioned = False
+        self.local_stage_module: Optional[nn.Module] = None
+        self._stage_specs = None
+
+        if isinstance(module, PipelineStageModule) or manual_pp_partition:
+            self.local_stage_module = module
+            self.partitioned = True
+
+    # ------------------------------------------------------------------
+    # partition
+    # ------------------------------------------------------------------
+    def _maybe_partition(self):
+        if self.partitioned:
+            return
+        split, stages = partition_model(
+            self.original_torch_module, self.pp_size,
+            pipeline_cuts=self.pipeline_cuts,
+            transformer_layer_cls=self.transformer_layer_cls,
+            input_names=self.input_names,
+            leaf_modules=self.leaf_module_cls)
+        self._analyze_stage_io(split)
+        self._split_root = split  # holds get_attr targets (params shared)
+        # defaults for concretized (non-input) forward args: FX keeps
+        # placeholder nodes for them (renamed with _N suffixes)
+        import inspect
+
+        self._arg_defaults = {
+            name: par.default
+            for name, par in inspect.signature(
+                self.original_torch_module.forward).parameters.items()
+            if par.default is not inspect.Parameter.empty
+        }
+        self.local_stage_module = stages[self.pp_rank]
+        self.partitioned = True
+
+    def _analyze_stage_io(self, split):
+        """From the split top-level graph, derive for every stage: which
+        placeholders come from original inputs and which from the previous
+        stage (reference partition.py:132-223 stage IO analysis)."""
+        specs = []
+        producer: Dict[Any, Tuple[int, int]] = {}  # node -> (stage, out_idx)
+        stage_idx = -1
+        stage_nouts: Dict[int, int] = {}
+        for node in split.graph.nodes:
+            if node.op == "call_module" and node.target.startswith("submod_"):
+                stage_idx = int(node.target.split("_")[1])
+                spec = []
+                for arg in node.args:
+                    if arg.op == "placeholder":
+                        spec.append(("input", arg.target))
+                    elif arg.op == "call_module":
+                        st, _ = producer[arg]
+                        if st != stage_idx - 1:
+                            raise NotImplementedError(
+                                "cross-stage skip connections not supported")
+                        spec.append(("prev", 0))
+                    elif arg.op == "call_function":  # getitem
+                        src = arg.args[0]
+                        idx = arg.args[1]
+                        st, _ = producer[src]
+                        if st != stage_idx - 1:
+                            raise NotImplementedError(
+                                "cross-stage skip connections not supported")
+                        spec.append(("prev", idx))
+                    elif arg.op == "get_attr":
+                        spec.append(("attr", arg.target))
+                    else:
+                        raise NotImplementedError(f"stage arg {arg.op}")
+                producer[node] = (stage_idx, 0)
+                specs.append(spec)
+        self._stage_specs = specs
+
+    def local_module(self):
+        self._maybe_partition()
+        return self.local_stage_module
+
+    def local_parameters(self):
+        self._maybe_partition()
+        return self.local_stage_module.parameters()
+
+    def local_named_parameters(self):
+        self._maybe_partition()
+        return self.local_stage_module.named_parameters()
+
+    def state_dict(self, *args, **kwargs):
+        self._maybe_partition()
+        return self.local_stage_module.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, sd, strict=True):
+        self._maybe_partition()
+        return self.local_stage_module.load_state_dict(sd, strict=strict)
+
+    # ------------------------------------------------------------------
+    # execution
+    # ------------------------------------------------------------------
+    def _split_microbatches(self, kwargs):
+        n = self.num_microbatches
+        mbs = [dict() for _ in range(n)]
+        for k, v in kwargs.items():
+            if isinstance(v, torch.Tensor):
+                assert v.shape[0] % n == 0, (
+                    f"batch dim {v.shape[0]} of '{k}' not divisible by "
+                    f"num_microbatches {n}")
+                for i, c in enumerate(v.chunk(n, dim=0)):
+                    mbs[i][k] = c
+            else:
+                for i in range(n):
+                    mbs[i][k] = v
+        return mbs
+
+    def _stage_forward(self, mb_kwargs, recvd: Optional[List[torch.Tensor]]):
+        if isinstance(self.local_stage_module, PipelineStageModule):
+            if self.pp_rank == 0:
+                args = [mb_kwargs[k] for k in sorted(mb_kwargs)] \
+                    if self.input_names is None else \
+                    [mb_kwargs[k] for k in self.input_names]
+                out = self.local_stage_module(*args)
+            else:
+                out = self.local_stage_module(*recvd)
+            return out
+        spec = self._stage_specs[self.pp_rank]
+        args = []
+        for kind, key in spec:
+            if kind == "input":
+                if key in mb_kwargs:
+                    args.append(mb_kwargs[key])
+                else:
+                    base = key.rsplit("_", 1)[0] \
+                        if key.rsplit("_", 1)[-1].isdigit() else key
+                    args.append(mb_kwargs.get(base,
+                                              self._arg_defaults.get(base)))
+            elif kind == "attr":
+                obj = self._split_root
+                for part in key.split("."):
+                    obj = getattr(obj, part)
+                args.append(obj)
+            else:
+                args.append(recvd[key])
+        return self.local_stage_module(*args)
+
+    @staticmethod
+    def _as_list(out):
+        if isinstance(out, (tuple, list)):
+            return list(out)
+        return [out]
+
+    def _run_schedule(self, schedule, kwargs, train: bool):
+        self._maybe_partition()
+        mbs = self._split_microbatches(kwargs)
+        recvd_inputs: Dict[int, List[torch.Tensor]] = {}
+        outputs: Dict[int, List[torch.Tensor]] = {}
+        losses: List[torch.Tensor] = []
+
+        for task in schedule.steps():
+            if isinstance(task, RecvForward):
+                tensors = ppcomm.recv_from(self.prev_rank)
+                for t in tensors:
+                    if t.is_floating_point():
+                        t.requires_grad_(True)
+                recvd_inputs[task.mb] = tensors
+            elif isinstance(task, ForwardStep):
+                with torch.enable_grad() if train else torch.no_grad():
+                    out = self._stage_forward(mbs[task.mb],
+                                              recvd_inputs.get(task.mb))
+                out_list = self._as_list(out)
+                outputs[task.mb] = out_list
+                if self.pp_rank == self.pp_size - 1 and \
+                        self.output_loss_value_spec:
+                    losses.append(out_list[0])
+            elif isinstance(task, SendForward):
+                ppcomm.send(outputs[task.mb], self.next_rank)
+            elif isinstance(task, SendForwardRecvBackward):
+                # both directions batched; grads arrive in mb order and
+                # attach to the oldest un-backwarded microbatch (FIFO)
+                grads = ppcomm.send_recv(outputs[task.mb], self.next_rank,
+                                         self.next_rank)
+                self._pending_grads.append(grads)
+            elif isinstance(task, RecvBackward):
+                self._pending_grads.append(ppcomm.recv_from(self.next_rank))
+            elif isinstance(task, BackwardStep):
+                mb = task.mb
+                if self.pp_rank == self.pp_size - 1 and \
+                        self.output_loss_value_spec:
+                    loss = outputs[mb][0]
+                    (loss / self.num_microbatches).backward()
+                else:
+                    grads = self._pending_grads.pop(0)
+                    # downstream sent one grad per FLOAT output, in order
+                    pairs = []
+                    gi = 0
+                    for t in outputs[mb]:
+                        if t.is_floating_point():
+                            if t.requires_grad:
+                                pairs.append((t, grads[gi]))
+                            gi += 1
+                    torch.autograd.backward([p[0] for p in pairs],
+                                            [p[1] for p in pairs])
+                # free the graph/output refs
+                outputs[mb] = [t.detach() for t in outputs[mb]]
+            elif isinstance(task, SendBackward):
+                mb = task.mb
+                grads = [t.grad if t.grad is not None
+                         else torch.zeros_like(t)
+                         for t in recvd_inputs[mb] if t.is_floating_point()]
+                ppcomm.send(grads, self.prev_rank)
+                del recvd_inputs[mb]
+            elif isinstance(task, ReduceGrads):
+                pass  # DP grad sync happens in the optimizer step
+        return losses
+
+    def run_train(self, **kwargs):
+        self.train()
+        self._pending_grads = []
+        schedule = Train1F1BSchedule(self.num_microbatches, self.pp_rank,
+                                     self.pp_size) \
+            if self.virtual_pipeline_size == 1 else \
+            TrainInterleavedSchedule(self.num_microbatches, self.pp_rank,
+                                     self.pp_size, self.virtual_pipeline_size)
+        losses = self._run_schedule(schedule, kwargs, train=True)
+        return self._process_loss(losses)
+
+    def run_eval(self, **kwargs):
+        self.eval()
+        self._pending_grads = []
+        schedule = InferenceSchedule(self.num_microbatches, self.pp_rank,
+                                     self.pp_size)
+        losses = self._run_schedule(schedule, kwargs, train=False)
+        if self.pp_rank == self.pp_size - 1:
+            if self.output_loss_value_spec:
+                return self._process_loss(losses)
+            return losses
+        return self._process_loss(losses) if self.output_loss_value_spec else None
+
+    def _process_loss(self, losses):
+        """Average microbatch losses; broadcast over PP (+ average over DP
+        when configured) — reference pipeline/model.py:1974-2028."""
+        device = torch.device("cuda", torch.cuda.current_device()) \
+            if torch.cuda.is_available() else torch.device("cpu")
+        if self.pp_rank == self.pp_size - 1 and losses:
+            loss = torch.stack([l.detach().float() for l in losses]).mean()
+            loss = loss.to(device)
+        else:
+            loss = torch.zeros((), dtype=torch.float32, device=device)
+        if self.broadcast_and_average_loss and self.pp_size > 1:
+            # all-reduce-as-broadcast over the PP group (only last stage
+            # contributes)
+            pcomm.all_reduce(loss, group=ps.get_group_info("pp"))
+        return loss
+
+    def forward(self, *args, **kwargs):
+        raise RuntimeError("NxDPPModel: use run_train()/run_eval()")

@@ -1,8 +1,10 @@
 """Multi-process gloo test harness: run a function under world_size ranks on
-CPU (the reference's NXD_CPU_MODE test vehicle)."""
+CPU (the reference's NXD_CPU_MODE test vehicle).  Hang-proof:
+a deadline kills stuck ranks and fails the test instead of blocking pytest."""
 
 import os
-import pickle
+import queue as pyqueue
+import time
 import traceback
 
 import torch
@@ -26,28 +28,40 @@ def _worker(rank, world_size, port, fn, args, q):
             torch.distributed.destroy_process_group()
 
 
-def run_distributed(fn, world_size=2, args=(), timeout=180):
+def run_distributed(fn, world_size=2, args=(), timeout=150):
     """Spawn world_size processes running fn(rank, world_size, *args);
-    returns list of per-rank results; raises on any rank error."""
+    returns list of per-rank results; raises on any rank error or hang."""
     import random
 
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
     port = random.randint(29600, 39999)
     procs = [
-        ctx.Process(target=_worker, args=(r, world_size, port, fn, args, q))
+        ctx.Process(target=_worker, args=(r, world_size, port, fn, args, q),
+                    daemon=True)
         for r in range(world_size)
     ]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(world_size):
-        rank, status, payload = q.get()
-        results[rank] = (status, payload)
+    deadline = time.time() + timeout
+    while len(results) < world_size and time.time() < deadline:
+        try:
+            rank, status, payload = q.get(timeout=1.0)
+            results[rank] = (status, payload)
+        except pyqueue.Empty:
+            if all(not p.is_alive() for p in procs) and q.empty():
+                break
     for p in procs:
-        p.join(timeout)
         if p.is_alive():
             p.terminate()
+    for p in procs:
+        p.join(10)
+    if len(results) < world_size:
+        missing = [r for r in range(world_size) if r not in results]
+        raise RuntimeError(
+            f"ranks {missing} hung or died without reporting; "
+            f"got: { {r: s for r, (s, _) in results.items()} }")
     errs = {r: p for r, (s, p) in results.items() if s == "err"}
     if errs:
         raise RuntimeError(f"rank failures: {errs}")

@@ -1,0 +1,147 @@
+"""Pipeline engine: schedule streams (pure python), FX partition on CPU,
+and a full pp=2 1F1B training step over gloo vs the non-PP golden."""
+
+import pytest
+import torch
+
+from dist_utils import run_distributed
+
+from neuronx_distributed_amd.pipeline.scheduler import (
+    Train1F1BSchedule, InferenceSchedule, ForwardStep, BackwardStep,
+    RecvForward, SendForward, RecvBackward, SendBackward,
+    SendForwardRecvBackward, ReduceGrads,
+)
+
+
+def _sched_counts(stream):
+    from collections import Counter
+
+    return Counter(type(t).__name__ for t in stream)
+
+
+def test_1f1b_stream_structure():
+    for n_mb, pp, rank in [(4, 2, 0), (4, 2, 1), (8, 4, 0), (8, 4, 2),
+                           (8, 4, 3), (2, 4, 1)]:
+        stream = list(Train1F1BSchedule(n_mb, rank, pp).steps())
+        c = _sched_counts(stream)
+        assert c["ForwardStep"] == n_mb
+        assert c["BackwardStep"] == n_mb
+        # every fwd before its own bwd
+        fwd_pos = {t.mb: i for i, t in enumerate(stream)
+                   if isinstance(t, ForwardStep)}
+        bwd_pos = {t.mb: i for i, t in enumerate(stream)
+                   if isinstance(t, BackwardStep)}
+        for mb in range(n_mb):
+            assert fwd_pos[mb] < bwd_pos[mb]
+        # first stage never receives fwd / sends bwd
+        if rank == 0:
+            assert c.get("RecvForward", 0) == 0
+            assert c.get("SendBackward", 0) == 0
+        if rank == pp - 1:
+            assert c.get("SendForward", 0) == 0
+            assert c.get("SendForwardRecvBackward", 0) == 0
+            assert c.get("RecvBackward", 0) == 0
+
+
+def test_1f1b_neighbor_consistency():
+    """Sends from stage s must match receives at s+1 in count."""
+    n_mb, pp = 6, 3
+    streams = [list(Train1F1BSchedule(n_mb, r, pp).steps()) for r in range(pp)]
+    for s in range(pp - 1):
+        sends = sum(1 for t in streams[s]
+                    if isinstance(t, (SendForward, SendForwardRecvBackward)))
+        recvs = sum(1 for t in streams[s + 1] if isinstance(t, RecvForward))
+        assert sends == recvs == n_mb
+        bsends = sum(1 for t in streams[s + 1] if isinstance(t, SendBackward))
+        brecvs = sum(1 for t in streams[s]
+                     if isinstance(t, (RecvBackward, SendForwardRecvBackward)))
+        assert bsends == brecvs == n_mb
+
+
+def test_inference_stream():
+    stream = list(InferenceSchedule(3, 1, 2).steps())
+    c = _sched_counts(stream)
+    assert c["ForwardStep"] == 3 and c["RecvForward"] == 3
+
+
+def _fx_partition_worker(rank, world):
+        from neuronx_distributed_amd.parallel import parallel_state as ps
+        from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+        from neuronx_distributed_amd.pipeline.partition import partition_model
+        from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+        torch.manual_seed(0)
+        model = LlamaForCausalLM(get_config("tiny"))
+        split, stages = partition_model(
+            model, 2, transformer_layer_cls=LlamaDecoderLayer,
+            input_names=["input_ids", "labels"])
+        assert len(stages) == 2
+        # stage params partition the model params
+        n0 = sum(p.numel() for p in stages[0].parameters())
+        n1 = sum(p.numel() for p in stages[1].parameters())
+        ntot = sum(p.numel() for p in model.parameters())
+        assert n0 + n1 == ntot
+        # stage0 output feeds stage1; run manually and compare with direct
+        torch.manual_seed(1)
+        x = torch.randint(0, 256, (2, 16))
+        ref = model(x, labels=x)
+        out = split(x, x)
+        assert torch.allclose(out, ref, atol=1e-6)
+        return True
+
+
+def test_fx_partition_tiny_llama():
+    """FX trace + split of the tiny Llama into 2 stages on CPU."""
+    run_distributed(_fx_partition_worker, world_size=1)
+
+
+def _pp2_train_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+    from neuronx_distributed_amd.pipeline import NxDPPModel
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 pipeline_model_parallel_size=world)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(get_config("tiny"))
+
+    # golden: same model, no PP, batch = all microbatches
+    torch.manual_seed(0)
+    golden = LlamaForCausalLM(get_config("tiny"))
+
+    pp_model = NxDPPModel(model, transformer_layer_cls=LlamaDecoderLayer,
+                          num_microbatches=4,
+                          input_names=["input_ids", "labels"])
+    torch.manual_seed(42)
+    x = torch.randint(0, 256, (8, 16))
+    loss = pp_model.run_train(input_ids=x, labels=x)
+
+    ref_loss = golden(x, labels=x)
+    # PP loss = mean over microbatch means == overall mean (equal sizes)
+    assert abs(loss.item() - ref_loss.item()) < 1e-4, (loss, ref_loss)
+
+    # gradient check: each stage's grads match the golden model's
+    ref_loss.backward()
+    golden_grads = {n: p.grad for n, p in golden.named_parameters()}
+    my_params = dict(pp_model.local_named_parameters())
+    matched = 0
+    for name, p in my_params.items():
+        if p.grad is None:
+            continue
+        # split_module prefixes differ; match by shape+values over candidates
+        for gn, gg in golden_grads.items():
+            if gg is not None and gg.shape == p.grad.shape and \
+                    torch.allclose(p.grad, gg, atol=2e-4):
+                matched += 1
+                break
+    assert matched >= len([p for p in my_params.values()
+                           if p.grad is not None]) * 0.9, \
+        f"only {matched} grads matched"
+    return loss.item()
+
+
+def test_pp2_train_matches_dense():
+    out = run_distributed(_pp2_train_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-6  # loss broadcast to all pp ranks
