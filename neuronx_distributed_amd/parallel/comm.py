@@ -132,8 +132,17 @@ def all_to_all(tensor: torch.Tensor, split_dim: int, concat_dim: int,
         return tensor
     assert tensor.shape[split_dim] % world == 0
     chunks = [c.contiguous() for c in tensor.chunk(world, dim=split_dim)]
-    outs = [torch.empty_like(c) for c in chunks]
-    dist.all_to_all(outs, chunks, group=g)
+    if _backend_is_gloo(g):
+        # gloo has no alltoall: emulate with all_gather + chunk select
+        rank = dist.get_rank(group=g)
+        full = tensor.contiguous()
+        gathered = [torch.empty_like(full) for _ in range(world)]
+        dist.all_gather(gathered, full, group=g)
+        outs = [gathered[r].chunk(world, dim=split_dim)[rank].contiguous()
+                for r in range(world)]
+    else:
+        outs = [torch.empty_like(c) for c in chunks]
+        dist.all_to_all(outs, chunks, group=g)
     return torch.cat(outs, dim=concat_dim)
 
 
