@@ -1,0 +1,44 @@
+"""Autoregressive generation loop (reference examples/inference runner +
+utils/sampling): prefill through the flash kernel, decode through the
+KV-cache path, sampling over vocab-parallel logits."""
+
+from typing import Optional
+
+import torch
+
+from ..utils.sampling import Sampler
+from .kv_cache import build_kv_caches
+
+
+@torch.no_grad()
+def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
+             sampler: Optional[Sampler] = None, eos_token_id: int = -1):
+    """model: LlamaForCausalLM-compatible (vocab-parallel logits out).
+    input_ids (B, S) on the model's device; returns (B, S+new)."""
+    from ..parallel import parallel_state as ps
+
+    sampler = sampler or Sampler(do_sample=False)
+    cfg = model.config
+    tp = ps.get_tensor_model_parallel_size()
+    B, S = input_ids.shape
+    kv_mult = max(1, tp // cfg.num_key_value_heads)
+    n_kv_local = cfg.num_key_value_heads * kv_mult // tp
+    caches = build_kv_caches(cfg.num_hidden_layers, B, n_kv_local,
+                             S + max_new_tokens, cfg.head_dim,
+                             device=input_ids.device)
+
+    # prefill
+    logits = model(input_ids, kv_caches=caches, pos_offset=0)
+    next_tok = sampler(logits[:, -1, :])
+    out = [input_ids, next_tok.unsqueeze(1)]
+    pos = S
+
+    for _ in range(max_new_tokens - 1):
+        step_in = next_tok.unsqueeze(1)
+        logits = model(step_in, kv_caches=caches, pos_offset=pos)
+        next_tok = sampler(logits[:, -1, :])
+        out.append(next_tok.unsqueeze(1))
+        pos += 1
+        if eos_token_id >= 0 and bool((next_tok == eos_token_id).all()):
+            break
+    return torch.cat(out, dim=1)

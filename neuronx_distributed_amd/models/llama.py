@@ -154,7 +154,7 @@ class LlamaAttention(nn.Module):
         v = v.transpose(1, 2)
         if kv_cache is not None:
             k, v = kv_cache.update(k, v, pos_offset)
-        out = flash_attn_func(q, k, v, causal=kv_cache is None)
+        out = flash_attn_func(q, k, v, causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
         if sp:
             out = out.transpose(0, 1)
