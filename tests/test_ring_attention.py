@@ -1,0 +1,73 @@
+"""Ring attention (CP) over gloo cp=2: forward and backward must match the
+single-rank full attention on the concatenated sequence."""
+
+import torch
+
+from dist_utils import run_distributed
+
+
+def _ring_worker(rank, world, causal):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.kernels.ring_attn import ring_attn_func
+    from neuronx_distributed_amd.kernels.flash_attn import _torch_reference
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 context_parallel_size=world)
+    torch.manual_seed(0)
+    B, H, S, D = 2, 4, 32, 128
+    q = torch.randn(B, H, S, D) * 0.5
+    k = torch.randn(B, H, S, D) * 0.5
+    v = torch.randn(B, H, S, D) * 0.5
+    Sl = S // world
+    ql = q[:, :, rank * Sl:(rank + 1) * Sl].clone().requires_grad_(True)
+    kl = k[:, :, rank * Sl:(rank + 1) * Sl].clone().requires_grad_(True)
+    vl = v[:, :, rank * Sl:(rank + 1) * Sl].clone().requires_grad_(True)
+
+    out = ring_attn_func(ql, kl, vl, causal=causal)
+
+    qf = q.clone().requires_grad_(True)
+    kf = k.clone().requires_grad_(True)
+    vf = v.clone().requires_grad_(True)
+    ref = _torch_reference(qf, kf, vf, causal=causal)
+    ref_l = ref[:, :, rank * Sl:(rank + 1) * Sl]
+    assert torch.allclose(out, ref_l, atol=1e-4), \
+        (out - ref_l).abs().max().item()
+
+    torch.manual_seed(7)
+    dy_full = torch.randn_like(ref)
+    out.backward(dy_full[:, :, rank * Sl:(rank + 1) * Sl])
+    ref.backward(dy_full)
+    for g, gf_full, name in ((ql.grad, qf.grad, "dq"), (kl.grad, kf.grad, "dk"),
+                             (vl.grad, vf.grad, "dv")):
+        gf = gf_full[:, :, rank * Sl:(rank + 1) * Sl]
+        assert torch.allclose(g, gf, atol=1e-4), \
+            f"{name}: {(g - gf).abs().max().item()}"
+    return True
+
+
+def test_ring_attention_causal():
+    run_distributed(_ring_worker, world_size=2, args=(True,))
+
+
+def test_ring_attention_full():
+    run_distributed(_ring_worker, world_size=2, args=(False,))
+
+
+def _batch_slice_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.utils.batch_utils import (
+        get_batch_on_this_context_parallel_rank)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 context_parallel_size=world)
+    batch = {"input_ids": torch.arange(16).reshape(1, 16),
+             "meta": "keep"}
+    sliced, off = get_batch_on_this_context_parallel_rank(batch)
+    assert sliced["input_ids"].shape == (1, 8)
+    assert off == rank * 8
+    assert sliced["input_ids"][0, 0].item() == rank * 8
+    return True
+
+
+def test_cp_batch_slicing():
+    run_distributed(_batch_slice_worker, world_size=2)
