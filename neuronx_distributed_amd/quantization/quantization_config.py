@@ -1,0 +1,47 @@
+"""Quantization configuration (reference quantization/quantization_config.py:
+25-35,100-125).
+
+CDNA4-native dtypes: INT8, FP8 OCP ``e4m3fn`` and ``e5m2`` (gfx950 uses the
+OCP formats, NOT MI300X's fnuz — cdna_hip_programming.md §4).  The
+reference's ±240 E4M3 clamp is a Neuron-hardware range limit and does not
+carry over."""
+
+import enum
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+
+class QuantizedDtype(enum.Enum):
+    INT8 = "int8"
+    F8E4M3 = "f8e4m3fn"   # OCP e4m3fn (gfx950-native MFMA input)
+    F8E5M2 = "f8e5m2"
+
+    @property
+    def torch_dtype(self):
+        return {
+            QuantizedDtype.INT8: torch.int8,
+            QuantizedDtype.F8E4M3: torch.float8_e4m3fn,
+            QuantizedDtype.F8E5M2: torch.float8_e5m2,
+        }[self]
+
+    @property
+    def max_value(self):
+        return {
+            QuantizedDtype.INT8: 127.0,
+            QuantizedDtype.F8E4M3: 448.0,
+            QuantizedDtype.F8E5M2: 57344.0,
+        }[self]
+
+
+class QuantizationType(enum.Enum):
+    PER_TENSOR_SYMMETRIC = "per_tensor_symmetric"
+    PER_CHANNEL_SYMMETRIC = "per_channel_symmetric"
+
+
+@dataclass
+class QuantizationConfig:
+    quantized_dtype: QuantizedDtype = QuantizedDtype.INT8
+    quantization_type: QuantizationType = QuantizationType.PER_CHANNEL_SYMMETRIC
+    quantization_per_channel_axis: int = 0

@@ -1,0 +1,107 @@
+"""Quantized tensor-parallel linears (reference
+quantization/quantization_layers.py:465,744): mirror Column/RowParallel
+config exactly (same shapes / partition dims :591-603) with the weight kept
+in the quantized dtype + scale parameters; the GEMM runs dequant->bf16
+hipBLASLt (CDNA4 fp8 MFMA GEMM path is a later optimization — the layout
+and scale plumbing here is the contract)."""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ..parallel import parallel_state as ps
+from ..parallel.layers import (
+    linear_with_async_allreduce,
+    BaseParallelLinear,
+)
+from ..parallel.mappings import (
+    copy_to_tensor_model_parallel_region,
+    gather_from_tensor_model_parallel_region,
+    reduce_from_tensor_model_parallel_region,
+    scatter_to_tensor_model_parallel_region,
+)
+from ..parallel.utils import divide, set_tensor_model_parallel_attributes
+from .quantization_config import (QuantizationConfig, QuantizationType,
+                                  QuantizedDtype)
+from .quantization_utils import dequantize, quantize_symmetric
+
+
+class _QuantizedParallelLinearBase(BaseParallelLinear):
+    def _make_scale(self, out_rows: int, cfg: QuantizationConfig, shard: bool):
+        if cfg.quantization_type == QuantizationType.PER_TENSOR_SYMMETRIC:
+            scale = nn.Parameter(torch.ones(1), requires_grad=False)
+        else:
+            scale = nn.Parameter(torch.ones(out_rows, 1), requires_grad=False)
+            if shard:
+                set_tensor_model_parallel_attributes(
+                    scale, ps.get_tensor_model_parallel_size() > 1, 0)
+        return scale
+
+    @classmethod
+    def from_float(cls, float_layer, quantization_config=None):
+        """Convert an existing Column/RowParallel layer (reference
+        quantize.convert per-module path)."""
+        cfg = quantization_config or QuantizationConfig()
+        layer = cls.__new__(cls)
+        BaseParallelLinear.__init__(layer)
+        layer.quantization_config = cfg
+        q, s = quantize_symmetric(float_layer.weight.detach(), cfg)
+        layer.weight = nn.Parameter(q, requires_grad=False)
+        for attr in ("tensor_model_parallel", "partition_dim",
+                     "partition_stride", "num_partitions"):
+            if hasattr(float_layer.weight, attr):
+                setattr(layer.weight, attr, getattr(float_layer.weight, attr))
+        layer.scale = nn.Parameter(
+            s if s.dim() else s.reshape(1), requires_grad=False)
+        layer.bias = float_layer.bias
+        layer._copy_cfg(float_layer)
+        return layer
+
+
+class QuantizedColumnParallel(_QuantizedParallelLinearBase):
+    def _copy_cfg(self, fl):
+        self.gather_output = fl.gather_output
+        self.sequence_parallel_enabled = fl.sequence_parallel_enabled
+        self.compute_dtype = fl.dtype
+
+    def forward(self, input_):
+        w = dequantize(self.weight, self.scale, self.compute_dtype)
+        if not self.sequence_parallel_enabled and \
+                ps.get_tensor_model_parallel_size() > 1:
+            input_parallel = input_
+            async_ar = True
+        else:
+            input_parallel = copy_to_tensor_model_parallel_region(input_)
+            async_ar = False
+        out = linear_with_async_allreduce(
+            input_parallel, w, self.bias, async_grad_allreduce=async_ar,
+            sequence_parallel_enabled=self.sequence_parallel_enabled)
+        if self.gather_output:
+            out = gather_from_tensor_model_parallel_region(out)
+        return out
+
+
+class QuantizedRowParallel(_QuantizedParallelLinearBase):
+    def _copy_cfg(self, fl):
+        self.input_is_parallel = fl.input_is_parallel
+        self.sequence_parallel_enabled = fl.sequence_parallel_enabled
+        self.compute_dtype = fl.dtype
+
+    def forward(self, input_):
+        w = dequantize(self.weight, self.scale, self.compute_dtype)
+        if not self.input_is_parallel:
+            input_ = scatter_to_tensor_model_parallel_region(input_)
+        out = linear_with_async_allreduce(
+            input_, w, None, async_grad_allreduce=False,
+            sequence_parallel_enabled=False)
+        if self.sequence_parallel_enabled:
+            from ..parallel.mappings import (
+                reduce_scatter_to_sequence_parallel_region)
+
+            out = reduce_scatter_to_sequence_parallel_region(out, seq_dim=0)
+        else:
+            out = reduce_from_tensor_model_parallel_region(out)
+        if self.bias is not None:
+            out = out + self.bias
+        return out

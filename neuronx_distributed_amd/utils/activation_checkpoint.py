@@ -11,10 +11,32 @@ import torch.nn as nn
 from torch.utils.checkpoint import checkpoint
 
 
+_PREFIX = "_checkpoint_wrapped_module."
+
+
 class CheckpointWrapper(nn.Module):
     def __init__(self, module: nn.Module):
         super().__init__()
         self._checkpoint_wrapped_module = module
+        # keep state-dict keys identical to the unwrapped module
+        self._register_state_dict_hook(self._strip_prefix_hook)
+        self._register_load_state_dict_pre_hook(self._add_prefix_hook,
+                                                with_module=True)
+
+    @staticmethod
+    def _strip_prefix_hook(module, state_dict, prefix, local_metadata):
+        for key in list(state_dict.keys()):
+            if key.startswith(prefix + _PREFIX):
+                new = prefix + key[len(prefix) + len(_PREFIX):]
+                state_dict[new] = state_dict.pop(key)
+        return state_dict
+
+    @staticmethod
+    def _add_prefix_hook(module, state_dict, prefix, *args):
+        for key in list(state_dict.keys()):
+            if key.startswith(prefix) and not key.startswith(prefix + _PREFIX):
+                new = prefix + _PREFIX + key[len(prefix):]
+                state_dict[new] = state_dict.pop(key)
 
     @property
     def module(self):
@@ -23,15 +45,6 @@ class CheckpointWrapper(nn.Module):
     def forward(self, *args, **kwargs):
         return checkpoint(self._checkpoint_wrapped_module, *args,
                           use_reentrant=False, **kwargs)
-
-    def named_parameters(self, *args, **kwargs):
-        return self._checkpoint_wrapped_module.named_parameters(*args, **kwargs)
-
-    def state_dict(self, *args, **kwargs):
-        return self._checkpoint_wrapped_module.state_dict(*args, **kwargs)
-
-    def load_state_dict(self, *args, **kwargs):
-        return self._checkpoint_wrapped_module.load_state_dict(*args, **kwargs)
 
 
 def apply_activation_checkpointing(model: nn.Module,
