@@ -74,6 +74,10 @@ CONFIGS = {
                               num_key_value_heads=8, vocab_size=128256,
                               rope_theta=500000.0,
                               max_position_embeddings=8192),
+    "test-d128": LlamaConfig(hidden_size=512, intermediate_size=1024,
+                             num_hidden_layers=2, num_attention_heads=4,
+                             num_key_value_heads=2, vocab_size=1024,
+                             max_position_embeddings=512),
     "tiny": LlamaConfig(hidden_size=64, intermediate_size=128,
                         num_hidden_layers=2, num_attention_heads=4,
                         num_key_value_heads=2, vocab_size=256,

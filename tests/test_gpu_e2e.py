@@ -1,0 +1,105 @@
+"""GPU end-to-end: tiny training step + KV-cache generation on MI355X with
+the HIP kernels mandatory (no torch fallbacks)."""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _dist():
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29766")
+    assert torch.cuda.is_available()
+    torch.cuda.set_device(0)
+    if not dist.is_initialized():
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    yield
+
+
+def _build(cfg_name="test-d128"):
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.bfloat16)
+    try:
+        with torch.device("cuda"):
+            model = LlamaForCausalLM(get_config(cfg_name))
+    finally:
+        torch.set_default_dtype(prev)
+    return model.cuda()
+
+
+def test_train_step_gpu():
+    from neuronx_distributed_amd import ops
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+
+    assert ops.is_available(), "HIP kernels must be loaded on GPU"
+    model = _build()
+    opt = NeuronZero1Optimizer(model.parameters(), torch.optim.AdamW, lr=1e-3)
+    x = torch.randint(0, 1024, (2, 256), device="cuda")
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        loss = model(x, labels=x)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(l == l for l in losses)
+    assert losses[-1] < losses[0], losses
+
+
+def test_generation_gpu():
+    from neuronx_distributed_amd.inference import generate
+    from neuronx_distributed_amd.utils.sampling import Sampler
+    from neuronx_distributed_amd.operators import argmax as dargmax
+
+    model = _build().eval()
+    torch.manual_seed(3)
+    x = torch.randint(0, 1024, (2, 256), device="cuda")
+    with torch.no_grad():
+        out = generate(model, x, max_new_tokens=8)
+        assert out.shape == (2, 264)
+        # golden greedy: full re-forward each step
+        cur = x.clone()
+        for _ in range(8):
+            logits = model(cur)
+            nxt = dargmax(logits[:, -1, :], dim=-1, gather_dim=-1)
+            cur = torch.cat([cur, nxt.unsqueeze(1)], dim=1)
+    # bf16 prefill kernel vs decode bmm may differ on near-ties; require
+    # the large majority of tokens to agree
+    agree = (out == cur).float().mean().item()
+    assert agree > 0.95, agree
+
+
+def test_sp_gpu_matches_dense():
+    """Sequence-parallel path on GPU (world 1 -> SP is identity comms but
+    exercises the (S,B,H) layout + kernels)."""
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.bfloat16)
+    try:
+        with torch.device("cuda"):
+            dense = LlamaForCausalLM(get_config("test-d128"))
+        torch.manual_seed(0)
+        with torch.device("cuda"):
+            sp = LlamaForCausalLM(get_config("test-d128",
+                                             sequence_parallel_enabled=True))
+    finally:
+        torch.set_default_dtype(prev)
+    x = torch.randint(0, 1024, (2, 256), device="cuda")
+    l1 = dense.cuda()(x, labels=x)
+    l2 = sp.cuda()(x, labels=x)
+    assert abs(l1.item() - l2.item()) < 5e-2, (l1.item(), l2.item())

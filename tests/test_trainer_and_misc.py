@@ -157,3 +157,39 @@ def test_tensor_capture():
     cap = get_captured_tensors()
     assert "0" in cap and cap["0"].shape == (2, 4)
     disable_tensor_capture()
+
+
+def test_checkpoint_converter_roundtrip():
+    from neuronx_distributed_amd.scripts import CheckpointConverterBase
+
+    torch.manual_seed(0)
+    full = {
+        "model.layers.0.self_attn.q_proj.weight": torch.randn(16, 8),
+        "model.layers.0.self_attn.o_proj.weight": torch.randn(8, 16),
+        "model.layers.0.mlp.gate_up_proj.weight": torch.randn(32, 8),
+        "model.norm.weight": torch.randn(8),
+    }
+    c = CheckpointConverterBase()
+    shards = c.shard_full_checkpoint(full, tp_degree=2)
+    assert shards[0]["model.layers.0.self_attn.q_proj.weight"].shape == (8, 8)
+    assert shards[0]["model.layers.0.self_attn.o_proj.weight"].shape == (8, 8)
+    assert shards[0]["model.layers.0.mlp.gate_up_proj.weight"].shape == (16, 8)
+    assert shards[0]["model.norm.weight"].shape == (8,)
+    merged = c.merge_sharded_checkpoints(shards)
+    for k in full:
+        assert torch.equal(merged[k], full[k]), k
+
+
+def test_top_level_api_surface():
+    import neuronx_distributed_amd as nxd
+
+    for attr in ("parallel_layers", "pipeline", "kernels", "utils",
+                 "ModelBuilder", "NxDModel", "shard_checkpoint",
+                 "NxDParallelState", "neuronx_distributed_config",
+                 "initialize_parallel_model", "initialize_parallel_optimizer",
+                 "save_checkpoint", "load_checkpoint"):
+        assert hasattr(nxd, attr), attr
+    from neuronx_distributed_amd.parallel_layers import (
+        ColumnParallelLinear, RowParallelLinear, ParallelEmbedding,
+        parallel_cross_entropy, initialize_model_parallel, clip_grad_norm,
+        PARALLEL_MODULES, PARALLEL_FUNCTIONS)
