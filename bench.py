@@ -44,6 +44,10 @@ def main():
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29777")
 
+    # synthetic random-init bench: shard-local fast init (skips the
+    # TP-invariant master-weight protocol, irrelevant for synthetic data)
+    os.environ.setdefault("NXDA_FAST_INIT", "1")
+
     on_gpu = torch.cuda.is_available()
     if on_gpu:
         torch.cuda.set_device(local_rank)

@@ -19,6 +19,7 @@ MI355X design notes:
 """
 
 import math
+import os
 from typing import Callable, Optional
 
 import torch
@@ -77,6 +78,16 @@ def _initialize_affine_weight(
                                          world)
     if ps.is_aot_mode():
         return None  # AOT tracing skips weight init (reference layers.py:138-140)
+
+    if os.environ.get("NXDA_FAST_INIT", "0") == "1":
+        # bench/synthetic mode: init the SHARD directly (per-rank RNG, not
+        # TP-degree-invariant) — skips the full master-weight materialization
+        with torch.no_grad():
+            tmp = torch.empty(weight.shape, dtype=torch.float32,
+                              device=weight.device)
+            init_method(tmp)
+            weight.data.copy_(tmp.to(weight.dtype))
+        return None
 
     master = torch.empty(out_features, in_features, dtype=torch.float32,
                          device="cpu", requires_grad=False)

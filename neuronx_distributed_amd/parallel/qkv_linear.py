@@ -181,6 +181,16 @@ class GQAQKVColumnParallelLinear(BaseParallelLinear):
     def _deterministic_init(self, input_size, q_out, kv_out, init_method):
         if self.weight_q.device.type == "meta":
             return
+        import os
+
+        if os.environ.get("NXDA_FAST_INIT", "0") == "1":
+            with torch.no_grad():
+                for w in (self.weight_q, self.weight_k, self.weight_v):
+                    tmp = torch.empty(w.shape, dtype=torch.float32,
+                                      device=w.device)
+                    init_method(tmp)
+                    w.data.copy_(tmp.to(self.dtype))
+            return
         world = ps.get_tensor_model_parallel_size()
         rank = ps.get_tensor_model_parallel_rank()
         with torch.no_grad():
