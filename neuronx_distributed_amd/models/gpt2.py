@@ -136,6 +136,6 @@ class GPT2LMHeadModel(nn.Module):
         logits = self.lm_head(x)
         if labels is None:
             return logits
-        loss = parallel_cross_entropy(logits[:, :-1, :].float(),
+        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(),
                                       labels[:, 1:].contiguous())
         return loss.mean()

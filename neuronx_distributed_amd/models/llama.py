@@ -252,5 +252,5 @@ class LlamaForCausalLM(nn.Module):
         if labels is None:
             return logits
         loss = parallel_cross_entropy(
-            logits[:, :-1, :].float(), labels[:, 1:].contiguous())
+            logits[:, :-1, :].contiguous(), labels[:, 1:].contiguous())
         return loss.mean()

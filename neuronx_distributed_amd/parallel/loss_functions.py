@@ -11,11 +11,64 @@ HIP kernel (ops/cross_entropy) to avoid materializing exp(logits) extra
 passes; the fallback below is plain torch and is the CPU reference.
 """
 
+import ctypes
+
 import torch
 import torch.distributed as dist
 
 from . import comm
 from . import parallel_state as ps
+
+
+class _FusedParallelCrossEntropy(torch.autograd.Function):
+    """HIP-fused path (ops/csrc/cross_entropy.hip): bf16 logits shard, no
+    fp32/softmax materialization; backward recomputes exp."""
+
+    @staticmethod
+    def forward(ctx, vocab_parallel_logits, target):
+        from .. import ops
+
+        lib = ops._require_lib()
+        tp = ps.get_group_info("tp")
+        tp_rank = comm.group_rank(tp)
+        V = vocab_parallel_logits.shape[-1]
+        logits = vocab_parallel_logits.contiguous()
+        N = logits.numel() // V
+        dev = logits.device
+        t64 = target.reshape(-1).to(torch.int64).contiguous()
+
+        rowmax = torch.empty(N, dtype=torch.float32, device=dev)
+        lib.ce_rowmax(ops._ptr(logits), ops._ptr(rowmax), ctypes.c_long(N),
+                      ctypes.c_int(V), ops._stream())
+        comm.all_reduce(rowmax, op=dist.ReduceOp.MAX, group=tp)
+
+        se_pred = torch.empty(2, N, dtype=torch.float32, device=dev)
+        lib.ce_sumexp(ops._ptr(logits), ops._ptr(rowmax), ops._ptr(t64),
+                      ops._ptr(se_pred[0]), ops._ptr(se_pred[1]),
+                      ctypes.c_long(N), ctypes.c_int(V),
+                      ctypes.c_long(tp_rank * V), ops._stream())
+        comm.all_reduce(se_pred, group=tp)
+        sumexp, predicted = se_pred[0], se_pred[1]
+        loss = (sumexp.log() - predicted).reshape(target.shape)
+        ctx.save_for_backward(logits, rowmax, sumexp, t64)
+        ctx.vocab_start = tp_rank * V
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        from .. import ops
+
+        lib = ops._require_lib()
+        logits, rowmax, sumexp, t64 = ctx.saved_tensors
+        V = logits.shape[-1]
+        N = logits.numel() // V
+        g = grad_output.reshape(-1).to(torch.float32).contiguous()
+        dlogits = torch.empty_like(logits)
+        lib.ce_bwd(ops._ptr(logits), ops._ptr(rowmax), ops._ptr(sumexp),
+                   ops._ptr(t64), ops._ptr(g), ops._ptr(dlogits),
+                   ctypes.c_long(N), ctypes.c_int(V),
+                   ctypes.c_long(ctx.vocab_start), ops._stream())
+        return dlogits, None
 
 
 class _ParallelCrossEntropy(torch.autograd.Function):
@@ -79,7 +132,19 @@ class _ParallelCrossEntropy(torch.autograd.Function):
 
 
 def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0):
-    """Per-token loss on vocab-sharded logits (reference loss_functions.py:217)."""
+    """Per-token loss on vocab-sharded logits (reference loss_functions.py:217).
+    GPU bf16 + no smoothing dispatches to the fused HIP kernels."""
+    if (vocab_parallel_logits.is_cuda
+            and vocab_parallel_logits.dtype == torch.bfloat16
+            and label_smoothing == 0.0
+            and vocab_parallel_logits.shape[-1] % 8 == 0):
+        from .. import ops
+
+        if ops.is_available():
+            return _FusedParallelCrossEntropy.apply(vocab_parallel_logits,
+                                                    target)
+    if vocab_parallel_logits.dtype in (torch.bfloat16, torch.float16):
+        vocab_parallel_logits = vocab_parallel_logits.float()
     return _ParallelCrossEntropy.apply(vocab_parallel_logits, target,
                                        label_smoothing)
 

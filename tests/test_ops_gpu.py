@@ -191,3 +191,35 @@ def test_fused_adamw_matches_torch():
     _cmp(m, mm, atol=1e-6, name="fused adamw m")
     _cmp(v, vv, atol=1e-7, name="fused adamw v")
     _cmp(param, ww.to(torch.bfloat16), atol=1e-6, name="fused adamw param")
+
+
+def test_fused_cross_entropy():
+    import torch.distributed as dist
+    import os
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29755")
+    if not dist.is_initialized():
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.loss_functions import (
+        parallel_cross_entropy)
+
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    B, S, V = 2, 64, 1024
+    logits = (torch.randn(B, S, V, device="cuda") * 2).to(torch.bfloat16)
+    logits.requires_grad_(True)
+    target = torch.randint(0, V, (B, S), device="cuda")
+    loss = parallel_cross_entropy(logits, target)
+
+    ref_in = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(
+        ref_in.reshape(-1, V), target.reshape(-1),
+        reduction="none").reshape(B, S)
+    _cmp(loss, ref, atol=1e-2, name="fused CE fwd")
+    g = torch.randn(B, S, device="cuda")
+    loss.backward(g)
+    ref.backward(g)
+    _cmp(logits.grad, ref_in.grad, atol=2e-3, rtol=5e-2, name="fused CE bwd")
