@@ -139,9 +139,11 @@ class _RoPEFn(torch.autograd.Function):
             lib = _require_lib()
             B, S, Hq, D = q.shape
             Hk = k.shape[2]
-            qo = q.contiguous().clone()
-            ko = k.contiguous().clone()
-            lib.rope_fwd(_ptr(qo), _ptr(ko), _ptr(cos), _ptr(sin),
+            qc, kc = q.contiguous(), k.contiguous()
+            qo = torch.empty_like(qc)
+            ko = torch.empty_like(kc)
+            lib.rope_fwd(_ptr(qc), _ptr(kc), _ptr(qo), _ptr(ko), _ptr(cos),
+                         _ptr(sin),
                          ctypes.c_int(B), ctypes.c_int(S), ctypes.c_int(Hq),
                          ctypes.c_int(Hk), ctypes.c_int(D),
                          ctypes.c_int(pos_offset), ctypes.c_int(0), _stream())
@@ -159,9 +161,11 @@ class _RoPEFn(torch.autograd.Function):
             lib = _require_lib()
             B, S, Hq, D = dq.shape
             Hk = dk.shape[2]
-            dqo = dq.contiguous().clone()
-            dko = dk.contiguous().clone()
-            lib.rope_fwd(_ptr(dqo), _ptr(dko), _ptr(cos), _ptr(sin),
+            dqc, dkc = dq.contiguous(), dk.contiguous()
+            dqo = torch.empty_like(dqc)
+            dko = torch.empty_like(dkc)
+            lib.rope_fwd(_ptr(dqc), _ptr(dkc), _ptr(dqo), _ptr(dko),
+                         _ptr(cos), _ptr(sin),
                          ctypes.c_int(B), ctypes.c_int(S), ctypes.c_int(Hq),
                          ctypes.c_int(Hk), ctypes.c_int(D), ctypes.c_int(off),
                          ctypes.c_int(1), _stream())

@@ -12,10 +12,12 @@
 #include "common.h"
 
 extern "C" __global__ void __launch_bounds__(256)
-rope_kernel(short* __restrict__ q, short* __restrict__ k,
+rope_kernel(const short* __restrict__ q, const short* __restrict__ k,
+            short* __restrict__ qo, short* __restrict__ ko,
             const float* __restrict__ cos_t, const float* __restrict__ sin_t,
             int B, int S, int Hq, int Hk, int D, float sign, int pos_offset) {
-  // one wave handles one (b, s, h) row's D elements; 4 waves/block
+  // out-of-place (halves the traffic vs clone+inplace); one wave per
+  // (b, s, h) row; 4 waves/block
   long total = (long)B * S * (Hq + Hk);
   int half = D >> 1;
   for (long row = blockIdx.x * 4 + (threadIdx.x >> 6); row < total;
@@ -24,29 +26,37 @@ rope_kernel(short* __restrict__ q, short* __restrict__ k,
     long bs = row / (Hq + Hk);
     int h = (int)(row % (Hq + Hk));
     int s = (int)(bs % S);
-    short* base = (h < Hq)
-        ? q + ((bs * Hq + h) * (long)D)
-        : k + ((bs * Hk + (h - Hq)) * (long)D);
+    const short* src;
+    short* dst;
+    if (h < Hq) {
+      src = q + ((bs * Hq + h) * (long)D);
+      dst = qo + ((bs * Hq + h) * (long)D);
+    } else {
+      src = k + ((bs * Hk + (h - Hq)) * (long)D);
+      dst = ko + ((bs * Hk + (h - Hq)) * (long)D);
+    }
     const float* cr = cos_t + (long)(s + pos_offset) * half;
     const float* sr = sin_t + (long)(s + pos_offset) * half;
     for (int d = lane; d < half; d += 64) {
       float c = cr[d];
       float sn = sr[d] * sign;
-      float x0 = bits2f(base[d]);
-      float x1 = bits2f(base[d + half]);
-      base[d] = f2bits(x0 * c - x1 * sn);
-      base[d + half] = f2bits(x1 * c + x0 * sn);
+      float x0 = bits2f(src[d]);
+      float x1 = bits2f(src[d + half]);
+      dst[d] = f2bits(x0 * c - x1 * sn);
+      dst[d + half] = f2bits(x1 * c + x0 * sn);
     }
   }
 }
 
-extern "C" void rope_fwd(void* q, void* k, const void* cos_t, const void* sin_t,
+extern "C" void rope_fwd(const void* q, const void* k, void* qo, void* ko,
+                         const void* cos_t, const void* sin_t,
                          int B, int S, int Hq, int Hk, int D, int pos_offset,
                          int backward, hipStream_t stream) {
   long total = (long)B * S * (Hq + Hk);
   int blocks = (int)((total + 3) / 4);
   if (blocks > 4096) blocks = 4096;
   rope_kernel<<<blocks, 256, 0, stream>>>(
-      (short*)q, (short*)k, (const float*)cos_t, (const float*)sin_t, B, S, Hq,
+      (const short*)q, (const short*)k, (short*)qo, (short*)ko,
+      (const float*)cos_t, (const float*)sin_t, B, S, Hq,
       Hk, D, backward ? -1.f : 1.f, pos_offset);
 }

@@ -201,10 +201,13 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
         works = []
         for b in self.buckets:
             world = b.group_info.size
-            b.flat_grad.mul_(1.0 / world)  # grad mean over the shard group
             if world == 1:
-                b.grad_shard.copy_(b.flat_grad[b.shard_lo:b.shard_hi])
-            elif comm._backend_is_gloo(b.group_info.group):
+                # single-rank shard group: the shard IS the flat grad —
+                # no scale, no copy
+                b.grad_shard = b.flat_grad
+                continue
+            b.flat_grad.mul_(1.0 / world)  # grad mean over the shard group
+            if comm._backend_is_gloo(b.group_info.group):
                 dist.all_reduce(b.flat_grad, group=b.group_info.group)
                 b.grad_shard.copy_(b.flat_grad[b.shard_lo:b.shard_hi])
             else:
