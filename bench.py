@@ -85,7 +85,7 @@ def main():
                                lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
                                grad_clipping=True, max_norm=1.0)
 
-    B = args.batch or max(1, world)
+    B = args.batch or 2 * max(1, world)
     S = args.seq
     mbs = args.microbatch
     assert B % mbs == 0

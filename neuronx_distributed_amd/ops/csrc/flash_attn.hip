@@ -29,9 +29,13 @@
 #define LOG2E 1.4426950408889634f
 #define NEG_INF (-1e30f)
 
-// LDS: double-buffered K [64][128] and Vt [128][64], both bf16, XOR-swizzled
+// LDS: double-buffered K [64][128] bf16 (256B rows, swz16 = conflict-free
+// b128 reads) and Vt [128][72] bf16 (144B padded rows: 16B-aligned, and
+// 36*d = 4*(9d mod 16) banks are distinct within every b128 lane group
+// since 9 is coprime to 16 -> conflict-free, no swizzle needed)
 #define K_TILE_B (FA_KV * FA_D * 2)    // 16 KB
-#define VT_TILE_B (FA_D * FA_KV * 2)   // 16 KB
+#define VT_PITCH 72
+#define VT_TILE_B (FA_D * VT_PITCH * 2)  // 18 KB
 
 extern "C" __global__ void __launch_bounds__(512, 2)
 flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
@@ -111,15 +115,15 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       // K row-major [64][128], swizzled 16B slots
-      *(uint4v*)(kbuf(buf) + swz(st_row[i], st_col[i] * 16)
+      *(uint4v*)(kbuf(buf) + swz16(st_row[i], st_col[i] * 16)
                  + st_row[i] * (FA_D * 2)) = kreg[i];
-      // Vt transposed [128][64]: element j -> row d = col*8+j, col k = row
+      // Vt transposed [128][VT_PITCH]: element j -> row d = col*8+j, col k
       union { uint4v u; short s[8]; } vv;
       vv.u = vreg[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int d = st_col[i] * 8 + j;
-        *(short*)(vbuf(buf) + d * (FA_KV * 2) + swz(d, st_row[i] * 2)) = vv.s[j];
+        *(short*)(vbuf(buf) + d * (VT_PITCH * 2) + st_row[i] * 2) = vv.s[j];
       }
     }
   };
@@ -144,7 +148,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           int row = col + 32 * kb;
           frag_u kf;
           kf.u4 = *(const uint4v*)(kbuf(cur) + row * (FA_D * 2)
-                                   + swz(row, (c * 16 + hi * 8) * 2));
+                                   + swz16(row, (c * 16 + hi * 8) * 2));
           acc[kb] = mfma_bf16(kf.bf, qf[c].bf, acc[kb]);
         }
       }
@@ -224,8 +228,8 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         for (int c16 = 0; c16 < 4; ++c16) {
           int d = 32 * nb + col;
           frag_u vf;
-          vf.u4 = *(const uint4v*)(vbuf(cur) + d * (FA_KV * 2)
-                                   + swz(d, (c16 * 16 + hi * 8) * 2));
+          vf.u4 = *(const uint4v*)(vbuf(cur) + d * (VT_PITCH * 2)
+                                   + (c16 * 16 + hi * 8) * 2);
           ot[nb] = mfma_bf16(vf.bf, pf[c16].bf, ot[nb]);
         }
       }

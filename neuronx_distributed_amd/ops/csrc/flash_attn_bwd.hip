@@ -21,10 +21,10 @@
 #include "common.h"
 #include "mfma.h"
 
-// swizzle for 64-byte-row tiles (transposed [128][32] bf16): only bits 4-5
-__device__ __forceinline__ int swz32(int row, int byte_in_row) {
-  return byte_in_row ^ ((row & 3) << 4);
-}
+// transposed tiles use a PADDED pitch instead of a swizzle: 80-byte rows
+// (40 bf16) are 16B-aligned and put row d on bank 4*(5d mod 16) -> all
+// distinct within a b128 lane group (5 coprime to 16) -> conflict-free.
+#define TR_PITCH 40
 
 #define FA_D 128
 #define LOG2E 1.4426950408889634f
@@ -84,8 +84,8 @@ __device__ __forceinline__ void write_tile32(StageRegs r, char* lds_rm,
   int rp = threadIdx.x >> 4;
   int c16 = threadIdx.x & 15;
   int r0 = 2 * rp, r1 = 2 * rp + 1;
-  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz(r0, c16 * 16)) = r.v0;
-  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz(r1, c16 * 16)) = r.v1;
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz16(r0, c16 * 16)) = r.v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz16(r1, c16 * 16)) = r.v1;
   union { uint4v u; short s[8]; } a, b;
   a.u = r.v0; b.u = r.v1;
 #pragma unroll
@@ -93,7 +93,7 @@ __device__ __forceinline__ void write_tile32(StageRegs r, char* lds_rm,
     int d = c16 * 8 + j;
     uint pair = ((uint)(unsigned short)a.s[j]) |
                 (((uint)(unsigned short)b.s[j]) << 16);
-    *(uint*)(lds_tr + d * 64 + swz32(d, r0 * 2)) = pair;
+    *(uint*)(lds_tr + d * (TR_PITCH * 2) + r0 * 2) = pair;
   }
 }
 
@@ -102,8 +102,8 @@ __device__ __forceinline__ void write_tile32_rm(StageRegs r, char* lds_rm) {
   int rp = threadIdx.x >> 4;
   int c16 = threadIdx.x & 15;
   int r0 = 2 * rp, r1 = 2 * rp + 1;
-  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz(r0, c16 * 16)) = r.v0;
-  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz(r1, c16 * 16)) = r.v1;
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz16(r0, c16 * 16)) = r.v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz16(r1, c16 * 16)) = r.v1;
 }
 
 __device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
@@ -123,8 +123,8 @@ __device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
   int rr1 = r1 < rows_valid ? r1 : (rows_valid > 0 ? rows_valid - 1 : 0);
   uint4v v0 = *(const uint4v*)(src + (src_row0 + rr0) * src_stride + c16 * 8);
   uint4v v1 = *(const uint4v*)(src + (src_row0 + rr1) * src_stride + c16 * 8);
-  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz(r0, c16 * 16)) = v0;
-  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz(r1, c16 * 16)) = v1;
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz16(r0, c16 * 16)) = v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz16(r1, c16 * 16)) = v1;
   union { uint4v u; short s[8]; } a, b;
   a.u = v0; b.u = v1;
 #pragma unroll
@@ -132,7 +132,7 @@ __device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
     int d = c16 * 8 + j;
     uint pair = ((uint)(unsigned short)a.s[j]) |
                 (((uint)(unsigned short)b.s[j]) << 16);
-    *(uint*)(lds_tr + d * 64 + swz32(d, r0 * 2)) = pair;
+    *(uint*)(lds_tr + d * (TR_PITCH * 2) + r0 * 2) = pair;
   }
 }
 
@@ -166,11 +166,12 @@ __device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
 // ---------------------------------------------------------------------------
 // LDS: Q rm 16KB?? 32x128x2 = 8KB; dO rm 8KB; Qt 8KB; dOt 8KB;
 //      per-wave P^T + dS^T tiles 2*4*32*40*2 = 20KB  => ~52KB
+#define TR_TILE_B (FA_D * TR_PITCH * 2)  // 10240 B
 #define BW_LDS_Q 0
 #define BW_LDS_DO (32 * FA_D * 2)
 #define BW_LDS_QT (2 * 32 * FA_D * 2)
-#define BW_LDS_DOT (3 * 32 * FA_D * 2)
-#define BW_LDS_PW (4 * 32 * FA_D * 2)
+#define BW_LDS_DOT (2 * 32 * FA_D * 2 + TR_TILE_B)
+#define BW_LDS_PW (2 * 32 * FA_D * 2 + 2 * TR_TILE_B)
 #define PW_BYTES (32 * PW_PITCH * 2)
 
 extern "C" __global__ void __launch_bounds__(512, 2)
@@ -248,10 +249,10 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       for (int c = 0; c < 8; ++c) {
         frag_u qfr, dofr;
         qfr.u4 = *(const uint4v*)(smem + BW_LDS_Q + col * (FA_D * 2)
-                                  + swz(col, (c * 16 + hi * 8) * 2));
+                                  + swz16(col, (c * 16 + hi * 8) * 2));
         st = mfma_bf16(kf[c].bf, qfr.bf, st);
         dofr.u4 = *(const uint4v*)(smem + BW_LDS_DO + col * (FA_D * 2)
-                                   + swz(col, (c * 16 + hi * 8) * 2));
+                                   + swz16(col, (c * 16 + hi * 8) * 2));
         dpt = mfma_bf16(vf[c].bf, dofr.bf, dpt);
       }
 
@@ -276,11 +277,11 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         for (int nb = 0; nb < 2; ++nb) {
           int d = dhalf * 64 + nb * 32 + col;
           frag_u dofr, qfr;
-          dofr.u4 = *(const uint4v*)(smem + BW_LDS_DOT + d * 64
-                                     + swz32(d, (cq * 16 + hi * 8) * 2));
+          dofr.u4 = *(const uint4v*)(smem + BW_LDS_DOT + d * (TR_PITCH * 2)
+                                     + (cq * 16 + hi * 8) * 2);
           dv_acc[nb] = mfma_bf16(pa.bf, dofr.bf, dv_acc[nb]);
-          qfr.u4 = *(const uint4v*)(smem + BW_LDS_QT + d * 64
-                                    + swz32(d, (cq * 16 + hi * 8) * 2));
+          qfr.u4 = *(const uint4v*)(smem + BW_LDS_QT + d * (TR_PITCH * 2)
+                                    + (cq * 16 + hi * 8) * 2);
           dk_acc[nb] = mfma_bf16(da.bf, qfr.bf, dk_acc[nb]);
         }
       }
@@ -345,8 +346,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 // ---------------------------------------------------------------------------
 #define DQ_LDS_K 0
 #define DQ_LDS_KT (32 * FA_D * 2)
-#define DQ_LDS_V (2 * 32 * FA_D * 2)
-#define DQ_LDS_PW (3 * 32 * FA_D * 2)
+#define DQ_LDS_V (32 * FA_D * 2 + TR_TILE_B)
+#define DQ_LDS_PW (2 * 32 * FA_D * 2 + TR_TILE_B)
 
 extern "C" __global__ void __launch_bounds__(256, 2)
 fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
@@ -409,11 +410,11 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       for (int c = 0; c < 8; ++c) {
         frag_u kfr;
         kfr.u4 = *(const uint4v*)(smem + DQ_LDS_K + col * (FA_D * 2)
-                                  + swz(col, (c * 16 + hi * 8) * 2));
+                                  + swz16(col, (c * 16 + hi * 8) * 2));
         st = mfma_bf16(kfr.bf, qf[c].bf, st);
         frag_u vfr;
         vfr.u4 = *(const uint4v*)(smem + DQ_LDS_V + col * (FA_D * 2)
-                                  + swz(col, (c * 16 + hi * 8) * 2));
+                                  + swz16(col, (c * 16 + hi * 8) * 2));
         dpt = mfma_bf16(vfr.bf, dof[c].bf, dpt);
       }
 
@@ -445,8 +446,8 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         for (int nb = 0; nb < 4; ++nb) {
           int d = nb * 32 + col;
           frag_u kfr;
-          kfr.u4 = *(const uint4v*)(smem + DQ_LDS_KT + d * 64
-                                    + swz32(d, (ck * 16 + hi * 8) * 2));
+          kfr.u4 = *(const uint4v*)(smem + DQ_LDS_KT + d * (TR_PITCH * 2)
+                                    + (ck * 16 + hi * 8) * 2);
           dq_acc[nb] = mfma_bf16(da.bf, kfr.bf, dq_acc[nb]);
         }
       }
@@ -496,13 +497,13 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
                                               (const short*)out,
                                               (float*)delta, rows);
   dim3 gkv((S + 127) / 128, Hq, B);
-  size_t lds1 = 4 * 32 * FA_D * 2 + 8 * 2 * PW_BYTES;
+  size_t lds1 = 2 * 32 * FA_D * 2 + 2 * TR_TILE_B + 8 * 2 * PW_BYTES;
   fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
       Hkv, S, scale, causal);
   dim3 gq((S + 127) / 128, Hq, B);
-  size_t lds2 = 3 * 32 * FA_D * 2 + 4 * PW_BYTES;
+  size_t lds2 = 2 * 32 * FA_D * 2 + TR_TILE_B + 4 * PW_BYTES;
   fa_bwd_dq_kernel<<<gq, 256, lds2, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dq, B, Hq, Hkv, S,

@@ -43,3 +43,9 @@ __device__ __forceinline__ uint pack_bf16x2(float lo, float hi) {
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return byte_in_row ^ ((row & 7) << 4);
 }
+
+// full swizzle for 256-byte rows: conflict-FREE when a b128 lane group's
+// rows are distinct mod 16 (they are: the 4x16 groups of ds_read_b128)
+__device__ __forceinline__ int swz16(int row, int byte_in_row) {
+  return byte_in_row ^ ((row & 15) << 4);
+}
