@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument("--layers", type=int, default=0,
                    help="override layer count (debug only; overridden runs "
                         "are marked invalid in the output)")
-    p.add_argument("--microbatch", type=int, default=1)
+    p.add_argument("--microbatch", type=int, default=4)
     return p.parse_args()
 
 
@@ -85,7 +85,7 @@ def main():
                                lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
                                grad_clipping=True, max_norm=1.0)
 
-    B = args.batch or 2 * max(1, world)
+    B = args.batch or 4 * max(1, world)
     S = args.seq
     mbs = args.microbatch
     assert B % mbs == 0

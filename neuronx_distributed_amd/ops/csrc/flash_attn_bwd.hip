@@ -174,7 +174,7 @@ __device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
 #define BW_LDS_PW (2 * 32 * FA_D * 2 + 2 * TR_TILE_B)
 #define PW_BYTES (32 * PW_PITCH * 2)
 
-extern "C" __global__ void __launch_bounds__(512, 2)
+extern "C" __global__ void __launch_bounds__(256, 2)
 fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const short* __restrict__ vp,
                    const short* __restrict__ dop,
@@ -182,16 +182,18 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const float* __restrict__ deltap,
                    short* __restrict__ dkp, short* __restrict__ dvp,
                    int B, int Hq, int Hkv, int S, float scale, int causal) {
-  // 8 waves: wave w -> kv-row group (w>>1, 32 rows), d-half (w&1, 64 cols).
-  // The two waves of a pair recompute the same S^T/dP^T (16 extra mfmas)
-  // but halve the dK/dV accumulators, reaching 2 waves/SIMD instead of 1 -
-  // a net win since the kernel is latency-bound, not mfma-bound.
+  // 4 waves / 64-row kv block: wave w -> kv-row group (w>>1, 32 rows),
+  // d-half (w&1, 64 cols).  The d-half pair recomputes the same S^T/dP^T
+  // but halves the dK/dV accumulators (fits 2 waves/SIMD), and the SMALL
+  // workgroup lets TWO independent WGs co-reside per CU so one computes
+  // while the other sits at its barrier (the dq kernel showed ~2x from
+  // exactly this co-residency).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int col = lane & 31;
   const int hi = lane >> 5;
-  const int kvg = wid >> 1;        // kv row group 0..3
+  const int kvg = wid >> 1;        // kv row group 0..1
   const int dhalf = wid & 1;       // d half 0..1
 
   const int kvblk = blockIdx.x;
@@ -202,7 +204,7 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
   const long lse_base = (long)(b * Hq + h) * S;
 
-  const int kv0 = kvblk * 128 + kvg * 32;   // this wave's kv rows
+  const int kv0 = kvblk * 64 + kvg * 32;    // this wave's kv rows
   const int my_k = kv0 + col;
 
   frag_u kf[8], vf[8];
@@ -222,7 +224,7 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   char* pw_ds = pw_p + PW_BYTES;
 
   const float s2 = scale * LOG2E;
-  int q_start = causal ? kvblk * 128 : 0;
+  int q_start = causal ? kvblk * 64 : 0;
 
   stage_tile32(qp + q_base, q_start, FA_D, S - q_start, smem + BW_LDS_Q,
                smem + BW_LDS_QT);
@@ -496,9 +498,9 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
   fa_bwd_delta_kernel<<<nb, 256, 0, stream>>>((const short*)dout,
                                               (const short*)out,
                                               (float*)delta, rows);
-  dim3 gkv((S + 127) / 128, Hq, B);
-  size_t lds1 = 2 * 32 * FA_D * 2 + 2 * TR_TILE_B + 8 * 2 * PW_BYTES;
-  fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
+  dim3 gkv((S + 63) / 64, Hq, B);
+  size_t lds1 = 2 * 32 * FA_D * 2 + 2 * TR_TILE_B + 4 * 2 * PW_BYTES;
+  fa_bwd_dkdv_kernel<<<gkv, 256, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
       Hkv, S, scale, causal);
