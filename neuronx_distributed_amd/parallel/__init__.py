@@ -77,3 +77,7 @@ PARALLEL_FUNCTIONS = [
     reduce_from_tensor_model_parallel_region,
     scatter_to_tensor_model_parallel_region,
 ]
+
+from .conv import OutputChannelParallelConv2d, InputChannelParallelConv2d
+PARALLEL_MODULES.append(OutputChannelParallelConv2d)
+PARALLEL_MODULES.append(InputChannelParallelConv2d)

@@ -3,3 +3,4 @@ from .quantization_config import (QuantizationConfig, QuantizationType,
 from .quantization_layers import QuantizedColumnParallel, QuantizedRowParallel
 from .quantization_utils import quantize_symmetric, dequantize
 from . import quantize
+from . import microscaling

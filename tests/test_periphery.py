@@ -1,0 +1,85 @@
+"""Conv2d parallel layers, medusa buffers, safetensors dedup, MX quant."""
+
+import torch
+
+from dist_utils import run_distributed
+
+
+def _conv_worker(rank, world):
+    import torch.nn.functional as F
+    from neuronx_distributed_amd.parallel import (
+        OutputChannelParallelConv2d, InputChannelParallelConv2d,
+        parallel_state as ps, comm)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    oc = OutputChannelParallelConv2d(4, 8, 3, padding=1, bias=False,
+                                     gather_output=True, dtype=torch.float32)
+    torch.manual_seed(1)
+    x = torch.randn(2, 4, 8, 8)
+    out = oc(x)
+    wfull = comm.all_gather(oc.weight.detach(), dim=0,
+                            group=ps.get_group_info("tp"))
+    ref = F.conv2d(x, wfull, padding=1)
+    assert torch.allclose(out, ref, atol=1e-5)
+
+    torch.manual_seed(0)
+    ic = InputChannelParallelConv2d(4, 8, 3, padding=1, bias=False,
+                                    input_is_parallel=False,
+                                    dtype=torch.float32)
+    out2 = ic(x)
+    wfull2 = comm.all_gather(ic.weight.detach(), dim=1,
+                             group=ps.get_group_info("tp"))
+    ref2 = F.conv2d(x, wfull2, padding=1)
+    assert torch.allclose(out2, ref2, atol=1e-5)
+    return True
+
+
+def test_parallel_conv2d_tp2():
+    run_distributed(_conv_worker, world_size=2)
+
+
+def test_medusa_buffers():
+    from neuronx_distributed_amd.utils.medusa_utils import (
+        generate_medusa_buffers)
+
+    buf = generate_medusa_buffers([[0], [1], [0, 0], [0, 1], [1, 0]])
+    n = 6
+    assert buf["medusa_attn_mask"].shape == (n, n)
+    # every node attends to root and itself
+    assert buf["medusa_attn_mask"][:, 0].all()
+    assert buf["medusa_attn_mask"].diagonal().all()
+    # node [0,0] attends to [0]
+    assert buf["medusa_attn_mask"][3, 1]
+    assert not buf["medusa_attn_mask"][3, 2]
+    assert buf["medusa_position_ids"].tolist() == [0, 1, 1, 2, 2, 2]
+    # leaves: [0,0],[0,1],[1,0] -> 3 retrieve paths
+    assert buf["retrieve_indices"].shape[0] == 3
+
+
+def test_safetensors_dedup(tmp_path):
+    from neuronx_distributed_amd.utils.safetensors_utils import (
+        save_safetensors, load_safetensors)
+
+    w = torch.randn(4, 4)
+    sd = {"a.weight": w, "tied.weight": w, "b": torch.ones(2)}
+    p = str(tmp_path / "x.safetensors")
+    save_safetensors(sd, p)
+    out = load_safetensors(p)
+    assert torch.equal(out["a.weight"], w)
+    assert torch.equal(out["tied.weight"], w)
+    assert set(out) == set(sd)
+
+
+def test_mx_quantization():
+    from neuronx_distributed_amd.quantization.microscaling import (
+        quantize_mx, dequantize_mx)
+
+    torch.manual_seed(0)
+    w = torch.randn(64, 128)
+    # e4m3: 3 mantissa bits -> ~6% per-element; e2m1 grid -> ~17%
+    for fmt, tol in (("fp8_e4m3", 0.08), ("fp4_e2m1", 0.25)):
+        q, s = quantize_mx(w, fmt)
+        wd = dequantize_mx(q, s, dtype=torch.float32)
+        rel = (wd - w).abs().max() / w.abs().max()
+        assert rel < tol, (fmt, rel.item())
