@@ -42,8 +42,10 @@ def quantize_mx(x: torch.Tensor, fmt: str = "fp8_e4m3", axis: int = -1
     assert orig[-1] % MX_BLOCK == 0
     xb = x.reshape(*orig[:-1], orig[-1] // MX_BLOCK, MX_BLOCK).float()
     amax = xb.abs().amax(dim=-1, keepdim=True).clamp(min=2.0 ** -126)
-    # e8m0: power-of-two scale so block max maps into the element range
-    scales = torch.floor(torch.log2(amax / emax)).clamp(-127, 127)
+    # e8m0: power-of-two scale so the block max lands INSIDE the element
+    # range (ceil: scaled max in (emax/2, emax], never overflowing e4m3fn's
+    # finite-only encoding)
+    scales = torch.ceil(torch.log2(amax / emax)).clamp(-127, 127)
     scaled = xb / torch.exp2(scales)
     if fmt == "fp4_e2m1":
         q = _round_fp4(scaled)
