@@ -159,12 +159,32 @@ class ExpertMLPs(nn.Module):
                        gathered * aff[keep].unsqueeze(-1).float())
         return out.to(hidden.dtype)
 
+    def forward_blockwise(self, hidden, expert_affinities, expert_index):
+        """No-drop blockwise strategy (reference expert_mlps_v2.py:691 +
+        blockwise.py K4): fixed-size expert blocks, grouped GEMMs."""
+        from .blockwise import blockwise_mm, compute_block_indices, \
+            DEFAULT_BLOCK_SIZE
+
+        assert self.ep_size == 1, "blockwise with EP: use capacity_factor"
+        block_size = min(DEFAULT_BLOCK_SIZE,
+                         max(32, hidden.shape[0] // 4))
+        tpi, b2e, _ = compute_block_indices(expert_index, self.num_experts,
+                                            block_size)
+        return blockwise_mm(hidden, expert_affinities,
+                            self.gate_up_proj.weight.to(hidden.dtype),
+                            self.down_proj.weight.to(hidden.dtype),
+                            tpi, b2e, expert_index, block_size,
+                            glu=self.glu_mlp)
+
     def forward(self, hidden, expert_affinities, expert_index):
         """Dispatch (reference :1407-1499): training -> capacity_factor if
-        set else all-experts; inference -> all-experts for small T else
-        capacity path."""
+        set (>0); <=0 -> blockwise; None -> all-experts."""
         hidden = copy_to_tensor_model_parallel_region(hidden)
         if self.capacity_factor is not None and self.capacity_factor > 0:
             return self.forward_capacity_factor(hidden, expert_affinities,
                                                 expert_index)
+        if self.capacity_factor is not None and self.capacity_factor <= 0 \
+                and self.ep_size == 1:
+            return self.forward_blockwise(hidden, expert_affinities,
+                                          expert_index)
         return self.forward_all_experts(hidden, expert_affinities)

@@ -159,3 +159,41 @@ def test_group_limited_router():
     # affinities normalized per token
     s = aff.sum(-1)
     assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
+
+
+def _blockwise_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    moe_bw = _make_moe(cf=-1.0)     # capacity_factor <= 0 -> blockwise
+    torch.manual_seed(0)
+    moe_ref = _make_moe(cf=None)    # all-experts (no dropping) golden
+    torch.manual_seed(1)
+    x = torch.randn(2, 40, 16, requires_grad=True)
+    out_bw, _ = moe_bw(x)
+    x2 = x.detach().clone().requires_grad_(True)
+    out_ref, _ = moe_ref(x2)
+    assert torch.allclose(out_bw, out_ref, atol=1e-4), \
+        (out_bw - out_ref).abs().max()
+    out_bw.sum().backward()
+    out_ref.sum().backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4)
+    g1 = moe_bw.expert_mlps.gate_up_proj.weight.grad
+    g2 = moe_ref.expert_mlps.gate_up_proj.weight.grad
+    assert torch.allclose(g1, g2, atol=1e-4), (g1 - g2).abs().max()
+    return True
+
+
+def test_moe_blockwise_matches_all_experts():
+    run_distributed(_blockwise_worker, world_size=1)
+
+
+def test_block_indices():
+    from neuronx_distributed_amd.moe.blockwise import compute_block_indices
+
+    idx = torch.tensor([[0], [1], [0], [0], [1]])  # counts: e0=3, e1=2
+    tpi, b2e, n = compute_block_indices(idx, num_experts=2, block_size=2)
+    assert n == 3  # e0 -> 2 blocks, e1 -> 1 block
+    assert b2e.tolist() == [0, 0, 1]
+    assert tpi.tolist() == [0, 2, 3, -1, 1, 4]
