@@ -58,10 +58,10 @@ ce_sumexp_kernel(const short* __restrict__ logits,
     for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
       s8v v = *(const s8v*)(r + i * 8);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) s += __expf(bits2f(v[j]) - m);
+      for (int j = 0; j < 8; ++j) s += __builtin_amdgcn_exp2f((bits2f(v[j]) - m) * 1.4426950408889634f);
     }
     for (int i = (nvec << 3) + threadIdx.x; i < V; i += blockDim.x)
-      s += __expf(bits2f(r[i]) - m);
+      s += __builtin_amdgcn_exp2f((bits2f(r[i]) - m) * 1.4426950408889634f);
     s = block_reduce_sum(s, scratch);
     if (threadIdx.x == 0) {
       sumexp[row] = s;
@@ -92,7 +92,7 @@ ce_bwd_kernel(const short* __restrict__ logits,
       s8v o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float p = __expf(bits2f(v[j]) - m) * inv;
+        float p = __builtin_amdgcn_exp2f((bits2f(v[j]) - m) * 1.4426950408889634f) * inv;
         long col = i * 8 + j;
         if (col == t) p -= 1.0f;
         o[j] = f2bits(p * g);
@@ -100,7 +100,7 @@ ce_bwd_kernel(const short* __restrict__ logits,
       *(s8v*)(d + i * 8) = o;
     }
     for (int i = (nvec << 3) + threadIdx.x; i < V; i += blockDim.x) {
-      float p = __expf(bits2f(r[i]) - m) * inv;
+      float p = __builtin_amdgcn_exp2f((bits2f(r[i]) - m) * 1.4426950408889634f) * inv;
       if (i == t) p -= 1.0f;
       d[i] = f2bits(p * g);
     }

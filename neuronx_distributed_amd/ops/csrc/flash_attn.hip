@@ -49,7 +49,9 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   };
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  // readfirstlane: provably wave-uniform -> scalar branches for the
+  // per-wave activity guards instead of exec-mask divergence (T20)
+  const int wid = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
   const int col = lane & 31;       // q column owned by this lane
   const int hi = lane >> 5;
 
@@ -141,6 +143,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     if (wave_active) {
       // ---- QK^T: S^T[k][q] = sum_d K[k][d] Q^T[d][q] ------------------
       f32x16 acc[2] = {};
+      __builtin_amdgcn_s_setprio(1);  // T5: keep the matrix pipe fed
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
 #pragma unroll
@@ -152,6 +155,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           acc[kb] = mfma_bf16(kf.bf, qf[c].bf, acc[kb]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- online softmax (exp2 domain), lane-local per q row ---------
       float sc[2][16];
@@ -181,7 +185,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       // -1e30 and exp2(sc - m_new) would be exp2(0)=1 for masked scores;
       // the floor keeps those at exp2(-9.9e29) = 0.
       float m_eff = fmaxf(m_new, -1e28f);
-      float alpha = __builtin_exp2f(m_run - m_eff);  // 0 when m_run=-1e30
+      float alpha = __builtin_amdgcn_exp2f(m_run - m_eff);  // 0 when m_run=-1e30
       m_run = m_new;
 
       float psum = 0.f;
@@ -189,7 +193,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       for (int kb = 0; kb < 2; ++kb)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          float p = __builtin_exp2f(sc[kb][r] - m_eff);
+          float p = __builtin_amdgcn_exp2f(sc[kb][r] - m_eff);
           sc[kb][r] = p;
           psum += p;
         }
@@ -222,6 +226,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         }
 
       // ---- PV: O^T[d][q] += V^T[d][k] P^T[k][q] -----------------------
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int nb = 0; nb < 4; ++nb) {
 #pragma unroll
@@ -233,6 +238,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           ot[nb] = mfma_bf16(vf.bf, pf[c16].bf, ot[nb]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     __syncthreads();

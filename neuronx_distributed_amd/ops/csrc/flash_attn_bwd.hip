@@ -174,7 +174,59 @@ __device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
 #define BW_LDS_PW (2 * 32 * FA_D * 2 + 2 * TR_TILE_B)
 #define PW_BYTES (32 * PW_PITCH * 2)
 
-extern "C" __global__ void __launch_bounds__(256, 2)
+
+// ---- 64-row tile staging (512 threads, 2-row pairs) ----------------------
+#define TR64_PITCH 72   // 144-byte rows: 16B-aligned, banks 4*(9d mod 16)
+#define TR64_TILE_B (FA_D * TR64_PITCH * 2)   // 18432 B
+#define BW64_LDS_Q 0
+#define BW64_LDS_DO (64 * FA_D * 2)
+#define BW64_LDS_QT (2 * 64 * FA_D * 2)
+#define BW64_LDS_DOT (2 * 64 * FA_D * 2 + TR64_TILE_B)
+#define BW64_LDS_PW (2 * 64 * FA_D * 2 + 2 * TR64_TILE_B)
+
+struct Stage64Regs { uint4v v0, v1; };
+
+__device__ __forceinline__ Stage64Regs load_tile64(
+    const short* __restrict__ src, long src_row0, long src_stride,
+    int rows_valid) {
+  int rp = threadIdx.x >> 4;        // 0..31 -> rows 2rp, 2rp+1 (0..63)
+  int c16 = threadIdx.x & 15;
+  int r0 = 2 * rp, r1 = 2 * rp + 1;
+  int rr0 = r0 < rows_valid ? r0 : (rows_valid > 0 ? rows_valid - 1 : 0);
+  int rr1 = r1 < rows_valid ? r1 : (rows_valid > 0 ? rows_valid - 1 : 0);
+  Stage64Regs r;
+  r.v0 = *(const uint4v*)(src + (src_row0 + rr0) * src_stride + c16 * 8);
+  r.v1 = *(const uint4v*)(src + (src_row0 + rr1) * src_stride + c16 * 8);
+  return r;
+}
+
+__device__ __forceinline__ void write_tile64(Stage64Regs r, char* lds_rm,
+                                             char* lds_tr) {
+  int rp = threadIdx.x >> 4;
+  int c16 = threadIdx.x & 15;
+  int r0 = 2 * rp, r1 = 2 * rp + 1;
+  *(uint4v*)(lds_rm + r0 * (FA_D * 2) + swz16(r0, c16 * 16)) = r.v0;
+  *(uint4v*)(lds_rm + r1 * (FA_D * 2) + swz16(r1, c16 * 16)) = r.v1;
+  union { uint4v u; short s[8]; } a, b;
+  a.u = r.v0; b.u = r.v1;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int d = c16 * 8 + j;
+    uint pair = ((uint)(unsigned short)a.s[j]) |
+                (((uint)(unsigned short)b.s[j]) << 16);
+    *(uint*)(lds_tr + d * (TR64_PITCH * 2) + r0 * 2) = pair;
+  }
+}
+
+__device__ __forceinline__ void stage_tile64(const short* __restrict__ src,
+                                             long src_row0, long src_stride,
+                                             int rows_valid, char* lds_rm,
+                                             char* lds_tr) {
+  write_tile64(load_tile64(src, src_row0, src_stride, rows_valid), lds_rm,
+               lds_tr);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2)
 fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const short* __restrict__ vp,
                    const short* __restrict__ dop,
@@ -182,18 +234,19 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const float* __restrict__ deltap,
                    short* __restrict__ dkp, short* __restrict__ dvp,
                    int B, int Hq, int Hkv, int S, float scale, int causal) {
-  // 4 waves / 64-row kv block: wave w -> kv-row group (w>>1, 32 rows),
-  // d-half (w&1, 64 cols).  The d-half pair recomputes the same S^T/dP^T
-  // but halves the dK/dV accumulators (fits 2 waves/SIMD), and the SMALL
-  // workgroup lets TWO independent WGs co-reside per CU so one computes
-  // while the other sits at its barrier (the dq kernel showed ~2x from
-  // exactly this co-residency).
+  // 8 waves / 128-row kv block: wave w -> kv rows (w>>1)*32, d-half w&1.
+  // Q-TILE = 64 rows per iteration, processed as two 32-q halves that
+  // REUSE the st/dpt accumulators and pw tiles: one barrier pair, one
+  // staging round and one lse/delta fetch per 64 q rows - the measured
+  // per-iteration overhead (~6.4us vs a ~2us compute floor) amortizes 2x.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  // readfirstlane: provably wave-uniform -> scalar branches for the
+  // per-wave activity guards instead of exec-mask divergence (T20)
+  const int wid = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
   const int col = lane & 31;
   const int hi = lane >> 5;
-  const int kvg = wid >> 1;        // kv row group 0..1
+  const int kvg = wid >> 1;        // kv row group 0..3
   const int dhalf = wid & 1;       // d half 0..1
 
   const int kvblk = blockIdx.x;
@@ -204,7 +257,7 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
   const long lse_base = (long)(b * Hq + h) * S;
 
-  const int kv0 = kvblk * 64 + kvg * 32;    // this wave's kv rows
+  const int kv0 = kvblk * 128 + kvg * 32;   // this wave's kv rows
   const int my_k = kv0 + col;
 
   frag_u kf[8], vf[8];
@@ -220,79 +273,89 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   f32x16 dv_acc[2] = {};
   f32x16 dk_acc[2] = {};
 
-  char* pw_p = smem + BW_LDS_PW + wid * 2 * PW_BYTES;
+  char* pw_p = smem + BW64_LDS_PW + wid * 2 * PW_BYTES;
   char* pw_ds = pw_p + PW_BYTES;
 
   const float s2 = scale * LOG2E;
-  int q_start = causal ? kvblk * 64 : 0;
+  int q_start = causal ? kvblk * 128 : 0;
 
-  stage_tile32(qp + q_base, q_start, FA_D, S - q_start, smem + BW_LDS_Q,
-               smem + BW_LDS_QT);
-  stage_tile32(dop + q_base, q_start, FA_D, S - q_start, smem + BW_LDS_DO,
-               smem + BW_LDS_DOT);
+  stage_tile64(qp + q_base, q_start, FA_D, S - q_start, smem + BW64_LDS_Q,
+               smem + BW64_LDS_QT);
+  stage_tile64(dop + q_base, q_start, FA_D, S - q_start, smem + BW64_LDS_DO,
+               smem + BW64_LDS_DOT);
 
-  StageRegs nq, ndo;
-  for (int q0 = q_start; q0 < S; q0 += 32) {
+  Stage64Regs nq, ndo;
+  for (int q0 = q_start; q0 < S; q0 += 64) {
     __syncthreads();
-    if (q0 + 32 < S) {
-      nq = load_tile32(qp + q_base, q0 + 32, FA_D, S - q0 - 32);
-      ndo = load_tile32(dop + q_base, q0 + 32, FA_D, S - q0 - 32);
+    if (q0 + 64 < S) {
+      nq = load_tile64(qp + q_base, q0 + 64, FA_D, S - q0 - 64);
+      ndo = load_tile64(dop + q_base, q0 + 64, FA_D, S - q0 - 64);
     }
-    const int my_q = q0 + col;
-    const bool wave_active = !causal || (q0 + 31 >= kv0);
+    const bool any_active = !causal || (q0 + 63 >= kv0);
 
-    if (wave_active) {
-      const float lse2 = lsep[lse_base + (my_q < S ? my_q : S - 1)] * LOG2E;
-      const float dlt = deltap[lse_base + (my_q < S ? my_q : S - 1)];
+    if (any_active) {
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int qh0 = q0 + half * 32;
+        const int my_q = qh0 + col;
+        const bool half_active = !causal || (qh0 + 31 >= kv0);
+        if (!half_active) continue;
 
-      f32x16 st = {};
-      f32x16 dpt = {};
+        const float lse2 = lsep[lse_base + (my_q < S ? my_q : S - 1)] * LOG2E;
+        const float dlt = deltap[lse_base + (my_q < S ? my_q : S - 1)];
+        const int rm_row = half * 32 + col;  // row inside the 64-row tile
+
+        f32x16 st = {};
+        f32x16 dpt = {};
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        frag_u qfr, dofr;
-        qfr.u4 = *(const uint4v*)(smem + BW_LDS_Q + col * (FA_D * 2)
-                                  + swz16(col, (c * 16 + hi * 8) * 2));
-        st = mfma_bf16(kf[c].bf, qfr.bf, st);
-        dofr.u4 = *(const uint4v*)(smem + BW_LDS_DO + col * (FA_D * 2)
-                                   + swz16(col, (c * 16 + hi * 8) * 2));
-        dpt = mfma_bf16(vf[c].bf, dofr.bf, dpt);
-      }
+        for (int c = 0; c < 8; ++c) {
+          frag_u qfr, dofr;
+          qfr.u4 = *(const uint4v*)(smem + BW64_LDS_Q + rm_row * (FA_D * 2)
+                                    + swz16(rm_row, (c * 16 + hi * 8) * 2));
+          st = mfma_bf16(kf[c].bf, qfr.bf, st);
+          dofr.u4 = *(const uint4v*)(smem + BW64_LDS_DO + rm_row * (FA_D * 2)
+                                     + swz16(rm_row, (c * 16 + hi * 8) * 2));
+          dpt = mfma_bf16(vf[c].bf, dofr.bf, dpt);
+        }
 
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int kg = kv0 + acc_row(r, hi);
-        bool masked = (causal && kg > my_q) || kg >= S || my_q >= S;
-        float p = masked ? 0.f : __builtin_exp2f(st[r] * s2 - lse2);
-        st[r] = p;
-        dpt[r] = p * (dpt[r] - dlt);
-      }
-      write_acc_tile(pw_p, st, lane);
-      write_acc_tile(pw_ds, dpt, lane);
+        for (int r = 0; r < 16; ++r) {
+          int kg = kv0 + acc_row(r, hi);
+          bool masked = (causal && kg > my_q) || kg >= S || my_q >= S;
+          float p = masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse2);
+          st[r] = p;
+          dpt[r] = p * (dpt[r] - dlt);
+        }
+        write_acc_tile(pw_p, st, lane);
+        write_acc_tile(pw_ds, dpt, lane);
 
-      // dV/dK over this wave's 64-column d-half
+        // dV/dK over this wave's 64-column d-half, q chunk = this half
 #pragma unroll
-      for (int cq = 0; cq < 2; ++cq) {
-        frag_u pa, da;
-        pa.bf = read_pw_row(pw_p, col, cq * 16 + hi * 8);
-        da.bf = read_pw_row(pw_ds, col, cq * 16 + hi * 8);
+        for (int cq = 0; cq < 2; ++cq) {
+          frag_u pa, da;
+          pa.bf = read_pw_row(pw_p, col, cq * 16 + hi * 8);
+          da.bf = read_pw_row(pw_ds, col, cq * 16 + hi * 8);
 #pragma unroll
-        for (int nb = 0; nb < 2; ++nb) {
-          int d = dhalf * 64 + nb * 32 + col;
-          frag_u dofr, qfr;
-          dofr.u4 = *(const uint4v*)(smem + BW_LDS_DOT + d * (TR_PITCH * 2)
-                                     + (cq * 16 + hi * 8) * 2);
-          dv_acc[nb] = mfma_bf16(pa.bf, dofr.bf, dv_acc[nb]);
-          qfr.u4 = *(const uint4v*)(smem + BW_LDS_QT + d * (TR_PITCH * 2)
-                                    + (cq * 16 + hi * 8) * 2);
-          dk_acc[nb] = mfma_bf16(da.bf, qfr.bf, dk_acc[nb]);
+          for (int nb = 0; nb < 2; ++nb) {
+            int d = dhalf * 64 + nb * 32 + col;
+            frag_u dofr, qfr;
+            dofr.u4 = *(const uint4v*)(smem + BW64_LDS_DOT
+                                       + d * (TR64_PITCH * 2)
+                                       + (half * 32 + cq * 16 + hi * 8) * 2);
+            dv_acc[nb] = mfma_bf16(pa.bf, dofr.bf, dv_acc[nb]);
+            qfr.u4 = *(const uint4v*)(smem + BW64_LDS_QT
+                                      + d * (TR64_PITCH * 2)
+                                      + (half * 32 + cq * 16 + hi * 8) * 2);
+            dk_acc[nb] = mfma_bf16(da.bf, qfr.bf, dk_acc[nb]);
+          }
         }
       }
     }
 
     __syncthreads();
-    if (q0 + 32 < S) {
-      write_tile32(nq, smem + BW_LDS_Q, smem + BW_LDS_QT);
-      write_tile32(ndo, smem + BW_LDS_DO, smem + BW_LDS_DOT);
+    if (q0 + 64 < S) {
+      write_tile64(nq, smem + BW64_LDS_Q, smem + BW64_LDS_QT);
+      write_tile64(ndo, smem + BW64_LDS_DO, smem + BW64_LDS_DOT);
     }
   }
 
@@ -304,12 +367,12 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int k = acc_row(r, hi);
-      int dd = nb * 32 + col;  // within the 64-col half
+      int dd = nb * 32 + col;
       *(short*)(otile + (k * 64 + dd) * 2) = f2bits(dv_acc[nb][r]);
     }
   __builtin_amdgcn_s_waitcnt(0);
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {   // 64 lanes * 4 = 256 pieces of 16B
+  for (int i = 0; i < 4; ++i) {
     int p = lane + i * 64;
     int row = p >> 3, c16 = p & 7;
     int kg = kv0 + row;
@@ -359,7 +422,9 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                  int B, int Hq, int Hkv, int S, float scale, int causal) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  // readfirstlane: provably wave-uniform -> scalar branches for the
+  // per-wave activity guards instead of exec-mask divergence (T20)
+  const int wid = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
   const int col = lane & 31;
   const int hi = lane >> 5;
 
@@ -425,7 +490,7 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       for (int r = 0; r < 16; ++r) {
         int kg = kv0 + acc_row(r, hi);
         bool masked = (causal && kg > my_q) || kg >= S || my_q >= S;
-        float p = masked ? 0.f : __builtin_exp2f(st[r] * s2 - lse2);
+        float p = masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse2);
         dst[r] = p * (dpt[r] - dlt);
       }
       // write dS as [q][k] (row q = col): cols k = acc_row pairs packed b32
@@ -498,9 +563,9 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
   fa_bwd_delta_kernel<<<nb, 256, 0, stream>>>((const short*)dout,
                                               (const short*)out,
                                               (float*)delta, rows);
-  dim3 gkv((S + 63) / 64, Hq, B);
-  size_t lds1 = 2 * 32 * FA_D * 2 + 2 * TR_TILE_B + 4 * 2 * PW_BYTES;
-  fa_bwd_dkdv_kernel<<<gkv, 256, lds1, stream>>>(
+  dim3 gkv((S + 127) / 128, Hq, B);
+  size_t lds1 = 2 * 64 * FA_D * 2 + 2 * TR64_TILE_B + 8 * 2 * PW_BYTES;
+  fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
       Hkv, S, scale, causal);

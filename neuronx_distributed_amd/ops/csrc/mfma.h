@@ -29,13 +29,14 @@ __device__ __forceinline__ int acc_row(int r, int hi) {
   return (r & 3) + 8 * (r >> 2) + 4 * hi;
 }
 
-// pack two f32 into one u32 of 2 bf16 (compiler emits v_cvt_pk_bf16_f32)
+// pack two f32 into one u32 of 2 bf16: use the compiler's native casts so
+// it can emit v_cvt_pk_bf16_f32 (a hand-rolled RNE costs ~5 VALU per pair,
+// cdna_hip_programming.md T12)
 __device__ __forceinline__ uint pack_bf16x2(float lo, float hi) {
-  union { float f; uint u; } a, b;
-  a.f = lo; b.f = hi;
-  uint ra = a.u + 0x7FFF + ((a.u >> 16) & 1);
-  uint rb = b.u + 0x7FFF + ((b.u >> 16) & 1);
-  return (ra >> 16) | (rb & 0xFFFF0000u);
+  union { __bf16 b[2]; uint u; } r;
+  r.b[0] = (__bf16)lo;
+  r.b[1] = (__bf16)hi;
+  return r.u;
 }
 
 // XOR swizzle: spread 16B slots of a row-major LDS tile over banks

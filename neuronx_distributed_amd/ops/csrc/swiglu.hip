@@ -22,7 +22,7 @@ swiglu_fwd_kernel(const short* __restrict__ x, short* __restrict__ out,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float gf = bits2f(gv[j]);
-      float s = gf / (1.f + __expf(-gf));
+      float s = gf / (1.f + __builtin_amdgcn_exp2f(-gf * 1.4426950408889634f));
       o[j] = f2bits(s * bits2f(uv[j]));
     }
     *(s8v*)(out + row * I + i * 8) = o;
@@ -48,7 +48,7 @@ swiglu_bwd_kernel(const short* __restrict__ x, const short* __restrict__ dy,
     for (int j = 0; j < 8; ++j) {
       float gf = bits2f(gv[j]);
       float df = bits2f(dv[j]);
-      float sig = 1.f / (1.f + __expf(-gf));
+      float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-gf * 1.4426950408889634f));
       float s = gf * sig;
       dg[j] = f2bits(df * bits2f(uv[j]) * (sig + s * (1.f - sig)));
       du[j] = f2bits(df * s);
