@@ -43,9 +43,12 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                  float* __restrict__ lsep, int B, int Hq, int Hkv, int S,
                  float scale, int causal) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // THREE rotating K/Vt buffers: tile t is read from buf t%3 while t+1 is
+  // written into (t+1)%3, whose previous readers (tile t-2) finished two
+  // barriers ago -> ONE barrier per tile instead of two.
   auto kbuf = [&](int i) { return smem + (size_t)i * K_TILE_B; };
   auto vbuf = [&](int i) {
-    return smem + 2 * K_TILE_B + (size_t)i * VT_TILE_B;
+    return smem + 3 * K_TILE_B + (size_t)i * VT_TILE_B;
   };
 
   const int lane = threadIdx.x & 63;
@@ -88,45 +91,40 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
   const float s2 = scale * LOG2E;
 
-  // ---- staging: thread t loads pieces t, t+512 of each tile -----------
-  // tile = 64 rows x 128 cols bf16 = 1024 16B pieces, 16 pieces per row:
-  // piece p -> (row p>>4, 16B slot p&15).  K: ds_write_b128 swizzled;
-  // V: same global piece, scatter-written transposed into Vt.
+  // ---- staging: thread t owns K/V rows {2rp, 2rp+1} at 16B slot c16 ----
+  // (rp = t>>4 in 0..31, c16 = t&15).  K: 2 x ds_write_b128 swizzled;
+  // V: the row PAIR transposes into 8 x b32 writes (two k-columns per
+  // write) instead of 16 scalar b16 scatters.
   const int tid = threadIdx.x;
-  int st_row[2], st_col[2];
-#pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    int p = tid + i * 512;
-    st_row[i] = p >> 4;
-    st_col[i] = p & 15;
-  }
+  const int st_rp = tid >> 4;
+  const int st_c16 = tid & 15;
+  const int st_r0 = 2 * st_rp, st_r1 = 2 * st_rp + 1;
 
   uint4v kreg[2], vreg[2];
   auto issue_loads = [&](int t) {
     int kv0 = t * FA_KV;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int row = kv0 + st_row[i];
-      int rr = row < S ? row : S - 1;  // clamp; masked later
-      kreg[i] = *(const uint4v*)(kp + kv_base + (long)rr * FA_D + st_col[i] * 8);
-      vreg[i] = *(const uint4v*)(vp + kv_base + (long)rr * FA_D + st_col[i] * 8);
-    }
+    int rr0 = kv0 + st_r0 < S ? kv0 + st_r0 : S - 1;
+    int rr1 = kv0 + st_r1 < S ? kv0 + st_r1 : S - 1;
+    kreg[0] = *(const uint4v*)(kp + kv_base + (long)rr0 * FA_D + st_c16 * 8);
+    kreg[1] = *(const uint4v*)(kp + kv_base + (long)rr1 * FA_D + st_c16 * 8);
+    vreg[0] = *(const uint4v*)(vp + kv_base + (long)rr0 * FA_D + st_c16 * 8);
+    vreg[1] = *(const uint4v*)(vp + kv_base + (long)rr1 * FA_D + st_c16 * 8);
   };
 
   auto write_tile = [&](int buf) {
+    *(uint4v*)(kbuf(buf) + swz16(st_r0, st_c16 * 16)
+               + st_r0 * (FA_D * 2)) = kreg[0];
+    *(uint4v*)(kbuf(buf) + swz16(st_r1, st_c16 * 16)
+               + st_r1 * (FA_D * 2)) = kreg[1];
+    union { uint4v u; short s[8]; } a, b;
+    a.u = vreg[0];
+    b.u = vreg[1];
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      // K row-major [64][128], swizzled 16B slots
-      *(uint4v*)(kbuf(buf) + swz16(st_row[i], st_col[i] * 16)
-                 + st_row[i] * (FA_D * 2)) = kreg[i];
-      // Vt transposed [128][VT_PITCH]: element j -> row d = col*8+j, col k
-      union { uint4v u; short s[8]; } vv;
-      vv.u = vreg[i];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int d = st_col[i] * 8 + j;
-        *(short*)(vbuf(buf) + d * (VT_PITCH * 2) + st_row[i] * 2) = vv.s[j];
-      }
+    for (int j = 0; j < 8; ++j) {
+      int d = st_c16 * 8 + j;
+      uint pair = ((uint)(unsigned short)a.s[j]) |
+                  (((uint)(unsigned short)b.s[j]) << 16);
+      *(uint*)(vbuf(buf) + d * (VT_PITCH * 2) + st_r0 * 2) = pair;
     }
   };
 
@@ -136,7 +134,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
   for (int t = 0; t < ntiles; ++t) {
     const int kv0 = t * FA_KV;
-    const int cur = t & 1;
+    const int cur = t % 3;
     if (t + 1 < ntiles) issue_loads(t + 1);
 
     const bool wave_active = kv0 < my_kv_end;
@@ -241,11 +239,8 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       __builtin_amdgcn_s_setprio(0);
     }
 
+    if (t + 1 < ntiles) write_tile((t + 1) % 3);
     __syncthreads();
-    if (t + 1 < ntiles) {
-      write_tile((t + 1) & 1);
-      __syncthreads();
-    }
   }
 
   // ---- epilogue: normalize, store O (transposed back) and LSE ---------
@@ -277,7 +272,7 @@ extern "C" void flash_attn_fwd(const void* q, const void* k, const void* v,
                                int S, float scale, int causal,
                                hipStream_t stream) {
   dim3 grid((S + FA_QBLK - 1) / FA_QBLK, Hq, B);
-  size_t lds = 2 * (K_TILE_B + VT_TILE_B);
+  size_t lds = 3 * (K_TILE_B + VT_TILE_B);
   flash_fwd_kernel<<<grid, 512, lds, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out,
       (float*)lse, B, Hq, Hkv, S, scale, causal);
