@@ -51,6 +51,18 @@ def main():
     on_gpu = torch.cuda.is_available()
     if on_gpu:
         torch.cuda.set_device(local_rank)
+        try:
+            # pre-tuned hipBLASLt/rocBLAS GEMM selections (profiles/, gfx950);
+            # tuning itself stays off - unknown shapes use defaults
+            tun = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                               "profiles", "tunableop_gfx950.csv")
+            if os.path.exists(tun) and \
+                    os.environ.get("NXDA_TUNABLEOP", "1") == "1":
+                torch.cuda.tunable.set_filename(tun, insert_device_ordinal=False)
+                torch.cuda.tunable.tuning_enable(False)
+                torch.cuda.tunable.enable(True)
+        except Exception:
+            pass
     backend = "nccl" if on_gpu else "gloo"
     if not dist.is_initialized():
         dist.init_process_group(backend, rank=rank, world_size=world)
