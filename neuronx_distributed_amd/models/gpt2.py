@@ -16,6 +16,11 @@ from ..parallel.layers import (
 )
 from ..parallel.layer_norm import LayerNorm
 from ..parallel.loss_functions import parallel_cross_entropy
+
+# fx.wrap patches name lookups in THIS module's globals (the defining
+# module's wrap does not cover imported references) so the pipeline tracer
+# treats the collective-bearing loss as a leaf call
+torch.fx.wrap("parallel_cross_entropy")
 from ..parallel import parallel_state as ps
 from ..kernels.flash_attn import flash_attn_func
 

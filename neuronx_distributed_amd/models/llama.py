@@ -29,6 +29,11 @@ from ..parallel.layers import (
 )
 from ..parallel.qkv_linear import GQAQKVColumnParallelLinear
 from ..parallel.loss_functions import parallel_cross_entropy
+
+# fx.wrap patches name lookups in THIS module's globals (the defining
+# module's wrap does not cover imported references) so the pipeline tracer
+# treats the collective-bearing loss as a leaf call
+torch.fx.wrap("parallel_cross_entropy")
 from ..parallel.mappings import (
     scatter_to_sequence_parallel_region,
     gather_from_sequence_parallel_region,

@@ -134,7 +134,8 @@ class _ParallelCrossEntropy(torch.autograd.Function):
 def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0):
     """Per-token loss on vocab-sharded logits (reference loss_functions.py:217).
     GPU bf16 + no smoothing dispatches to the fused HIP kernels."""
-    if (vocab_parallel_logits.is_cuda
+    if (not isinstance(vocab_parallel_logits, torch.fx.Proxy)
+            and vocab_parallel_logits.is_cuda
             and vocab_parallel_logits.dtype == torch.bfloat16
             and label_smoothing == 0.0
             and vocab_parallel_logits.shape[-1] % 8 == 0):
@@ -143,7 +144,8 @@ def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0):
         if ops.is_available():
             return _FusedParallelCrossEntropy.apply(vocab_parallel_logits,
                                                     target)
-    if vocab_parallel_logits.dtype in (torch.bfloat16, torch.float16):
+    if not isinstance(vocab_parallel_logits, torch.fx.Proxy) and \
+            vocab_parallel_logits.dtype in (torch.bfloat16, torch.float16):
         vocab_parallel_logits = vocab_parallel_logits.float()
     return _ParallelCrossEntropy.apply(vocab_parallel_logits, target,
                                        label_smoothing)
