@@ -58,8 +58,14 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int col = lane & 31;       // q column owned by this lane
   const int hi = lane >> 5;
 
-  const int qblk = blockIdx.x;
-  const int h = blockIdx.y;
+  // Dispatch mapping (causal LPT): heads on x, q-blocks on y REVERSED so
+  // the largest-causal-work blocks launch first.  With x = q-block the
+  // hardware pairs CU c with the SAME q-block index every occupancy round
+  // (512 WGs / 256 CUs at 1 WG/CU) -> worst CUs do ~1.9x the mean causal
+  // work while others idle; biggest-first lets early finishers absorb the
+  // small diagonal blocks (greedy LPT, near-ideal pairing).
+  const int qblk = (int)(gridDim.y - 1 - blockIdx.y);
+  const int h = blockIdx.x;
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
@@ -140,19 +146,27 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     const bool wave_active = kv0 < my_kv_end;
     if (wave_active) {
       // ---- QK^T: S^T[k][q] = sum_d K[k][d] Q^T[d][q] ------------------
+      // All 16 K fragments are prefetched into registers BEFORE the mfma
+      // burst (T3/T4): with ds_reads issued only 1-2 mfmas ahead the
+      // compiler emits lgkmcnt(0) waits before every other mfma and the
+      // ~64-cycle LDS latency stalls the matrix pipe; batching the reads
+      // turns those into counted waits that are already satisfied.
       f32x16 acc[2] = {};
+      frag_u kfr[2][8];
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+        int row = col + 32 * kb;
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+          kfr[kb][c].u4 = *(const uint4v*)(kbuf(cur) + row * (FA_D * 2)
+                                           + swz16(row, (c * 16 + hi * 8) * 2));
+      }
       __builtin_amdgcn_s_setprio(1);  // T5: keep the matrix pipe fed
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
+      for (int kb = 0; kb < 2; ++kb)
 #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
-          int row = col + 32 * kb;
-          frag_u kf;
-          kf.u4 = *(const uint4v*)(kbuf(cur) + row * (FA_D * 2)
-                                   + swz16(row, (c * 16 + hi * 8) * 2));
-          acc[kb] = mfma_bf16(kf.bf, qf[c].bf, acc[kb]);
-        }
-      }
+        for (int c = 0; c < 8; ++c)
+          acc[kb] = mfma_bf16(kfr[kb][c].bf, qf[c].bf, acc[kb]);
       __builtin_amdgcn_s_setprio(0);
 
       // ---- online softmax (exp2 domain), lane-local per q row ---------
@@ -224,19 +238,28 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         }
 
       // ---- PV: O^T[d][q] += V^T[d][k] P^T[k][q] -----------------------
-      __builtin_amdgcn_s_setprio(1);
+      // Same batched-prefetch structure as QK, two d-blocks at a time
+      // (8 fragments = 32 VGPRs in flight).
 #pragma unroll
-      for (int nb = 0; nb < 4; ++nb) {
+      for (int np = 0; np < 2; ++np) {
+        frag_u vfr[2][4];
 #pragma unroll
-        for (int c16 = 0; c16 < 4; ++c16) {
-          int d = 32 * nb + col;
-          frag_u vf;
-          vf.u4 = *(const uint4v*)(vbuf(cur) + d * (VT_PITCH * 2)
-                                   + (c16 * 16 + hi * 8) * 2);
-          ot[nb] = mfma_bf16(vf.bf, pf[c16].bf, ot[nb]);
+        for (int ni = 0; ni < 2; ++ni) {
+          int d = 32 * (2 * np + ni) + col;
+#pragma unroll
+          for (int c16 = 0; c16 < 4; ++c16)
+            vfr[ni][c16].u4 = *(const uint4v*)(vbuf(cur) + d * (VT_PITCH * 2)
+                                               + (c16 * 16 + hi * 8) * 2);
         }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+          for (int c16 = 0; c16 < 4; ++c16)
+            ot[2 * np + ni] =
+                mfma_bf16(vfr[ni][c16].bf, pf[c16].bf, ot[2 * np + ni]);
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
 
     if (t + 1 < ntiles) write_tile((t + 1) % 3);
@@ -271,7 +294,7 @@ extern "C" void flash_attn_fwd(const void* q, const void* k, const void* v,
                                void* out, void* lse, int B, int Hq, int Hkv,
                                int S, float scale, int causal,
                                hipStream_t stream) {
-  dim3 grid((S + FA_QBLK - 1) / FA_QBLK, Hq, B);
+  dim3 grid(Hq, (S + FA_QBLK - 1) / FA_QBLK, B);
   size_t lds = 3 * (K_TILE_B + VT_TILE_B);
   flash_fwd_kernel<<<grid, 512, lds, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out,

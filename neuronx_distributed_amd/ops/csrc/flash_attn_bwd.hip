@@ -249,8 +249,11 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int kvg = wid >> 1;        // kv row group 0..3
   const int dhalf = wid & 1;       // d half 0..1
 
-  const int kvblk = blockIdx.x;
-  const int h = blockIdx.y;
+  // heads on x, kv-blocks on y: causal work DEcreases with kv-block index,
+  // so ascending y already dispatches biggest-first (greedy LPT; see the
+  // fwd kernel's dispatch note)
+  const int kvblk = blockIdx.y;
+  const int h = blockIdx.x;
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
   const long q_base = ((long)(b * Hq + h) * S) * FA_D;
@@ -428,14 +431,17 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int col = lane & 31;
   const int hi = lane >> 5;
 
-  const int h = blockIdx.y;
+  // heads on x, q-blocks on y reversed: biggest-causal-work first (LPT,
+  // see the fwd kernel's dispatch note)
+  const int qblk = (int)(gridDim.y - 1 - blockIdx.y);
+  const int h = blockIdx.x;
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
   const long q_base = ((long)(b * Hq + h) * S) * FA_D;
   const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
   const long lse_base = (long)(b * Hq + h) * S;
 
-  const int q0 = blockIdx.x * 128 + wid * 32;
+  const int q0 = qblk * 128 + wid * 32;
   const int my_q = q0 + col;
   const int q_ld = my_q < S ? my_q : S - 1;
 
@@ -454,7 +460,7 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
   // WG-uniform loop bound (all waves share barriers); per-wave causal
   // skipping happens via wave_active below
-  const int kv_end = causal ? min(S, (int)blockIdx.x * 128 + 128) : S;
+  const int kv_end = causal ? min(S, qblk * 128 + 128) : S;
   const float s2 = scale * LOG2E;
 
   stage_tile32(kp + kv_base, 0, FA_D, S, smem + DQ_LDS_K, smem + DQ_LDS_KT);
@@ -563,13 +569,13 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
   fa_bwd_delta_kernel<<<nb, 256, 0, stream>>>((const short*)dout,
                                               (const short*)out,
                                               (float*)delta, rows);
-  dim3 gkv((S + 127) / 128, Hq, B);
+  dim3 gkv(Hq, (S + 127) / 128, B);
   size_t lds1 = 2 * 64 * FA_D * 2 + 2 * TR64_TILE_B + 8 * 2 * PW_BYTES;
   fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
       Hkv, S, scale, causal);
-  dim3 gq((S + 127) / 128, Hq, B);
+  dim3 gq(Hq, (S + 127) / 128, B);
   size_t lds2 = 2 * 32 * FA_D * 2 + TR_TILE_B + 4 * PW_BYTES;
   fa_bwd_dq_kernel<<<gq, 256, lds2, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
