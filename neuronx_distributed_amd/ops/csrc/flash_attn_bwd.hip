@@ -182,7 +182,8 @@ __device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
 #define BW64_LDS_DO (64 * FA_D * 2)
 #define BW64_LDS_QT (2 * 64 * FA_D * 2)
 #define BW64_LDS_DOT (2 * 64 * FA_D * 2 + TR64_TILE_B)
-#define BW64_LDS_PW (2 * 64 * FA_D * 2 + 2 * TR64_TILE_B)
+// 128 floats: lse*log2(e) for the 64 q rows at [0..64), delta at [64..128)
+#define BW64_LDS_LD (2 * 64 * FA_D * 2 + 2 * TR64_TILE_B)
 
 struct Stage64Regs { uint4v v0, v1; };
 
@@ -236,9 +237,15 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    int B, int Hq, int Hkv, int S, float scale, int causal) {
   // 8 waves / 128-row kv block: wave w -> kv rows (w>>1)*32, d-half w&1.
   // Q-TILE = 64 rows per iteration, processed as two 32-q halves that
-  // REUSE the st/dpt accumulators and pw tiles: one barrier pair, one
-  // staging round and one lse/delta fetch per 64 q rows - the measured
-  // per-iteration overhead (~6.4us vs a ~2us compute floor) amortizes 2x.
+  // REUSE the st/dpt accumulators: one barrier pair and one staging round
+  // per 64 q rows.
+  //
+  // The score/dS tiles never touch LDS: S^T/dP are computed with Q on the
+  // A operand (q in accumulator REGISTER rows, kv in lanes), so the
+  // pack_bf16x2 + permlane32_swap repack from the forward kernel turns
+  // them directly into the A fragments of the dV/dK mfmas (k-dim = q,
+  // lane = kv).  The softmax's per-q lse/delta become per-register-row
+  // values read from a 128-float LDS tile staged once per q-tile.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
   // readfirstlane: provably wave-uniform -> scalar branches for the
@@ -276,8 +283,18 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   f32x16 dv_acc[2] = {};
   f32x16 dk_acc[2] = {};
 
-  char* pw_p = smem + BW64_LDS_PW + wid * 2 * PW_BYTES;
-  char* pw_ds = pw_p + PW_BYTES;
+  float* ld_sm = (float*)(smem + BW64_LDS_LD);
+  // threads 0..63 fetch lse (pre-scaled by log2 e), 64..127 delta
+  auto load_ld = [&](int qt0) -> float {
+    int qg = qt0 + (threadIdx.x & 63);
+    int qc = qg < S ? qg : S - 1;
+    if (threadIdx.x < 64) return lsep[lse_base + qc] * LOG2E;
+    if (threadIdx.x < 128) return deltap[lse_base + qc];
+    return 0.f;
+  };
+  auto write_ld = [&](float v) {
+    if (threadIdx.x < 128) ld_sm[threadIdx.x] = v;
+  };
 
   const float s2 = scale * LOG2E;
   int q_start = causal ? kvblk * 128 : 0;
@@ -286,13 +303,16 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                smem + BW64_LDS_QT);
   stage_tile64(dop + q_base, q_start, FA_D, S - q_start, smem + BW64_LDS_DO,
                smem + BW64_LDS_DOT);
+  write_ld(load_ld(q_start));
 
   Stage64Regs nq, ndo;
+  float nld = 0.f;
   for (int q0 = q_start; q0 < S; q0 += 64) {
     __syncthreads();
     if (q0 + 64 < S) {
       nq = load_tile64(qp + q_base, q0 + 64, FA_D, S - q0 - 64);
       ndo = load_tile64(dop + q_base, q0 + 64, FA_D, S - q0 - 64);
+      nld = load_ld(q0 + 64);
     }
     const bool any_active = !causal || (q0 + 63 >= kv0);
 
@@ -300,14 +320,14 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
         const int qh0 = q0 + half * 32;
-        const int my_q = qh0 + col;
         const bool half_active = !causal || (qh0 + 31 >= kv0);
         if (!half_active) continue;
 
-        const float lse2 = lsep[lse_base + (my_q < S ? my_q : S - 1)] * LOG2E;
-        const float dlt = deltap[lse_base + (my_q < S ? my_q : S - 1)];
         const int rm_row = half * 32 + col;  // row inside the 64-row tile
+        const int my_kv = kv0 + col;         // this lane's kv column
 
+        // S^T2[q][kv]: A = Q rows (B-layout frag doubles as A: lane=q),
+        // B = K rows (lane=kv) -> q in accumulator rows, kv in lanes
         f32x16 st = {};
         f32x16 dpt = {};
 #pragma unroll
@@ -315,43 +335,97 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           frag_u qfr, dofr;
           qfr.u4 = *(const uint4v*)(smem + BW64_LDS_Q + rm_row * (FA_D * 2)
                                     + swz16(rm_row, (c * 16 + hi * 8) * 2));
-          st = mfma_bf16(kf[c].bf, qfr.bf, st);
+          st = mfma_bf16(qfr.bf, kf[c].bf, st);
           dofr.u4 = *(const uint4v*)(smem + BW64_LDS_DO + rm_row * (FA_D * 2)
                                      + swz16(rm_row, (c * 16 + hi * 8) * 2));
-          dpt = mfma_bf16(vf[c].bf, dofr.bf, dpt);
+          dpt = mfma_bf16(dofr.bf, vf[c].bf, dpt);
         }
 
+        // p / dS per element; lse & delta indexed by the REGISTER q row.
+        // acc rows r = 4g..4g+3 are CONSECUTIVE q rows 8g+4*hi.. so the
+        // 32 per-row values batch into 8 b128 broadcast reads (one wait)
+        // instead of 32 serialized b32 reads.
+        float lse_v[16], dlt_v[16];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          *(float4*)(lse_v + 4 * g) =
+              *(const float4*)(ld_sm + half * 32 + 8 * g + 4 * hi);
+          *(float4*)(dlt_v + 4 * g) =
+              *(const float4*)(ld_sm + 64 + half * 32 + 8 * g + 4 * hi);
+        }
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          int kg = kv0 + acc_row(r, hi);
-          bool masked = (causal && kg > my_q) || kg >= S || my_q >= S;
-          float p = masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse2);
+          int ql = half * 32 + acc_row(r, hi);
+          int qg = q0 + ql;
+          bool masked = (causal && my_kv > qg) || my_kv >= S || qg >= S;
+          float p =
+              masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse_v[r]);
           st[r] = p;
-          dpt[r] = p * (dpt[r] - dlt);
+          dpt[r] = p * (dpt[r] - dlt_v[r]);
         }
-        write_acc_tile(pw_p, st, lane);
-        write_acc_tile(pw_ds, dpt, lane);
 
-        // dV/dK over this wave's 64-column d-half, q chunk = this half
+        // prefetch the dV/dK B fragments NOW — they are independent of the
+        // pack below, so the 8 b128 LDS reads land under the pack VALU
+        frag_u dofr[2][2], qfr2[2][2];
 #pragma unroll
-        for (int cq = 0; cq < 2; ++cq) {
-          frag_u pa, da;
-          pa.bf = read_pw_row(pw_p, col, cq * 16 + hi * 8);
-          da.bf = read_pw_row(pw_ds, col, cq * 16 + hi * 8);
+        for (int cq = 0; cq < 2; ++cq)
 #pragma unroll
           for (int nb = 0; nb < 2; ++nb) {
             int d = dhalf * 64 + nb * 32 + col;
-            frag_u dofr, qfr;
-            dofr.u4 = *(const uint4v*)(smem + BW64_LDS_DOT
-                                       + d * (TR64_PITCH * 2)
-                                       + (half * 32 + cq * 16 + hi * 8) * 2);
-            dv_acc[nb] = mfma_bf16(pa.bf, dofr.bf, dv_acc[nb]);
-            qfr.u4 = *(const uint4v*)(smem + BW64_LDS_QT
-                                      + d * (TR64_PITCH * 2)
-                                      + (half * 32 + cq * 16 + hi * 8) * 2);
-            dk_acc[nb] = mfma_bf16(da.bf, qfr.bf, dk_acc[nb]);
+            dofr[cq][nb].u4 = *(const uint4v*)(smem + BW64_LDS_DOT
+                                               + d * (TR64_PITCH * 2)
+                                               + (half * 32 + cq * 16 + hi * 8) * 2);
+            qfr2[cq][nb].u4 = *(const uint4v*)(smem + BW64_LDS_QT
+                                               + d * (TR64_PITCH * 2)
+                                               + (half * 32 + cq * 16 + hi * 8) * 2);
           }
+
+        // repack accumulator rows (q) into A-fragment k-dim in REGISTERS
+        // (same pack pairs + permlane32_swap as the forward kernel's P
+        // repack; lane dim = kv is already in place)
+        frag_u pA[2], dA[2];
+#pragma unroll
+        for (int cc = 0; cc < 2; ++cc) {
+          uint b0 = pack_bf16x2(st[8 * cc + 0], st[8 * cc + 1]);
+          uint b1 = pack_bf16x2(st[8 * cc + 2], st[8 * cc + 3]);
+          uint b2 = pack_bf16x2(st[8 * cc + 4], st[8 * cc + 5]);
+          uint b3 = pack_bf16x2(st[8 * cc + 6], st[8 * cc + 7]);
+          {
+            auto r01 = __builtin_amdgcn_permlane32_swap(b0, b2, false, false);
+            b0 = r01[0]; b2 = r01[1];
+          }
+          {
+            auto r23 = __builtin_amdgcn_permlane32_swap(b1, b3, false, false);
+            b1 = r23[0]; b3 = r23[1];
+          }
+          pA[cc].u[0] = b0; pA[cc].u[1] = b1;
+          pA[cc].u[2] = b2; pA[cc].u[3] = b3;
+          uint c0 = pack_bf16x2(dpt[8 * cc + 0], dpt[8 * cc + 1]);
+          uint c1 = pack_bf16x2(dpt[8 * cc + 2], dpt[8 * cc + 3]);
+          uint c2 = pack_bf16x2(dpt[8 * cc + 4], dpt[8 * cc + 5]);
+          uint c3 = pack_bf16x2(dpt[8 * cc + 6], dpt[8 * cc + 7]);
+          {
+            auto r01 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+            c0 = r01[0]; c2 = r01[1];
+          }
+          {
+            auto r23 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+            c1 = r23[0]; c3 = r23[1];
+          }
+          dA[cc].u[0] = c0; dA[cc].u[1] = c1;
+          dA[cc].u[2] = c2; dA[cc].u[3] = c3;
         }
+
+        // dV/dK over this wave's 64-column d-half, q chunk = this half
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int cq = 0; cq < 2; ++cq)
+#pragma unroll
+          for (int nb = 0; nb < 2; ++nb) {
+            dv_acc[nb] = mfma_bf16(pA[cq].bf, dofr[cq][nb].bf, dv_acc[nb]);
+            dk_acc[nb] = mfma_bf16(dA[cq].bf, qfr2[cq][nb].bf, dk_acc[nb]);
+          }
+        __builtin_amdgcn_s_setprio(0);
       }
     }
 
@@ -359,6 +433,7 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     if (q0 + 64 < S) {
       write_tile64(nq, smem + BW64_LDS_Q, smem + BW64_LDS_QT);
       write_tile64(ndo, smem + BW64_LDS_DO, smem + BW64_LDS_DOT);
+      write_ld(nld);
     }
   }
 
@@ -570,7 +645,7 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
                                               (const short*)out,
                                               (float*)delta, rows);
   dim3 gkv(Hq, (S + 127) / 128, B);
-  size_t lds1 = 2 * 64 * FA_D * 2 + 2 * TR64_TILE_B + 8 * 2 * PW_BYTES;
+  size_t lds1 = 2 * 64 * FA_D * 2 + 2 * TR64_TILE_B + 128 * 4;
   fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
