@@ -62,6 +62,27 @@ def send(tensors: List[torch.Tensor], dst: int):
         dist.send(t.detach().contiguous().to(dev), dst=dst)
 
 
+def send_async(tensors: List[torch.Tensor], dst: int):
+    """Post the whole message (count, headers, payloads) as isends and
+    return (works, refs).  The caller must keep ``refs`` alive and wait the
+    works before reusing/freeing — the engine drains them at schedule end.
+    Blocking sends can mutually deadlock in the interleaved schedule where
+    two neighbours send to each other concurrently (fwd one way, bwd the
+    other); isends progress on the transport's own streams."""
+    dev = _p2p_device()
+    n = torch.tensor([len(tensors)], dtype=torch.int64,
+                     device=_header_device())
+    works, refs = [], [n]
+    works.append(dist.isend(n, dst))
+    for t in tensors:
+        h = _header_from(t).to(_header_device())
+        p = t.detach().contiguous().to(dev)
+        refs += [h, p]
+        works.append(dist.isend(h, dst))
+        works.append(dist.isend(p, dst))
+    return works, refs
+
+
 def recv_from(src: int) -> List[torch.Tensor]:
     dev = _p2p_device()
     n = torch.empty(1, dtype=torch.int64, device=_header_device())

@@ -83,8 +83,12 @@ class NxDPPModel(nn.Module):
     def _maybe_partition(self):
         if self.partitioned:
             return
+        # virtual/interleaved PP: the model splits into pp_size * C stages;
+        # rank r owns chunks c at global stage c*pp_size + r (Megatron
+        # mapping; reference scheduler.py:256-541)
+        n_stages = self.pp_size * self.virtual_pipeline_size
         split, stages = partition_model(
-            self.original_torch_module, self.pp_size,
+            self.original_torch_module, n_stages,
             pipeline_cuts=self.pipeline_cuts,
             transformer_layer_cls=self.transformer_layer_cls,
             input_names=self.input_names,
@@ -101,7 +105,12 @@ class NxDPPModel(nn.Module):
                 self.original_torch_module.forward).parameters.items()
             if par.default is not inspect.Parameter.empty
         }
-        self.local_stage_module = stages[self.pp_rank]
+        if self.virtual_pipeline_size == 1:
+            self.local_stage_module = stages[self.pp_rank]
+        else:
+            self.local_stage_module = nn.ModuleList(
+                [stages[c * self.pp_size + self.pp_rank]
+                 for c in range(self.virtual_pipeline_size)])
         self.partitioned = True
 
     def _analyze_stage_io(self, split):
@@ -179,7 +188,8 @@ class NxDPPModel(nn.Module):
                     mbs[i][k] = v
         return mbs
 
-    def _stage_forward(self, mb_kwargs, recvd: Optional[List[torch.Tensor]]):
+    def _stage_forward(self, mb_kwargs, recvd: Optional[List[torch.Tensor]],
+                       chunk: int = 0):
         if isinstance(self.local_stage_module, PipelineStageModule):
             if self.pp_rank == 0:
                 args = [mb_kwargs[k] for k in sorted(mb_kwargs)] \
@@ -189,7 +199,12 @@ class NxDPPModel(nn.Module):
             else:
                 out = self.local_stage_module(*recvd)
             return out
-        spec = self._stage_specs[self.pp_rank]
+        if self.virtual_pipeline_size > 1:
+            spec = self._stage_specs[chunk * self.pp_size + self.pp_rank]
+            module = self.local_stage_module[chunk]
+        else:
+            spec = self._stage_specs[self.pp_rank]
+            module = self.local_stage_module
         args = []
         for kind, key in spec:
             if kind == "input":
@@ -207,7 +222,7 @@ class NxDPPModel(nn.Module):
                 args.append(obj)
             else:
                 args.append(recvd[key])
-        return self.local_stage_module(*args)
+        return module(*args)
 
     @staticmethod
     def _as_list(out):
@@ -218,48 +233,55 @@ class NxDPPModel(nn.Module):
     def _run_schedule(self, schedule, kwargs, train: bool):
         self._maybe_partition()
         mbs = self._split_microbatches(kwargs)
-        recvd_inputs: Dict[int, List[torch.Tensor]] = {}
-        outputs: Dict[int, List[torch.Tensor]] = {}
+        # state keyed by (mb, chunk); chunk is 0 throughout for C == 1
+        recvd_inputs: Dict[Tuple[int, int], List[torch.Tensor]] = {}
+        outputs: Dict[Tuple[int, int], List[torch.Tensor]] = {}
         losses: List[torch.Tensor] = []
+        self._pending_sends = []
+        C = self.virtual_pipeline_size
+
+        def is_loss_stage(chunk):
+            return self.pp_rank == self.pp_size - 1 and chunk == C - 1 and \
+                self.output_loss_value_spec
 
         for task in schedule.steps():
+            key = (task.mb, task.chunk)
             if isinstance(task, RecvForward):
                 tensors = ppcomm.recv_from(self.prev_rank)
                 for t in tensors:
                     if t.is_floating_point():
                         t.requires_grad_(True)
-                recvd_inputs[task.mb] = tensors
+                recvd_inputs[key] = tensors
             elif isinstance(task, ForwardStep):
                 with torch.enable_grad() if train else torch.no_grad():
                     out = self._stage_forward(mbs[task.mb],
-                                              recvd_inputs.get(task.mb))
+                                              recvd_inputs.get(key),
+                                              task.chunk)
                 out_list = self._as_list(out)
-                outputs[task.mb] = out_list
-                if self.pp_rank == self.pp_size - 1 and \
-                        self.output_loss_value_spec:
+                outputs[key] = out_list
+                if is_loss_stage(task.chunk):
                     losses.append(out_list[0])
             elif isinstance(task, SendForward):
-                ppcomm.send(outputs[task.mb], self.next_rank)
+                self._pending_sends.append(
+                    ppcomm.send_async(outputs[key], self.next_rank))
             elif isinstance(task, SendForwardRecvBackward):
                 # both directions batched; grads arrive in mb order and
                 # attach to the oldest un-backwarded microbatch (FIFO)
-                grads = ppcomm.send_recv(outputs[task.mb], self.next_rank,
+                grads = ppcomm.send_recv(outputs[key], self.next_rank,
                                          self.next_rank)
                 self._pending_grads.append(grads)
             elif isinstance(task, RecvBackward):
                 self._pending_grads.append(ppcomm.recv_from(self.next_rank))
             elif isinstance(task, BackwardStep):
-                mb = task.mb
-                if self.pp_rank == self.pp_size - 1 and \
-                        self.output_loss_value_spec:
-                    loss = outputs[mb][0]
+                if is_loss_stage(task.chunk):
+                    loss = outputs[key][0]
                     (loss / self.num_microbatches).backward()
                 else:
                     grads = self._pending_grads.pop(0)
                     # downstream sent one grad per FLOAT output, in order
                     pairs = []
                     gi = 0
-                    for t in outputs[mb]:
+                    for t in outputs[key]:
                         if t.is_floating_point():
                             if t.requires_grad:
                                 pairs.append((t, grads[gi]))
@@ -267,16 +289,21 @@ class NxDPPModel(nn.Module):
                     torch.autograd.backward([p[0] for p in pairs],
                                             [p[1] for p in pairs])
                 # free the graph/output refs
-                outputs[mb] = [t.detach() for t in outputs[mb]]
+                outputs[key] = [t.detach() for t in outputs[key]]
             elif isinstance(task, SendBackward):
-                mb = task.mb
                 grads = [t.grad if t.grad is not None
                          else torch.zeros_like(t)
-                         for t in recvd_inputs[mb] if t.is_floating_point()]
-                ppcomm.send(grads, self.prev_rank)
-                del recvd_inputs[mb]
+                         for t in recvd_inputs[key] if t.is_floating_point()]
+                self._pending_sends.append(
+                    ppcomm.send_async(grads, self.prev_rank))
+                del recvd_inputs[key]
             elif isinstance(task, ReduceGrads):
                 pass  # DP grad sync happens in the optimizer step
+        # drain outstanding isends (payload refs held in _pending_sends)
+        for works, _refs in self._pending_sends:
+            for w in works:
+                w.wait()
+        self._pending_sends = []
         return losses
 
     def run_train(self, **kwargs):
