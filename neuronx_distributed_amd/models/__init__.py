@@ -1,1 +1,3 @@
 from .llama import LlamaConfig, LlamaForCausalLM, LlamaModel, get_config, CONFIGS
+from .mixtral import (MixtralConfig, MixtralForCausalLM, MixtralModel,
+                      get_moe_config, MOE_CONFIGS)

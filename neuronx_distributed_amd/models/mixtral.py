@@ -1,0 +1,204 @@
+"""Mixtral-style MoE causal LM on the MI355X parallel layers.
+
+Model-zoo parity with the reference's MoE helpers
+(``experimental/nxd_model_zoo/mixtral_model.py``, ``llama4_moe.py``; see
+SURVEY.md §2.4): GQA attention + top-k routed expert MLPs built from this
+repo's MoE stack (RouterTopK, ExpertMLPs with all-experts /
+capacity-factor / blockwise dispatch, optional SharedExperts, EP/TP
+sharding).  ``moe_frequency`` interleaves dense and MoE layers
+(llama4-style: every Nth layer is MoE); ``num_shared_experts`` adds an
+always-on shared expert branch (llama4/deepseek style).
+
+Aux (load-balancing) loss: router logits from every MoE layer are
+collected by the model forward and added to the CE loss scaled by
+``router_aux_loss_coef`` (reference loss_function.py parity).  For
+pipeline partitioning run with ``router_aux_loss_coef = 0`` so layers
+return plain tensors.
+"""
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ..moe.expert_mlps import ExpertMLPs
+from ..moe.loss_function import load_balancing_loss_func
+from ..moe.model import MoE
+from ..moe.routing import RouterTopK
+from ..moe.shared_experts import SharedExperts
+from ..parallel.layers import ColumnParallelLinear, ParallelEmbedding
+from ..parallel.loss_functions import parallel_cross_entropy
+from ..parallel.mappings import (
+    gather_from_sequence_parallel_region,
+    scatter_to_sequence_parallel_region,
+)
+from .. import ops
+from .llama import LlamaAttention, LlamaMLP, RMSNorm, _init_method
+
+torch.fx.wrap("parallel_cross_entropy")
+
+
+@dataclass
+class MixtralConfig:
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 8
+    vocab_size: int = 32000
+    max_position_embeddings: int = 4096
+    rope_theta: float = 1e6
+    rms_norm_eps: float = 1e-5
+    initializer_range: float = 0.02
+    sequence_parallel_enabled: bool = False
+    tie_word_embeddings: bool = False
+    # MoE
+    num_local_experts: int = 8
+    num_experts_per_tok: int = 2
+    router_aux_loss_coef: float = 0.02
+    capacity_factor: Optional[float] = None  # None = all-experts dispatch
+    moe_frequency: int = 1       # every Nth layer is MoE (1 = all layers)
+    num_shared_experts: int = 0  # >0 adds a SharedExperts branch
+    glu_mlp: bool = True
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.num_attention_heads
+
+
+MOE_CONFIGS = {
+    "mixtral-8x7b": MixtralConfig(),
+    "tiny-moe": MixtralConfig(
+        hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, vocab_size=256,
+        max_position_embeddings=128, num_local_experts=4,
+        num_experts_per_tok=2),
+}
+
+
+def get_moe_config(name: str, **overrides) -> MixtralConfig:
+    import dataclasses
+
+    return dataclasses.replace(MOE_CONFIGS[name], **overrides)
+
+
+def build_moe_layer(config: MixtralConfig) -> MoE:
+    """Assemble router + expert MLPs (+ shared experts) — the reference's
+    ``initialize_moe_module`` equivalent (mixtral_model.py)."""
+    sp = config.sequence_parallel_enabled
+    router = RouterTopK(config.num_local_experts, config.num_experts_per_tok,
+                        config.hidden_size, sequence_parallel_enabled=sp)
+    expert_mlps = ExpertMLPs(
+        config.num_local_experts, config.hidden_size,
+        config.intermediate_size, config.num_experts_per_tok,
+        capacity_factor=config.capacity_factor, glu_mlp=config.glu_mlp,
+        init_method=_init_method(config.initializer_range))
+    shared = None
+    if config.num_shared_experts > 0:
+        shared = SharedExperts(config.hidden_size, config.intermediate_size,
+                               config.num_shared_experts,
+                               init_method=_init_method(
+                                   config.initializer_range))
+    return MoE(router, expert_mlps, shared_experts=shared,
+               return_router_logits=config.router_aux_loss_coef > 0,
+               sequence_parallel_enabled=sp)
+
+
+class MixtralDecoderLayer(nn.Module):
+    def __init__(self, config: MixtralConfig, layer_idx: int = 0):
+        super().__init__()
+        sp = config.sequence_parallel_enabled
+        self.self_attn = LlamaAttention(config)
+        self.is_moe = (layer_idx + 1) % config.moe_frequency == 0
+        if self.is_moe:
+            self.block_sparse_moe = build_moe_layer(config)
+        else:  # llama4-style interleaving: dense layers in between
+            self.mlp = LlamaMLP(config)
+        self.input_layernorm = RMSNorm(config.hidden_size,
+                                       config.rms_norm_eps,
+                                       sequence_parallel_enabled=sp)
+        self.post_attention_layernorm = RMSNorm(
+            config.hidden_size, config.rms_norm_eps,
+            sequence_parallel_enabled=sp)
+
+    def forward(self, hidden, cos, sin, pos_offset=0, kv_cache=None):
+        h = hidden + self.self_attn(self.input_layernorm(hidden), cos, sin,
+                                    pos_offset, kv_cache)
+        normed = self.post_attention_layernorm(h)
+        if not self.is_moe:
+            return h + self.mlp(normed), None
+        out = self.block_sparse_moe(normed)
+        if isinstance(out, tuple):
+            moe_out, router_logits = out
+        else:
+            moe_out, router_logits = out, None
+        return h + moe_out, router_logits
+
+
+class MixtralModel(nn.Module):
+    def __init__(self, config: MixtralConfig):
+        super().__init__()
+        self.config = config
+        self.embed_tokens = ParallelEmbedding(
+            config.vocab_size, config.hidden_size,
+            init_method=_init_method(config.initializer_range),
+            sequence_parallel_enabled=False)
+        self.layers = nn.ModuleList(
+            MixtralDecoderLayer(config, i)
+            for i in range(config.num_hidden_layers))
+        self.norm = RMSNorm(
+            config.hidden_size, config.rms_norm_eps,
+            sequence_parallel_enabled=config.sequence_parallel_enabled)
+        cos, sin = ops.precompute_rope_freqs(config.max_position_embeddings,
+                                             config.head_dim,
+                                             config.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def forward(self, input_ids, pos_offset=0, kv_caches=None):
+        hidden = self.embed_tokens(input_ids)  # (B,S,H)
+        if self.config.sequence_parallel_enabled:
+            hidden = hidden.transpose(0, 1).contiguous()
+            hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
+        all_router_logits = []
+        for i, layer in enumerate(self.layers):
+            kc = kv_caches[i] if kv_caches is not None else None
+            hidden, router_logits = layer(hidden, self.rope_cos,
+                                          self.rope_sin, pos_offset, kc)
+            if router_logits is not None:
+                all_router_logits.append(router_logits)
+        hidden = self.norm(hidden)
+        if self.config.sequence_parallel_enabled:
+            hidden = gather_from_sequence_parallel_region(
+                hidden, seq_dim=0, to_model_parallel=True)
+            hidden = hidden.transpose(0, 1)
+        return hidden, all_router_logits
+
+
+class MixtralForCausalLM(nn.Module):
+    def __init__(self, config: MixtralConfig):
+        super().__init__()
+        self.config = config
+        self.model = MixtralModel(config)
+        self.lm_head = ColumnParallelLinear(
+            config.hidden_size, config.vocab_size, bias=False,
+            gather_output=False,
+            init_method=_init_method(config.initializer_range))
+        if config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+
+    def forward(self, input_ids, labels=None, pos_offset=0, kv_caches=None):
+        hidden, router_logits = self.model(input_ids, pos_offset, kv_caches)
+        logits = self.lm_head(hidden)  # (B,S,V/tp)
+        if labels is None:
+            return logits
+        loss = parallel_cross_entropy(
+            logits[:, :-1, :].contiguous(), labels[:, 1:].contiguous()).mean()
+        if router_logits and self.config.router_aux_loss_coef > 0:
+            aux = load_balancing_loss_func(
+                torch.cat(router_logits, dim=0),
+                self.config.num_local_experts,
+                self.config.num_experts_per_tok)
+            loss = loss + self.config.router_aux_loss_coef * aux
+        return loss

@@ -197,3 +197,79 @@ def test_block_indices():
     assert n == 3  # e0 -> 2 blocks, e1 -> 1 block
     assert b2e.tolist() == [0, 0, 1]
     assert tpi.tolist() == [0, 2, 3, -1, 1, 4]
+
+
+# ---------------------------------------------------------------------------
+# Mixtral / MoE model family
+# ---------------------------------------------------------------------------
+def _mixtral_train_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(get_moe_config("tiny-moe"))
+    opt = torch.optim.AdamW(m.parameters(), lr=1e-2)
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        loss = m(x, labels=x)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0] - 0.3, losses
+    return losses[-1]
+
+
+def test_mixtral_training_loss_decreases():
+    run_distributed(_mixtral_train_worker, world_size=1)
+
+
+def _mixtral_tp_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(get_moe_config("tiny-moe"))
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    loss = m(x, labels=x)
+    return float(loss)
+
+
+def test_mixtral_tp2_matches_tp1():
+    tp1 = run_distributed(_mixtral_tp_worker, world_size=1)[0]
+    tp2 = run_distributed(_mixtral_tp_worker, world_size=2)
+    assert abs(tp2[0] - tp2[1]) < 1e-5
+    assert abs(tp1 - tp2[0]) < 5e-3, (tp1, tp2)
+
+
+def _mixtral_llama4_worker(rank, world):
+    """moe_frequency=2 + shared experts (llama4/deepseek style) trains."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(get_moe_config("tiny-moe", moe_frequency=2,
+                                          num_shared_experts=1))
+    dense = [l for l in m.model.layers if not l.is_moe]
+    moe = [l for l in m.model.layers if l.is_moe]
+    assert len(dense) == 1 and len(moe) == 1
+    assert moe[0].block_sparse_moe.shared_experts is not None
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    loss = m(x, labels=x)
+    loss.backward()
+    assert all(p.grad is not None for p in m.parameters())
+    return float(loss)
+
+
+def test_mixtral_llama4_style():
+    run_distributed(_mixtral_llama4_worker, world_size=1)
