@@ -193,3 +193,31 @@ def test_top_level_api_surface():
         ColumnParallelLinear, RowParallelLinear, ParallelEmbedding,
         parallel_cross_entropy, initialize_model_parallel, clip_grad_norm,
         PARALLEL_MODULES, PARALLEL_FUNCTIONS)
+
+
+def test_training_metrics(tmp_path):
+    import time
+    from neuronx_distributed_amd.utils.training_metrics import (
+        Metric, Throughput, TrainingMetrics)
+
+    tp = Throughput(batch_size=4, world_size=2, grad_accum_usteps=2,
+                    moving_avg_window_size=4)
+    time.sleep(0.01)
+    v1 = tp.get_throughput()
+    assert v1 > 0
+    for _ in range(4):
+        time.sleep(0.002)
+        v = tp.get_throughput()
+    assert v > 0
+
+    f = str(tmp_path / "results.json")
+    tm = TrainingMetrics(f)
+    tm.store_parameters({"Model": "llama2-7b", "World size": 2})
+    tm.store_metrics([Metric("Throughput", 123.4, "seq/s"),
+                      Metric("FinalLoss", 2.5)])
+    tm.store_metrics([Metric("Throughput", 125.0, "seq/s")])
+    import json
+    d = json.load(open(f))
+    assert d["parameters"]["Model"] == "llama2-7b"
+    assert len(d["metrics"]) == 3
+    assert d["metrics"][0]["MetricName"] == "Throughput"
