@@ -45,3 +45,8 @@ class QuantizationConfig:
     quantized_dtype: QuantizedDtype = QuantizedDtype.INT8
     quantization_type: QuantizationType = QuantizationType.PER_CHANNEL_SYMMETRIC
     quantization_per_channel_axis: int = 0
+    # W8A8: dynamically quantize activations per-token to fp8 and run the
+    # GEMM on the gfx950 fp8 MFMA pipe via torch._scaled_mm (~2.1x the
+    # bf16 GEMM rate measured on MI355X).  False = weight-only (W8A16),
+    # dequant->bf16 GEMM.
+    quantize_activation: bool = False

@@ -24,10 +24,22 @@ from ..parallel.mappings import (
 from ..parallel.utils import divide, set_tensor_model_parallel_attributes
 from .quantization_config import (QuantizationConfig, QuantizationType,
                                   QuantizedDtype)
-from .quantization_utils import dequantize, quantize_symmetric
+from .quantization_utils import (dequantize, fp8_scaled_linear,
+                                 quantize_symmetric)
 
 
 class _QuantizedParallelLinearBase(BaseParallelLinear):
+    def _use_fp8_mm(self, input_: torch.Tensor) -> bool:
+        """fp8 MFMA path: W8A8 e4m3fn via torch._scaled_mm (inference —
+        the quantized weight carries no grad).  SP still runs the dequant
+        path (the all-gather is fused into the bf16 linear there)."""
+        cfg = self.quantization_config
+        return (cfg.quantize_activation
+                and cfg.quantized_dtype == QuantizedDtype.F8E4M3
+                and input_.is_cuda
+                and not self.sequence_parallel_enabled
+                and not input_.requires_grad)
+
     def _make_scale(self, out_rows: int, cfg: QuantizationConfig, shard: bool):
         if cfg.quantization_type == QuantizationType.PER_TENSOR_SYMMETRIC:
             scale = nn.Parameter(torch.ones(1), requires_grad=False)
@@ -66,6 +78,14 @@ class QuantizedColumnParallel(_QuantizedParallelLinearBase):
         self.compute_dtype = fl.dtype
 
     def forward(self, input_):
+        if self._use_fp8_mm(input_):
+            out = fp8_scaled_linear(input_, self.weight, self.scale,
+                                    self.compute_dtype)
+            if self.bias is not None:
+                out = out + self.bias
+            if self.gather_output:
+                out = gather_from_tensor_model_parallel_region(out)
+            return out
         w = dequantize(self.weight, self.scale, self.compute_dtype)
         if not self.sequence_parallel_enabled and \
                 ps.get_tensor_model_parallel_size() > 1:
@@ -89,12 +109,16 @@ class QuantizedRowParallel(_QuantizedParallelLinearBase):
         self.compute_dtype = fl.dtype
 
     def forward(self, input_):
-        w = dequantize(self.weight, self.scale, self.compute_dtype)
         if not self.input_is_parallel:
             input_ = scatter_to_tensor_model_parallel_region(input_)
-        out = linear_with_async_allreduce(
-            input_, w, None, async_grad_allreduce=False,
-            sequence_parallel_enabled=False)
+        if self._use_fp8_mm(input_):
+            out = fp8_scaled_linear(input_, self.weight, self.scale,
+                                    self.compute_dtype)
+        else:
+            w = dequantize(self.weight, self.scale, self.compute_dtype)
+            out = linear_with_async_allreduce(
+                input_, w, None, async_grad_allreduce=False,
+                sequence_parallel_enabled=False)
         if self.sequence_parallel_enabled:
             from ..parallel.mappings import (
                 reduce_scatter_to_sequence_parallel_region)

@@ -29,3 +29,26 @@ def quantize_symmetric(weight: torch.Tensor, cfg: QuantizationConfig):
 def dequantize(q_weight: torch.Tensor, scale: torch.Tensor,
                dtype=torch.bfloat16) -> torch.Tensor:
     return (q_weight.float() * scale.float()).to(dtype)
+
+
+def fp8_scaled_linear(x: torch.Tensor, q_weight: torch.Tensor,
+                      w_scale: torch.Tensor,
+                      out_dtype=torch.bfloat16) -> torch.Tensor:
+    """W8A8 fp8 GEMM on the gfx950 fp8 MFMA pipe: dynamic per-token
+    activation quantization (e4m3fn, rowwise scale) + per-channel weight
+    scales through ``torch._scaled_mm`` (hipBLASLt fp8; ~2.1x the bf16
+    rate on MI355X).  x (..., K) @ q_weight (N, K)^T -> (..., N)."""
+    lead = x.shape[:-1]
+    K = x.shape[-1]
+    x2 = x.reshape(-1, K)
+    amax = x2.abs().amax(dim=1, keepdim=True).float().clamp(min=1e-6)
+    xs = amax / 448.0
+    xq = (x2 / xs.to(x2.dtype)).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    N = q_weight.shape[0]
+    ws = w_scale.float().reshape(-1)
+    if ws.numel() == 1:
+        ws = ws.expand(N)
+    sb = ws.reshape(1, N).contiguous()
+    out = torch._scaled_mm(xq, q_weight.t(), scale_a=xs, scale_b=sb,
+                           out_dtype=out_dtype)
+    return out.reshape(*lead, N)
