@@ -178,8 +178,18 @@ class ExpertMLPs(nn.Module):
 
     def forward(self, hidden, expert_affinities, expert_index):
         """Dispatch (reference :1407-1499): training -> capacity_factor if
-        set (>0); <=0 -> blockwise; None -> all-experts."""
+        set (>0); <=0 -> blockwise; None -> all-experts.  Small inference
+        batches take the fused HIP decode path (K9): per-expert slot
+        blocks, weights streamed once, gather/SwiGLU/scatter fused."""
         hidden = copy_to_tensor_model_parallel_region(hidden)
+        if (not isinstance(hidden, torch.fx.Proxy) and not self.training
+                and self.glu_mlp and self.ep_size == 1 and hidden.is_cuda
+                and hidden.dtype == torch.bfloat16
+                and self.gate_up_proj.weight.dtype == torch.bfloat16
+                and hidden.shape[0] <= 512 and ops.moe_decode_available()):
+            return ops.moe_decode_glu(hidden, self.gate_up_proj.weight,
+                                      self.down_proj.weight,
+                                      expert_affinities, expert_index)
         if self.capacity_factor is not None and self.capacity_factor > 0:
             return self.forward_capacity_factor(hidden, expert_affinities,
                                                 expert_index)

@@ -330,3 +330,75 @@ def adamw_step(master: torch.Tensor, m: torch.Tensor, v: torch.Tensor,
                    ctypes.c_float(beta2), ctypes.c_float(eps),
                    ctypes.c_float(weight_decay), ctypes.c_float(inv_bc1),
                    ctypes.c_float(inv_bc2), _stream())
+
+
+# ---------------------------------------------------------------------------
+# Fused MoE decode (K9 parity; csrc/moe_decode.hip)
+# ---------------------------------------------------------------------------
+
+_MOE_MB = 16
+
+
+def moe_decode_available() -> bool:
+    lib = _load()
+    return lib is not None and hasattr(lib, "moe_decode_glu")
+
+
+def moe_decode_glu(hidden: torch.Tensor, gate_up_w: torch.Tensor,
+                   down_w: torch.Tensor, expert_affinities: torch.Tensor,
+                   expert_index: torch.Tensor) -> torch.Tensor:
+    """Fused decode MoE (inference): hidden (T,H) bf16, gate_up_w
+    (E,H,2I) fused [gate|up], down_w (E,I,H), affinities (T,E) float,
+    expert_index (T,k) long -> (T,H) bf16 (TP-partial, like ExpertMLPs).
+
+    Slots (token, expert hit) are grouped per expert into blocks of 16 so
+    each block streams its expert's weights once; gather, SwiGLU, affinity
+    scaling and the scatter-add all happen inside the two kernels."""
+    lib = _require_lib()
+    T, H = hidden.shape
+    E, _, twoI = gate_up_w.shape
+    I = twoI // 2
+    assert H % 64 == 0 and I % 64 == 0, (H, I)
+    k = expert_index.shape[1]
+    dev = hidden.device
+
+    flat_e = expert_index.reshape(-1)
+    order = torch.argsort(flat_e, stable=True)
+    sorted_e = flat_e[order].to(torch.long)
+    sorted_tok = (order // k).to(torch.int32)
+
+    counts = torch.bincount(flat_e, minlength=E)
+    blocks_per_e = (counts + _MOE_MB - 1) // _MOE_MB
+    nb = int(blocks_per_e.sum().item())
+    if nb == 0:
+        return hidden.new_zeros(T, H)
+    blockbase = torch.cumsum(torch.nn.functional.pad(blocks_per_e, (1, 0)),
+                             0)  # (E+1,)
+    countbase = torch.cumsum(torch.nn.functional.pad(counts, (1, 0)), 0)
+
+    within = torch.arange(T * k, device=dev) - countbase[sorted_e]
+    padded_idx = (blockbase[sorted_e] + within // _MOE_MB) * _MOE_MB + \
+        within % _MOE_MB
+
+    slot_token = torch.zeros(nb * _MOE_MB, dtype=torch.int32, device=dev)
+    slot_token[padded_idx] = sorted_tok
+    aff = torch.zeros(nb * _MOE_MB, dtype=torch.float32, device=dev)
+    aff[padded_idx] = expert_affinities.float()[
+        sorted_tok.long(), sorted_e]
+
+    block_expert = torch.repeat_interleave(
+        torch.arange(E, device=dev), blocks_per_e).to(torch.int32)
+    bi = torch.arange(nb, device=dev)
+    block_ord = bi - blockbase[block_expert.long()]
+    block_len = torch.clamp(counts[block_expert.long()] -
+                            _MOE_MB * block_ord, 0, _MOE_MB).to(torch.int32)
+
+    act = torch.empty(nb * _MOE_MB, I, dtype=torch.bfloat16, device=dev)
+    out = torch.zeros(T, H, dtype=torch.float32, device=dev)
+    hidden = hidden.contiguous()
+    gate_up_w = gate_up_w.contiguous()
+    down_w = down_w.contiguous()
+    lib.moe_decode_glu(_ptr(hidden), _ptr(gate_up_w), _ptr(down_w),
+                       _ptr(slot_token), _ptr(block_expert), _ptr(block_len),
+                       _ptr(aff), _ptr(act), _ptr(out), nb, H, I, _stream())
+    return out.to(hidden.dtype)
