@@ -111,7 +111,51 @@ class NxDPPModel(nn.Module):
             self.local_stage_module = nn.ModuleList(
                 [stages[c * self.pp_size + self.pp_rank]
                  for c in range(self.virtual_pipeline_size)])
+        self._setup_shared_weights(stages)
         self.partitioned = True
+
+    def _setup_shared_weights(self, stages):
+        """Tied parameters (e.g. embed_tokens/lm_head) that land in stages
+        on DIFFERENT pp ranks each accumulate only their stage's grad
+        contribution — they need a grad all-reduce over the owning ranks
+        (reference pipeline/model.py:750-832 shared-weight groups).
+
+        Same-rank sharing (C>1 chunks on one rank) needs nothing: the
+        stages hold the SAME Parameter object and autograd accumulates."""
+        import torch.distributed as dist
+
+        self._shared_weight_syncs = []
+        if not dist.is_initialized() or self.pp_size == 1:
+            return
+        P = self.pp_size
+        by_id: Dict[int, dict] = {}
+        for s, st in enumerate(stages):
+            for n, p in st.named_parameters():
+                ent = by_id.setdefault(id(p), {"stages": set(), "param": p})
+                ent["stages"].add(s % P)
+        pp_info = ps.get_group_info("pp")
+        my_rank = dist.get_rank()
+        for ent in by_id.values():  # deterministic order on every rank
+            owners = sorted(ent["stages"])
+            if len(owners) < 2:
+                continue
+            # one group per pp mesh row — new_group is collective over the
+            # WHOLE world, so every rank creates every row's group
+            for row in pp_info.mesh:
+                granks = [row[r] for r in owners]
+                group = dist.new_group(granks)
+                if my_rank in granks:
+                    self._shared_weight_syncs.append((ent["param"], group))
+
+    def _sync_shared_weight_grads(self):
+        import torch.distributed as dist
+
+        for p, group in getattr(self, "_shared_weight_syncs", []):
+            if not p.requires_grad:
+                continue
+            if p.grad is None:
+                p.grad = torch.zeros_like(p)
+            dist.all_reduce(p.grad, group=group)
 
     def _analyze_stage_io(self, split):
         """From the split top-level graph, derive for every stage: which
@@ -298,7 +342,10 @@ class NxDPPModel(nn.Module):
                     ppcomm.send_async(grads, self.prev_rank))
                 del recvd_inputs[key]
             elif isinstance(task, ReduceGrads):
-                pass  # DP grad sync happens in the optimizer step
+                # DP grad sync happens in the optimizer step; here only the
+                # cross-stage tied-weight grads are combined
+                if train:
+                    self._sync_shared_weight_grads()
         # drain outstanding isends (payload refs held in _pending_sends)
         for works, _refs in self._pending_sends:
             for w in works:

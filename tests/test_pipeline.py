@@ -251,3 +251,43 @@ def _interleaved_train_worker(rank, world):
 def test_interleaved_train_matches_dense():
     out = run_distributed(_interleaved_train_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-6
+
+
+def _tied_weights_worker(rank, world):
+    """tie_word_embeddings + PP2: the tied weight lives on stage 0 (embed)
+    and stage 1 (lm_head); its grad must equal the golden combined grad
+    after the shared-weight all-reduce."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+    from neuronx_distributed_amd.pipeline import NxDPPModel
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 pipeline_model_parallel_size=world)
+    cfg = get_config("tiny", tie_word_embeddings=True)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(cfg)
+    torch.manual_seed(0)
+    golden = LlamaForCausalLM(cfg)
+
+    pp_model = NxDPPModel(model, transformer_layer_cls=LlamaDecoderLayer,
+                          num_microbatches=2,
+                          input_names=["input_ids", "labels"])
+    torch.manual_seed(42)
+    x = torch.randint(0, 256, (4, 16))
+    loss = pp_model.run_train(input_ids=x, labels=x)
+    assert len(pp_model._shared_weight_syncs) == 1
+
+    ref_loss = golden(x, labels=x)
+    assert abs(loss.item() - ref_loss.item()) < 1e-4
+    ref_loss.backward()
+    gtied = golden.model.embed_tokens.weight.grad
+
+    tied_p = pp_model._shared_weight_syncs[0][0]
+    assert torch.allclose(tied_p.grad, gtied, atol=2e-4), \
+        (tied_p.grad - gtied).abs().max()
+    return loss.item()
+
+
+def test_pp2_tied_embeddings():
+    run_distributed(_tied_weights_worker, world_size=2)
