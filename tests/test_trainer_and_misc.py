@@ -221,3 +221,29 @@ def test_training_metrics(tmp_path):
     assert d["parameters"]["Model"] == "llama2-7b"
     assert len(d["metrics"]) == 3
     assert d["metrics"][0]["MetricName"] == "Throughput"
+
+
+def _neox_worker(rank, world):
+    """GPT-NeoX (parallel residual, partial rotary, LN+bias) trains at TP=world
+    and TP2 matches TP1."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (GPTNeoXForCausalLM,
+                                                get_neox_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    m = GPTNeoXForCausalLM(get_neox_config("gpt-neox-tiny"))
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    loss = m(x, labels=x)
+    loss.backward()
+    assert all(p.grad is not None for p in m.parameters())
+    return float(loss.detach())
+
+
+def test_gpt_neox_tp2_matches_tp1():
+    from tests.dist_utils import run_distributed
+    tp1 = run_distributed(_neox_worker, world_size=1)[0]
+    tp2 = run_distributed(_neox_worker, world_size=2)
+    assert abs(tp2[0] - tp2[1]) < 1e-5
+    assert abs(tp1 - tp2[0]) < 5e-3, (tp1, tp2)
