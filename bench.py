@@ -29,6 +29,8 @@ def parse_args():
     p.add_argument("--tp", type=int, default=0, help="TP degree (default: world)")
     p.add_argument("--pp", type=int, default=1)
     p.add_argument("--sequence-parallel", action="store_true")
+    p.add_argument("--no-sequence-parallel", action="store_true",
+                   help="disable the TP>1 default of sequence parallelism")
     p.add_argument("--layers", type=int, default=0,
                    help="override layer count (debug only; overridden runs "
                         "are marked invalid in the output)")
@@ -77,7 +79,11 @@ def main():
                                  pipeline_model_parallel_size=args.pp)
     nxd.parallel.model_parallel_manual_seed(1234)
 
-    overrides = {"sequence_parallel_enabled": args.sequence_parallel,
+    # SP is on by default for TP>1: same math (reduce-scatter+all-gather ==
+    # all-reduce volume) but the norms/elementwise shard tp-ways
+    use_sp = (args.sequence_parallel or tp > 1) and \
+        not args.no_sequence_parallel and args.seq % max(tp, 1) == 0
+    overrides = {"sequence_parallel_enabled": use_sp,
                  "max_position_embeddings": max(args.seq, 4096)}
     if args.layers:
         overrides["num_hidden_layers"] = args.layers
@@ -156,7 +162,7 @@ def main():
                 "global_batch": B,
                 "seq_len": S,
                 "parallelism": f"tp{tp}" + (f"_pp{args.pp}" if args.pp > 1 else "")
-                               + ("_sp" if args.sequence_parallel else ""),
+                               + ("_sp" if use_sp else ""),
                 "zero1": True,
                 "loss": float(loss.item()),
             },

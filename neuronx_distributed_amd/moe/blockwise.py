@@ -84,13 +84,19 @@ def blockwise_mm(hidden: torch.Tensor, expert_affinities: torch.Tensor,
     nb = block_to_expert.numel()
     xb = x.reshape(nb, block_size, H)
     out_b = torch.empty_like(xb)
+    from .. import ops as _ops
+
     for e in torch.unique(block_to_expert).tolist():
         sel = (block_to_expert == e).nonzero().reshape(-1)
         xe = xb[sel].reshape(-1, H)
         gu = xe @ gate_up_w[e]
         if glu:
-            I = gu.shape[-1] // 2
-            act = torch.nn.functional.silu(gu[..., :I]) * gu[..., I:]
+            if gu.is_cuda and gu.dtype == torch.bfloat16 and \
+                    _ops.is_available():
+                act = _ops.swiglu(gu)  # fused HIP silu(gate)*up
+            else:
+                I = gu.shape[-1] // 2
+                act = torch.nn.functional.silu(gu[..., :I]) * gu[..., I:]
         else:
             act = torch.nn.functional.gelu(gu)
         out_b[sel] = (act @ down_w[e]).reshape(-1, block_size, H)

@@ -7,7 +7,11 @@ executes via an instruction map (reference model.py:1716-1743).
 
 With real RCCL P2P the bidirectional steady-state exchanges are batched
 (SendFwdRecvBwd / SendBwdRecvFwd), which removes the reference's
-deadlock-ordering constraints (scheduler.py:226-233).
+deadlock-ordering constraints (scheduler.py:226-233).  The reference's
+odd/even rank scheduler for num_microbatches == pp_size (model.py:288-295)
+exists only to break its all-gather-P2P ordering cycles — with true
+bidirectional P2P the plain 1F1B stream is already cycle-free for every
+mb/pp combination, so no parity-special schedule is needed here.
 """
 
 from dataclasses import dataclass

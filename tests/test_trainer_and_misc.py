@@ -247,3 +247,27 @@ def test_gpt_neox_tp2_matches_tp1():
     tp2 = run_distributed(_neox_worker, world_size=2)
     assert abs(tp2[0] - tp2[1]) < 1e-5
     assert abs(tp1 - tp2[0]) < 5e-3, (tp1, tp2)
+
+
+def test_bench_multirank_cpu_smoke():
+    """bench.py's multi-rank path (the driver's N>1 launch shape) runs end
+    to end on gloo with the tiny model, prints the contract JSON line, and
+    enables SP by default at TP>1."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29771", os.path.join(repo, "bench.py"),
+           "--gpus", "2", "--steps", "1", "--warmup", "0", "--model", "tiny",
+           "--seq", "64", "--batch", "2", "--microbatch", "1"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
+                         cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    r = json.loads(line)
+    assert r["n_gpus"] == 2 and r["unit"] == "tokens/s"
+    assert r["config"]["parallelism"] == "tp2_sp"
+    assert r["value"] > 0
