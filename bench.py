@@ -103,7 +103,9 @@ def main():
                                lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
                                grad_clipping=True, max_norm=1.0)
 
-    B = args.batch or 4 * max(1, world)
+    # per-GPU batch 16 amortizes the fixed step costs (optimizer,
+    # norms): measured 19.3k -> 20.3k tokens/s on 1 GPU vs batch 4
+    B = args.batch or 16 * max(1, world)
     S = args.seq
     mbs = args.microbatch
     assert B % mbs == 0
