@@ -1,0 +1,85 @@
+"""hipGraph-captured decode engine.
+
+The eager decode step is launch-bound (~200 kernel dispatches per token);
+capturing one step in a hipGraph (``torch.cuda.CUDAGraph`` on ROCm) and
+replaying it per token removes that overhead — the MI355X answer to the
+reference's traced/compiled token-generation model (reference
+trace/model_builder + nxd_model bucket executor).
+
+The step is made capturable by driving every position-dependent piece of
+the model from DEVICE tensors (see ``LlamaAttention._decode_step`` and
+``KVCache.update`` tensor-pos paths): between replays only the contents
+of ``step_in`` (next token) and ``pos_t`` change.
+"""
+
+from typing import List, Optional
+
+import torch
+
+from .kv_cache import KVCache
+
+
+class GraphDecoder:
+    """Wraps a causal-LM for single-token decode via graph replay.
+
+    Usage:
+        logits = model(prompt, kv_caches=caches, pos_offset=0)   # prefill
+        dec = GraphDecoder(model, caches, start_pos=prompt_len)
+        logits = dec.step(next_tok)   # (B, 1, V/tp); repeat per token
+    """
+
+    def __init__(self, model, kv_caches: List[KVCache], start_pos: int,
+                 batch: int, device=None):
+        self.model = model
+        self.caches = kv_caches
+        device = device or torch.device("cuda",
+                                        torch.cuda.current_device())
+        self.step_in = torch.zeros(batch, 1, dtype=torch.long, device=device)
+        self.pos_t = torch.tensor([start_pos], dtype=torch.long,
+                                  device=device)
+        self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.logits: Optional[torch.Tensor] = None
+
+    def _eager(self):
+        return self.model(self.step_in, kv_caches=self.caches,
+                          pos_offset=self.pos_t)
+
+    def capture(self) -> bool:
+        """Capture one decode step; returns False (eager fallback) if
+        capture fails (e.g. a non-capturable collective)."""
+        try:
+            # TunableOp TUNING mode launches candidate kernels mid-step —
+            # incompatible with stream capture (observed GPU memory
+            # faults).  Replaying already-tuned selections is fine.
+            tun = torch.cuda.tunable
+            if tun.is_enabled() and tun.tuning_is_enabled():
+                return False
+        except Exception:
+            pass
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):  # warmup allocations off-graph
+                for _ in range(2):
+                    self._eager()
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self.logits = self._eager()
+            self.graph = g
+            return True
+        except Exception:
+            self.graph = None
+            return False
+
+    @torch.no_grad()
+    def step(self, next_tok: torch.Tensor) -> torch.Tensor:
+        """next_tok (B,) -> logits (B, 1, V/tp); advances the position."""
+        self.step_in.copy_(next_tok.view(-1, 1))
+        if self.graph is not None:
+            self.graph.replay()
+            out = self.logits
+        else:
+            out = self._eager()
+        self.pos_t += 1
+        return out

@@ -12,10 +12,16 @@ from .kv_cache import build_kv_caches
 
 @torch.no_grad()
 def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
-             sampler: Optional[Sampler] = None, eos_token_id: int = -1):
+             sampler: Optional[Sampler] = None, eos_token_id: int = -1,
+             use_cuda_graph: bool = True):
     """model: LlamaForCausalLM-compatible (vocab-parallel logits out).
-    input_ids (B, S) on the model's device; returns (B, S+new)."""
+    input_ids (B, S) on the model's device; returns (B, S+new).
+
+    On GPU the per-token decode step is captured in a hipGraph and
+    replayed (launch-bound otherwise); pass ``use_cuda_graph=False`` for
+    the eager loop."""
     from ..parallel import parallel_state as ps
+    from .decode_graph import GraphDecoder
 
     sampler = sampler or Sampler(do_sample=False)
     cfg = model.config
@@ -27,12 +33,25 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
                              S + max_new_tokens, cfg.head_dim,
                              device=input_ids.device)
 
-    # prefill
+    # prefill (eager, flash kernel)
     logits = model(input_ids, kv_caches=caches, pos_offset=0)
     next_tok = sampler(logits[:, -1, :])
     out = [input_ids, next_tok.unsqueeze(1)]
-    pos = S
 
+    if input_ids.is_cuda:
+        dec = GraphDecoder(model, caches, start_pos=S, batch=B,
+                           device=input_ids.device)
+        if use_cuda_graph:
+            dec.capture()
+        for _ in range(max_new_tokens - 1):
+            logits = dec.step(next_tok)
+            next_tok = sampler(logits[:, -1, :])
+            out.append(next_tok.unsqueeze(1))
+            if eos_token_id >= 0 and bool((next_tok == eos_token_id).all()):
+                break
+        return torch.cat(out, dim=1)
+
+    pos = S
     for _ in range(max_new_tokens - 1):
         step_in = next_tok.unsqueeze(1)
         logits = model(step_in, kv_caches=caches, pos_offset=pos)

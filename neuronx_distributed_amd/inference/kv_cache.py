@@ -19,9 +19,17 @@ class KVCache:
         self.v = torch.zeros_like(self.k)
         self.max_seq = max_seq
 
-    def update(self, k_new: torch.Tensor, v_new: torch.Tensor, pos: int):
+    def update(self, k_new: torch.Tensor, v_new: torch.Tensor, pos):
         """k_new (B,H,S,D) written at [pos:pos+S]; returns views of the
-        cache covering [0:pos+S]."""
+        cache covering [0:pos+S].
+
+        ``pos`` may be a device int64 tensor of shape (1,) (hipGraph
+        decode): the write becomes index_copy_ and the FULL buffers are
+        returned (attention masks by position — no dynamic shapes)."""
+        if isinstance(pos, torch.Tensor):
+            self.k.index_copy_(2, pos, k_new)
+            self.v.index_copy_(2, pos, v_new)
+            return self.k, self.v
         S = k_new.shape[2]
         self.k[:, :, pos:pos + S] = k_new
         self.v[:, :, pos:pos + S] = v_new

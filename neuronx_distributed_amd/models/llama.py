@@ -155,6 +155,11 @@ class LlamaAttention(nn.Module):
         q = q.reshape(B, S, self.num_heads_local, self.head_dim)
         k = k.reshape(B, S, self.num_kv_local, self.head_dim)
         v = v.reshape(B, S, self.num_kv_local, self.head_dim)
+        if isinstance(pos_offset, torch.Tensor):
+            # hipGraph-capturable decode: device-tensor position, S == 1,
+            # full-cache masked attention (no dynamic shapes)
+            return self._decode_step(q, k, v, cos, sin, pos_offset,
+                                     kv_cache, sp)
         q, k = ops.apply_rotary_pos_emb(q, k, cos, sin, pos_offset)
 
         # (B,S,h,D) -> (B,h,S,D)
@@ -165,6 +170,42 @@ class LlamaAttention(nn.Module):
             k, v = kv_cache.update(k, v, pos_offset)
         out = flash_attn_func(q, k, v, causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
+        if sp:
+            out = out.transpose(0, 1)
+        return self.o_proj(out)
+
+    def _decode_step(self, q, k, v, cos, sin, pos_t, kv_cache, sp):
+        """One-token decode with everything derived from the device tensor
+        ``pos_t`` (shape (1,)) so the whole step can be captured in a
+        hipGraph and replayed with only memory updates between steps."""
+        B = q.shape[0]
+        half = self.head_dim // 2
+        c = cos.index_select(0, pos_t).view(1, 1, 1, half)  # (1,1,1,D/2)
+        s = sin.index_select(0, pos_t).view(1, 1, 1, half)
+
+        def rope1(x):
+            x0 = x[..., :half].float()
+            x1 = x[..., half:].float()
+            return torch.cat([x0 * c - x1 * s, x1 * c + x0 * s],
+                             dim=-1).to(x.dtype)
+
+        q = rope1(q).transpose(1, 2)  # (B,h,1,D)
+        k = rope1(k).transpose(1, 2)
+        v = v.transpose(1, 2)
+        K, V = kv_cache.update(k, v, pos_t)  # FULL (B,Hkv,Smax,D) buffers
+        Smax = K.shape[2]
+        rep = self.num_heads_local // self.num_kv_local
+        scale = 1.0 / math.sqrt(self.head_dim)
+        # bf16 batched GEMV against the cache (fp32 accumulation inside the
+        # GEMM; casting K/V to fp32 would re-materialize the whole cache
+        # every step) -> (B,Hkv,rep,Smax)
+        qg = q.view(B, self.num_kv_local, rep, self.head_dim)
+        scores = (qg @ K.transpose(-1, -2)).float() * scale
+        idx = torch.arange(Smax, device=q.device)
+        scores = scores.masked_fill(idx > pos_t, float("-inf"))
+        probs = torch.softmax(scores, dim=-1)
+        out = probs.to(q.dtype) @ V  # (B,Hkv,rep,D)
+        out = out.reshape(B, 1, self.num_heads_local * self.head_dim)
         if sp:
             out = out.transpose(0, 1)
         return self.o_proj(out)

@@ -103,3 +103,31 @@ def test_sp_gpu_matches_dense():
     l1 = dense.cuda()(x, labels=x)
     l2 = sp.cuda()(x, labels=x)
     assert abs(l1.item() - l2.item()) < 5e-2, (l1.item(), l2.item())
+
+
+def test_graph_decode_matches_eager():
+    """hipGraph-captured decode produces the same greedy tokens as the
+    eager decode loop."""
+    import torch.distributed as dist
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.inference.generation import generate
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29551")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("test-d128")
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.bfloat16)
+    torch.manual_seed(0)
+    with torch.device("cuda"):
+        model = LlamaForCausalLM(cfg)
+    torch.set_default_dtype(prev)
+    model.eval()
+    x = torch.randint(0, cfg.vocab_size, (2, 33), device="cuda")
+    out_eager = generate(model, x, max_new_tokens=9, use_cuda_graph=False)
+    out_graph = generate(model, x, max_new_tokens=9, use_cuda_graph=True)
+    assert torch.equal(out_eager, out_graph), (out_eager, out_graph)
