@@ -136,44 +136,12 @@ __device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
   }
 }
 
-// per-wave [32][40] bf16 tile (rows k or q, padded cols): write a 32x32
-// accumulator-layout matrix (col = l&31, row = acc_row(r,hi)) as PAIRS
-// (consecutive regs are consecutive rows) -> 2 cols packed... rows differ,
-// so write scalar b16: 16 writes per lane.
-#define PW_PITCH 40  // elements; 80 B rows -> bank stride 20 (conflict-lite)
-__device__ __forceinline__ void write_acc_tile(char* tile, const f32x16& a,
-                                               int lane) {
-  int col = lane & 31, hi = lane >> 5;
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    int row = acc_row(r, hi);
-    *(short*)(tile + (row * PW_PITCH + col) * 2) = f2bits(a[r]);
-  }
-}
-
-// read an 8-element fragment (row fixed = l&31 style caller-supplied, cols
-// contiguous) from a PW_PITCH tile
-__device__ __forceinline__ bf16x8 read_pw_row(const char* tile, int row,
-                                              int col0) {
-  // (row*40 + col0)*2 with col0 in {0,8,16,24} is 16-byte aligned
-  frag_u f;
-  f.u4 = *(const uint4v*)(tile + (row * PW_PITCH + col0) * 2);
-  return f.bf;
-}
-
 // ---------------------------------------------------------------------------
-// dK/dV kernel: 4 waves, wave owns kv rows [kv0 + 32*wid, +32)
+// dK/dV kernel (8 waves over a 128-row kv block; see kernel comment)
+// LDS: Q rm 32KB + Qt/dOt transposed 36.9KB + lse/delta 512B — the P/dS
+// tiles live entirely in registers (pack + permlane32_swap repack).
 // ---------------------------------------------------------------------------
-// LDS: Q rm 16KB?? 32x128x2 = 8KB; dO rm 8KB; Qt 8KB; dOt 8KB;
-//      per-wave P^T + dS^T tiles 2*4*32*40*2 = 20KB  => ~52KB
 #define TR_TILE_B (FA_D * TR_PITCH * 2)  // 10240 B
-#define BW_LDS_Q 0
-#define BW_LDS_DO (32 * FA_D * 2)
-#define BW_LDS_QT (2 * 32 * FA_D * 2)
-#define BW_LDS_DOT (2 * 32 * FA_D * 2 + TR_TILE_B)
-#define BW_LDS_PW (2 * 32 * FA_D * 2 + 2 * TR_TILE_B)
-#define PW_BYTES (32 * PW_PITCH * 2)
-
 
 // ---- 64-row tile staging (512 threads, 2-row pairs) ----------------------
 #define TR64_PITCH 72   // 144-byte rows: 16B-aligned, banks 4*(9d mod 16)
@@ -490,7 +458,6 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 #define DQ_LDS_K 0
 #define DQ_LDS_KT (32 * FA_D * 2)
 #define DQ_LDS_V (32 * FA_D * 2 + TR_TILE_B)
-#define DQ_LDS_PW (2 * 32 * FA_D * 2 + TR_TILE_B)
 
 extern "C" __global__ void __launch_bounds__(256, 2)
 fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
@@ -531,7 +498,6 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const float dlt = deltap[lse_base + q_ld];
 
   f32x16 dq_acc[4] = {};
-  char* pw_ds = smem + DQ_LDS_PW + wid * PW_BYTES;
 
   // WG-uniform loop bound (all waves share barriers); per-wave causal
   // skipping happens via wave_active below
@@ -574,31 +540,53 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         float p = masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse2);
         dst[r] = p * (dpt[r] - dlt);
       }
-      // write dS as [q][k] (row q = col): cols k = acc_row pairs packed b32
-      {
-#pragma unroll
-        for (int r = 0; r < 16; r += 2) {
-          int k0 = acc_row(r, hi);  // r even: k0, r+1 -> k0+1 (consecutive)
-          uint pk = pack_bf16x2(dst[r], dst[r + 1]);
-          *(uint*)(pw_ds + (col * PW_PITCH + k0) * 2) = pk;
-        }
-      }
 
-      // dQ[q][d] += sum_k dS[q][k] K[k][d]
-      //   A = dS rows q=col (pw tile), B = K[k][d] via Kt rows d
+      // prefetch the 8 Kt B-fragments (independent of the pack below —
+      // the b128 LDS reads land under the pack VALU)
+      frag_u kfr2[2][4];
 #pragma unroll
-      for (int ck = 0; ck < 2; ++ck) {
-        frag_u da;
-        da.bf = read_pw_row(pw_ds, col, ck * 16 + hi * 8);
+      for (int ck = 0; ck < 2; ++ck)
 #pragma unroll
         for (int nb = 0; nb < 4; ++nb) {
           int d = nb * 32 + col;
-          frag_u kfr;
-          kfr.u4 = *(const uint4v*)(smem + DQ_LDS_KT + d * (TR_PITCH * 2)
-                                    + (ck * 16 + hi * 8) * 2);
-          dq_acc[nb] = mfma_bf16(da.bf, kfr.bf, dq_acc[nb]);
+          kfr2[ck][nb].u4 = *(const uint4v*)(smem + DQ_LDS_KT
+                                             + d * (TR_PITCH * 2)
+                                             + (ck * 16 + hi * 8) * 2);
+        }
+
+      // repack dS accumulator rows (kv) into A-fragment k-dim in
+      // REGISTERS (lane dim q already in place) — same pack pairs +
+      // permlane32_swap as the forward P repack; no pw LDS round-trip
+      frag_u dA[2];
+#pragma unroll
+      for (int cc = 0; cc < 2; ++cc) {
+        uint b0 = pack_bf16x2(dst[8 * cc + 0], dst[8 * cc + 1]);
+        uint b1 = pack_bf16x2(dst[8 * cc + 2], dst[8 * cc + 3]);
+        uint b2 = pack_bf16x2(dst[8 * cc + 4], dst[8 * cc + 5]);
+        uint b3 = pack_bf16x2(dst[8 * cc + 6], dst[8 * cc + 7]);
+        {
+          auto r01 = __builtin_amdgcn_permlane32_swap(b0, b2, false, false);
+          b0 = r01[0]; b2 = r01[1];
+        }
+        {
+          auto r23 = __builtin_amdgcn_permlane32_swap(b1, b3, false, false);
+          b1 = r23[0]; b3 = r23[1];
+        }
+        dA[cc].u[0] = b0; dA[cc].u[1] = b1;
+        dA[cc].u[2] = b2; dA[cc].u[3] = b3;
+      }
+
+      // dQ[q][d] += sum_k dS[q][k] K[k][d]
+      //   A = dS (registers, lane=q), B = K[k][d] via Kt rows d
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ck = 0; ck < 2; ++ck) {
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb) {
+          dq_acc[nb] = mfma_bf16(dA[ck].bf, kfr2[ck][nb].bf, dq_acc[nb]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     __syncthreads();
@@ -651,7 +639,9 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
       Hkv, S, scale, causal);
   dim3 gq(Hq, (S + 127) / 128, B);
-  size_t lds2 = 2 * 32 * FA_D * 2 + TR_TILE_B + 4 * PW_BYTES;
+  // staging layout needs 26.6 KB; the epilogue reuses LDS as 4 per-wave
+  // 32x128 transpose tiles = 32 KB, which dominates
+  size_t lds2 = 4 * 32 * FA_D * 2;
   fa_bwd_dq_kernel<<<gq, 256, lds2, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dq, B, Hq, Hkv, S,
