@@ -80,9 +80,12 @@ def main():
     nxd.parallel.model_parallel_manual_seed(1234)
 
     # SP is on by default for TP>1: same math (reduce-scatter+all-gather ==
-    # all-reduce volume) but the norms/elementwise shard tp-ways
+    # all-reduce volume) but the norms/elementwise shard tp-ways.
+    # (PP runs without SP: the FX pipeline partition traces the dense
+    # activation layout.)
     use_sp = (args.sequence_parallel or tp > 1) and \
-        not args.no_sequence_parallel and args.seq % max(tp, 1) == 0
+        not args.no_sequence_parallel and args.seq % max(tp, 1) == 0 and \
+        args.pp == 1
     overrides = {"sequence_parallel_enabled": use_sp,
                  "max_position_embeddings": max(args.seq, 4096)}
     if args.layers:
@@ -99,10 +102,6 @@ def main():
     torch.set_default_dtype(prev)
     model = model.to(device)
 
-    opt = NeuronZero1Optimizer(model.parameters(), torch.optim.AdamW,
-                               lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
-                               grad_clipping=True, max_norm=1.0)
-
     # per-GPU batch 16 amortizes the fixed step costs (optimizer,
     # norms): measured 19.3k -> 20.3k tokens/s on 1 GPU vs batch 4
     B = args.batch or 16 * max(1, world)
@@ -110,17 +109,44 @@ def main():
     mbs = args.microbatch
     assert B % mbs == 0
     n_micro = B // mbs
-    torch.manual_seed(4321)
-    data = [torch.randint(0, cfg.vocab_size, (mbs, S), device=device)
-            for _ in range(n_micro)]
 
-    def step():
-        opt.zero_grad()
-        for x in data:
-            loss = model(x, labels=x)
-            (loss / n_micro).backward()
-        opt.step()
-        return loss
+    if args.pp > 1:
+        # 1F1B pipeline over FX-partitioned stages (BASELINE config #3
+        # shape, e.g. --tp 2 --pp 4 on 8 GPUs)
+        from neuronx_distributed_amd.pipeline import NxDPPModel
+        from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+
+        model = NxDPPModel(model, transformer_layer_cls=LlamaDecoderLayer,
+                           num_microbatches=n_micro,
+                           input_names=["input_ids", "labels"])
+        params = list(model.local_parameters())
+    else:
+        params = list(model.parameters())
+
+    opt = NeuronZero1Optimizer(params, torch.optim.AdamW,
+                               lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
+                               grad_clipping=True, max_norm=1.0)
+
+    torch.manual_seed(4321)
+    if args.pp > 1:
+        full = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+
+        def step():
+            opt.zero_grad()
+            loss = model.run_train(input_ids=full, labels=full)
+            opt.step()
+            return loss
+    else:
+        data = [torch.randint(0, cfg.vocab_size, (mbs, S), device=device)
+                for _ in range(n_micro)]
+
+        def step():
+            opt.zero_grad()
+            for x in data:
+                loss = model(x, labels=x)
+                (loss / n_micro).backward()
+            opt.step()
+            return loss
 
     for _ in range(args.warmup):
         step()

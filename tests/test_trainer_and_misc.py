@@ -271,3 +271,26 @@ def test_bench_multirank_cpu_smoke():
     assert r["n_gpus"] == 2 and r["unit"] == "tokens/s"
     assert r["config"]["parallelism"] == "tp2_sp"
     assert r["value"] > 0
+
+
+def test_bench_pp_cpu_smoke():
+    """bench.py --pp 2: the 1F1B pipeline path runs end to end (tiny model,
+    gloo) and reports the contract line with parallelism tp1_pp2."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29772", os.path.join(repo, "bench.py"),
+           "--gpus", "2", "--steps", "1", "--warmup", "0", "--model", "tiny",
+           "--seq", "64", "--batch", "4", "--microbatch", "1",
+           "--tp", "1", "--pp", "2"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
+                         cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    r = json.loads(line)
+    assert r["config"]["parallelism"] == "tp1_pp2"
+    assert r["value"] > 0
