@@ -38,9 +38,10 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None):
     if q.is_cuda:
         from .. import ops
 
-        if q.shape[2] != k.shape[2]:
-            # rectangular attention (KV-cache decode): memory-bound batched
-            # GEMV via rocBLAS — the HIP prefill kernel requires Sq == Sk
+        if q.shape[2] != k.shape[2] or q.shape[3] != 128:
+            # rectangular attention (KV-cache decode) or head_dim != 128:
+            # composed path — the HIP MFMA kernel is D=128 (the production
+            # model shapes); other head dims run batched GEMMs + softmax
             return _torch_reference(q, k, v, causal, softmax_scale)
         if hasattr(ops, "flash_attn") and ops.flash_attn_available():
             return ops.flash_attn(q, k, v, causal=causal,
