@@ -177,8 +177,25 @@ class LlamaAttention(nn.Module):
     def _decode_step(self, q, k, v, cos, sin, pos_t, kv_cache, sp):
         """One-token decode with everything derived from the device tensor
         ``pos_t`` (shape (1,)) so the whole step can be captured in a
-        hipGraph and replayed with only memory updates between steps."""
+        hipGraph and replayed with only memory updates between steps.
+
+        On GPU with D=128 the whole chain (RoPE + cache append +
+        flash-decode + GQA) is ONE fused HIP kernel (ops.decode_attn_step);
+        the torch chain below is the CPU/odd-shape reference."""
         B = q.shape[0]
+        rep = self.num_heads_local // self.num_kv_local
+        if (q.is_cuda and self.head_dim == 128 and rep in (1, 2, 4, 8)
+                and q.dtype == torch.bfloat16 and kv_cache is not None
+                and ops.decode_attn_available()):
+            out = ops.decode_attn_step(
+                q.reshape(B, -1).contiguous(), k.reshape(B, -1).contiguous(),
+                v.reshape(B, -1).contiguous(), kv_cache.k, kv_cache.v,
+                cos, sin, pos_t.reshape(1), self.num_heads_local,
+                self.num_kv_local, 1.0 / math.sqrt(self.head_dim))
+            out = out.unsqueeze(1)  # (B, 1, H*D)
+            if sp:
+                out = out.transpose(0, 1)
+            return self.o_proj(out)
         half = self.head_dim // 2
         c = cos.index_select(0, pos_t).view(1, 1, 1, half)  # (1,1,1,D/2)
         s = sin.index_select(0, pos_t).view(1, 1, 1, half)

@@ -402,3 +402,70 @@ def moe_decode_glu(hidden: torch.Tensor, gate_up_w: torch.Tensor,
                        _ptr(slot_token), _ptr(block_expert), _ptr(block_len),
                        _ptr(aff), _ptr(act), _ptr(out), nb, H, I, _stream())
     return out.to(hidden.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Skinny-M decode GEMM (csrc/skinny_gemm.hip)
+# ---------------------------------------------------------------------------
+
+def skinny_gemm_available() -> bool:
+    lib = _load()
+    return lib is not None and hasattr(lib, "skinny_gemm")
+
+
+def skinny_linear(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """x2 (M<=32, K) bf16 @ w (N, K)^T -> (M, N) bf16 via the split-K
+    weight-streaming kernel (decode shapes run ~10x hipBLASLt's skinny
+    kernels; see profiles/README.md)."""
+    lib = _require_lib()
+    M, K = x2.shape
+    N = w.shape[0]
+    out = torch.zeros(M, N, dtype=torch.float32, device=x2.device)
+    lib.skinny_gemm(_ptr(x2), _ptr(w), _ptr(out), M, N, K, _stream())
+    return out.to(torch.bfloat16)
+
+
+def use_skinny_linear(inp: torch.Tensor, weight: torch.Tensor,
+                      sequence_parallel_enabled: bool = False) -> bool:
+    """Gate for the decode GEMM fast path (opt-in via NXDA_SKINNY_GEMM=1):
+    measured on MI355X, hipBLASLt's skinny kernels already run the decode
+    shapes at 1.8-6.6 TB/s — this kernel is kept for shapes/layouts where
+    the library falls over, not as the default."""
+    if os.environ.get("NXDA_SKINNY_GEMM", "0") != "1":
+        return False
+    return (not torch.is_grad_enabled() and not sequence_parallel_enabled
+            and inp.is_cuda and inp.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and inp.numel() // inp.shape[-1] <= 32
+            and weight.shape[-1] % 8 == 0 and inp.shape[-1] == weight.shape[-1]
+            and skinny_gemm_available())
+
+
+# ---------------------------------------------------------------------------
+# Fused decode attention (csrc/decode_attn.hip)
+# ---------------------------------------------------------------------------
+
+def decode_attn_available() -> bool:
+    lib = _load()
+    return lib is not None and hasattr(lib, "decode_attn")
+
+
+def decode_attn_step(q2: torch.Tensor, k2: torch.Tensor, v2: torch.Tensor,
+                     kcache: torch.Tensor, vcache: torch.Tensor,
+                     cos: torch.Tensor, sin: torch.Tensor,
+                     pos_t: torch.Tensor, Hq: int, Hkv: int,
+                     scale: float) -> torch.Tensor:
+    """One fused decode-attention step: RoPE(q,k) + cache append at the
+    device position ``pos_t`` + flash-decode over the cache + GQA.
+    q2 (B, Hq*128), k2/v2 (B, Hkv*128) bf16 pre-rope; kcache/vcache
+    (B, Hkv, Smax, 128); cos/sin (Smax, 64) fp32 -> out (B, Hq*128)."""
+    lib = _require_lib()
+    B = q2.shape[0]
+    Smax = kcache.shape[2]
+    assert Hq // Hkv in (1, 2, 4, 8), "GQA rep must be 1/2/4/8"
+    assert cos.dtype == torch.float32 and pos_t.dtype == torch.int64
+    out = torch.empty(B, Hq * 128, dtype=torch.bfloat16, device=q2.device)
+    lib.decode_attn(_ptr(q2), _ptr(k2), _ptr(v2), _ptr(kcache), _ptr(vcache),
+                    _ptr(cos), _ptr(sin), _ptr(pos_t), _ptr(out), B, Hq, Hkv,
+                    Smax, ctypes.c_float(scale), _stream())
+    return out

@@ -197,6 +197,16 @@ class LinearWithAsyncCommunication(torch.autograd.Function):
 def linear_with_async_allreduce(input_, weight, bias=None,
                                 async_grad_allreduce=False,
                                 sequence_parallel_enabled=False):
+    from .. import ops as _ops
+
+    if _ops.use_skinny_linear(input_, weight, sequence_parallel_enabled):
+        # decode fast path: M <= 32 tokens, inference — the split-K
+        # weight-streaming HIP kernel instead of hipBLASLt skinny kernels
+        flat = input_.reshape(-1, input_.shape[-1]).contiguous()
+        out = _ops.skinny_linear(flat, weight)
+        if bias is not None:
+            out = out + bias
+        return out.reshape(*input_.shape[:-1], weight.shape[0])
     args = cast_if_autocast_enabled(input_, weight, bias)
     with torch.amp.autocast("cuda", enabled=False):
         return LinearWithAsyncCommunication.apply(

@@ -221,6 +221,18 @@ class GQAQKVColumnParallelLinear(BaseParallelLinear):
                 model_state_dict[key] = self._kv_local_rows(model_state_dict[key])
 
     def forward(self, input_):
+        from .. import ops as _ops
+
+        if _ops.use_skinny_linear(input_, self.weight_q,
+                                  self.sequence_parallel_enabled) and \
+                self.bias_q is None:
+            flat = input_.reshape(-1, input_.shape[-1]).contiguous()
+            lead = input_.shape[:-1]
+            q = _ops.skinny_linear(flat, self.weight_q)
+            k = _ops.skinny_linear(flat, self.weight_k)
+            v = _ops.skinny_linear(flat, self.weight_v)
+            return (q.reshape(*lead, -1), k.reshape(*lead, -1),
+                    v.reshape(*lead, -1))
         q, k, v = _QKVLinearWithAsyncCommunication.apply(
             input_, self.weight_q, self.weight_k, self.weight_v,
             self.bias_q, self.bias_k, self.bias_v,

@@ -323,3 +323,16 @@ def test_moe_decode_imbalanced_routing():
     act = torch.nn.functional.silu(gu[:, :I]) * gu[:, I:]
     ref = act @ w_d[2].float()
     _cmp(out, ref.to(torch.bfloat16), atol=2e-2, name="moe_decode_imb")
+
+
+def test_skinny_gemm_matches_matmul():
+    """Split-K skinny-M decode GEMM vs torch matmul (fp32 reference)."""
+    torch.manual_seed(0)
+    for M, N, K in [(32, 4096, 4096), (8, 1024, 11008), (1, 512, 256),
+                    (32, 1000, 4096)]:
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.05
+        out = ops.skinny_linear(x, w)
+        ref = (x.float() @ w.float().t())
+        _cmp(out, ref.to(torch.bfloat16), atol=2e-2,
+             name=f"skinny_{M}x{N}x{K}")
