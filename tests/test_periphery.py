@@ -83,3 +83,36 @@ def test_mx_quantization():
         wd = dequantize_mx(q, s, dtype=torch.float32)
         rel = (wd - w).abs().max() / w.abs().max()
         assert rel < tol, (fmt, rel.item())
+
+
+def test_quantization_observer():
+    import torch.nn as nn
+    from neuronx_distributed_amd.quantization.observer import (
+        attach_observers, collect_activation_scales)
+
+    m = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    obs = attach_observers(m, nn.Linear)
+    assert len(obs) == 2
+    x = torch.randn(32, 8) * 3
+    m(x)
+    scales = collect_activation_scales(obs, qmax=448.0)
+    assert set(scales) == set(obs)
+    assert abs(scales["0"].item() - x.abs().max().item() / 448.0) < 1e-5
+
+
+def test_tensor_replacement():
+    import torch.nn as nn
+    from neuronx_distributed_amd.utils.tensor_replacement import TensorReplacer
+
+    m = nn.Sequential(nn.Linear(4, 4), nn.Linear(4, 4))
+    x = torch.randn(2, 4)
+    base = m(x)
+    fixed = torch.ones(2, 4)
+    with TensorReplacer(m).replace("0", fixed):
+        out = m(x)
+    assert torch.allclose(out, m[1](fixed))
+    # scaling callable + restoration after exit
+    with TensorReplacer(m).replace("0", lambda o: o * 0):
+        out0 = m(x)
+    assert torch.allclose(out0, m[1](torch.zeros(2, 4)))
+    assert torch.allclose(m(x), base)
