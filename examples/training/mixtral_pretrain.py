@@ -9,6 +9,10 @@ reference's MoE training examples).  Launch:
 
 import argparse
 import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                "..", ".."))
 
 import torch
 import torch.distributed as dist
