@@ -317,3 +317,26 @@ def test_bench_tp_pp_3d_cpu_smoke():
     r = json.loads(line)
     assert r["config"]["parallelism"] == "tp2_pp2"
     assert r["value"] > 0
+
+
+def test_bench_tp4_cpu_smoke():
+    """bench.py at TP=4 (world 4, gloo): exercises KV-head replication
+    (tiny model: 2 kv heads x tp4 -> kv_size_multiplier 2) through the
+    full training step."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+           "--master-port", "29774", os.path.join(repo, "bench.py"),
+           "--gpus", "4", "--steps", "1", "--warmup", "0", "--model", "tiny",
+           "--seq", "64", "--batch", "4", "--microbatch", "2"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
+                         cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    r = json.loads(line)
+    assert r["config"]["parallelism"] == "tp4_sp"
+    assert r["value"] > 0
