@@ -366,28 +366,46 @@ class RowParallelLinear(BaseParallelLinear):
 
 
 class ParallelEmbedding(BaseParallelLinear):
-    """Embedding with the vocab dim sharded over TP (reference
-    layers.py:186,334-378): out-of-shard ids are masked to 0, looked up
-    locally, masked out, and the partial embeddings all-reduced (or
-    reduce-scattered along seq when sequence-parallel)."""
+    """Embedding sharded over TP (reference layers.py:186,334-431).
+
+    Default: VOCAB-dim shard — out-of-shard ids are masked to 0, looked
+    up locally, masked out, and the partial embeddings all-reduced (or
+    reduce-scattered along seq when sequence-parallel).
+    ``shard_along_embedding=True``: EMBEDDING-dim shard — every rank
+    holds the full vocab over H/tp columns; the local lookup is
+    all-gathered along the last dim."""
 
     def __init__(self, num_embeddings, embedding_dim, init_method=None,
                  dtype=None, device=None, padding_idx=None,
                  sequence_parallel_enabled=False, shard_along_embedding=False,
                  pad=False):
         super().__init__()
-        assert not shard_along_embedding, "embedding-dim sharding: TODO"
         self.num_embeddings = num_embeddings
         self.embedding_dim = embedding_dim
         self.padding_idx = padding_idx
         self.sequence_parallel_enabled = sequence_parallel_enabled
+        self.shard_along_embedding = shard_along_embedding
         world = ps.get_tensor_model_parallel_size()
         tp_rank = ps.get_tensor_model_parallel_rank()
+        dtype = dtype or torch.get_default_dtype()
+        init_method = init_method or _init_normal(1.0)
+
+        if shard_along_embedding:
+            self.embedding_dim_per_partition = divide(embedding_dim, world)
+            self.num_embeddings_per_partition = num_embeddings
+            self.start_index = 0
+            self.end_index = num_embeddings
+            self.weight = nn.Parameter(
+                torch.empty(num_embeddings, self.embedding_dim_per_partition,
+                            dtype=dtype, device=device))
+            _initialize_affine_weight(
+                self.weight, num_embeddings, embedding_dim,
+                self.embedding_dim_per_partition, partition_dim=1,
+                init_method=init_method, dtype=dtype)
+            return
         self.num_embeddings_per_partition = divide(num_embeddings, world)
         self.start_index = tp_rank * self.num_embeddings_per_partition
         self.end_index = self.start_index + self.num_embeddings_per_partition
-        dtype = dtype or torch.get_default_dtype()
-        init_method = init_method or _init_normal(1.0)
 
         self.weight = nn.Parameter(
             torch.empty(self.num_embeddings_per_partition, embedding_dim,
@@ -399,6 +417,12 @@ class ParallelEmbedding(BaseParallelLinear):
 
     def forward(self, input_):
         world = ps.get_tensor_model_parallel_size()
+        if self.shard_along_embedding:
+            local = F.embedding(input_, self.weight,
+                                padding_idx=self.padding_idx)
+            if world == 1:
+                return local
+            return gather_from_tensor_model_parallel_region(local)
         if world > 1:
             input_mask = (input_ >= self.start_index) & (input_ < self.end_index)
             masked_input = (input_ - self.start_index) * input_mask

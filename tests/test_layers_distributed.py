@@ -196,3 +196,27 @@ def _deterministic_init_worker(rank, world):
 ])
 def test_distributed_layer(worker):
     run_distributed(worker, world_size=2)
+
+
+def _embed_dim_shard_worker(rank, world):
+    """shard_along_embedding=True: full vocab per rank, H/tp columns,
+    all-gathered output matches the dense embedding."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.layers import ParallelEmbedding
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    emb = ParallelEmbedding(32, 16, shard_along_embedding=True)
+    torch.manual_seed(0)
+    dense = ParallelEmbedding(32, 16)  # world>1 vocab-shard; compare paths
+    x = torch.randint(0, 32, (2, 5))
+    out = emb(x)
+    ref = dense(x)
+    assert out.shape == ref.shape == (2, 5, 16)
+    assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
+    return float(out.sum())
+
+
+def test_embedding_dim_sharding():
+    outs = run_distributed(_embed_dim_shard_worker, world_size=2)
+    assert abs(outs[0] - outs[1]) < 1e-4
