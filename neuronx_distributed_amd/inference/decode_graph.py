@@ -68,7 +68,12 @@ class GraphDecoder:
                 self.logits = self._eager()
             self.graph = g
             return True
-        except Exception:
+        except Exception as e:
+            from ..utils.logger import get_logger
+
+            get_logger(__name__).warning(
+                "hipGraph capture failed (%s: %s) — decoding eagerly",
+                type(e).__name__, e)
             self.graph = None
             return False
 
