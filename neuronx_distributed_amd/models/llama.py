@@ -55,9 +55,14 @@ class LlamaConfig:
     initializer_range: float = 0.02
     sequence_parallel_enabled: bool = False
     tie_word_embeddings: bool = False
+    # set by pad_llama_config: padded head counts keep the ORIGINAL head
+    # size (hidden_size // original_num_heads)
+    head_dim_override: Optional[int] = None
 
     @property
     def head_dim(self):
+        if self.head_dim_override is not None:
+            return self.head_dim_override
         return self.hidden_size // self.num_attention_heads
 
 

@@ -220,3 +220,47 @@ def _embed_dim_shard_worker(rank, world):
 def test_embedding_dim_sharding():
     outs = run_distributed(_embed_dim_shard_worker, world_size=2)
     assert abs(outs[0] - outs[1]) < 1e-4
+
+
+def _pad_worker(rank, world):
+    """Head padding: a 5-head model's checkpoint padded to 6 heads (zero
+    extra qkv rows + zero o_proj columns) computes the SAME function; the
+    padded config divides tp=2."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.parallel.pad import (pad_attention_state_dict,
+                                                      pad_llama_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    cfg = get_config("tiny", hidden_size=40, num_attention_heads=5,
+                     num_key_value_heads=1, intermediate_size=64)
+    padded = pad_llama_config(cfg, 2)
+    assert padded.num_attention_heads == 6 and padded.head_dim == 8
+
+    if world == 1:
+        torch.manual_seed(0)
+        dense = LlamaForCausalLM(cfg)
+        sd = pad_attention_state_dict(dense.state_dict(), cfg, padded)
+        pm = LlamaForCausalLM(padded)
+        pm.load_state_dict(sd)
+        x = torch.randint(0, cfg.vocab_size, (2, 8))
+        ref = dense(x)
+        out = pm(x)
+        assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
+        return float(out.sum())
+    # tp2: padded config constructs and runs (5 heads would not divide)
+    torch.manual_seed(0)
+    pm = LlamaForCausalLM(padded)
+    x = torch.randint(0, cfg.vocab_size, (2, 8))
+    loss = pm(x, labels=x)
+    loss.backward()
+    return float(loss.detach())
+
+
+def test_head_padding_equivalence():
+    run_distributed(_pad_worker, world_size=1)
+
+
+def test_head_padding_tp2():
+    out = run_distributed(_pad_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-5
