@@ -109,3 +109,40 @@ def _sampler_worker(rank, world):
 
 def test_sampler_topk():
     run_distributed(_sampler_worker, world_size=1)
+
+
+def _spec_worker(rank, world):
+    """Greedy speculative decoding returns EXACTLY the target-only greedy
+    tokens — with a perfect draft (the target itself, rate 1.0) and with a
+    mismatched draft (random init, partial acceptance)."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.inference.generation import generate
+    from neuronx_distributed_amd.inference.speculation import (
+        speculative_generate)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("tiny")
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(cfg).eval()
+    torch.manual_seed(99)
+    draft = LlamaForCausalLM(cfg).eval()
+
+    torch.manual_seed(1)
+    x = torch.randint(0, cfg.vocab_size, (2, 9))
+    ref = generate(target, x, max_new_tokens=12)
+
+    out_self, rate_self = speculative_generate(target, target, x,
+                                               max_new_tokens=12, spec_len=3)
+    assert torch.equal(out_self, ref), (out_self, ref)
+    assert rate_self == 1.0, rate_self
+
+    out_mix, rate_mix = speculative_generate(target, draft, x,
+                                             max_new_tokens=12, spec_len=3)
+    assert torch.equal(out_mix, ref), (out_mix, ref)
+    assert 0.0 <= rate_mix <= 1.0
+    return rate_mix
+
+
+def test_speculative_decoding():
+    run_distributed(_spec_worker, world_size=1)
