@@ -99,3 +99,64 @@ def speculative_generate(target, draft, input_ids: torch.Tensor,
     tokens = torch.cat(out, dim=1)[:, : S + max_new_tokens]
     rate = n_accepted / max(1, n_rounds * spec_len)
     return tokens, rate
+
+
+@torch.no_grad()
+def medusa_generate(target, medusa_heads, input_ids: torch.Tensor,
+                    max_new_tokens: int = 32):
+    """Medusa-style speculation: ``medusa_heads`` (iterable of
+    utils.medusa_utils.MedusaHead, head i predicting the token at offset
+    i+2) propose a chain from the LAST HIDDEN STATE in one shot — no
+    autoregressive draft model — and the target verifies the chunk exactly
+    like :func:`speculative_generate`.  Greedy output is identical to
+    plain generation."""
+    from ..utils.sampling import Sampler
+
+    sampler = Sampler(do_sample=False)
+    heads = list(medusa_heads)
+    k = len(heads)
+    B, S = input_ids.shape
+
+    from ..parallel import parallel_state as ps
+
+    cfg = target.config
+    tp = ps.get_tensor_model_parallel_size()
+    kv_mult = max(1, tp // cfg.num_key_value_heads)
+    n_kv_local = cfg.num_key_value_heads * kv_mult // tp
+    caches = build_kv_caches(cfg.num_hidden_layers, B, n_kv_local,
+                             S + max_new_tokens + k + 1, cfg.head_dim,
+                             device=input_ids.device)
+
+    hidden = target.model(input_ids, pos_offset=0, kv_caches=caches)
+    tok = sampler(target.lm_head(hidden)[:, -1, :])
+    h_last = hidden[:, -1, :]
+    out = [input_ids, tok.unsqueeze(1)]
+    pos = S
+    produced = 1
+    n_rounds = 0
+    n_accepted = 0
+
+    while produced < max_new_tokens:
+        kk = min(k, max_new_tokens - produced)
+        props = [sampler(heads[i](h_last)) for i in range(kk)]
+        dmat = torch.stack(props, dim=1)  # (B, kk)
+        chunk = torch.cat([tok.unsqueeze(1), dmat], dim=1)
+        hidden = target.model(chunk, pos_offset=pos, kv_caches=caches)
+        tl = target.lm_head(hidden)
+        tmat = torch.stack([sampler(tl[:, i, :])
+                            for i in range(kk + 1)], dim=1)
+        agree = (dmat == tmat[:, :-1]).all(dim=0)
+        j = int(agree.cumprod(dim=0).sum().item())
+        for i in range(j):
+            out.append(dmat[:, i].unsqueeze(1))
+        out.append(tmat[:, j].unsqueeze(1))
+        h_last = hidden[:, j, :]
+        tok = tmat[:, j]
+        produced += j + 1
+        pos += j + 1
+        n_rounds += 1
+        n_accepted += j
+
+    tokens = torch.cat(out, dim=1)[:, : S + max_new_tokens]
+    rate = n_accepted / max(1, n_rounds * k)
+    return tokens, rate

@@ -146,3 +146,32 @@ def _spec_worker(rank, world):
 
 def test_speculative_decoding():
     run_distributed(_spec_worker, world_size=1)
+
+
+def _medusa_worker(rank, world):
+    """Medusa-head speculation: greedy output equals plain generation."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.inference.generation import generate
+    from neuronx_distributed_amd.inference.speculation import medusa_generate
+    from neuronx_distributed_amd.utils.medusa_utils import MedusaHead
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("tiny")
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(cfg).eval()
+    torch.manual_seed(7)
+    heads = torch.nn.ModuleList(
+        [MedusaHead(cfg.hidden_size, cfg.vocab_size) for _ in range(3)])
+
+    torch.manual_seed(1)
+    x = torch.randint(0, cfg.vocab_size, (2, 9))
+    ref = generate(target, x, max_new_tokens=10)
+    out, rate = medusa_generate(target, heads, x, max_new_tokens=10)
+    assert torch.equal(out, ref), (out, ref)
+    assert 0.0 <= rate <= 1.0
+    return rate
+
+
+def test_medusa_generation():
+    run_distributed(_medusa_worker, world_size=1)
