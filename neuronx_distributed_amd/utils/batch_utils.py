@@ -22,3 +22,21 @@ def get_batch_on_this_context_parallel_rank(batch: dict, seq_dim: int = 1):
         else:
             out[k] = v
     return out, r * (chunk_len or 0)
+
+
+def create_dp_dataloader(dataset, batch_size: int, shuffle: bool = True,
+                         seed: int = 0, drop_last: bool = True, **kwargs):
+    """DataLoader sharded over the DATA-parallel group (reference
+    examples/training create_pretraining_dataset pattern): every TP/PP/CP
+    rank inside one DP replica sees the SAME batches (they compute one
+    model replica together); different DP ranks see disjoint shards."""
+    from torch.utils.data import DataLoader
+    from torch.utils.data.distributed import DistributedSampler
+
+    dp = ps.get_data_parallel_size()
+    dp_rank = ps.get_data_parallel_rank()
+    sampler = DistributedSampler(dataset, num_replicas=dp, rank=dp_rank,
+                                 shuffle=shuffle, seed=seed,
+                                 drop_last=drop_last)
+    return DataLoader(dataset, batch_size=batch_size, sampler=sampler,
+                      drop_last=drop_last, **kwargs)

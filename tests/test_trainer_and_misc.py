@@ -340,3 +340,35 @@ def test_bench_tp4_cpu_smoke():
     r = json.loads(line)
     assert r["config"]["parallelism"] == "tp4_sp"
     assert r["value"] > 0
+
+
+def _dp_loader_worker(rank, world):
+    """create_dp_dataloader: TP ranks share batches; DP shards them."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.utils.batch_utils import create_dp_dataloader
+
+    # tp=2 at world 2 -> dp=1: both ranks see identical batches
+    ps.initialize_model_parallel(tensor_model_parallel_size=2)
+    data = torch.arange(32).reshape(16, 2)
+    dl = create_dp_dataloader(list(data), batch_size=4, shuffle=True, seed=1)
+    first = next(iter(dl))
+    return [int(x) for x in first.reshape(-1)]
+
+
+def _dp_loader_worker_dp2(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.utils.batch_utils import create_dp_dataloader
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)  # dp=2
+    data = list(range(16))
+    dl = create_dp_dataloader(data, batch_size=2, shuffle=False)
+    seen = [int(x) for b in dl for x in b]
+    return seen
+
+
+def test_dp_dataloader():
+    out = run_distributed(_dp_loader_worker, world_size=2)
+    assert out[0] == out[1]  # same tp replica -> same data
+    shards = run_distributed(_dp_loader_worker_dp2, world_size=2)
+    assert set(shards[0]).isdisjoint(shards[1])
+    assert len(shards[0]) == len(shards[1]) == 8
