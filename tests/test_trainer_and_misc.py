@@ -372,3 +372,39 @@ def test_dp_dataloader():
     shards = run_distributed(_dp_loader_worker_dp2, world_size=2)
     assert set(shards[0]).isdisjoint(shards[1])
     assert len(shards[0]) == len(shards[1]) == 8
+
+
+def _trainer_pp_worker(rank, world):
+    """High-level trainer facade at PP=2: neuronx_distributed_config ->
+    initialize_parallel_model (pipeline wrap) -> parallel optimizer ->
+    run_train steps reduce the loss."""
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+
+    cfg = nxd.neuronx_distributed_config(
+        tensor_parallel_size=1, pipeline_parallel_size=world,
+        pipeline_config={"transformer_layer_cls": LlamaDecoderLayer,
+                         "num_microbatches": 2,
+                         "input_names": ["input_ids", "labels"]},
+        optimizer_config={"zero_one_enabled": True, "grad_clipping": True,
+                          "max_grad_norm": 1.0})
+    model = nxd.initialize_parallel_model(
+        cfg, lambda: LlamaForCausalLM(get_config("tiny")))
+    opt = nxd.initialize_parallel_optimizer(cfg, torch.optim.AdamW,
+                                            model.parameters(), lr=1e-2)
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (4, 16))
+    losses = []
+    for _ in range(4):
+        opt.zero_grad()
+        loss = model.run_train(input_ids=x, labels=x)
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+    return losses[-1]
+
+
+def test_trainer_facade_pp2():
+    out = run_distributed(_trainer_pp_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-5
