@@ -204,7 +204,9 @@ def ring_attn_func(q, k, v, causal=True, softmax_scale=None):
         from .flash_attn import flash_attn_func
 
         return flash_attn_func(q, k, v, causal=causal, softmax_scale=scale)
-    return _RingAttnFn.apply(q, k, v, causal, scale)
+    # the K/V ring (irecv) and the block kernels need contiguous tensors
+    return _RingAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                             causal, scale)
 
 
 # reference-compatible name (kernels/ring_attention_kernel.py)

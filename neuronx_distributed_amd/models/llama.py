@@ -173,7 +173,15 @@ class LlamaAttention(nn.Module):
         v = v.transpose(1, 2)
         if kv_cache is not None:
             k, v = kv_cache.update(k, v, pos_offset)
-        out = flash_attn_func(q, k, v, causal=True)
+        if kv_cache is None and not isinstance(pos_offset, torch.Tensor) \
+                and ps.get_context_model_parallel_size() > 1:
+            # context parallelism: q/k/v are this rank's contiguous S/cp
+            # slice (utils.batch_utils); K/V ring-rotate over the CP group
+            from ..kernels.ring_attn import ring_attn_func
+
+            out = ring_attn_func(q, k, v, causal=True)
+        else:
+            out = flash_attn_func(q, k, v, causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
         if sp:
             out = out.transpose(0, 1)
