@@ -25,7 +25,7 @@ import torch.nn as nn
 from ..moe.expert_mlps import ExpertMLPs
 from ..moe.loss_function import load_balancing_loss_func
 from ..moe.model import MoE
-from ..moe.routing import RouterTopK
+from ..moe.routing import (GroupLimitedRouter, RouterSinkhorn, RouterTopK)
 from ..moe.shared_experts import SharedExperts
 from ..parallel.layers import ColumnParallelLinear, ParallelEmbedding
 from ..parallel.loss_functions import parallel_cross_entropy
@@ -61,6 +61,10 @@ class MixtralConfig:
     moe_frequency: int = 1       # every Nth layer is MoE (1 = all layers)
     num_shared_experts: int = 0  # >0 adds a SharedExperts branch
     glu_mlp: bool = True
+    # "topk" (Mixtral) | "group_limited" (DeepSeek-V3 style) | "sinkhorn"
+    router_type: str = "topk"
+    n_groups: int = 4
+    topk_group: int = 2
 
     @property
     def head_dim(self):
@@ -87,8 +91,19 @@ def build_moe_layer(config: MixtralConfig) -> MoE:
     """Assemble router + expert MLPs (+ shared experts) — the reference's
     ``initialize_moe_module`` equivalent (mixtral_model.py)."""
     sp = config.sequence_parallel_enabled
-    router = RouterTopK(config.num_local_experts, config.num_experts_per_tok,
-                        config.hidden_size, sequence_parallel_enabled=sp)
+    if config.router_type == "group_limited":
+        router = GroupLimitedRouter(
+            config.num_local_experts, config.num_experts_per_tok,
+            config.hidden_size, n_groups=config.n_groups,
+            topk_group=config.topk_group, sequence_parallel_enabled=sp)
+    elif config.router_type == "sinkhorn":
+        router = RouterSinkhorn(config.num_local_experts, 1,
+                                config.hidden_size,
+                                sequence_parallel_enabled=sp)
+    else:
+        router = RouterTopK(config.num_local_experts,
+                            config.num_experts_per_tok, config.hidden_size,
+                            sequence_parallel_enabled=sp)
     expert_mlps = ExpertMLPs(
         config.num_local_experts, config.hidden_size,
         config.intermediate_size, config.num_experts_per_tok,

@@ -273,3 +273,29 @@ def _mixtral_llama4_worker(rank, world):
 
 def test_mixtral_llama4_style():
     run_distributed(_mixtral_llama4_worker, world_size=1)
+
+
+def _mixtral_router_worker(rank, world, rtype):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    kw = {"router_type": rtype}
+    if rtype == "sinkhorn":
+        kw["num_experts_per_tok"] = 1
+    if rtype == "group_limited":
+        kw.update(n_groups=2, topk_group=1)  # 4 experts -> 2 per group
+    m = MixtralForCausalLM(get_moe_config("tiny-moe", **kw))
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    loss = m(x, labels=x)
+    loss.backward()
+    assert torch.isfinite(loss)
+    return float(loss.detach())
+
+
+def test_mixtral_router_variants():
+    for rtype in ("group_limited", "sinkhorn"):
+        run_distributed(_mixtral_router_worker, world_size=1, args=(rtype,))
