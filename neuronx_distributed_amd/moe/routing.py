@@ -136,7 +136,8 @@ class GroupLimitedRouter(RouterBase):
 
         T = scores.shape[0]
         grouped = scores_for_choice.view(T, self.n_groups, -1)
-        group_scores = grouped.topk(2, dim=-1)[0].sum(dim=-1)  # (T, G)
+        gk = min(2, grouped.shape[-1])
+        group_scores = grouped.topk(gk, dim=-1)[0].sum(dim=-1)  # (T, G)
         group_idx = torch.topk(group_scores, self.topk_group, dim=-1)[1]
         group_mask = torch.zeros_like(group_scores).scatter(-1, group_idx, 1.0)
         expert_mask = group_mask.unsqueeze(-1).expand(

@@ -264,3 +264,30 @@ def test_head_padding_equivalence():
 def test_head_padding_tp2():
     out = run_distributed(_pad_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-5
+
+
+def _dist_ops_worker(rank, world):
+    """Distributed topk/argmax over the TP-sharded dim match single-rank
+    torch results."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.operators import argmax as dargmax
+    from neuronx_distributed_amd.operators import topk as dtopk
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    full = torch.randn(4, 64)
+    shard = full.chunk(world, dim=-1)[rank]
+
+    am = dargmax(shard, dim=-1, gather_dim=-1)
+    assert torch.equal(am.reshape(-1), full.argmax(-1).reshape(-1)), am
+
+    vals, idx = dtopk(shard, k=3, dim=-1, gather_dim=-1)
+    rv, ri = full.topk(3, dim=-1)
+    assert torch.allclose(vals, rv, atol=1e-6)
+    assert torch.equal(idx, ri)
+    return float(vals.sum())
+
+
+def test_distributed_topk_argmax():
+    out = run_distributed(_dist_ops_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-5
