@@ -341,3 +341,37 @@ def _pp_ckpt_worker(rank, world):
 
 def test_pp2_with_activation_checkpointing():
     run_distributed(_pp_ckpt_worker, world_size=2)
+
+
+def _neox_pp_worker(rank, world):
+    """FX partition + 1F1B on the GPT-NeoX architecture (parallel residual,
+    LN biases): loss matches dense."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (GPTNeoXForCausalLM,
+                                                get_neox_config)
+    from neuronx_distributed_amd.models.gpt_neox import GPTNeoXLayer
+    from neuronx_distributed_amd.pipeline import NxDPPModel
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 pipeline_model_parallel_size=world)
+    cfg = get_neox_config("gpt-neox-tiny")
+    torch.manual_seed(0)
+    model = GPTNeoXForCausalLM(cfg)
+    torch.manual_seed(0)
+    golden = GPTNeoXForCausalLM(cfg)
+
+    pp_model = NxDPPModel(model, transformer_layer_cls=GPTNeoXLayer,
+                          num_microbatches=2,
+                          input_names=["input_ids", "labels"],
+                          leaf_module_cls=(GPTNeoXLayer,))
+    torch.manual_seed(42)
+    x = torch.randint(0, 256, (4, 16))
+    loss = pp_model.run_train(input_ids=x, labels=x)
+    ref = golden(x, labels=x)
+    assert abs(loss.item() - ref.item()) < 1e-4, (loss, ref)
+    return loss.item()
+
+
+def test_neox_pp2():
+    out = run_distributed(_neox_pp_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-6
