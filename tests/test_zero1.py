@@ -125,3 +125,42 @@ def _dcp_worker(rank, world):
 
 def test_zero1_dcp_roundtrip():
     run_distributed(_dcp_worker, world_size=2)
+
+
+def _multi_group_worker(rank, world):
+    """Two param groups (different lr / weight_decay) through zero1 match
+    plain AdamW with the same groups."""
+    import torch
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m1 = torch.nn.Sequential(torch.nn.Linear(32, 32),
+                             torch.nn.LayerNorm(32))
+    torch.manual_seed(0)
+    m2 = torch.nn.Sequential(torch.nn.Linear(32, 32),
+                             torch.nn.LayerNorm(32))
+
+    def groups(m):
+        decay = [p for n, p in m.named_parameters() if "0." in n]
+        no_decay = [p for n, p in m.named_parameters() if "1." in n]
+        return [{"params": decay, "weight_decay": 0.1, "lr": 1e-2},
+                {"params": no_decay, "weight_decay": 0.0, "lr": 5e-3}]
+
+    o1 = NeuronZero1Optimizer(groups(m1), torch.optim.AdamW,
+                              grad_clipping=False)
+    o2 = torch.optim.AdamW(groups(m2))
+    for _ in range(3):
+        x = torch.randn(8, 32)
+        for m, o in ((m1, o1), (m2, o2)):
+            o.zero_grad()
+            m(x).pow(2).mean().backward()
+            o.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
+    return 0.0
+
+
+def test_zero1_param_groups():
+    run_distributed(_multi_group_worker, world_size=2)
