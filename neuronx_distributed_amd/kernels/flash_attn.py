@@ -16,7 +16,7 @@ import os
 import torch
 
 
-def _torch_reference(q, k, v, causal=True, scale=None):
+def _torch_reference(q, k, v, causal=True, scale=None, window=None):
     B, Hq, S, D = q.shape
     Hkv = k.shape[1]
     rep = Hq // Hkv
@@ -25,24 +25,31 @@ def _torch_reference(q, k, v, causal=True, scale=None):
         v = v.repeat_interleave(rep, dim=1)
     scale = scale or 1.0 / math.sqrt(D)
     scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    Sk = k.shape[2]
     if causal:
-        Sk = k.shape[2]
         mask = torch.ones(S, Sk, dtype=torch.bool, device=q.device).tril(
             diagonal=Sk - S)
+        if window is not None:
+            # sliding window: row i (global Sk-S+i) attends the last
+            # ``window`` positions only
+            mask &= torch.ones(S, Sk, dtype=torch.bool,
+                               device=q.device).triu(
+                                   diagonal=Sk - S - window + 1)
         scores = scores.masked_fill(~mask, float("-inf"))
     probs = torch.softmax(scores, dim=-1)
     return torch.matmul(probs, v.float()).to(q.dtype)
 
 
-def flash_attn_func(q, k, v, causal=True, softmax_scale=None):
+def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
     if q.is_cuda:
         from .. import ops
 
-        if q.shape[2] != k.shape[2] or q.shape[3] != 128:
-            # rectangular attention (KV-cache decode) or head_dim != 128:
-            # composed path — the HIP MFMA kernel is D=128 (the production
-            # model shapes); other head dims run batched GEMMs + softmax
-            return _torch_reference(q, k, v, causal, softmax_scale)
+        if q.shape[2] != k.shape[2] or q.shape[3] != 128 or \
+                window is not None:
+            # rectangular attention (KV-cache decode), head_dim != 128, or
+            # sliding-window masking: composed path — the HIP MFMA kernel
+            # is D=128 full-causal (the production training shapes)
+            return _torch_reference(q, k, v, causal, softmax_scale, window)
         if hasattr(ops, "flash_attn") and ops.flash_attn_available():
             return ops.flash_attn(q, k, v, causal=causal,
                                   softmax_scale=softmax_scale)
@@ -52,7 +59,7 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None):
             "HIP flash-attention kernel unavailable on GPU; build "
             "neuronx_distributed_amd.ops (or set NXDA_ALLOW_TORCH_FALLBACK=1 "
             "for bring-up)")
-    return _torch_reference(q, k, v, causal, softmax_scale)
+    return _torch_reference(q, k, v, causal, softmax_scale, window)
 
 
 # reference-compatible name (kernels/flash_attn.py:162)

@@ -58,6 +58,10 @@ class LlamaConfig:
     # set by pad_llama_config: padded head counts keep the ORIGINAL head
     # size (hidden_size // original_num_heads)
     head_dim_override: Optional[int] = None
+    # Mistral-style sliding-window attention (None = full causal).  Runs
+    # on the composed/batched-GEMM paths; the MFMA flash kernel is
+    # full-causal (window support is round-2 kernel work).
+    sliding_window: Optional[int] = None
 
     @property
     def head_dim(self):
@@ -79,6 +83,11 @@ CONFIGS = {
                              num_key_value_heads=8, vocab_size=128256,
                              rope_theta=500000.0,
                              max_position_embeddings=8192),
+    "mistral-7b": LlamaConfig(hidden_size=4096, intermediate_size=14336,
+                              num_hidden_layers=32, num_attention_heads=32,
+                              num_key_value_heads=8, vocab_size=32000,
+                              sliding_window=4096,
+                              max_position_embeddings=8192),
     "llama3-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672,
                               num_hidden_layers=80, num_attention_heads=64,
                               num_key_value_heads=8, vocab_size=128256,
@@ -181,7 +190,8 @@ class LlamaAttention(nn.Module):
 
             out = ring_attn_func(q, k, v, causal=True)
         else:
-            out = flash_attn_func(q, k, v, causal=True)
+            out = flash_attn_func(q, k, v, causal=True,
+                              window=self.config.sliding_window)
         out = out.transpose(1, 2).reshape(B, S, -1)
         if sp:
             out = out.transpose(0, 1)
@@ -199,6 +209,7 @@ class LlamaAttention(nn.Module):
         rep = self.num_heads_local // self.num_kv_local
         if (q.is_cuda and self.head_dim == 128 and rep in (1, 2, 4, 8)
                 and q.dtype == torch.bfloat16 and kv_cache is not None
+                and self.config.sliding_window is None
                 and ops.decode_attn_available()):
             out = ops.decode_attn_step(
                 q.reshape(B, -1).contiguous(), k.reshape(B, -1).contiguous(),
@@ -232,7 +243,10 @@ class LlamaAttention(nn.Module):
         qg = q.view(B, self.num_kv_local, rep, self.head_dim)
         scores = (qg @ K.transpose(-1, -2)).float() * scale
         idx = torch.arange(Smax, device=q.device)
-        scores = scores.masked_fill(idx > pos_t, float("-inf"))
+        invalid = idx > pos_t
+        if self.config.sliding_window is not None:
+            invalid = invalid | (idx <= pos_t - self.config.sliding_window)
+        scores = scores.masked_fill(invalid, float("-inf"))
         probs = torch.softmax(scores, dim=-1)
         out = probs.to(q.dtype) @ V  # (B,Hkv,rep,D)
         out = out.reshape(B, 1, self.num_heads_local * self.head_dim)
