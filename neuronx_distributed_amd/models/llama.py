@@ -138,6 +138,8 @@ class LlamaAttention(nn.Module):
         self.config = config
         tp = ps.get_tensor_model_parallel_size()
         self.head_dim = config.head_dim
+        # shared by the MoE/other families whose configs lack the field
+        self.sliding_window = getattr(config, "sliding_window", None)
         kv_mult = max(1, tp // config.num_key_value_heads)
         self.kv_mult = kv_mult
         self.num_heads_local = config.num_attention_heads // tp
@@ -191,7 +193,7 @@ class LlamaAttention(nn.Module):
             out = ring_attn_func(q, k, v, causal=True)
         else:
             out = flash_attn_func(q, k, v, causal=True,
-                              window=self.config.sliding_window)
+                                  window=self.sliding_window)
         out = out.transpose(1, 2).reshape(B, S, -1)
         if sp:
             out = out.transpose(0, 1)
@@ -209,7 +211,7 @@ class LlamaAttention(nn.Module):
         rep = self.num_heads_local // self.num_kv_local
         if (q.is_cuda and self.head_dim == 128 and rep in (1, 2, 4, 8)
                 and q.dtype == torch.bfloat16 and kv_cache is not None
-                and self.config.sliding_window is None
+                and self.sliding_window is None
                 and ops.decode_attn_available()):
             out = ops.decode_attn_step(
                 q.reshape(B, -1).contiguous(), k.reshape(B, -1).contiguous(),
@@ -244,8 +246,8 @@ class LlamaAttention(nn.Module):
         scores = (qg @ K.transpose(-1, -2)).float() * scale
         idx = torch.arange(Smax, device=q.device)
         invalid = idx > pos_t
-        if self.config.sliding_window is not None:
-            invalid = invalid | (idx <= pos_t - self.config.sliding_window)
+        if self.sliding_window is not None:
+            invalid = invalid | (idx <= pos_t - self.sliding_window)
         scores = scores.masked_fill(invalid, float("-inf"))
         probs = torch.softmax(scores, dim=-1)
         out = probs.to(q.dtype) @ V  # (B,Hkv,rep,D)
