@@ -359,3 +359,36 @@ def _mixtral_generate_worker(rank, world):
 
 def test_mixtral_generate():
     run_distributed(_mixtral_generate_worker, world_size=1)
+
+
+def _token_shuffle_worker(rank, world):
+    """token_shuffle o token_unshuffle is the identity (incl. the
+    cross-rank all-to-all), and a shuffled all-experts MoE equals the
+    unshuffled one (per-token computation is permutation-invariant)."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.moe import MoE
+    from neuronx_distributed_amd.moe.token_shuffling import (token_shuffle,
+                                                             token_unshuffle)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 token_shuffle_group_size=world)
+    torch.manual_seed(10 + rank)
+    x = torch.randn(8, 16)
+    h, perm = token_shuffle(x, seed=5)
+    back = token_unshuffle(h, perm)
+    assert torch.allclose(back, x, atol=1e-6)
+
+    torch.manual_seed(0)
+    moe = _make_moe()
+    moe.token_shuffle_group_size = world
+    moe.train()
+    out_shuf, _ = moe(x.unsqueeze(0))
+    moe.token_shuffle_group_size = 1
+    out_plain, _ = moe(x.unsqueeze(0))
+    assert torch.allclose(out_shuf, out_plain, atol=1e-5), \
+        (out_shuf - out_plain).abs().max()
+    return float(out_shuf.sum())
+
+
+def test_token_shuffling():
+    run_distributed(_token_shuffle_worker, world_size=2)
