@@ -291,3 +291,38 @@ def _dist_ops_worker(rank, world):
 def test_distributed_topk_argmax():
     out = run_distributed(_dist_ops_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-5
+
+
+def _reshard_worker(rank, world):
+    """FULL (unsharded) tensors reshard onto TP2 layers through
+    _reshard_full_state_dict: Column rows, Row columns, and the GQAQKV
+    preshard_hook's KV replication — the public sharded-checkpoint
+    contract."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.checkpointing import (
+        _reshard_full_state_dict)
+    from neuronx_distributed_amd.parallel.layers import (ColumnParallelLinear,
+                                                         RowParallelLinear)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+
+    col = ColumnParallelLinear(8, 12, bias=False, gather_output=False)
+    full_c = torch.arange(96, dtype=torch.float32).reshape(12, 8)
+    sd = {"weight": full_c.clone()}
+    _reshard_full_state_dict(col, sd)
+    col.load_state_dict(sd)
+    assert torch.equal(col.weight.detach(),
+                       full_c[rank * 6:(rank + 1) * 6])
+
+    row = RowParallelLinear(12, 8, bias=False, input_is_parallel=True)
+    full_r = torch.arange(96, dtype=torch.float32).reshape(8, 12)
+    sd = {"weight": full_r.clone()}
+    _reshard_full_state_dict(row, sd)
+    row.load_state_dict(sd)
+    assert torch.equal(row.weight.detach(),
+                       full_r[:, rank * 6:(rank + 1) * 6])
+    return 0.0
+
+
+def test_full_checkpoint_resharding():
+    run_distributed(_reshard_worker, world_size=2)
