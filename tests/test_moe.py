@@ -392,3 +392,39 @@ def _token_shuffle_worker(rank, world):
 
 def test_token_shuffling():
     run_distributed(_token_shuffle_worker, world_size=2)
+
+
+def _ep_zero1_worker(rank, world):
+    """EP=2 MoE + zero1: expert params shard over EDP, dense over DPxCP;
+    3 steps reduce the loss and expert weights stay consistent with the
+    EP layout (each rank trains only its local experts)."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+    from neuronx_distributed_amd.optimizer import NeuronEPZero1Optimizer
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 expert_model_parallel_size=world)
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(get_moe_config("tiny-moe"))
+    n_local = m.model.layers[0].block_sparse_moe.expert_mlps.num_experts_local
+    assert n_local == 4 // world, n_local
+
+    opt = NeuronEPZero1Optimizer(m.parameters(), torch.optim.AdamW, lr=1e-2,
+                                 grad_clipping=True, max_norm=1.0)
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    losses = []
+    for _ in range(3):
+        opt.zero_grad()
+        loss = m(x, labels=x)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0], losses
+    return losses[-1]
+
+
+def test_ep_zero1():
+    out = run_distributed(_ep_zero1_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-5
