@@ -375,3 +375,37 @@ def _neox_pp_worker(rank, world):
 def test_neox_pp2():
     out = run_distributed(_neox_pp_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-6
+
+
+def _pp_eval_worker(rank, world):
+    """InferenceSchedule (forward-only) through PP2 matches the dense
+    model's eval loss."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+    from neuronx_distributed_amd.pipeline import NxDPPModel
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 pipeline_model_parallel_size=world)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(get_config("tiny"))
+    torch.manual_seed(0)
+    golden = LlamaForCausalLM(get_config("tiny"))
+
+    pp_model = NxDPPModel(model, transformer_layer_cls=LlamaDecoderLayer,
+                          num_microbatches=2,
+                          input_names=["input_ids", "labels"])
+    torch.manual_seed(42)
+    x = torch.randint(0, 256, (4, 16))
+    loss = pp_model.run_eval(input_ids=x, labels=x)
+    with torch.no_grad():
+        ref = golden(x, labels=x)
+    assert abs(float(loss) - float(ref)) < 1e-4, (loss, ref)
+    # eval must not build grads
+    assert all(p.grad is None for p in pp_model.local_parameters())
+    return float(loss)
+
+
+def test_pp2_eval():
+    out = run_distributed(_pp_eval_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-6
