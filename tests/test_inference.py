@@ -175,3 +175,41 @@ def _medusa_worker(rank, world):
 
 def test_medusa_generation():
     run_distributed(_medusa_worker, world_size=1)
+
+
+def _builder_worker(rank, world):
+    """ModelBuilder e2e: trace two shape buckets, compile, route inputs to
+    the right bucket, checkpoint_loader applied."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.inference.model_builder import ModelBuilder
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("tiny")
+    torch.manual_seed(0)
+    ref = LlamaForCausalLM(cfg).eval()
+    sd = ref.state_dict()
+
+    builder = ModelBuilder(model_fn=lambda: LlamaForCausalLM(cfg),
+                           checkpoint_loader=lambda: sd)
+    builder.trace({"input_ids": torch.zeros(1, 8, dtype=torch.long)},
+                  tag="short")
+    builder.trace({"input_ids": torch.zeros(1, 16, dtype=torch.long)},
+                  tag="long")
+    nxd_model = builder.compile(use_hip_graphs=False)
+
+    torch.manual_seed(1)
+    x8 = torch.randint(0, 256, (1, 8))
+    x16 = torch.randint(0, 256, (1, 16))
+    with torch.no_grad():
+        out8 = nxd_model(input_ids=x8)
+        out16 = nxd_model(input_ids=x16)
+        r8 = ref(x8)
+        r16 = ref(x16)
+    assert torch.allclose(out8, r8, atol=1e-6)
+    assert torch.allclose(out16, r16, atol=1e-6)
+    return 0.0
+
+
+def test_model_builder_e2e():
+    run_distributed(_builder_worker, world_size=1)
