@@ -50,10 +50,12 @@ class _Bucket:
     of ``flat_grad``; this rank owns fp32 master shard ``master``."""
 
     def __init__(self, params: List[torch.nn.Parameter], group_info,
-                 group_index: int, optimizer_dtype):
+                 group_index: int, optimizer_dtype,
+                 fp32_grad_acc: bool = False):
         self.params = params
         self.group_info = group_info
         self.group_index = group_index
+        self.fp32_grad_acc = fp32_grad_acc
         world = group_info.size
         device = params[0].device
         dtype = params[0].dtype
@@ -61,7 +63,13 @@ class _Bucket:
         numel = sum(p.numel() for p in params)
         self.padded = int(math.ceil(numel / world) * world)
         self.flat_param = torch.zeros(self.padded, dtype=dtype, device=device)
-        self.flat_grad = torch.zeros(self.padded, dtype=dtype, device=device)
+        # fp32 grad accumulation (reference mixed_precision_config
+        # use_fp32_grad_acc): grads accumulate into an fp32 flat buffer via
+        # post-accumulate hooks (p.grad stays param-dtype per microbatch
+        # and is folded in + freed); comm then runs in fp32
+        grad_dtype = torch.float32 if fp32_grad_acc else dtype
+        self.flat_grad = torch.zeros(self.padded, dtype=grad_dtype,
+                                     device=device)
 
         # segment bookkeeping: (param, start, end)
         self.segments = []
@@ -72,7 +80,17 @@ class _Bucket:
                 self.flat_param[off:off + n].copy_(p.data.reshape(-1))
                 new_data = self.flat_param[off:off + n].view(p.shape)
                 p.data = new_data
-                p.grad = self.flat_grad[off:off + n].view(p.shape)
+                if fp32_grad_acc:
+                    seg = self.flat_grad[off:off + n]
+
+                    def _acc(param, _seg=seg):
+                        if param.grad is not None:
+                            _seg.add_(param.grad.detach().reshape(-1).float())
+                            param.grad = None
+
+                    p.register_post_accumulate_grad_hook(_acc)
+                else:
+                    p.grad = self.flat_grad[off:off + n].view(p.shape)
                 self.segments.append((p, off, off + n))
                 off += n
 
@@ -87,7 +105,8 @@ class _Bucket:
         self.master.requires_grad_(True)
         # reduce-scatter lands in the comm dtype (= param dtype); the fp32
         # cast happens when attaching to the master shard
-        self.grad_shard = torch.zeros(self.shard_size, dtype=dtype, device=device)
+        self.grad_shard = torch.zeros(self.shard_size, dtype=grad_dtype,
+                                      device=device)
 
         # which elements of MY shard must be excluded from the grad norm
         # (TP-duplicated params counted only on tp_rank 0)
@@ -103,7 +122,15 @@ class _Bucket:
 
     def relink_grads(self):
         """Re-point p.grad into the flat buffer if something (zero_grad
-        set_to_none, checkpoint load) broke the linkage."""
+        set_to_none, checkpoint load) broke the linkage.  In fp32-acc mode
+        fold in any grads the hooks have not consumed."""
+        if self.fp32_grad_acc:
+            for p, s, e in self.segments:
+                if p.grad is not None:
+                    self.flat_grad[s:e].add_(
+                        p.grad.detach().reshape(-1).float())
+                    p.grad = None
+            return
         for p, s, e in self.segments:
             g = p.grad
             view = self.flat_grad[s:e].view(p.shape)
@@ -128,6 +155,7 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
                  max_norm: float = 1.0, pin_layout: bool = False,
                  sharding_groups=None, grad_norm_groups=None,
                  lazy_init: bool = False, use_fused_kernel: bool = True,
+                 use_fp32_grad_acc: bool = False,
                  **defaults):
         if isinstance(params, torch.Tensor):
             raise TypeError("params must be an iterable")
@@ -142,6 +170,7 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
         self.grad_clipping = grad_clipping
         self.max_norm = max_norm
         self.use_fused_kernel = use_fused_kernel
+        self.use_fp32_grad_acc = use_fp32_grad_acc
         self._sharding_group = (sharding_groups
                                 if isinstance(sharding_groups, ps.GroupInfo)
                                 else None)
@@ -187,12 +216,14 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
                     cur.append(p)
                     size += p.numel() * p.element_size()
                     if size >= cap:
-                        self.buckets.append(_Bucket(cur, ginfo, gi,
-                                                    self.optimizer_dtype))
+                        self.buckets.append(_Bucket(
+                            cur, ginfo, gi, self.optimizer_dtype,
+                            self.use_fp32_grad_acc))
                         cur, size = [], 0
                 if cur:
-                    self.buckets.append(_Bucket(cur, ginfo, gi,
-                                                self.optimizer_dtype))
+                    self.buckets.append(_Bucket(
+                        cur, ginfo, gi, self.optimizer_dtype,
+                        self.use_fp32_grad_acc))
 
     # -- step -------------------------------------------------------------
 
@@ -262,6 +293,7 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
         from .. import ops
 
         return (self.use_fused_kernel
+                and not self.use_fp32_grad_acc
                 and self.optimizer_class is torch.optim.AdamW
                 and self.optimizer_dtype == torch.float32
                 and self.buckets and self.buckets[0].master.is_cuda
@@ -322,6 +354,10 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
     def zero_grad(self, set_to_none: bool = False):
         for b in self.buckets:
             b.flat_grad.zero_()
+            if b.fp32_grad_acc:
+                for p, _, _ in b.segments:
+                    p.grad = None
+                continue
             for p, s, e in b.segments:
                 if p.grad is None or p.grad.data_ptr() != b.flat_grad[s:e].data_ptr():
                     p.grad = b.flat_grad[s:e].view(p.shape)

@@ -141,11 +141,13 @@ def initialize_parallel_optimizer(nxd_config: Dict, optimizer_class,
                                   parameters, **defaults) -> "NxDOptimizer":
     """reference trainer.py:237-315."""
     opt_cfg = nxd_config["optimizer_config"]
+    mp_cfg = nxd_config.get("mixed_precision_config") or {}
     if opt_cfg.get("zero_one_enabled", True):
         optimizer = NeuronZero1Optimizer(
             parameters, optimizer_class,
             grad_clipping=opt_cfg.get("grad_clipping", True),
             max_norm=opt_cfg.get("max_grad_norm", 1.0),
+            use_fp32_grad_acc=mp_cfg.get("use_fp32_grad_acc", False),
             **defaults)
     else:
         optimizer = optimizer_class(parameters, **defaults)

