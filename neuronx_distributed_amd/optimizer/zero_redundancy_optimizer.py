@@ -364,7 +364,11 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
 
     # -- checkpoint -------------------------------------------------------
 
-    def state_dict(self):
+    def state_dict(self, include_masters: bool = True):
+        """``include_masters=False`` (mixed_precision_config
+        use_master_weights_in_ckpt=False) drops the fp32 master shards —
+        smaller checkpoints; on load they are rebuilt from the bf16
+        params (one-time precision round-trip)."""
         return {
             "base_optimizer": self.base_optimizer.state_dict(),
             "step_count": getattr(self, "_step_count", 0),
@@ -372,7 +376,8 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
                 {"m": b.fused_m.cpu(), "v": b.fused_v.cpu()}
                 if hasattr(b, "fused_m") else None for b in self.buckets
             ],
-            "masters": [b.master.detach().cpu() for b in self.buckets],
+            "masters": [b.master.detach().cpu() for b in self.buckets]
+            if include_masters else None,
             "shard_meta": [
                 {"padded": b.padded, "rank": b.rank, "world": b.group_info.size}
                 for b in self.buckets
@@ -386,8 +391,16 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
             if fs is not None:
                 b.fused_m = fs["m"].to(b.master.device)
                 b.fused_v = fs["v"].to(b.master.device)
-        for b, m in zip(self.buckets, state_dict["masters"]):
-            b.master.data.copy_(m.to(b.master.device))
+        masters = state_dict.get("masters")
+        if masters is None:
+            # masters were excluded from the checkpoint: rebuild from the
+            # (already-loaded) bf16 params
+            for b in self.buckets:
+                b.master.data.copy_(
+                    b.flat_param[b.shard_lo:b.shard_hi].to(b.master.dtype))
+        else:
+            for b, m in zip(self.buckets, masters):
+                b.master.data.copy_(m.to(b.master.device))
         self._all_gather_params()
 
 

@@ -31,6 +31,10 @@ class NxDOptimizer(torch.optim.Optimizer):
         return self.optimizer.state
 
     def state_dict(self):
+        mp = self.nxd_config.get("mixed_precision_config") or {}
+        if isinstance(self.optimizer, NeuronZero1Optimizer):
+            return self.optimizer.state_dict(
+                include_masters=mp.get("use_master_weights_in_ckpt", True))
         return self.optimizer.state_dict()
 
     def load_state_dict(self, sd):
