@@ -106,13 +106,18 @@ def initialize_parallel_model(nxd_config: Dict, model_fn: Callable, *model_args,
         pipeline_cfg = dict(nxd_config.get("pipeline_config") or {})
         model = NxDPPModel(model, **pipeline_cfg)
 
-    # phase 3: materialize + move
+    # phase 3: materialize + move — host-RAM-bounded: ranks move in
+    # groups of sequential_move_factor with a rendezvous between groups
+    # (reference model_utils.py:335-358)
     if init_cfg.get("meta_device_init"):
         from ..utils.model_utils import reinit_model
 
         model = reinit_model(model, device, init_cfg.get("param_init_fn"))
-    else:
-        model = model.to(device) if pp_size == 1 else model
+    elif pp_size == 1:
+        model = get_model_sequential(
+            lambda: model, device,
+            sequential_move_factor=init_cfg.get("sequential_move_factor",
+                                                11))
 
     # phase 4: LoRA
     if nxd_config.get("lora_config") is not None:
