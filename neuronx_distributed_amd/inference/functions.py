@@ -14,13 +14,12 @@ On MI355X the reference's trace->HLO->compile->NEFF chain collapses to
 """
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Callable, Dict, Optional
 
 import torch
 
 from ..parallel import parallel_state as ps
-from ..parallel.utils import create_local_weight
 from .parallel_context import NxDParallelState
 
 

@@ -2,7 +2,7 @@
 INPUT_STATE tensors, nxd_model.py:123-139 — here plain device-resident
 buffers updated in place, hipGraph-compatible)."""
 
-from typing import List, Optional
+from typing import List
 
 import torch
 

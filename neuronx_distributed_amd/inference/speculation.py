@@ -12,7 +12,6 @@ KV-cache bookkeeping: both caches may hold STALE rows beyond the accepted
 position after a rejection; every attention path here masks by position
 and rewrites rows in place, so stale rows are simply overwritten."""
 
-from typing import Optional
 
 import torch
 

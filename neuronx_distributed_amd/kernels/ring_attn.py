@@ -16,7 +16,6 @@ arrive home after cp steps — no all-reduce, no extra memory.
 """
 
 import math
-from typing import List, Optional
 
 import torch
 import torch.distributed as dist

@@ -1,6 +1,6 @@
 """LoRA configuration (reference modules/lora/config.py)."""
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 

@@ -2,7 +2,7 @@
 modules for LoRA-adapted versions, freezes the base, exposes adapter
 save/load/merge."""
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 import torch.nn as nn

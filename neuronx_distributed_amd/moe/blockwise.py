@@ -14,7 +14,6 @@ a persistent-CTA HIP grouped GEMM is the planned follow-up and slots in
 behind the same interface)."""
 
 import math
-from typing import Tuple
 
 import torch
 

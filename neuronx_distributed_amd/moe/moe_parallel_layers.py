@@ -10,13 +10,12 @@ EDP group (reference :140-162 re-tagging)."""
 import torch
 import torch.nn as nn
 
-from ..parallel import comm, parallel_state as ps
+from ..parallel import parallel_state as ps
 from ..parallel.utils import (
     divide,
     set_tensor_model_parallel_attributes,
     EXPERT_PARALLEL_ATTR,
 )
-from ..parallel.random import get_rng_state_tracker
 
 
 class _ExpertFusedLinearBase(nn.Module):

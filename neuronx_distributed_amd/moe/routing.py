@@ -8,7 +8,6 @@ fp32 suffices and is what the math needs), Sinkhorn runs a FIXED iteration
 count in fp32 under no_grad (:224,283-314).
 """
 
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -26,7 +26,7 @@ elementwise kernels.
 
 import math
 import os
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist

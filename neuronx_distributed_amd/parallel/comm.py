@@ -15,7 +15,6 @@ IS RCCL on ROCm) — plus:
 Every function takes a :class:`GroupInfo` or raw ProcessGroup.
 """
 
-from typing import List, Optional
 
 import torch
 import torch.distributed as dist

@@ -5,7 +5,6 @@ Output-channel sharding mirrors ColumnParallel (weight (O/tp, I, kH, kW));
 input-channel sharding mirrors RowParallel (weight (O, I/tp, kH, kW), output
 all-reduced).  Conv itself runs on MIOpen via F.conv2d."""
 
-from typing import Optional, Tuple, Union
 
 import torch
 import torch.nn as nn

@@ -20,9 +20,8 @@ ascending-descending rings, TP4 interleave) have no analogue here; plain
 row-major placement (the reference's LOGIC1) is the xGMI-native layout.
 """
 
-import itertools
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import torch

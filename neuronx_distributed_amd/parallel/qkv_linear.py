@@ -24,7 +24,6 @@ import torch.distributed as dist
 from . import comm
 from . import parallel_state as ps
 from .layers import BaseParallelLinear, default_init_method
-from .random import get_rng_state_tracker
 from .utils import divide, set_tensor_model_parallel_attributes
 
 

@@ -15,7 +15,7 @@ mb/pp combination, so no parity-special schedule is needed here.
 """
 
 from dataclasses import dataclass
-from typing import Iterator, List
+from typing import Iterator
 
 
 @dataclass(frozen=True)

@@ -3,7 +3,7 @@
 Attach to float layers, run calibration batches, then convert with the
 observed activation/weight ranges (static quantization)."""
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 import torch.nn as nn

@@ -8,7 +8,6 @@ carry over."""
 
 import enum
 from dataclasses import dataclass
-from typing import Optional
 
 import torch
 

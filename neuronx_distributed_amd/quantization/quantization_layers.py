@@ -5,7 +5,6 @@ in the quantized dtype + scale parameters; the GEMM runs dequant->bf16
 hipBLASLt (CDNA4 fp8 MFMA GEMM path is a later optimization — the layout
 and scale plumbing here is the contract)."""
 
-from typing import Optional
 
 import torch
 import torch.nn as nn
@@ -21,7 +20,7 @@ from ..parallel.mappings import (
     reduce_from_tensor_model_parallel_region,
     scatter_to_tensor_model_parallel_region,
 )
-from ..parallel.utils import divide, set_tensor_model_parallel_attributes
+from ..parallel.utils import set_tensor_model_parallel_attributes
 from .quantization_config import (QuantizationConfig, QuantizationType,
                                   QuantizedDtype)
 from .quantization_utils import (dequantize, fp8_scaled_linear,

@@ -10,7 +10,6 @@ Implements the reference's public behavior:
   broadcast-free per-rank loads.
 """
 
-import json
 import os
 import shutil
 from concurrent.futures import ThreadPoolExecutor

@@ -3,8 +3,6 @@ wraps given module classes with the torch checkpoint fn).  On MI355X we use
 torch.utils.checkpoint with non-reentrant mode (plays well with RCCL async
 collectives and the flash kernels' saved tensors)."""
 
-from functools import partial
-from typing import Iterable, Optional, Type, Union
 
 import torch
 import torch.nn as nn

@@ -1,7 +1,7 @@
 """Medusa speculative-decoding buffers (reference utils/medusa_utils.py):
 tree attention masks + candidate gathering for multi-head draft decoding."""
 
-from typing import List, Optional, Tuple
+from typing import List
 
 import torch
 

@@ -7,7 +7,6 @@ context manager (build on meta to avoid host OOM), ``get_model_sequential``
 """
 
 import contextlib
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -2,7 +2,7 @@
 safetensors refuses aliased storages (tied embeddings); dedup before save and
 re-tie on load."""
 
-from typing import Dict, List, Tuple
+from typing import Dict, List
 
 import torch
 

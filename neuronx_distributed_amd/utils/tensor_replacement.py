@@ -3,7 +3,7 @@ swap named modules' outputs (or inputs) with provided tensors during
 forward — the counterpart of tensor_capture for fault-injection and
 debug-divergence experiments."""
 
-from typing import Any, Callable, Dict, Optional
+from typing import Any, Dict
 
 import torch
 import torch.nn as nn
