@@ -38,7 +38,8 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
     next_tok = sampler(logits[:, -1, :])
     out = [input_ids, next_tok.unsqueeze(1)]
 
-    if input_ids.is_cuda:
+    if input_ids.is_cuda and getattr(model, "supports_tensor_position",
+                                     False):
         dec = GraphDecoder(model, caches, start_pos=S, batch=B,
                            device=input_ids.device)
         if use_cuda_graph:

@@ -43,6 +43,10 @@ class GPTNeoXConfig:
     def head_dim(self):
         return self.hidden_size // self.num_attention_heads
 
+    @property
+    def num_key_value_heads(self):
+        return self.num_attention_heads  # no GQA in NeoX
+
 
 NEOX_CONFIGS = {
     "gpt-neox-20b": GPTNeoXConfig(),
@@ -68,16 +72,19 @@ def _rotate_half(x):
     return torch.cat((-x2, x1), dim=-1)
 
 
-def _partial_rope(q, k, cos, sin, rot_dim):
+def _partial_rope(q, k, cos, sin, rot_dim, pos_offset=0):
     """Apply rotary embedding to the first ``rot_dim`` dims of the head,
     pass the rest through (NeoX rotary_pct).  cos/sin are the (S, rot/2)
-    half tables from ops.precompute_rope_freqs."""
+    half tables from ops.precompute_rope_freqs; rows taken at
+    [pos_offset, pos_offset+S) for KV-cache decode."""
     q_rot, q_pass = q[..., :rot_dim], q[..., rot_dim:]
     k_rot, k_pass = k[..., :rot_dim], k[..., rot_dim:]
     S = q.shape[2]
     half = rot_dim // 2
-    c = torch.cat([cos[:S, :half], cos[:S, :half]], -1).to(q.dtype)
-    s = torch.cat([sin[:S, :half], sin[:S, :half]], -1).to(q.dtype)
+    cs = cos[pos_offset:pos_offset + S, :half]
+    sn = sin[pos_offset:pos_offset + S, :half]
+    c = torch.cat([cs, cs], -1).to(q.dtype)
+    s = torch.cat([sn, sn], -1).to(q.dtype)
     q_rot = q_rot * c + _rotate_half(q_rot) * s
     k_rot = k_rot * c + _rotate_half(k_rot) * s
     return (torch.cat([q_rot, q_pass], dim=-1),
@@ -99,23 +106,20 @@ class GPTNeoXAttention(nn.Module):
             cfg.hidden_size, cfg.hidden_size, bias=True,
             input_is_parallel=True, init_method=_init(cfg.initializer_range))
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, pos_offset=0, kv_cache=None):
         B, S, _ = x.shape
         qkv = self.query_key_value(x)
         q, k, v = qkv.chunk(3, dim=-1)
         q = q.reshape(B, S, self.n_local, self.head_dim).transpose(1, 2)
         k = k.reshape(B, S, self.n_local, self.head_dim).transpose(1, 2)
         v = v.reshape(B, S, self.n_local, self.head_dim).transpose(1, 2)
-        q, k = _partial_rope(q, k, cos, sin, self.rot_dim)
-        if self.head_dim == 128 and q.is_cuda and q.dtype == torch.bfloat16:
-            out = flash_attn_func(q.contiguous(), k.contiguous(),
-                                  v.contiguous(), causal=True)
-        else:
-            scale = 1.0 / math.sqrt(self.head_dim)
-            scores = (q.float() @ k.float().transpose(-1, -2)) * scale
-            mask = torch.ones(S, S, dtype=torch.bool, device=x.device).tril()
-            scores = scores.masked_fill(~mask, float("-inf"))
-            out = (torch.softmax(scores, -1) @ v.float()).to(x.dtype)
+        q, k = _partial_rope(q, k, cos, sin, self.rot_dim, pos_offset)
+        if kv_cache is not None:
+            k, v = kv_cache.update(k.contiguous(), v.contiguous(), pos_offset)
+        # flash dispatcher routes D != 128 / rectangular shapes through the
+        # composed batched-GEMM reference path
+        out = flash_attn_func(q.contiguous(), k.contiguous(), v.contiguous(),
+                              causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
         return self.dense(out)
 
@@ -146,8 +150,9 @@ class GPTNeoXLayer(nn.Module):
         self.attention = GPTNeoXAttention(cfg)
         self.mlp = GPTNeoXMLP(cfg)
 
-    def forward(self, x, cos, sin):
-        attn_out = self.attention(self.input_layernorm(x), cos, sin)
+    def forward(self, x, cos, sin, pos_offset=0, kv_cache=None):
+        attn_out = self.attention(self.input_layernorm(x), cos, sin,
+                                  pos_offset, kv_cache)
         if self.use_parallel_residual:
             # x + attn(ln1(x)) + mlp(ln2(x))  — NeoX parallel residual
             return x + attn_out + self.mlp(self.post_attention_layernorm(x))
@@ -176,10 +181,11 @@ class GPTNeoXForCausalLM(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids, labels=None):
+    def forward(self, input_ids, labels=None, pos_offset=0, kv_caches=None):
         x = self.embed_in(input_ids)
-        for layer in self.layers:
-            x = layer(x, self.rope_cos, self.rope_sin)
+        for i, layer in enumerate(self.layers):
+            kc = kv_caches[i] if kv_caches is not None else None
+            x = layer(x, self.rope_cos, self.rope_sin, pos_offset, kc)
         x = self.final_layer_norm(x)
         logits = self.embed_out(x)
         if labels is None:

@@ -327,6 +327,9 @@ class LlamaModel(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
+    # decode path accepts a device-tensor position (hipGraph decode engine)
+    supports_tensor_position = True
+
     def __init__(self, config: LlamaConfig):
         super().__init__()
         self.config = config

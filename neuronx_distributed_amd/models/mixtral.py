@@ -192,6 +192,9 @@ class MixtralModel(nn.Module):
 
 
 class MixtralForCausalLM(nn.Module):
+    # attention is LlamaAttention -> tensor-position decode works
+    supports_tensor_position = True
+
     def __init__(self, config: MixtralConfig):
         super().__init__()
         self.config = config

@@ -408,3 +408,29 @@ def _trainer_pp_worker(rank, world):
 def test_trainer_facade_pp2():
     out = run_distributed(_trainer_pp_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-5
+
+
+def _neox_generate_worker(rank, world):
+    """GPT-NeoX KV-cache generation (partial rotary + cache) matches full
+    re-forward greedy decoding."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (GPTNeoXForCausalLM,
+                                                get_neox_config)
+    from neuronx_distributed_amd.inference.generation import generate
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m = GPTNeoXForCausalLM(get_neox_config("gpt-neox-tiny")).eval()
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 9))
+    out = generate(m, x, max_new_tokens=6)
+    seq = x
+    for _ in range(6):
+        logits = m(seq)
+        seq = torch.cat([seq, logits[:, -1, :].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(out, seq), (out, seq)
+    return 0.0
+
+
+def test_neox_generate():
+    run_distributed(_neox_generate_worker, world_size=1)
