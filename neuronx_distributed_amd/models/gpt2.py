@@ -35,6 +35,14 @@ class GPT2Config:
     layer_norm_epsilon: float = 1e-5
     initializer_range: float = 0.02
 
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.num_attention_heads
+
+    @property
+    def num_key_value_heads(self):
+        return self.num_attention_heads  # MHA
+
 
 CONFIGS = {
     "gpt2-small": GPT2Config(),
@@ -69,22 +77,19 @@ class GPT2Attention(nn.Module):
             cfg.hidden_size, cfg.hidden_size, bias=True,
             input_is_parallel=True, init_method=_init(cfg.initializer_range))
 
-    def forward(self, x):
+    def forward(self, x, pos_offset=0, kv_cache=None):
         B, S, _ = x.shape
         qkv = self.c_attn(x)
         q, k, v = qkv.chunk(3, dim=-1)
         q = q.reshape(B, S, self.n_local, self.head_dim).transpose(1, 2)
         k = k.reshape(B, S, self.n_local, self.head_dim).transpose(1, 2)
         v = v.reshape(B, S, self.n_local, self.head_dim).transpose(1, 2)
-        if self.head_dim == 128 and q.is_cuda and q.dtype == torch.bfloat16:
-            out = flash_attn_func(q, k, v, causal=True)
-        else:
-            # generic head_dim: composed attention (flash kernel is D=128)
-            scale = 1.0 / math.sqrt(self.head_dim)
-            scores = (q.float() @ k.float().transpose(-1, -2)) * scale
-            mask = torch.ones(S, S, dtype=torch.bool, device=x.device).tril()
-            scores = scores.masked_fill(~mask, float("-inf"))
-            out = (torch.softmax(scores, -1) @ v.float()).to(x.dtype)
+        if kv_cache is not None:
+            k, v = kv_cache.update(k.contiguous(), v.contiguous(), pos_offset)
+        # dispatcher: HIP kernel for D=128 square shapes, composed
+        # batched-GEMM reference otherwise (odd head dims, decode)
+        out = flash_attn_func(q.contiguous(), k.contiguous(), v.contiguous(),
+                              causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
         return self.c_proj(out)
 
@@ -111,8 +116,8 @@ class GPT2Block(nn.Module):
         self.ln_2 = LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_epsilon)
         self.mlp = GPT2MLP(cfg)
 
-    def forward(self, x):
-        x = x + self.attn(self.ln_1(x))
+    def forward(self, x, pos_offset=0, kv_cache=None):
+        x = x + self.attn(self.ln_1(x), pos_offset, kv_cache)
         return x + self.mlp(self.ln_2(x))
 
 
@@ -131,12 +136,14 @@ class GPT2LMHeadModel(nn.Module):
             cfg.hidden_size, cfg.vocab_size, bias=False, gather_output=False,
             init_method=_init(cfg.initializer_range))
 
-    def forward(self, input_ids, labels=None):
+    def forward(self, input_ids, labels=None, pos_offset=0, kv_caches=None):
         B, S = input_ids.shape
-        pos = torch.arange(S, device=input_ids.device)
+        pos = torch.arange(pos_offset, pos_offset + S,
+                           device=input_ids.device)
         x = self.wte(input_ids) + self.wpe(pos)
-        for block in self.h:
-            x = block(x)
+        for i, block in enumerate(self.h):
+            kc = kv_caches[i] if kv_caches is not None else None
+            x = block(x, pos_offset, kc)
         x = self.ln_f(x)
         logits = self.lm_head(x)
         if labels is None:

@@ -434,3 +434,27 @@ def _neox_generate_worker(rank, world):
 
 def test_neox_generate():
     run_distributed(_neox_generate_worker, world_size=1)
+
+
+def _gpt2_generate_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models.gpt2 import (GPT2LMHeadModel,
+                                                     get_gpt2_config)
+    from neuronx_distributed_amd.inference.generation import generate
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m = GPT2LMHeadModel(get_gpt2_config("gpt2-tiny")).eval()
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 9))
+    out = generate(m, x, max_new_tokens=6)
+    seq = x
+    for _ in range(6):
+        logits = m(seq)
+        seq = torch.cat([seq, logits[:, -1, :].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(out, seq), (out, seq)
+    return 0.0
+
+
+def test_gpt2_generate():
+    run_distributed(_gpt2_generate_worker, world_size=1)
