@@ -3,7 +3,6 @@ family (test vehicle for PP partitioning, test/unit_test/pipeline/
 test_base.py, and BASELINE.json config #1: 2-layer GPT-2-small TP=2
 CPU/gloo plumbing check)."""
 
-import math
 from dataclasses import dataclass
 
 import torch

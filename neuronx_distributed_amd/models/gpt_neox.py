@@ -5,7 +5,6 @@ NeoX specifics vs GPT-2/llama: PARALLEL residual
 (x + attn(ln1(x)) + mlp(ln2(x))), LayerNorm with bias, fused QKV with
 bias, partial rotary embeddings (``rotary_pct`` of head_dim), no GQA."""
 
-import math
 from dataclasses import dataclass
 
 import torch
