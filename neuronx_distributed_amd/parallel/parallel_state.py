@@ -490,6 +490,29 @@ def get_kv_shared_group_size() -> int:
     return _info("kv").size
 
 
+def initialize_speculative_draft_group(draft_tp_size: int):
+    """Spec-draft groups (reference parallel_state.py:1533): sub-groups of
+    ``draft_tp_size`` ranks inside each TP group — a draft model usually
+    runs at a smaller TP degree on a subset of the target's ranks.
+    Callable any time after initialize_model_parallel (collective: every
+    rank must call with the same size)."""
+    global _GROUPS
+    tp_mesh = _info("tp").mesh
+    assert len(tp_mesh[0]) % draft_tp_size == 0, (
+        f"draft tp {draft_tp_size} must divide tp {len(tp_mesh[0])}")
+    mesh = []
+    for row in tp_mesh:
+        for i in range(0, len(row), draft_tp_size):
+            mesh.append(row[i:i + draft_tp_size])
+    _GROUPS["spec_draft"] = _new_group(mesh, "spec_draft")
+    return _GROUPS["spec_draft"]
+
+
+def get_speculative_draft_group(as_list: bool = False):
+    g = _info("spec_draft")
+    return g.mesh if as_list else g.group
+
+
 def get_token_shuffle_group(as_list: bool = False):
     g = _info("token_shuffle")
     return g.mesh if as_list else g.group

@@ -82,3 +82,17 @@ def test_kv_shared_mesh():
 def test_token_shuffle_mesh():
     mesh = ps._build_token_shuffle_mesh([[0, 2, 4, 6], [1, 3, 5, 7]], 2)
     assert mesh == [[0, 2], [4, 6], [1, 3], [5, 7]]
+
+
+def test_spec_draft_mesh():
+    # draft tp 2 inside tp 4: sub-groups split each tp row
+    import neuronx_distributed_amd.parallel.parallel_state as ps
+    g = _mesh(8, tp=4)
+    # mesh-only mode: build the draft mesh by the same rule used in
+    # initialize_speculative_draft_group
+    tp_mesh = g.tp_groups
+    mesh = []
+    for row in tp_mesh:
+        for i in range(0, len(row), 2):
+            mesh.append(row[i:i + 2])
+    assert mesh == [[0, 1], [2, 3], [4, 5], [6, 7]]
