@@ -109,6 +109,33 @@ def _initialize_affine_weight(
 # Fused forward/backward with async TP communication
 # ---------------------------------------------------------------------------
 
+def _sp_overlapped_linear(input_, weight, bias, tp_info):
+    """Chunked SP forward (NXDA_SP_OVERLAP=1): instead of one blocking
+    all-gather followed by one GEMM, post per-source-rank async broadcasts
+    and GEMM each sequence chunk as it lands — the later chunks' transfers
+    overlap the earlier chunks' MFMA work (SURVEY §7 hard-parts: fused SP
+    producer; chunk granularity = 1/tp of the sequence)."""
+    world = tp_info.size
+    me = tp_info.rank_in_group(dist.get_rank())
+    ranks = tp_info.ranks_of(dist.get_rank())
+    bufs = []
+    works = []
+    src = input_.contiguous()
+    for r in range(world):
+        buf = src if r == me else torch.empty_like(src)
+        bufs.append(buf)
+        works.append(dist.broadcast(buf, src=ranks[r], group=tp_info.group,
+                                    async_op=True))
+    outs = []
+    for r in range(world):
+        works[r].wait()
+        outs.append(F.linear(bufs[r], weight))
+    out = torch.cat(outs, dim=0)
+    if bias is not None:
+        out = out + bias
+    return out
+
+
 class LinearWithAsyncCommunication(torch.autograd.Function):
     """F.linear with TP/SP collectives placed for overlap.
 
@@ -126,7 +153,17 @@ class LinearWithAsyncCommunication(torch.autograd.Function):
         ctx.compute_weight_gradient = weight.requires_grad
 
         if sequence_parallel_enabled:
-            total_input = comm.all_gather(input_, dim=0, group=ps.get_group_info("tp"))
+            tp_info = ps.get_group_info("tp")
+            if os.environ.get("NXDA_SP_OVERLAP", "0") == "1" and \
+                    tp_info.size > 1 and not ps.is_aot_mode():
+                output = _sp_overlapped_linear(input_, weight, bias, tp_info)
+                if save_for_backward:
+                    if ctx.compute_weight_gradient:
+                        ctx.save_for_backward(input_, weight)
+                    else:
+                        ctx.save_for_backward(weight)
+                return output
+            total_input = comm.all_gather(input_, dim=0, group=tp_info)
         else:
             total_input = input_
 

@@ -326,3 +326,45 @@ def _reshard_worker(rank, world):
 
 def test_full_checkpoint_resharding():
     run_distributed(_reshard_worker, world_size=2)
+
+
+def _sp_overlap_worker(rank, world):
+    """NXDA_SP_OVERLAP=1 (chunked broadcast + per-chunk GEMM) produces the
+    same forward AND backward as the plain all-gather SP path."""
+    import os
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.layers import ColumnParallelLinear
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    lin = ColumnParallelLinear(16, 32, bias=True, gather_output=False,
+                               sequence_parallel_enabled=True)
+    torch.manual_seed(3 + rank)
+    x = torch.randn(4, 2, 16, requires_grad=True)  # (S/tp, B, H)
+
+    os.environ["NXDA_SP_OVERLAP"] = "0"
+    out_ref = lin(x)
+    g = torch.randn_like(out_ref)
+    out_ref.backward(g)
+    grads_ref = [p.grad.clone() for p in lin.parameters()]
+    xg_ref = x.grad.clone()
+    for p in lin.parameters():
+        p.grad = None
+    x.grad = None
+
+    os.environ["NXDA_SP_OVERLAP"] = "1"
+    try:
+        out = lin(x)
+        out.backward(g)
+    finally:
+        os.environ["NXDA_SP_OVERLAP"] = "0"
+    assert torch.allclose(out, out_ref, atol=1e-6)
+    for p, gr in zip(lin.parameters(), grads_ref):
+        assert torch.allclose(p.grad, gr, atol=1e-5), (p.grad - gr).abs().max()
+    assert torch.allclose(x.grad, xg_ref, atol=1e-5)
+    return float(out.sum())
+
+
+def test_sp_overlap_matches_plain():
+    run_distributed(_sp_overlap_worker, world_size=2)
