@@ -16,6 +16,8 @@ Every function takes a :class:`GroupInfo` or raw ProcessGroup.
 """
 
 
+import os
+
 import torch
 import torch.distributed as dist
 
@@ -51,12 +53,28 @@ def group_rank(group) -> int:
 
 def all_reduce(tensor: torch.Tensor, op=dist.ReduceOp.SUM, group=None,
                async_op: bool = False):
-    """In-place all-reduce. Returns the async work handle if requested."""
+    """In-place all-reduce. Returns the async work handle if requested.
+
+    NXDA_ONESHOT_AR_MAX_BYTES > 0 routes SUM all-reduces up to that size
+    through a one-shot all-gather + local reduce: on point-to-point xGMI
+    a ring all-reduce costs 2(N-1) latency steps on the per-link bound;
+    one-shot is a single exchange and wins for small latency-bound
+    payloads (SURVEY §7 hard-parts; default off, tune on 8 GPUs)."""
     if ps.is_aot_mode():
         return None
     if isinstance(group, ps.GroupInfo) and group.size == 1:
         return None
     g = _unwrap(group)
+    limit = int(os.environ.get("NXDA_ONESHOT_AR_MAX_BYTES", "0"))
+    if (limit > 0 and not async_op and op == dist.ReduceOp.SUM
+            and tensor.numel() * tensor.element_size() <= limit):
+        world = dist.get_world_size(group=g)
+        flat = tensor.reshape(-1)
+        gathered = torch.empty(world * flat.numel(), dtype=tensor.dtype,
+                               device=tensor.device)
+        dist.all_gather_into_tensor(gathered, flat.contiguous(), group=g)
+        flat.copy_(gathered.view(world, -1).sum(dim=0))
+        return None
     return dist.all_reduce(tensor, op=op, group=g, async_op=async_op)
 
 

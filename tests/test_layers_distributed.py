@@ -368,3 +368,28 @@ def _sp_overlap_worker(rank, world):
 
 def test_sp_overlap_matches_plain():
     run_distributed(_sp_overlap_worker, world_size=2)
+
+
+def _oneshot_ar_worker(rank, world):
+    """One-shot all-reduce (all-gather + local sum) matches dist ring
+    all-reduce when enabled for small payloads."""
+    import os
+
+    from neuronx_distributed_amd.parallel import comm, parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    t1 = torch.arange(17, dtype=torch.float32) * (rank + 1)
+    t2 = t1.clone()
+    comm.all_reduce(t1, group=ps.get_group_info("tp"))
+    os.environ["NXDA_ONESHOT_AR_MAX_BYTES"] = "1048576"
+    try:
+        comm.all_reduce(t2, group=ps.get_group_info("tp"))
+    finally:
+        os.environ["NXDA_ONESHOT_AR_MAX_BYTES"] = "0"
+    assert torch.allclose(t1, t2), (t1 - t2).abs().max()
+    return float(t1.sum())
+
+
+def test_oneshot_allreduce():
+    out = run_distributed(_oneshot_ar_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-5
