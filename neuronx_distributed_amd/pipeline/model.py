@@ -147,6 +147,16 @@ class NxDPPModel(nn.Module):
                 if my_rank in granks:
                     self._shared_weight_syncs.append((ent["param"], group))
 
+    def _purge_pending_sends(self):
+        """Drop isend bookkeeping (and the payload refs) for sends that
+        already completed — bounds the engine's activation footprint to
+        in-flight messages instead of the whole schedule (the reference's
+        deallocate-pipeline-outputs concern, model.py:1163-1215)."""
+        self._pending_sends = [
+            (works, refs) for works, refs in self._pending_sends
+            if not all(w.is_completed() for w in works)
+        ]
+
     def _sync_shared_weight_grads(self):
         import torch.distributed as dist
 
@@ -306,6 +316,7 @@ class NxDPPModel(nn.Module):
                 if is_loss_stage(task.chunk):
                     losses.append(out_list[0])
             elif isinstance(task, SendForward):
+                self._purge_pending_sends()
                 self._pending_sends.append(
                     ppcomm.send_async(outputs[key], self.next_rank))
             elif isinstance(task, SendForwardRecvBackward):
@@ -338,6 +349,7 @@ class NxDPPModel(nn.Module):
                 grads = [t.grad if t.grad is not None
                          else torch.zeros_like(t)
                          for t in recvd_inputs[key] if t.is_floating_point()]
+                self._purge_pending_sends()
                 self._pending_sends.append(
                     ppcomm.send_async(grads, self.prev_rank))
                 del recvd_inputs[key]
