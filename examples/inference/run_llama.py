@@ -9,7 +9,11 @@ The MI355X analogue of the reference's examples/inference/run_llama.py:
 
 import argparse
 import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                "..", ".."))
 
 import torch
 import torch.distributed as dist
@@ -30,6 +34,10 @@ def main():
     p.add_argument("--top-k", type=int, default=0, help="0 = greedy")
     args = p.parse_args()
 
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29793")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if torch.cuda.is_available():

@@ -15,6 +15,11 @@ import argparse
 import os
 import time
 
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                "..", ".."))
+
 import torch
 import torch.distributed as dist
 
@@ -23,7 +28,16 @@ from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
 from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
 
 
+def _single_process_defaults():
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29792")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    os.environ.setdefault("LOCAL_RANK", "0")
+
+
 def main():
+    _single_process_defaults()
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="llama2-7b")
     p.add_argument("--tp", type=int, default=8)
