@@ -185,11 +185,13 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
 
         # base optimizer over master shards, preserving per-group options
         base_groups = []
+        self._base_group_index = []  # base group -> source param_group idx
         for gi, group in enumerate(self.param_groups):
             opts = {k: v for k, v in group.items() if k != "params"}
             shard_params = [b.master for b in self.buckets if b.group_index == gi]
             if shard_params:
                 base_groups.append({"params": shard_params, **opts})
+                self._base_group_index.append(gi)
         self.base_optimizer = optimizer_class(base_groups, **{})
         self._grad_norm = None
 
@@ -339,6 +341,13 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
 
         if self.grad_clipping:
             self._clip_grads()
+        # LR schedulers mutate OUR param_groups; mirror hyperparams onto
+        # the base optimizer's (copied) groups before stepping
+        for bg, gi in zip(self.base_optimizer.param_groups,
+                          self._base_group_index):
+            for k, v in self.param_groups[gi].items():
+                if k != "params":
+                    bg[k] = v
         for b in self.buckets:
             b.master.grad = b.grad_shard.to(self.optimizer_dtype)
         self.base_optimizer.step()

@@ -458,3 +458,31 @@ def _gpt2_generate_worker(rank, world):
 
 def test_gpt2_generate():
     run_distributed(_gpt2_generate_worker, world_size=1)
+
+
+def _lr_sched_worker(rank, world):
+    """torch LR schedulers drive NxDOptimizer/zero1 param groups."""
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    cfg = nxd.neuronx_distributed_config(tensor_parallel_size=1)
+    model = nxd.initialize_parallel_model(
+        cfg, lambda: LlamaForCausalLM(get_config("tiny")))
+    opt = nxd.initialize_parallel_optimizer(cfg, torch.optim.AdamW,
+                                            model.parameters(), lr=1e-2)
+    sched = torch.optim.lr_scheduler.LambdaLR(
+        opt, lr_lambda=lambda step: 1.0 / (1 + step))
+    x = torch.randint(0, 256, (2, 16))
+    lrs = []
+    for _ in range(3):
+        opt.zero_grad()
+        model(x, labels=x).backward()
+        opt.step()
+        sched.step()
+        lrs.append(opt.param_groups[0]["lr"])
+    assert abs(lrs[0] - 5e-3) < 1e-9 and abs(lrs[1] - 1e-2 / 3) < 1e-9, lrs
+    return lrs[-1]
+
+
+def test_lr_scheduler_integration():
+    run_distributed(_lr_sched_worker, world_size=1)

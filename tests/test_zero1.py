@@ -205,3 +205,30 @@ def _fp32_acc_worker(rank, world):
 
 def test_zero1_fp32_grad_acc():
     run_distributed(_fp32_acc_worker, world_size=1)
+
+
+def _lr_propagation_worker(rank, world):
+    """Scheduler-updated lr reaches BOTH the fused path (reads
+    param_groups directly) and the base optimizer (mirrored groups):
+    stepping with lr=0 must leave weights unchanged."""
+    import torch
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m = torch.nn.Linear(16, 16)
+    opt = NeuronZero1Optimizer(m.parameters(), torch.optim.AdamW, lr=1e-2,
+                               grad_clipping=False)
+    opt.param_groups[0]["lr"] = 0.0  # what a scheduler would do
+    before = [p.detach().clone() for p in m.parameters()]
+    opt.zero_grad()
+    m(torch.randn(4, 16)).pow(2).mean().backward()
+    opt.step()
+    for p, b in zip(m.parameters(), before):
+        assert torch.equal(p.detach(), b), (p - b).abs().max()
+    return 0.0
+
+
+def test_zero1_lr_propagation():
+    run_distributed(_lr_propagation_worker, world_size=1)
