@@ -40,3 +40,29 @@ def create_dp_dataloader(dataset, batch_size: int, shuffle: bool = True,
                                  drop_last=drop_last)
     return DataLoader(dataset, batch_size=batch_size, sampler=sampler,
                       drop_last=drop_last, **kwargs)
+
+
+def pad_batch_to_multiple(batch: dict, multiple: int, seq_dim: int = 1,
+                          pad_token_id: int = 0, label_pad: int = -100):
+    """Right-pad every seq-dim tensor to a multiple (SP needs seq % tp == 0,
+    CP seq % cp == 0; reference examples pad in the dataloader).  Labels
+    (any key containing 'label') pad with ``label_pad`` so CE ignores the
+    padding positions."""
+    out = {}
+    pad_len = None
+    for k, v in batch.items():
+        if isinstance(v, torch.Tensor) and v.dim() > seq_dim:
+            S = v.shape[seq_dim]
+            target = (S + multiple - 1) // multiple * multiple
+            pad_len = target - S
+            if pad_len == 0:
+                out[k] = v
+                continue
+            fill = label_pad if "label" in k else pad_token_id
+            pad_shape = list(v.shape)
+            pad_shape[seq_dim] = pad_len
+            out[k] = torch.cat(
+                [v, v.new_full(pad_shape, fill)], dim=seq_dim)
+        else:
+            out[k] = v
+    return out, (pad_len or 0)

@@ -116,3 +116,20 @@ def test_tensor_replacement():
         out0 = m(x)
     assert torch.allclose(out0, m[1](torch.zeros(2, 4)))
     assert torch.allclose(m(x), base)
+
+
+def test_pad_batch_to_multiple():
+    from neuronx_distributed_amd.utils.batch_utils import (
+        pad_batch_to_multiple)
+
+    b = {"input_ids": torch.ones(2, 10, dtype=torch.long),
+         "labels": torch.ones(2, 10, dtype=torch.long),
+         "meta": 3}
+    out, pad = pad_batch_to_multiple(b, 8)
+    assert pad == 6
+    assert out["input_ids"].shape == (2, 16)
+    assert (out["input_ids"][:, 10:] == 0).all()
+    assert (out["labels"][:, 10:] == -100).all()
+    assert out["meta"] == 3
+    out2, pad2 = pad_batch_to_multiple(b, 5)
+    assert pad2 == 0 and out2["input_ids"].shape == (2, 10)
