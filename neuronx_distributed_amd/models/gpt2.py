@@ -147,6 +147,6 @@ class GPT2LMHeadModel(nn.Module):
         logits = self.lm_head(x)
         if labels is None:
             return logits
-        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(),
-                                      labels[:, 1:].contiguous())
-        return loss.mean()
+        shift = labels[:, 1:].contiguous()
+        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(), shift)
+        return loss.sum() / (shift != -100).sum().clamp(min=1)

@@ -189,6 +189,6 @@ class GPTNeoXForCausalLM(nn.Module):
         logits = self.embed_out(x)
         if labels is None:
             return logits
-        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(),
-                                      labels[:, 1:].contiguous())
-        return loss.mean()
+        shift = labels[:, 1:].contiguous()
+        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(), shift)
+        return loss.sum() / (shift != -100).sum().clamp(min=1)

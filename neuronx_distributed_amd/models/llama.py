@@ -346,6 +346,7 @@ class LlamaForCausalLM(nn.Module):
         logits = self.lm_head(hidden)  # (B,S,V/tp)
         if labels is None:
             return logits
-        loss = parallel_cross_entropy(
-            logits[:, :-1, :].contiguous(), labels[:, 1:].contiguous())
-        return loss.mean()
+        shift = labels[:, 1:].contiguous()
+        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(), shift)
+        valid = (shift != -100).sum().clamp(min=1)
+        return loss.sum() / valid

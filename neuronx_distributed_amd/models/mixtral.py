@@ -211,8 +211,10 @@ class MixtralForCausalLM(nn.Module):
         logits = self.lm_head(hidden)  # (B,S,V/tp)
         if labels is None:
             return logits
-        loss = parallel_cross_entropy(
-            logits[:, :-1, :].contiguous(), labels[:, 1:].contiguous()).mean()
+        shift = labels[:, 1:].contiguous()
+        loss = parallel_cross_entropy(logits[:, :-1, :].contiguous(),
+                                      shift).sum() / \
+            (shift != -100).sum().clamp(min=1)
         if router_logits and self.config.router_aux_loss_coef > 0:
             aux = load_balancing_loss_func(
                 torch.cat(router_logits, dim=0),

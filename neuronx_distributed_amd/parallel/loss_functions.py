@@ -131,9 +131,15 @@ class _ParallelCrossEntropy(torch.autograd.Function):
         return grad_input, None, None
 
 
-def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0):
+def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0,
+                           ignore_index=-100):
     """Per-token loss on vocab-sharded logits (reference loss_functions.py:217).
-    GPU bf16 + no smoothing dispatches to the fused HIP kernels."""
+    GPU bf16 + no smoothing dispatches to the fused HIP kernels.
+
+    Targets equal to ``ignore_index`` (HF padding convention) contribute
+    ZERO loss and zero gradient (the mask multiply below routes a zero
+    grad_output into the CE backward); callers wanting mean-over-valid
+    divide by the valid count (see the model zoo loss heads)."""
     if (not isinstance(vocab_parallel_logits, torch.fx.Proxy)
             and vocab_parallel_logits.is_cuda
             and vocab_parallel_logits.dtype == torch.bfloat16
@@ -142,13 +148,15 @@ def parallel_cross_entropy(vocab_parallel_logits, target, label_smoothing=0.0):
         from .. import ops
 
         if ops.is_available():
-            return _FusedParallelCrossEntropy.apply(vocab_parallel_logits,
-                                                    target)
+            raw = _FusedParallelCrossEntropy.apply(vocab_parallel_logits,
+                                                   target)
+            return raw * (target != ignore_index).to(raw.dtype)
     if not isinstance(vocab_parallel_logits, torch.fx.Proxy) and \
             vocab_parallel_logits.dtype in (torch.bfloat16, torch.float16):
         vocab_parallel_logits = vocab_parallel_logits.float()
-    return _ParallelCrossEntropy.apply(vocab_parallel_logits, target,
-                                       label_smoothing)
+    raw = _ParallelCrossEntropy.apply(vocab_parallel_logits, target,
+                                      label_smoothing)
+    return raw * (target != ignore_index).to(raw.dtype)
 
 
 # keep the collective-bearing loss opaque to the pipeline FX tracer

@@ -393,3 +393,37 @@ def _oneshot_ar_worker(rank, world):
 def test_oneshot_allreduce():
     out = run_distributed(_oneshot_ar_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-5
+
+
+def _ce_ignore_worker(rank, world):
+    """parallel_cross_entropy ignores -100 targets (zero loss + zero grad)
+    and matches torch F.cross_entropy(ignore_index=-100) at tp2."""
+    import torch.nn.functional as F
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.loss_functions import (
+        parallel_cross_entropy)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    full = torch.randn(6, 16, requires_grad=True)
+    tgt = torch.tensor([1, 3, -100, 7, -100, 12])
+    shard = full.detach().chunk(world, dim=-1)[rank].requires_grad_(True)
+
+    loss_tok = parallel_cross_entropy(shard, tgt)
+    assert float(loss_tok[2]) == 0.0 and float(loss_tok[4]) == 0.0
+    loss = loss_tok.sum() / 4
+    loss.backward()
+
+    ref = F.cross_entropy(full, tgt, ignore_index=-100)
+    ref.backward()
+    assert abs(float(loss) - float(ref)) < 1e-5, (loss, ref)
+    gref = full.grad.chunk(world, dim=-1)[rank]
+    assert torch.allclose(shard.grad, gref, atol=1e-5), \
+        (shard.grad - gref).abs().max()
+    return float(loss)
+
+
+def test_ce_ignore_index():
+    out = run_distributed(_ce_ignore_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-6
