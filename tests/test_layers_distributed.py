@@ -427,3 +427,27 @@ def _ce_ignore_worker(rank, world):
 def test_ce_ignore_index():
     out = run_distributed(_ce_ignore_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-6
+
+
+def _logprobs_worker(rank, world):
+    """from_parallel_logits_to_logprobs == log_softmax gather of the
+    shifted target at tp2."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.loss_functions import (
+        from_parallel_logits_to_logprobs)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    full = torch.randn(2, 6, 16)
+    tgt = torch.randint(0, 16, (2, 6))
+    shard = full.chunk(world, dim=-1)[rank]
+    lp = from_parallel_logits_to_logprobs(shard, tgt, inference=True)
+    ref = torch.log_softmax(full[:, :-1, :], dim=-1).gather(
+        -1, tgt[:, 1:].unsqueeze(-1)).squeeze(-1)
+    assert torch.allclose(lp, ref, atol=1e-5), (lp - ref).abs().max()
+    return float(lp.sum())
+
+
+def test_parallel_logprobs():
+    out = run_distributed(_logprobs_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-5
