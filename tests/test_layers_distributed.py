@@ -451,3 +451,33 @@ def _logprobs_worker(rank, world):
 def test_parallel_logprobs():
     out = run_distributed(_logprobs_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-5
+
+
+def _lowlevel_ckpt_worker(rank, world):
+    """parallel.checkpointing save/load: tp_rank_xx_pp_rank_xx shard files
+    roundtrip with staggered (serial) loading."""
+    import os
+    import tempfile
+
+    from neuronx_distributed_amd.parallel import checkpointing
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.layers import ColumnParallelLinear
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    lin = ColumnParallelLinear(8, 12, bias=True, gather_output=False)
+    path = os.path.join(tempfile.gettempdir(), "nxda_lowlevel_ckpt")
+    checkpointing.save(lin.state_dict(), path, tag="model")
+
+    torch.manual_seed(99)
+    lin2 = ColumnParallelLinear(8, 12, bias=True, gather_output=False)
+    assert not torch.allclose(lin2.weight, lin.weight)
+    checkpointing.load(path, tag="model", model=lin2, load_serially=True)
+    assert torch.allclose(lin2.weight, lin.weight)
+    assert torch.allclose(lin2.bias, lin.bias)
+    return float(lin2.weight.sum())
+
+
+def test_lowlevel_checkpointing():
+    out = run_distributed(_lowlevel_ckpt_worker, world_size=2)
+    assert out[0] != out[1]  # different shards per tp rank
