@@ -121,6 +121,10 @@ class RMSNorm(nn.Module):
         self.sequence_parallel_enabled = sequence_parallel_enabled
         self.weight.sequence_parallel_enabled = sequence_parallel_enabled
 
+    def reset_parameters(self):
+        with torch.no_grad():
+            self.weight.fill_(1.0)
+
     def forward(self, x):
         return ops.rmsnorm(x, self.weight, self.variance_epsilon)
 
@@ -309,6 +313,15 @@ class LlamaModel(nn.Module):
                                              config.head_dim, config.rope_theta)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+
+    def reset_parameters(self):
+        # meta materialization clobbers non-persistent buffers: recompute
+        cos, sin = ops.precompute_rope_freqs(
+            self.config.max_position_embeddings, self.config.head_dim,
+            self.config.rope_theta, device=self.rope_cos.device)
+        with torch.no_grad():
+            self.rope_cos.copy_(cos)
+            self.rope_sin.copy_(sin)
 
     def forward(self, input_ids, pos_offset=0, kv_caches=None):
         hidden = self.embed_tokens(input_ids)  # (B,S,H)

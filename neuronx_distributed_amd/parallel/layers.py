@@ -251,6 +251,20 @@ def linear_with_async_allreduce(input_, weight, bias=None,
 
 
 class BaseParallelLinear(nn.Module):
+    def reset_parameters(self):
+        """Re-run the deterministic TP-invariant init (meta-device
+        materialization path, utils/model_utils.reinit_model)."""
+        spec = getattr(self, "_reset_spec", None)
+        if spec is None:
+            return
+        out_f, in_f, per_part, pdim, stride, init_method, dtype = spec
+        _initialize_affine_weight(self.weight, out_f, in_f, per_part,
+                                  partition_dim=pdim, init_method=init_method,
+                                  stride=stride, dtype=dtype)
+        if getattr(self, "bias", None) is not None:
+            with torch.no_grad():
+                self.bias.zero_()
+
     def _init_weight(self, weight, init_method):
         if getattr(weight, "device", torch.device("cpu")).type == "meta":
             return
@@ -302,6 +316,9 @@ class ColumnParallelLinear(BaseParallelLinear):
         self.weight = nn.Parameter(
             torch.empty(self.output_size_per_partition, input_size, dtype=dtype,
                         device=device))
+        self._reset_spec = (output_size, input_size,
+                            self.output_size_per_partition, 0, stride,
+                            init_method, dtype)
         self.master_weight = _initialize_affine_weight(
             self.weight, output_size, input_size, self.output_size_per_partition,
             partition_dim=0, init_method=init_method, stride=stride,
@@ -366,6 +383,9 @@ class RowParallelLinear(BaseParallelLinear):
         self.weight = nn.Parameter(
             torch.empty(output_size, self.input_size_per_partition, dtype=dtype,
                         device=device))
+        self._reset_spec = (output_size, input_size,
+                            self.input_size_per_partition, 1, stride,
+                            init_method, dtype)
         self.master_weight = _initialize_affine_weight(
             self.weight, output_size, input_size, self.input_size_per_partition,
             partition_dim=1, init_method=init_method, stride=stride,
@@ -435,6 +455,9 @@ class ParallelEmbedding(BaseParallelLinear):
             self.weight = nn.Parameter(
                 torch.empty(num_embeddings, self.embedding_dim_per_partition,
                             dtype=dtype, device=device))
+            self._reset_spec = (num_embeddings, embedding_dim,
+                                self.embedding_dim_per_partition, 1, 1,
+                                init_method, dtype)
             _initialize_affine_weight(
                 self.weight, num_embeddings, embedding_dim,
                 self.embedding_dim_per_partition, partition_dim=1,
@@ -447,6 +470,9 @@ class ParallelEmbedding(BaseParallelLinear):
         self.weight = nn.Parameter(
             torch.empty(self.num_embeddings_per_partition, embedding_dim,
                         dtype=dtype, device=device))
+        self._reset_spec = (num_embeddings, embedding_dim,
+                            self.num_embeddings_per_partition, 0, 1,
+                            init_method, dtype)
         _initialize_affine_weight(
             self.weight, num_embeddings, embedding_dim,
             self.num_embeddings_per_partition, partition_dim=0,

@@ -146,6 +146,7 @@ class GQAQKVColumnParallelLinear(BaseParallelLinear):
             set_tensor_model_parallel_attributes(w, world > 1, 0, 1, world)
             w.kv_shared = kv_size_multiplier > 1
 
+        self._reset_args = (input_size, q_out, kv_out, init_method)
         if not ps.is_aot_mode():
             self._deterministic_init(input_size, q_out, kv_out, init_method)
 
@@ -162,6 +163,14 @@ class GQAQKVColumnParallelLinear(BaseParallelLinear):
             self.register_parameter("bias_q", None)
             self.register_parameter("bias_k", None)
             self.register_parameter("bias_v", None)
+
+    def reset_parameters(self):
+        """Meta-materialization path: re-run the deterministic init."""
+        self._deterministic_init(*self._reset_args)
+        for b in (self.bias_q, self.bias_k, self.bias_v):
+            if b is not None:
+                with torch.no_grad():
+                    b.zero_()
 
     def _kv_local_rows(self, master: torch.Tensor) -> torch.Tensor:
         """Slice this rank's replicated-KV rows out of the full

@@ -486,3 +486,30 @@ def _lr_sched_worker(rank, world):
 
 def test_lr_scheduler_integration():
     run_distributed(_lr_sched_worker, world_size=1)
+
+
+def _meta_init_worker(rank, world):
+    """meta_device_init: model builds on meta (no host RAM for weights)
+    and materializes through the trainer phases; a train step runs."""
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    cfg = nxd.neuronx_distributed_config(
+        tensor_parallel_size=1,
+        model_init_config={"meta_device_init": True})
+    model = nxd.initialize_parallel_model(
+        cfg, lambda: LlamaForCausalLM(get_config("tiny")))
+    for p in model.parameters():
+        assert p.device.type != "meta"
+    opt = nxd.initialize_parallel_optimizer(cfg, torch.optim.AdamW,
+                                            model.parameters(), lr=1e-2)
+    x = torch.randint(0, 256, (2, 16))
+    loss = model(x, labels=x)
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss)
+    return float(loss.detach())
+
+
+def test_meta_device_init():
+    run_distributed(_meta_init_worker, world_size=1)
