@@ -180,6 +180,16 @@ class GPTNeoXForCausalLM(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
+    def reset_parameters(self):
+        cfg = self.config
+        cos, sin = ops.precompute_rope_freqs(
+            cfg.max_position_embeddings,
+            max(2, 2 * (int(cfg.head_dim * cfg.rotary_pct) // 2)),
+            cfg.rope_theta, device=self.rope_cos.device)
+        with torch.no_grad():
+            self.rope_cos.copy_(cos)
+            self.rope_sin.copy_(sin)
+
     def forward(self, input_ids, labels=None, pos_offset=0, kv_caches=None):
         x = self.embed_in(input_ids)
         for i, layer in enumerate(self.layers):

@@ -171,6 +171,15 @@ class MixtralModel(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
+    def reset_parameters(self):
+        # meta materialization clobbers non-persistent buffers: recompute
+        cos, sin = ops.precompute_rope_freqs(
+            self.config.max_position_embeddings, self.config.head_dim,
+            self.config.rope_theta, device=self.rope_cos.device)
+        with torch.no_grad():
+            self.rope_cos.copy_(cos)
+            self.rope_sin.copy_(sin)
+
     def forward(self, input_ids, pos_offset=0, kv_caches=None):
         hidden = self.embed_tokens(input_ids)  # (B,S,H)
         if self.config.sequence_parallel_enabled:

@@ -26,6 +26,16 @@ class _ExpertFusedLinearBase(nn.Module):
         if ps.get_expert_model_parallel_size() > 1:
             setattr(param, EXPERT_PARALLEL_ATTR, True)
 
+    def reset_parameters(self):
+        """Meta-materialization: re-run the deterministic per-expert init."""
+        args = getattr(self, "_reset_args", None)
+        if args is None or not hasattr(self, "_reset_init"):
+            return
+        out_size, in_size, pdim = args
+        init_method, dtype, stride = self._reset_init
+        self._init_expertwise(self.weight, out_size, in_size, pdim,
+                              init_method, dtype, stride=stride)
+
     def _init_expertwise(self, weight, full_out, full_in, partition_dim,
                          init_method, dtype, stride=1):
         """Deterministic per-expert init: full (H_in, H_out) master per
@@ -64,6 +74,8 @@ class ExpertFusedColumnParallelLinear(_ExpertFusedLinearBase):
         self._tag(self.weight, 2)
         self.weight.partition_stride = stride
         init_method = init_method or (lambda t: nn.init.normal_(t, std=0.02))
+        self._reset_args = (output_size, input_size, 2)
+        self._reset_init = (init_method, dtype, stride)
         self._init_expertwise(self.weight, output_size, input_size, 2,
                               init_method, dtype, stride=stride)
 
@@ -87,6 +99,8 @@ class ExpertFusedRowParallelLinear(_ExpertFusedLinearBase):
             dtype=dtype, device=device))
         self._tag(self.weight, 1)
         init_method = init_method or (lambda t: nn.init.normal_(t, std=0.02))
+        self._reset_args = (output_size, input_size, 1)
+        self._reset_init = (init_method, dtype, 1)
         self._init_expertwise(self.weight, output_size, input_size, 1,
                               init_method, dtype)
 

@@ -428,3 +428,38 @@ def _ep_zero1_worker(rank, world):
 def test_ep_zero1():
     out = run_distributed(_ep_zero1_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-5
+
+
+def _mixtral_meta_worker(rank, world):
+    """Mixtral under meta_device_init materializes with real (finite,
+    trained-able) weights."""
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+
+    cfg = nxd.neuronx_distributed_config(
+        tensor_parallel_size=1,
+        model_init_config={"meta_device_init": True})
+    model = nxd.initialize_parallel_model(
+        cfg, lambda: MixtralForCausalLM(get_moe_config("tiny-moe")))
+    # rope buffers must be REAL tables, not to_empty garbage
+    from neuronx_distributed_amd import ops as _ops
+    ref_cos, _ = _ops.precompute_rope_freqs(
+        model.module.config.max_position_embeddings
+        if hasattr(model, "module") else model.config.max_position_embeddings,
+        16 // 4 * 4, 1e6)
+    inner = model.module if hasattr(model, "module") else model
+    base = inner.model if hasattr(inner, "model") else inner
+    cfg2 = inner.config
+    ref_cos, _ = _ops.precompute_rope_freqs(cfg2.max_position_embeddings,
+                                            cfg2.head_dim, cfg2.rope_theta)
+    assert torch.allclose(base.rope_cos.float().cpu(), ref_cos, atol=1e-5)
+    x = torch.randint(0, 256, (2, 16))
+    loss = model(x, labels=x)
+    assert torch.isfinite(loss), loss
+    loss.backward()
+    return float(loss.detach())
+
+
+def test_mixtral_meta_init():
+    run_distributed(_mixtral_meta_worker, world_size=1)
