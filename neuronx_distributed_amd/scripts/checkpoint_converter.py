@@ -17,17 +17,24 @@ class CheckpointConverterBase:
     """Subclass and override the *_partition_dim tables per architecture.
     Keys are SUFFIX matches on parameter names."""
 
-    # name-suffix -> partition dim (column-parallel: 0, row-parallel: 1)
+    # name-suffix -> partition dim (column-parallel: 0, row-parallel: 1);
+    # 3-D expert-fused weights (E, in, out) map column->dim 2, row->dim 1
     COLUMN_PARALLEL_SUFFIXES = ["q_proj.weight", "k_proj.weight",
                                 "v_proj.weight", "gate_proj.weight",
                                 "up_proj.weight", "lm_head.weight",
                                 "embed_tokens.weight", "wte.weight",
-                                "c_fc.weight"]
+                                "c_fc.weight",
+                                # native layer names (this package)
+                                "qkv_proj.weight_q", "qkv_proj.weight_k",
+                                "qkv_proj.weight_v", "embed_in.weight",
+                                "embed_out.weight", "dense_h_to_4h.weight"]
     ROW_PARALLEL_SUFFIXES = ["o_proj.weight", "down_proj.weight",
-                             "c_proj.weight"]
+                             "c_proj.weight", "dense_4h_to_h.weight",
+                             "attention.dense.weight"]
     # fused [gate; up] / [q;k;v] weights: (suffix, num_blocks)
     STRIDED_COLUMN_SUFFIXES = [("gate_up_proj.weight", 2),
-                               ("c_attn.weight", 3)]
+                               ("c_attn.weight", 3),
+                               ("query_key_value.weight", 3)]
 
     def _dim_of(self, name: str):
         for suf, stride in self.STRIDED_COLUMN_SUFFIXES:
@@ -52,6 +59,8 @@ class CheckpointConverterBase:
                 if dim is None or not isinstance(w, torch.Tensor):
                     shard[name] = w
                 else:
+                    if w.dim() == 3:  # expert-fused (E, in, out)
+                        dim = 2 if dim == 0 else 1
                     per = w.shape[dim] // tp_degree
                     shard[name] = create_local_weight(w, dim, per, stride,
                                                       rank=r,
@@ -70,6 +79,8 @@ class CheckpointConverterBase:
                 full[name] = w0
                 continue
             parts = [s[name] for s in shards]
+            if w0.dim() == 3:  # expert-fused (E, in, out)
+                dim = 2 if dim == 0 else 1
             if stride == 1:
                 full[name] = torch.cat(parts, dim=dim)
             else:
