@@ -151,3 +151,34 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def convert_hf_llama_state_dict(hf_sd):
+    """HF LlamaForCausalLM checkpoint -> this package's native llama
+    layout: q/k/v_proj become qkv_proj.weight_q/k/v, gate_proj+up_proj
+    fuse into gate_up_proj.weight ([gate; up] rows — the stride-2 fused
+    FULL layout the converter shards), rotary inv_freq buffers drop
+    (recomputed at init).  Norms/o_proj/down_proj/embed/lm_head names are
+    already identical."""
+    out = {}
+    gates = {}
+    ups = {}
+    for k, v in hf_sd.items():
+        if k.endswith("rotary_emb.inv_freq"):
+            continue
+        if k.endswith("self_attn.q_proj.weight"):
+            out[k.replace("q_proj.weight", "qkv_proj.weight_q")] = v
+        elif k.endswith("self_attn.k_proj.weight"):
+            out[k.replace("k_proj.weight", "qkv_proj.weight_k")] = v
+        elif k.endswith("self_attn.v_proj.weight"):
+            out[k.replace("v_proj.weight", "qkv_proj.weight_v")] = v
+        elif k.endswith("mlp.gate_proj.weight"):
+            gates[k.rsplit("gate_proj.weight", 1)[0]] = v
+        elif k.endswith("mlp.up_proj.weight"):
+            ups[k.rsplit("up_proj.weight", 1)[0]] = v
+        else:
+            out[k] = v
+    assert set(gates) == set(ups), "gate/up projections must pair up"
+    for base, g in gates.items():
+        out[base + "gate_up_proj.weight"] = torch.cat([g, ups[base]], dim=0)
+    return out

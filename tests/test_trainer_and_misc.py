@@ -544,3 +544,47 @@ def _native_converter_worker(rank, world):
 
 def test_native_converter_roundtrip():
     run_distributed(_native_converter_worker, world_size=1)
+
+
+def _hf_convert_worker(rank, world):
+    """HF llama checkpoint names convert to the native layout, load into
+    the model, and the fused gate_up halves land in the right place."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.scripts.checkpoint_converter import (
+        convert_hf_llama_state_dict)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("tiny")
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(cfg)
+    native = m.state_dict()
+
+    # build an HF-shaped dict from the native one (inverse mapping)
+    hf = {}
+    for k, v in native.items():
+        if k.endswith("qkv_proj.weight_q"):
+            hf[k.replace("qkv_proj.weight_q", "q_proj.weight")] = v
+        elif k.endswith("qkv_proj.weight_k"):
+            hf[k.replace("qkv_proj.weight_k", "k_proj.weight")] = v
+        elif k.endswith("qkv_proj.weight_v"):
+            hf[k.replace("qkv_proj.weight_v", "v_proj.weight")] = v
+        elif k.endswith("gate_up_proj.weight"):
+            I = v.shape[0] // 2
+            hf[k.replace("gate_up_proj.weight", "gate_proj.weight")] = v[:I]
+            hf[k.replace("gate_up_proj.weight", "up_proj.weight")] = v[I:]
+        else:
+            hf[k] = v
+    hf["model.layers.0.self_attn.rotary_emb.inv_freq"] = torch.zeros(4)
+
+    back = convert_hf_llama_state_dict(hf)
+    torch.manual_seed(1)
+    m2 = LlamaForCausalLM(cfg)
+    m2.load_state_dict(back)
+    x = torch.randint(0, 256, (2, 8))
+    assert torch.allclose(m(x), m2(x), atol=1e-6)
+    return 0.0
+
+
+def test_hf_llama_conversion():
+    run_distributed(_hf_convert_worker, world_size=1)
