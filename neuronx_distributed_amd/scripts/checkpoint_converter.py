@@ -15,7 +15,13 @@ from ..parallel.utils import create_local_weight
 
 class CheckpointConverterBase:
     """Subclass and override the *_partition_dim tables per architecture.
-    Keys are SUFFIX matches on parameter names."""
+    Keys are SUFFIX matches on parameter names.
+
+    Limitation: GQA KV-head REPLICATION (tp > kv heads) is a runtime
+    resharding concern — load the FULL checkpoint through
+    ``parallel.checkpointing.load(sharded=False)`` whose preshard_hook
+    applies the replication layout; this offline converter shards evenly
+    and covers tp <= kv_heads."""
 
     # name-suffix -> partition dim (column-parallel: 0, row-parallel: 1);
     # 3-D expert-fused weights (E, in, out) map column->dim 2, row->dim 1
