@@ -188,3 +188,36 @@ def convert_hf_llama_state_dict(hf_sd):
     for base, g in gates.items():
         out[base + "gate_up_proj.weight"] = torch.cat([g, ups[base]], dim=0)
     return out
+
+
+def convert_hf_mixtral_state_dict(hf_sd, num_experts: int):
+    """HF MixtralForCausalLM -> native layout: per-expert w1/w3 (gate/up,
+    (I,H)) stack-transpose into the expert-fused gate_up (E, H, 2I)
+    [gate|up] on the last dim; w2 (down, (H,I)) into down (E, I, H);
+    block_sparse_moe.gate -> router.linear_router (fp32).  Attention and
+    norms follow the llama mapping."""
+    out = convert_hf_llama_state_dict(
+        {k: v for k, v in hf_sd.items() if ".block_sparse_moe." not in k})
+    layers = {}
+    for k, v in hf_sd.items():
+        if ".block_sparse_moe." not in k:
+            continue
+        base, rest = k.split(".block_sparse_moe.", 1)
+        layers.setdefault(base, {})[rest] = v
+    for base, params in layers.items():
+        if "gate.weight" in params:
+            out[f"{base}.block_sparse_moe.router.linear_router.weight"] = \
+                params["gate.weight"].float()
+        gate_up = []
+        down = []
+        for e in range(num_experts):
+            w1 = params[f"experts.{e}.w1.weight"]  # (I, H) gate
+            w3 = params[f"experts.{e}.w3.weight"]  # (I, H) up
+            w2 = params[f"experts.{e}.w2.weight"]  # (H, I) down
+            gate_up.append(torch.cat([w1.t(), w3.t()], dim=1))  # (H, 2I)
+            down.append(w2.t())  # (I, H)
+        out[f"{base}.block_sparse_moe.expert_mlps.gate_up_proj.weight"] = \
+            torch.stack(gate_up)
+        out[f"{base}.block_sparse_moe.expert_mlps.down_proj.weight"] = \
+            torch.stack(down)
+    return out

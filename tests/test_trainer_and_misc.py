@@ -588,3 +588,53 @@ def _hf_convert_worker(rank, world):
 
 def test_hf_llama_conversion():
     run_distributed(_hf_convert_worker, world_size=1)
+
+
+def _hf_mixtral_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+    from neuronx_distributed_amd.scripts.checkpoint_converter import (
+        convert_hf_mixtral_state_dict)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_moe_config("tiny-moe")
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(cfg)
+    native = m.state_dict()
+
+    hf = {}
+    for k, v in native.items():
+        if k.endswith("qkv_proj.weight_q"):
+            hf[k.replace("qkv_proj.weight_q", "q_proj.weight")] = v
+        elif k.endswith("qkv_proj.weight_k"):
+            hf[k.replace("qkv_proj.weight_k", "k_proj.weight")] = v
+        elif k.endswith("qkv_proj.weight_v"):
+            hf[k.replace("qkv_proj.weight_v", "v_proj.weight")] = v
+        elif k.endswith("router.linear_router.weight"):
+            hf[k.replace("router.linear_router.weight", "gate.weight")] = v
+        elif k.endswith("expert_mlps.gate_up_proj.weight"):
+            base = k.rsplit(".expert_mlps.gate_up_proj.weight", 1)[0]
+            I = v.shape[2] // 2
+            for e in range(v.shape[0]):
+                hf[f"{base}.experts.{e}.w1.weight"] = v[e, :, :I].t().contiguous()
+                hf[f"{base}.experts.{e}.w3.weight"] = v[e, :, I:].t().contiguous()
+        elif k.endswith("expert_mlps.down_proj.weight"):
+            base = k.rsplit(".expert_mlps.down_proj.weight", 1)[0]
+            for e in range(v.shape[0]):
+                hf[f"{base}.experts.{e}.w2.weight"] = v[e].t().contiguous()
+        else:
+            hf[k] = v
+
+    back = convert_hf_mixtral_state_dict(hf, cfg.num_local_experts)
+    torch.manual_seed(1)
+    m2 = MixtralForCausalLM(cfg)
+    m2.load_state_dict(back)
+    x = torch.randint(0, 256, (2, 8))
+    with torch.no_grad():
+        assert torch.allclose(m(x), m2(x), atol=1e-6)
+    return 0.0
+
+
+def test_hf_mixtral_conversion():
+    run_distributed(_hf_mixtral_worker, world_size=1)
