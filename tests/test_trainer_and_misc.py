@@ -638,3 +638,25 @@ def _hf_mixtral_worker(rank, world):
 
 def test_hf_mixtral_conversion():
     run_distributed(_hf_mixtral_worker, world_size=1)
+
+
+def test_serving_endpoint():
+    """The FastAPI serving demo answers /generate in-process."""
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location(
+        "serve_demo", os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), "examples", "inference", "serve.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    from starlette.testclient import TestClient
+
+    app = mod.build_app("tiny")
+    client = TestClient(app)
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+    r = client.post("/generate", json={"token_ids": [[1, 2, 3]],
+                                       "max_new_tokens": 4})
+    assert r.status_code == 200
+    ids = r.json()["token_ids"]
+    assert len(ids) == 1 and len(ids[0]) == 4
