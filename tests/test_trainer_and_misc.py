@@ -651,12 +651,21 @@ def test_serving_endpoint():
     spec.loader.exec_module(mod)
     from starlette.testclient import TestClient
 
-    app = mod.build_app("tiny")
-    client = TestClient(app)
-    r = client.get("/health")
-    assert r.status_code == 200 and r.json()["status"] == "ok"
-    r = client.post("/generate", json={"token_ids": [[1, 2, 3]],
-                                       "max_new_tokens": 4})
-    assert r.status_code == 200
-    ids = r.json()["token_ids"]
-    assert len(ids) == 1 and len(ids[0]) == 4
+    try:
+        app = mod.build_app("tiny")
+        client = TestClient(app)
+        r = client.get("/health")
+        assert r.status_code == 200 and r.json()["status"] == "ok"
+        r = client.post("/generate", json={"token_ids": [[1, 2, 3]],
+                                           "max_new_tokens": 4})
+        assert r.status_code == 200
+        ids = r.json()["token_ids"]
+        assert len(ids) == 1 and len(ids[0]) == 4
+    finally:
+        # leave the pytest process clean for later tests
+        import torch.distributed as dist
+        from neuronx_distributed_amd.parallel import parallel_state as ps
+        if ps.model_parallel_is_initialized():
+            ps.destroy_model_parallel()
+        if dist.is_initialized():
+            dist.destroy_process_group()
