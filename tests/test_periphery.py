@@ -133,3 +133,12 @@ def test_pad_batch_to_multiple():
     assert out["meta"] == 3
     out2, pad2 = pad_batch_to_multiple(b, 5)
     assert pad2 == 0 and out2["input_ids"].shape == (2, 10)
+
+
+def test_memory_stats_cpu():
+    from neuronx_distributed_amd.utils.memory import (log_memory_stats,
+                                                      memory_stats)
+
+    s = memory_stats()
+    assert set(s) >= {"allocated_gib", "reserved_gib", "peak_allocated_gib"}
+    log_memory_stats("test")
