@@ -57,3 +57,6 @@ from .inference import (
 )
 
 trainer = _trainer_pkg
+
+from .inference.generation import generate
+from .inference.speculation import medusa_generate, speculative_generate
