@@ -139,6 +139,9 @@ def initialize_parallel_model(nxd_config: Dict, model_fn: Callable, *model_args,
         apply_activation_checkpointing(target,
                                        activation_checkpoint_classes=classes)
 
+    from .post_partition_hooks import run_post_partition_hooks
+
+    run_post_partition_hooks(model, nxd_config)
     return NxDModel(model, nxd_config)
 
 

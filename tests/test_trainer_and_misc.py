@@ -669,3 +669,29 @@ def test_serving_endpoint():
             ps.destroy_model_parallel()
         if dist.is_initialized():
             dist.destroy_process_group()
+
+
+def _pph_worker(rank, world):
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.trainer.post_partition_hooks import (
+        clear_post_partition_hooks, register_post_partition_hook)
+
+    calls = []
+
+    @register_post_partition_hook
+    def mark(model, cfg):
+        calls.append(type(model).__name__)
+
+    try:
+        cfg = nxd.neuronx_distributed_config(tensor_parallel_size=1)
+        nxd.initialize_parallel_model(
+            cfg, lambda: LlamaForCausalLM(get_config("tiny")))
+    finally:
+        clear_post_partition_hooks()
+    assert calls == ["LlamaForCausalLM"], calls
+    return 0.0
+
+
+def test_post_partition_hooks():
+    run_distributed(_pph_worker, world_size=1)
