@@ -695,3 +695,22 @@ def _pph_worker(rank, world):
 
 def test_post_partition_hooks():
     run_distributed(_pph_worker, world_size=1)
+
+
+def test_checkpoint_storage_backend(tmp_path):
+    import pytest as _pytest
+    from neuronx_distributed_amd.trainer.checkpoint_storage import (
+        LocalStorage, get_storage)
+
+    st = get_storage(str(tmp_path))
+    assert isinstance(st, LocalStorage)
+    st.save_object({"a": torch.ones(3)}, "tag/model/x.pt")
+    assert st.exists("tag/model/x.pt")
+    obj = st.load_object("tag/model/x.pt")
+    assert torch.equal(obj["a"], torch.ones(3))
+    st.write_text("tag/done", "done")
+    assert "tag" in st.listdir()
+    st.remove_tree("tag")
+    assert not st.exists("tag/done")
+    with _pytest.raises(NotImplementedError):
+        get_storage("s3://bucket/prefix")
