@@ -463,3 +463,42 @@ def _mixtral_meta_worker(rank, world):
 
 def test_mixtral_meta_init():
     run_distributed(_mixtral_meta_worker, world_size=1)
+
+
+def _mixtral_ckpt_worker(rank, world):
+    """Activation checkpointing over MixtralDecoderLayer (tuple outputs
+    with router logits) recomputes correctly: grads match the unwrapped
+    model."""
+    import copy
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+    from neuronx_distributed_amd.models.mixtral import MixtralDecoderLayer
+    from neuronx_distributed_amd.utils.activation_checkpoint import (
+        apply_activation_checkpointing)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m = MixtralForCausalLM(get_moe_config("tiny-moe"))
+    ref = copy.deepcopy(m)
+    apply_activation_checkpointing(
+        m, activation_checkpoint_classes=(MixtralDecoderLayer,))
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    loss = m(x, labels=x)
+    loss.backward()
+    rl = ref(x, labels=x)
+    rl.backward()
+    assert abs(float(loss) - float(rl)) < 1e-6
+    for (n1, p1), (n2, p2) in zip(m.named_parameters(),
+                                  ref.named_parameters()):
+        if p1.grad is None:
+            assert p2.grad is None, n1
+            continue
+        assert torch.allclose(p1.grad, p2.grad, atol=1e-6), n1
+    return float(loss.detach())
+
+
+def test_mixtral_activation_checkpointing():
+    run_distributed(_mixtral_ckpt_worker, world_size=1)
