@@ -62,6 +62,8 @@ class LlamaConfig:
     # on the composed/batched-GEMM paths; the MFMA flash kernel is
     # full-causal (window support is round-2 kernel work).
     sliding_window: Optional[int] = None
+    # Qwen-2 style QKV bias
+    attention_bias: bool = False
 
     @property
     def head_dim(self):
@@ -88,6 +90,11 @@ CONFIGS = {
                               num_key_value_heads=8, vocab_size=32000,
                               sliding_window=4096,
                               max_position_embeddings=8192),
+    "qwen2-7b": LlamaConfig(hidden_size=3584, intermediate_size=18944,
+                            num_hidden_layers=28, num_attention_heads=28,
+                            num_key_value_heads=4, vocab_size=152064,
+                            rope_theta=1e6, attention_bias=True,
+                            max_position_embeddings=8192),
     "llama3-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672,
                               num_hidden_layers=80, num_attention_heads=64,
                               num_key_value_heads=8, vocab_size=128256,
@@ -154,7 +161,8 @@ class LlamaAttention(nn.Module):
             config.hidden_size,
             [config.num_attention_heads * self.head_dim,
              config.num_key_value_heads * self.head_dim],
-            bias=False, gather_output=False,
+            bias=getattr(config, "attention_bias", False),
+            gather_output=False,
             num_attention_heads=config.num_attention_heads,
             num_key_value_heads=config.num_key_value_heads,
             head_dim=self.head_dim, kv_size_multiplier=kv_mult,

@@ -32,7 +32,9 @@ class CheckpointConverterBase:
                                 "c_fc.weight",
                                 # native layer names (this package)
                                 "qkv_proj.weight_q", "qkv_proj.weight_k",
-                                "qkv_proj.weight_v", "embed_in.weight",
+                                "qkv_proj.weight_v", "qkv_proj.bias_q",
+                                "qkv_proj.bias_k", "qkv_proj.bias_v",
+                                "embed_in.weight",
                                 "embed_out.weight", "dense_h_to_4h.weight"]
     ROW_PARALLEL_SUFFIXES = ["o_proj.weight", "down_proj.weight",
                              "c_proj.weight", "dense_4h_to_h.weight",
@@ -178,6 +180,12 @@ def convert_hf_llama_state_dict(hf_sd):
             out[k.replace("k_proj.weight", "qkv_proj.weight_k")] = v
         elif k.endswith("self_attn.v_proj.weight"):
             out[k.replace("v_proj.weight", "qkv_proj.weight_v")] = v
+        elif k.endswith("self_attn.q_proj.bias"):  # qwen2-style
+            out[k.replace("q_proj.bias", "qkv_proj.bias_q")] = v
+        elif k.endswith("self_attn.k_proj.bias"):
+            out[k.replace("k_proj.bias", "qkv_proj.bias_k")] = v
+        elif k.endswith("self_attn.v_proj.bias"):
+            out[k.replace("v_proj.bias", "qkv_proj.bias_v")] = v
         elif k.endswith("mlp.gate_proj.weight"):
             gates[k.rsplit("gate_proj.weight", 1)[0]] = v
         elif k.endswith("mlp.up_proj.weight"):

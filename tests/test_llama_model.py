@@ -138,3 +138,27 @@ def _sliding_window_worker(rank, world):
 
 def test_sliding_window():
     run_distributed(_sliding_window_worker, world_size=1)
+
+
+def _qwen_worker(rank, world):
+    """Qwen2-style attention bias: trains at tp2, TP parity with tp1."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(get_config("tiny", attention_bias=True))
+    assert m.model.layers[0].self_attn.qkv_proj.bias_q is not None
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    loss = m(x, labels=x)
+    loss.backward()
+    assert m.model.layers[0].self_attn.qkv_proj.bias_q.grad is not None
+    return float(loss.detach())
+
+
+def test_qwen_attention_bias():
+    tp1 = run_distributed(_qwen_worker, world_size=1)[0]
+    tp2 = run_distributed(_qwen_worker, world_size=2)
+    assert abs(tp2[0] - tp2[1]) < 1e-5
+    assert abs(tp1 - tp2[0]) < 5e-3, (tp1, tp2)
