@@ -162,3 +162,41 @@ def test_qwen_attention_bias():
     tp2 = run_distributed(_qwen_worker, world_size=2)
     assert abs(tp2[0] - tp2[1]) < 1e-5
     assert abs(tp1 - tp2[0]) < 5e-3, (tp1, tp2)
+
+
+def _chunked_loss_worker(rank, world):
+    """chunked_lm_loss == plain lm_head+CE loss AND produces identical
+    grads (recompute correctness) at tp2."""
+    import copy
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.utils.chunked_loss import chunked_lm_loss
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(get_config("tiny"))
+    ref = copy.deepcopy(m)
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 17))
+    labels = x.clone()
+    labels[0, 5:8] = -100
+
+    hidden = m.model(x)
+    loss = chunked_lm_loss(hidden, m.lm_head, labels, num_chunks=3)
+    loss.backward()
+
+    rl = ref(x, labels=labels)
+    rl.backward()
+    assert abs(float(loss) - float(rl)) < 1e-5, (loss, rl)
+    for p1, p2 in zip(m.parameters(), ref.parameters()):
+        if p2.grad is None:
+            continue
+        assert torch.allclose(p1.grad, p2.grad, atol=1e-5), \
+            (p1.grad - p2.grad).abs().max()
+    return float(loss.detach())
+
+
+def test_chunked_lm_loss():
+    out = run_distributed(_chunked_loss_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-6
