@@ -64,6 +64,9 @@ class LlamaConfig:
     sliding_window: Optional[int] = None
     # Qwen-2 style QKV bias
     attention_bias: bool = False
+    # >0: compute lm_head+CE in this many sequence chunks under
+    # checkpointing (caps logits memory; utils/chunked_loss.py)
+    loss_chunks: int = 0
 
     @property
     def head_dim(self):
@@ -364,6 +367,11 @@ class LlamaForCausalLM(nn.Module):
 
     def forward(self, input_ids, labels=None, pos_offset=0, kv_caches=None):
         hidden = self.model(input_ids, pos_offset, kv_caches)
+        if labels is not None and self.config.loss_chunks > 0:
+            from ..utils.chunked_loss import chunked_lm_loss
+
+            return chunked_lm_loss(hidden, self.lm_head, labels,
+                                   num_chunks=self.config.loss_chunks)
         logits = self.lm_head(hidden)  # (B,S,V/tp)
         if labels is None:
             return logits

@@ -200,3 +200,22 @@ def _chunked_loss_worker(rank, world):
 def test_chunked_lm_loss():
     out = run_distributed(_chunked_loss_worker, world_size=2)
     assert abs(out[0] - out[1]) < 1e-6
+
+
+def _loss_chunks_cfg_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    m1 = LlamaForCausalLM(get_config("tiny"))
+    torch.manual_seed(0)
+    m2 = LlamaForCausalLM(get_config("tiny", loss_chunks=4))
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+    assert abs(float(m1(x, labels=x)) - float(m2(x, labels=x))) < 1e-5
+    return 0.0
+
+
+def test_loss_chunks_config():
+    run_distributed(_loss_chunks_cfg_worker, world_size=1)
