@@ -502,3 +502,47 @@ def _mixtral_ckpt_worker(rank, world):
 
 def test_mixtral_activation_checkpointing():
     run_distributed(_mixtral_ckpt_worker, world_size=1)
+
+
+def _ep2_parity_worker(rank, world):
+    """EP all-experts numerical parity: with controlled per-global-expert
+    weights, the EP path (all-gather tokens -> local experts ->
+    reduce-scatter) equals the dense sum over ALL experts for this rank's
+    tokens."""
+    from neuronx_distributed_amd.moe import ExpertMLPs
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 expert_model_parallel_size=world)
+    E, H, I, k = 4, 8, 16, 2
+    mlps = ExpertMLPs(E, H, I, k, capacity_factor=None,
+                      dtype=torch.float32)
+    # deterministic weights by GLOBAL expert id
+    e0 = rank * (E // world)
+    with torch.no_grad():
+        for j in range(E // world):
+            g = e0 + j
+            mlps.gate_up_proj.weight.data[j] = 0.01 * (g + 1) * torch.ones(
+                H, 2 * I)
+            mlps.down_proj.weight.data[j] = 0.01 * (g + 1) * torch.ones(I, H)
+
+    torch.manual_seed(10 + rank)
+    x = torch.randn(6, H)
+    aff = torch.softmax(torch.randn(6, E), dim=-1)
+    idx = aff.topk(k, dim=-1).indices
+    out = mlps(x, aff, idx)
+
+    # dense reference over all 4 experts for THIS rank's tokens
+    ref = torch.zeros_like(x)
+    for g in range(E):
+        w_gu = 0.01 * (g + 1) * torch.ones(H, 2 * I)
+        w_d = 0.01 * (g + 1) * torch.ones(I, H)
+        gu = x @ w_gu
+        act = torch.nn.functional.silu(gu[:, :I]) * gu[:, I:]
+        ref += (act @ w_d) * aff[:, g:g + 1]
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+    return float(out.sum())
+
+
+def test_moe_ep2_numerical_parity():
+    run_distributed(_ep2_parity_worker, world_size=2)
