@@ -26,7 +26,7 @@ _DEFAULT_BUCKET_CAP_MB = 512
 
 
 def _bucket_cap_bytes() -> int:
-    return int(os.environ.get("ALLREDUCE_BUCKET_CAP_MB", _DEFAULT_BUCKET_CAP_MB)) * 1024 * 1024
+    return float(os.environ.get("ALLREDUCE_BUCKET_CAP_MB", _DEFAULT_BUCKET_CAP_MB)) * 1024 * 1024
 
 
 def _flat_allreduce(grads: List[torch.Tensor], group, scale: float = 1.0):

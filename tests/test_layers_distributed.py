@@ -481,3 +481,38 @@ def _lowlevel_ckpt_worker(rank, world):
 def test_lowlevel_checkpointing():
     out = run_distributed(_lowlevel_ckpt_worker, world_size=2)
     assert out[0] != out[1]  # different shards per tp rank
+
+
+def _bucket_ar_worker(rank, world):
+    """bucket_allreduce_gradients: per-dtype buckets, cap-splitting, mean
+    over DP; grads match manual all-reduce."""
+    import os
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.grads import (
+        bucket_allreduce_gradients)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)  # dp = world
+    torch.manual_seed(5 + rank)
+    params = [torch.nn.Parameter(torch.randn(64, 64)) for _ in range(4)]
+    params.append(torch.nn.Parameter(torch.randn(16).double()))
+    for p in params:
+        p.grad = torch.randn_like(p)
+    manual = [p.grad.clone() for p in params]
+
+    os.environ["ALLREDUCE_BUCKET_CAP_MB"] = "0.01"  # force multiple buckets
+    try:
+        bucket_allreduce_gradients([p.grad for p in params])
+    finally:
+        os.environ.pop("ALLREDUCE_BUCKET_CAP_MB")
+
+    import torch.distributed as dist
+    for p, g in zip(params, manual):
+        dist.all_reduce(g)
+        g /= world
+        assert torch.allclose(p.grad, g, atol=1e-6), (p.grad - g).abs().max()
+    return 0.0
+
+
+def test_bucketed_dp_allreduce():
+    run_distributed(_bucket_ar_worker, world_size=2)
