@@ -516,3 +516,34 @@ def _bucket_ar_worker(rank, world):
 
 def test_bucketed_dp_allreduce():
     run_distributed(_bucket_ar_worker, world_size=2)
+
+
+def _gradnorm_worker(rank, world):
+    """clip_grad_norm at tp2: sharded params contribute their shards,
+    REPLICATED params count once (not tp times); the norm equals the dense
+    single-rank value."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.grads import clip_grad_norm
+    from neuronx_distributed_amd.parallel.utils import (
+        set_tensor_model_parallel_attributes)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    full = torch.randn(8, 4)     # sharded rows over tp
+    rep = torch.randn(5)         # replicated (e.g. a norm weight)
+
+    p_shard = torch.nn.Parameter(full.chunk(world, 0)[rank].clone())
+    set_tensor_model_parallel_attributes(p_shard, True, 0)
+    p_rep = torch.nn.Parameter(rep.clone())
+    p_shard.grad = p_shard.data.clone()
+    p_rep.grad = p_rep.data.clone()
+
+    total = clip_grad_norm([p_shard, p_rep], max_norm=1e9)
+    ref = torch.sqrt(full.pow(2).sum() + rep.pow(2).sum())
+    assert abs(float(total) - float(ref)) < 1e-4, (total, ref)
+    return float(total)
+
+
+def test_grad_norm_tp_duplicates():
+    out = run_distributed(_gradnorm_worker, world_size=2)
+    assert abs(out[0] - out[1]) < 1e-6
