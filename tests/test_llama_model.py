@@ -219,3 +219,40 @@ def _loss_chunks_cfg_worker(rank, world):
 
 def test_loss_chunks_config():
     run_distributed(_loss_chunks_cfg_worker, world_size=1)
+
+
+def _dp2_worker(rank, world):
+    """Pure DP (tp=1, world 2): bucketed grad averaging over different
+    per-rank batches equals the single-process run on the combined batch
+    (loss-mean scaling matched)."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.grads import (
+        bucket_allreduce_gradients)
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)  # dp=2
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(get_config("tiny"))
+    torch.manual_seed(0)
+    golden = LlamaForCausalLM(get_config("tiny"))
+
+    torch.manual_seed(42)
+    xfull = torch.randint(0, 256, (4, 16))
+    x = xfull.chunk(world, 0)[rank]
+    loss = m(x, labels=x)
+    loss.backward()
+    bucket_allreduce_gradients([p.grad for p in m.parameters()
+                                if p.grad is not None])
+
+    gl = golden(xfull, labels=xfull)
+    gl.backward()
+    for p, g in zip(m.parameters(), golden.parameters()):
+        if g.grad is None:
+            continue
+        assert torch.allclose(p.grad, g.grad, atol=2e-4), \
+            (p.grad - g.grad).abs().max()
+    return float(loss.detach())
+
+
+def test_dp2_grad_parity():
+    run_distributed(_dp2_worker, world_size=2)
