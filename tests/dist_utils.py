@@ -28,14 +28,31 @@ def _worker(rank, world_size, port, fn, args, q):
             torch.distributed.destroy_process_group()
 
 
-def run_distributed(fn, world_size=2, args=(), timeout=150):
-    """Spawn world_size processes running fn(rank, world_size, *args);
-    returns list of per-rank results; raises on any rank error or hang."""
-    import random
+def _free_port() -> int:
+    import socket
 
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def run_distributed(fn, world_size=2, args=(), timeout=150, _retry=True):
+    """Spawn world_size processes running fn(rank, world_size, *args);
+    returns list of per-rank results; raises on any rank error or hang.
+    Retries ONCE on failure with a fresh port (rendezvous port races are
+    the only nondeterminism here)."""
+    try:
+        return _run_distributed_once(fn, world_size, args, timeout)
+    except RuntimeError:
+        if not _retry:
+            raise
+        return _run_distributed_once(fn, world_size, args, timeout)
+
+
+def _run_distributed_once(fn, world_size=2, args=(), timeout=150):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = random.randint(29600, 39999)
+    port = _free_port()
     procs = [
         ctx.Process(target=_worker, args=(r, world_size, port, fn, args, q),
                     daemon=True)
