@@ -11,6 +11,14 @@ import torch
 from dist_utils import run_distributed
 
 
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _trainer_worker(rank, world, tmpdir):
     import neuronx_distributed_amd as nxd
     from neuronx_distributed_amd.trainer import (
@@ -260,7 +268,7 @@ def test_bench_multirank_cpu_smoke():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29771", os.path.join(repo, "bench.py"),
+           "--master-port", str(_free_port()), os.path.join(repo, "bench.py"),
            "--gpus", "2", "--steps", "1", "--warmup", "0", "--model", "tiny",
            "--seq", "64", "--batch", "2", "--microbatch", "1"]
     out = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
@@ -283,7 +291,7 @@ def test_bench_pp_cpu_smoke():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29772", os.path.join(repo, "bench.py"),
+           "--master-port", str(_free_port()), os.path.join(repo, "bench.py"),
            "--gpus", "2", "--steps", "1", "--warmup", "0", "--model", "tiny",
            "--seq", "64", "--batch", "4", "--microbatch", "1",
            "--tp", "1", "--pp", "2"]
@@ -306,7 +314,7 @@ def test_bench_tp_pp_3d_cpu_smoke():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
-           "--master-port", "29773", os.path.join(repo, "bench.py"),
+           "--master-port", str(_free_port()), os.path.join(repo, "bench.py"),
            "--gpus", "4", "--steps", "1", "--warmup", "0", "--model", "tiny",
            "--seq", "64", "--batch", "4", "--microbatch", "1",
            "--tp", "2", "--pp", "2"]
@@ -330,7 +338,7 @@ def test_bench_tp4_cpu_smoke():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
-           "--master-port", "29774", os.path.join(repo, "bench.py"),
+           "--master-port", str(_free_port()), os.path.join(repo, "bench.py"),
            "--gpus", "4", "--steps", "1", "--warmup", "0", "--model", "tiny",
            "--seq", "64", "--batch", "4", "--microbatch", "2"]
     out = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
