@@ -213,3 +213,29 @@ def _builder_worker(rank, world):
 
 def test_model_builder_e2e():
     run_distributed(_builder_worker, world_size=1)
+
+
+def _sampler_topp_worker(rank, world):
+    """top-p sampling: only tokens inside the nucleus can be drawn."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.utils.sampling import Sampler
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    logits = torch.full((4, 16), -10.0)
+    logits[:, 3] = 5.0   # p ~ 0.88
+    logits[:, 7] = 3.0   # p ~ 0.12 -> nucleus at top_p=0.5 is {3}
+    s = Sampler(do_sample=True, top_k=16, top_p=0.5, temperature=1.0)
+    for _ in range(10):
+        tok = s(logits)
+        assert (tok == 3).all(), tok
+    s2 = Sampler(do_sample=True, top_k=16, top_p=1.0, temperature=1.0)
+    seen = set()
+    for _ in range(50):
+        seen.update(s2(logits).tolist())
+    assert 3 in seen and seen <= {3, 7}, seen
+    return 0.0
+
+
+def test_sampler_top_p():
+    run_distributed(_sampler_topp_worker, world_size=1)
