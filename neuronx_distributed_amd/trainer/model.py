@@ -33,6 +33,20 @@ class NxDModel(nn.Module):
         with torch.no_grad():
             return self.module(*args, **kwargs)
 
+    @property
+    def supports_tensor_position(self):
+        return getattr(self.module, "supports_tensor_position", False) \
+            and not self.pp_enabled
+
+    @property
+    def config(self):
+        """Delegate to the wrapped model (generation/speculation utils
+        read model.config)."""
+        inner = self.module
+        if self.pp_enabled:
+            raise AttributeError("config is stage-local under PP")
+        return inner.config
+
     def local_module(self):
         return self.module.local_module() if self.pp_enabled else self.module
 

@@ -722,3 +722,22 @@ def test_checkpoint_storage_backend(tmp_path):
     assert not st.exists("tag/done")
     with _pytest.raises(NotImplementedError):
         get_storage("s3://bucket/prefix")
+
+
+def _nxdmodel_generate_worker(rank, world):
+    """generate() works directly on the trainer's NxDModel wrapper."""
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    cfg = nxd.neuronx_distributed_config(tensor_parallel_size=1)
+    model = nxd.initialize_parallel_model(
+        cfg, lambda: LlamaForCausalLM(get_config("tiny")))
+    model.eval()
+    x = torch.randint(0, 256, (1, 8))
+    out = nxd.generate(model, x, max_new_tokens=4)
+    assert out.shape == (1, 12)
+    return 0.0
+
+
+def test_nxdmodel_generate():
+    run_distributed(_nxdmodel_generate_worker, world_size=1)
