@@ -97,3 +97,31 @@ def _adapter_roundtrip_worker(rank, world):
 
 def test_lora_adapter_roundtrip():
     run_distributed(_adapter_roundtrip_worker, world_size=1)
+
+
+def _trainer_lora_worker(rank, world):
+    """lora_config flows through the trainer facade (phase 4): only
+    adapter params are trainable and a step runs."""
+    import neuronx_distributed_amd as nxd
+    from neuronx_distributed_amd.lora import LoraConfig
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+
+    cfg = nxd.neuronx_distributed_config(
+        tensor_parallel_size=1,
+        lora_config=LoraConfig(lora_rank=4, lora_alpha=8,
+                               target_modules=["o_proj"]))
+    model = nxd.initialize_parallel_model(
+        cfg, lambda: LlamaForCausalLM(get_config("tiny")))
+    trainable = [n for n, p in model.named_parameters() if p.requires_grad]
+    assert trainable and all("lora_" in n for n in trainable), trainable[:4]
+    opt = torch.optim.AdamW((p for p in model.parameters()
+                             if p.requires_grad), lr=1e-2)
+    x = torch.randint(0, 256, (2, 16))
+    loss = model.run_train(x, labels=x)
+    opt.step()
+    assert torch.isfinite(loss)
+    return float(loss.detach())
+
+
+def test_trainer_lora_integration():
+    run_distributed(_trainer_lora_worker, world_size=1)
