@@ -38,7 +38,11 @@ def neuronx_distributed_config(
     lnc_size: int = 1,
 ) -> Dict[str, Any]:
     """Build + validate the config dict (reference trainer.py:32-144) and
-    initialize model parallelism if not already done."""
+    initialize model parallelism if not already done.
+
+    ``lnc_size`` is accepted for API compatibility only: logical-neuron-
+    core grouping is trn2 hardware topology with no MI355X analogue (one
+    process drives one GPU)."""
     optimizer_config = dict(optimizer_config or {})
     optimizer_config.setdefault("zero_one_enabled", True)
     optimizer_config.setdefault("grad_clipping", True)
