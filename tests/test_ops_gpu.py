@@ -390,8 +390,4 @@ def test_fused_ce_ignore_index():
     g = logits.grad.float()
     assert g[2].abs().max().item() == 0.0
     assert g[4].abs().max().item() == 0.0
-    # valid rows match the fp32 torch CE gradient
-    ref = torch.nn.functional.cross_entropy(
-        logits.detach().float().requires_grad_(True), tgt,
-        ignore_index=-100)
     assert torch.isfinite(loss).all()
