@@ -133,7 +133,7 @@ def _to_cpu(obj):
 
 def save_checkpoint(path: str, tag, model=None, optimizer=None,
                     scheduler=None, user_content: Optional[Dict] = None,
-                    num_workers: int = 8, use_xser: bool = False,
+                    num_workers: int = 8, use_xser: bool = False,  # noqa: ARG001 — xser is an XLA host-RAM workaround; MI355X snapshots to CPU and torch.saves (API compat)
                     num_kept: Optional[int] = None, async_save: bool = False,
                     zero1_optimizer: Optional[bool] = None) -> None:
     """reference trainer/checkpoint.py:654-824."""
