@@ -28,7 +28,8 @@ def _needs_build():
     if not os.path.exists(LIB):
         return True
     lib_mtime = os.path.getmtime(LIB)
-    deps = _sources() + [os.path.join(CSRC, "common.h")]
+    deps = _sources() + [os.path.join(CSRC, h)
+                         for h in ("common.h", "mfma.h")]
     return any(os.path.getmtime(s) > lib_mtime for s in deps)
 
 
