@@ -659,6 +659,8 @@ def test_serving_endpoint():
     spec.loader.exec_module(mod)
     from starlette.testclient import TestClient
 
+    saved_env = {k: os.environ.get(k)
+                 for k in ("NXDA_FAST_INIT", "MASTER_ADDR", "MASTER_PORT")}
     try:
         app = mod.build_app("tiny")
         client = TestClient(app)
@@ -677,6 +679,11 @@ def test_serving_endpoint():
             ps.destroy_model_parallel()
         if dist.is_initialized():
             dist.destroy_process_group()
+        for k, v in saved_env.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
 
 
 def _pph_worker(rank, world):
