@@ -85,14 +85,15 @@ class ExpertMLPs(nn.Module):
         With EP, tokens are all-gathered over the EP group, local experts
         process the union, and the result is summed back via
         reduce-scatter (reference forward_all_experts_EP,
-        expert_mlps_v2.py:394)."""
-        from ..parallel import comm
+        expert_mlps_v2.py:394).  The collectives are autograd-aware so EP
+        training gradients flow back to the tokens and the router."""
+        from ..parallel.mappings import (gather_from_group,
+                                         reduce_scatter_to_group)
 
         if self.ep_size > 1:
-            ep = ps.get_group_info("ep")
-            hidden = comm.all_gather(hidden, dim=0, group=ep)
-            expert_affinities = comm.all_gather(expert_affinities, dim=0,
-                                                group=ep)
+            hidden = gather_from_group(hidden, dim=0, group_name="ep")
+            expert_affinities = gather_from_group(expert_affinities, dim=0,
+                                                  group_name="ep")
         T = hidden.shape[0]
         x = hidden.unsqueeze(0).expand(self.num_experts_local, T,
                                        self.hidden_size)
@@ -101,7 +102,7 @@ class ExpertMLPs(nn.Module):
         aff = expert_affinities[:, e0:e0 + self.num_experts_local]  # (T,El)
         out = torch.einsum("eth,te->th", out.float(), aff.float()).to(hidden.dtype)
         if self.ep_size > 1:
-            out = comm.reduce_scatter(out, dim=0, group=ps.get_group_info("ep"))
+            out = reduce_scatter_to_group(out, dim=0, group_name="ep")
         return out
 
     def forward_capacity_factor(self, hidden, expert_affinities, expert_index):

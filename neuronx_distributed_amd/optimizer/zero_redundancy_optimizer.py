@@ -229,17 +229,32 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
 
     # -- step -------------------------------------------------------------
 
+    def _bucket_grad_scale(self, b) -> float:
+        """Mean scale for one bucket's grads.  Dense buckets: 1/shard-group
+        size.  Expert buckets (sharded over EDP): the grads of an expert
+        already sum token contributions from all EP ranks (all-to-all
+        backward), so they need an extra 1/ep on top of the 1/edp reduce
+        scale to match the dense 1/(data-parallel) mean (reference
+        zero_redundancy_optimizer.py:241-281 ep scale_factor)."""
+        ep_size = ps._GROUPS["ep"].size if "ep" in ps._GROUPS else 1
+        if ep_size > 1 and b.group_info.name == "edp":
+            return 1.0 / (b.group_info.size * ep_size)
+        return 1.0 / b.group_info.size
+
     @torch.no_grad()
     def _reduce_scatter_grads(self):
         works = []
         for b in self.buckets:
             world = b.group_info.size
+            scale = self._bucket_grad_scale(b)
             if world == 1:
-                # single-rank shard group: the shard IS the flat grad —
-                # no scale, no copy
+                # single-rank shard group: the shard IS the flat grad
+                # (the 1/ep expert scale still applies when EP > 1)
+                if scale != 1.0:
+                    b.flat_grad.mul_(scale)
                 b.grad_shard = b.flat_grad
                 continue
-            b.flat_grad.mul_(1.0 / world)  # grad mean over the shard group
+            b.flat_grad.mul_(scale)  # grad mean over the shard group
             if comm._backend_is_gloo(b.group_info.group):
                 dist.all_reduce(b.flat_grad, group=b.group_info.group)
                 b.grad_shard.copy_(b.flat_grad[b.shard_lo:b.shard_hi])

@@ -67,7 +67,31 @@ def bucket_allreduce_gradients(grads: Iterable[torch.Tensor], group=None,
 
 
 def allreduce_gradients_for_parameters(parameters, group=None, scale=None):
-    grads = [p.grad for p in parameters if p.grad is not None]
+    """DP gradient sync (non-ZeRO path), expert-aware (reference
+    trainer/optimizer.py:132-141 + grads.py:284-298): dense grads all-reduce
+    over the DP group; expert-parallel grads are distinct per EP rank, so
+    they all-reduce over the expert-data-parallel (EDP) group only, with an
+    extra 1/ep scale so dense and expert grads see the same effective
+    1/(data-parallel-degree) mean."""
+    parameters = [p for p in parameters if p.grad is not None]
+    ep_size = (ps._GROUPS["ep"].size
+               if "ep" in ps._GROUPS and group is None else 1)
+    if ep_size > 1:
+        expert = [p.grad for p in parameters if param_is_expert_parallel(p)]
+        dense = [p.grad for p in parameters
+                 if not param_is_expert_parallel(p)]
+        if expert:
+            edp = ps.get_group_info("edp")
+            if edp.size == 1:
+                # no replicas to reduce, but the 1/ep mean scale still applies
+                for g in expert:
+                    g.mul_(1.0 / ep_size)
+            else:
+                bucket_allreduce_gradients(
+                    expert, group=edp, scale=1.0 / (edp.size * ep_size))
+        bucket_allreduce_gradients(dense, group=group, scale=scale)
+        return
+    grads = [p.grad for p in parameters]
     bucket_allreduce_gradients(grads, group=group, scale=scale)
 
 
@@ -106,56 +130,61 @@ def get_grad_norm(parameters, norm_type: float = 2.0,
         parameters = [parameters]
     params_with_grad = [p for p in parameters if p.grad is not None]
     tp_rank = ps.get_tensor_model_parallel_rank()
-    ep_rank = (ps.get_expert_model_parallel_rank()
-               if "ep" in ps._GROUPS else 0)
+    ep_size = ps._GROUPS["ep"].size if "ep" in ps._GROUPS else 1
 
     device = None
-    local_sq = None
-    use_inf = norm_type == float("inf")
-    local_max = None
     for p in params_with_grad:
-        if device is None:
-            device = p.grad.device
-            local_sq = torch.zeros(1, dtype=torch.float32, device=device)
-            local_max = torch.zeros(1, dtype=torch.float32, device=device)
-        is_tp = param_is_tensor_parallel(p)
-        is_ep = param_is_expert_parallel(p)
-        if not is_tp and tp_rank != 0:
-            continue  # TP-duplicated: count once
-        if is_ep and not is_tp and tp_rank != 0:
-            continue
-        g = p.grad.detach()
-        if use_inf:
-            local_max = torch.maximum(local_max, g.abs().max().reshape(1).float())
-        else:
-            local_sq += g.float().pow(norm_type).sum()
-
+        device = p.grad.device
+        break
     if device is None:
         device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
-        local_sq = torch.zeros(1, dtype=torch.float32, device=device)
-        local_max = torch.zeros(1, dtype=torch.float32, device=device)
+    # Expert params are DISTINCT per EP rank (sum their norm over EP);
+    # dense params are REPLICATED across EP (count once per EP group) —
+    # so the two accumulate separately (reference ep_total_norm,
+    # grads.py:41-189).
+    dense_acc = torch.zeros(1, dtype=torch.float32, device=device)
+    expert_acc = torch.zeros(1, dtype=torch.float32, device=device)
+    use_inf = norm_type == float("inf")
+    for p in params_with_grad:
+        is_tp = param_is_tensor_parallel(p)
+        if not is_tp and tp_rank != 0:
+            continue  # TP-duplicated: count once
+        acc = expert_acc if param_is_expert_parallel(p) else dense_acc
+        g = p.grad.detach()
+        if use_inf:
+            torch.maximum(acc, g.abs().max().reshape(1).float(), out=acc)
+        else:
+            acc += g.detach().abs().float().pow(norm_type).sum()
 
     # Reduce over every model-parallel dim; ZeRO-1 also reduces over the DP
-    # sharding group since each rank only holds a shard's grads.
-    reduce_groups = []
+    # sharding group since each rank only holds a shard's grads.  Expert
+    # norms additionally reduce over EP (before joining the dense part).
+    shared_groups = []
     for name in ("tp", "pp"):
         if name in ps._GROUPS and ps._GROUPS[name].size > 1:
-            reduce_groups.append(ps._GROUPS[name])
-    if "ep" in ps._GROUPS and ps._GROUPS["ep"].size > 1:
-        reduce_groups.append(ps._GROUPS["ep"])
+            shared_groups.append(ps._GROUPS[name])
     if zero1_optimizer:
         if zero1_groups is not None:
-            reduce_groups.append(zero1_groups)
+            shared_groups.append(zero1_groups)
         elif "dp" in ps._GROUPS and ps._GROUPS["dp"].size > 1:
-            reduce_groups.append(ps._GROUPS["dp"])
+            shared_groups.append(ps._GROUPS["dp"])
 
+    op = dist.ReduceOp.MAX if use_inf else dist.ReduceOp.SUM
+    # Under ZeRO-1 every rank's (dense + expert) shard pieces are mutually
+    # distinct across the whole dp*cp sharding group, so the sharding-group
+    # reduce already covers EP; reducing over EP again would count each
+    # expert piece ep times.
+    if ep_size > 1 and not zero1_optimizer:
+        comm.all_reduce(expert_acc, op=op, group=ps._GROUPS["ep"])
     if use_inf:
-        for g in reduce_groups:
-            comm.all_reduce(local_max, op=dist.ReduceOp.MAX, group=g)
-        return local_max.squeeze()
-    for g in reduce_groups:
-        comm.all_reduce(local_sq, group=g)
-    return local_sq.squeeze().pow(1.0 / norm_type)
+        local = torch.maximum(dense_acc, expert_acc)
+        for g in shared_groups:
+            comm.all_reduce(local, op=dist.ReduceOp.MAX, group=g)
+        return local.squeeze()
+    local = dense_acc + expert_acc
+    for g in shared_groups:
+        comm.all_reduce(local, group=g)
+    return local.squeeze().pow(1.0 / norm_type)
 
 
 def clip_grad_norm(parameters, max_norm: float, norm_type: float = 2.0,

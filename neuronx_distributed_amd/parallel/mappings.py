@@ -184,6 +184,50 @@ def all_to_all(input_, split_dim, concat_dim, group_name="ep"):
     return _AllToAll.apply(input_, split_dim, concat_dim, group_name)
 
 
+class _GatherFromGroup(torch.autograd.Function):
+    """All-gather along ``dim`` over an arbitrary group; backward
+    reduce-scatters the gradient (each rank's input chunk receives grad
+    contributions from every rank's use of the gathered tensor)."""
+
+    @staticmethod
+    def forward(ctx, input_, dim, group_name):
+        ctx.dim = dim
+        ctx.group_name = group_name
+        return comm.all_gather(input_, dim=dim,
+                               group=ps.get_group_info(group_name))
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        g = comm.reduce_scatter(grad_output.contiguous(), dim=ctx.dim,
+                                group=ps.get_group_info(ctx.group_name))
+        return g, None, None
+
+
+class _ReduceScatterToGroup(torch.autograd.Function):
+    """Reduce-scatter along ``dim``; backward all-gathers the gradient."""
+
+    @staticmethod
+    def forward(ctx, input_, dim, group_name):
+        ctx.dim = dim
+        ctx.group_name = group_name
+        return comm.reduce_scatter(input_.contiguous(), dim=dim,
+                                   group=ps.get_group_info(group_name))
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        g = comm.all_gather(grad_output.contiguous(), dim=ctx.dim,
+                            group=ps.get_group_info(ctx.group_name))
+        return g, None, None
+
+
+def gather_from_group(input_, dim=0, group_name="ep"):
+    return _GatherFromGroup.apply(input_, dim, group_name)
+
+
+def reduce_scatter_to_group(input_, dim=0, group_name="ep"):
+    return _ReduceScatterToGroup.apply(input_, dim, group_name)
+
+
 def enter_expert_parallel_region(hidden: torch.Tensor, scatter_gather: bool = False):
     """(E, C, H) -> (E/ep, ep*C, H): each EP rank ends up holding the full
     token set for its local experts (reference mappings.py:481-523)."""

@@ -389,8 +389,10 @@ class NxDPPModel(nn.Module):
         return self._process_loss(losses) if self.output_loss_value_spec else None
 
     def _process_loss(self, losses):
-        """Average microbatch losses; broadcast over PP (+ average over DP
-        when configured) — reference pipeline/model.py:1974-2028."""
+        """Average microbatch losses, then average over CP and DP and
+        broadcast over PP — reference pipeline/model.py:1974-2028 (the
+        reference all-reduces the scalar loss over CP and DP before the
+        PP broadcast so every rank reports the same global mean)."""
         device = torch.device("cuda", torch.cuda.current_device()) \
             if torch.cuda.is_available() else torch.device("cpu")
         if self.pp_rank == self.pp_size - 1 and losses:
@@ -398,10 +400,17 @@ class NxDPPModel(nn.Module):
             loss = loss.to(device)
         else:
             loss = torch.zeros((), dtype=torch.float32, device=device)
-        if self.broadcast_and_average_loss and self.pp_size > 1:
-            # all-reduce-as-broadcast over the PP group (only last stage
-            # contributes)
-            pcomm.all_reduce(loss, group=ps.get_group_info("pp"))
+        if self.broadcast_and_average_loss:
+            if self.pp_rank == self.pp_size - 1:
+                for name in ("cp", "dp"):
+                    g = ps.get_group_info(name) if name in ps._GROUPS else None
+                    if g is not None and g.size > 1:
+                        pcomm.all_reduce(loss, group=g)
+                        loss = loss / g.size
+            if self.pp_size > 1:
+                # all-reduce-as-broadcast over the PP group (only last
+                # stage contributes)
+                pcomm.all_reduce(loss, group=ps.get_group_info("pp"))
         return loss
 
     def forward(self, *args, **kwargs):

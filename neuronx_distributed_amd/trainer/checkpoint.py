@@ -51,14 +51,38 @@ def finalize_checkpoints():
     _PENDING = []
 
 
-def _rank_name() -> str:
-    return (f"dp_rank_{ps.get_data_parallel_rank():02d}"
-            f"_tp_rank_{ps.get_tensor_model_parallel_rank():02d}"
-            f"_pp_rank_{ps.get_pipeline_model_parallel_rank():02d}")
+def _ep_size() -> int:
+    return ps._GROUPS["ep"].size if "ep" in ps._GROUPS else 1
+
+
+def _cp_size() -> int:
+    return ps._GROUPS["cp"].size if "cp" in ps._GROUPS else 1
+
+
+def _rank_name(dp: bool = True) -> str:
+    """Shard filename stem (reference _get_path, trainer/checkpoint.py:54-63):
+    ``dp_rank_xx[_cp_rank_xx][_ep_rank_xx]_tp_rank_xx_pp_rank_xx``.
+    The cp/ep segments appear only when that parallelism is on, so the
+    tp/pp-only layout stays byte-identical to round-1 checkpoints.
+    ``dp=False`` pins the dp (and cp) fields to 0 — used for model shards,
+    which are deduplicated over data-parallel replicas."""
+    name = f"dp_rank_{ps.get_data_parallel_rank() if dp else 0:02d}"
+    if _cp_size() > 1:
+        name += f"_cp_rank_{ps.get_context_model_parallel_rank() if dp else 0:02d}"
+    if _ep_size() > 1:
+        name += f"_ep_rank_{ps.get_expert_model_parallel_rank():02d}"
+    name += (f"_tp_rank_{ps.get_tensor_model_parallel_rank():02d}"
+             f"_pp_rank_{ps.get_pipeline_model_parallel_rank():02d}")
+    return name
 
 
 def _is_model_writer() -> bool:
-    # model shards deduped over DP (and CP): only dp/cp rank 0 writes
+    """Model shards are deduped over data-parallel replicas.  Without EP
+    that is dp/cp rank 0.  With EP, each EP rank holds DIFFERENT experts,
+    so one writer per EP rank: the rank with expert-data-parallel rank 0
+    (the EDP dim spans DPxCP)."""
+    if _ep_size() > 1:
+        return ps.get_expert_data_parallel_rank() == 0
     return (ps.get_data_parallel_rank() == 0
             and ps.get_context_model_parallel_rank() == 0)
 
@@ -147,7 +171,7 @@ def save_checkpoint(path: str, tag, model=None, optimizer=None,
         sd = model.state_dict()
         if _is_model_writer():
             jobs.append((sd, os.path.join(ckpt_dir, "model",
-                                          _rank_name() + ".pt")))
+                                          _rank_name(dp=False) + ".pt")))
     if optimizer is not None:
         from ..optimizer import NeuronZero1Optimizer
         from .optimizer import NxDOptimizer
@@ -208,21 +232,16 @@ def load_checkpoint(path: str, tag=None, model=None, optimizer=None,
         raise FileNotFoundError(f"checkpoint {ckpt_dir} incomplete (no done tag)")
 
     if model is not None:
-        fname = os.path.join(ckpt_dir, "model", _rank_name() + ".pt")
-        if not os.path.exists(fname):
-            # model saved only by dp/cp rank 0 replicas: map to the rank-0
-            # replica's file
-            alt = (f"dp_rank_00_tp_rank_{ps.get_tensor_model_parallel_rank():02d}"
-                   f"_pp_rank_{ps.get_pipeline_model_parallel_rank():02d}.pt")
-            fname = os.path.join(ckpt_dir, "model", alt)
+        # model shards are written deduped with dp/cp pinned to 0
+        fname = os.path.join(ckpt_dir, "model", _rank_name(dp=False) + ".pt")
         sd = torch.load(fname, map_location="cpu", weights_only=False)
         model.load_state_dict(sd, strict=strict)
     if optimizer is not None:
         fname = os.path.join(ckpt_dir, "optim", _rank_name() + ".pt")
         if not os.path.exists(fname):
-            alt = (f"dp_rank_00_tp_rank_{ps.get_tensor_model_parallel_rank():02d}"
-                   f"_pp_rank_{ps.get_pipeline_model_parallel_rank():02d}.pt")
-            fname = os.path.join(ckpt_dir, "optim", alt)
+            # non-zero1 optimizers are saved by dp rank 0 only
+            fname = os.path.join(ckpt_dir, "optim",
+                                 _rank_name(dp=False) + ".pt")
         optimizer.load_state_dict(
             torch.load(fname, map_location="cpu", weights_only=False))
     if scheduler is not None:
