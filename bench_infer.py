@@ -36,6 +36,28 @@ def main():
     on_gpu = torch.cuda.is_available()
     if on_gpu:
         torch.cuda.set_device(local_rank)
+        try:
+            tun = os.environ.get(
+                "NXDA_TUNABLEOP_FILE",
+                os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                             "profiles", "tunableop_gfx950.csv"))
+            if os.environ.get("NXDA_TUNE", "0") == "1":
+                # tuning run: search hipBLASLt algos for this run's GEMM
+                # shapes, write the selections (GraphDecoder detects tuning
+                # mode and decodes eagerly so every shape gets tuned)
+                torch.cuda.tunable.set_filename(
+                    os.environ.get("NXDA_TUNE_OUT", "tunableop_out.csv"),
+                    insert_device_ordinal=False)
+                torch.cuda.tunable.tuning_enable(True)
+                torch.cuda.tunable.enable(True)
+            elif os.path.exists(tun) and \
+                    os.environ.get("NXDA_TUNABLEOP", "1") == "1":
+                torch.cuda.tunable.set_filename(tun,
+                                                insert_device_ordinal=False)
+                torch.cuda.tunable.tuning_enable(False)
+                torch.cuda.tunable.enable(True)
+        except Exception:
+            pass
     dist.init_process_group("nccl" if on_gpu else "gloo", rank=rank,
                             world_size=world)
 
@@ -111,6 +133,8 @@ def main():
             "decode_tokens_per_s": round(args.batch * steps / decode_s, 1),
             "dtype": "bf16", "data": "synthetic",
         }), flush=True)
+    if on_gpu and os.environ.get("NXDA_TUNE", "0") == "1":
+        torch.cuda.tunable.write_file()
     dist.destroy_process_group()
 
 

