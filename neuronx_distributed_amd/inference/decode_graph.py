@@ -41,8 +41,11 @@ class GraphDecoder:
         self.logits: Optional[torch.Tensor] = None
 
     def _eager(self):
-        return self.model(self.step_in, kv_caches=self.caches,
-                          pos_offset=self.pos_t)
+        # no_grad so the model's inference-only fused kernels (add_rmsnorm,
+        # fused-QKV single GEMM) are active in the CAPTURED graph too
+        with torch.no_grad():
+            return self.model(self.step_in, kv_caches=self.caches,
+                              pos_offset=self.pos_t)
 
     def capture(self) -> bool:
         """Capture one decode step; returns False (eager fallback) if

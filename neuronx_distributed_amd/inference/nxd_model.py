@@ -60,6 +60,15 @@ class NxDModel(nn.Module):
 
     @torch.no_grad()
     def forward(self, **kwargs):
+        # accept host inputs: route tensors to the module's device (the
+        # reference SPMDModel runtime also owns the host->device copy)
+        try:
+            dev = next(self.module.parameters()).device
+            kwargs = {k: (v.to(dev) if isinstance(v, torch.Tensor)
+                          and v.device != dev else v)
+                      for k, v in kwargs.items()}
+        except StopIteration:
+            pass
         tag = self.route(kwargs)
         if not self.use_hip_graphs or tag is None:
             return self.module(**kwargs)

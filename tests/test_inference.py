@@ -202,12 +202,13 @@ def _builder_worker(rank, world):
     x8 = torch.randint(0, 256, (1, 8))
     x16 = torch.randint(0, 256, (1, 16))
     with torch.no_grad():
-        out8 = nxd_model(input_ids=x8)
-        out16 = nxd_model(input_ids=x16)
+        out8 = nxd_model(input_ids=x8).cpu()
+        out16 = nxd_model(input_ids=x16).cpu()
         r8 = ref(x8)
         r16 = ref(x16)
-    assert torch.allclose(out8, r8, atol=1e-6)
-    assert torch.allclose(out16, r16, atol=1e-6)
+    # builder may place the model on GPU: allow fp32 cpu-vs-gpu drift
+    assert torch.allclose(out8, r8, atol=1e-4)
+    assert torch.allclose(out16, r16, atol=1e-4)
     return 0.0
 
 
