@@ -224,9 +224,8 @@ class _SwiGLUFn(torch.autograd.Function):
     def forward(ctx, x):
         ctx.save_for_backward(x)
         I = x.shape[-1] // 2
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16 and I % 8 == 0:
             lib = _require_lib()
-            assert x.dtype == torch.bfloat16 and I % 8 == 0
             x2 = x.contiguous()
             N = x2.numel() // x2.shape[-1]
             out = torch.empty(x2.shape[:-1] + (I,), dtype=x2.dtype,
@@ -241,7 +240,7 @@ class _SwiGLUFn(torch.autograd.Function):
     def backward(ctx, dy):
         (x,) = ctx.saved_tensors
         I = x.shape[-1] // 2
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16 and I % 8 == 0:
             lib = _require_lib()
             dy2 = dy.contiguous()
             N = x.numel() // x.shape[-1]
