@@ -6,11 +6,20 @@ hand-driven planner because DCP's collectives could not run under XLA.
 On MI355X the gloo/RCCL store works, so this is vanilla DCP over DTensors:
 each flat master/momentum shard is declared as a 1-D DTensor sharded over
 the zero1 group, and DCP's planner handles dedup, file layout and —
-the point of the exercise — RESHARDING on load when the world size
-changed between save and load.
+the point of the exercise — RESHARDING on load when the data-parallel
+world changed between save and load.
+
+tp/pp/ep > 1 (reference zero_dcp_utils.py:84-516 supports the full mesh):
+each (tp, pp, ep) slice holds DIFFERENT parameters, so its buckets are
+saved under a mesh-coordinate key prefix (``mp_tp00_pp00/``...) and the
+DTensor mesh is built per-rank FROM the existing zero1 process group
+(``DeviceMesh.from_group``) — ranks of different slices contribute
+disjoint keys to one global DCP plan.  Resharding on load is supported
+over the data-parallel (zero1) dim; changing tp/pp/ep between save and
+load requires the offline converter instead.
 
 The non-DCP path (per-rank files + offline merge in
-``scripts/checkpoint_converter.py``) remains the default engine;
+``scripts/convert_zero_checkpoints.py``) remains the default engine;
 this module is the DCP alternative.
 """
 
@@ -23,36 +32,60 @@ import torch.distributed as dist
 from ..parallel import parallel_state as ps
 
 
-def _zero1_mesh():
-    from torch.distributed.device_mesh import init_device_mesh
+def _zero1_mesh(optimizer):
+    """1-D device mesh over this rank's zero1 sharding group (dense
+    buckets), or None for a single-rank group (plain tensors then — the
+    tp/pp key prefix already makes them globally unique).  Uses the
+    already-created process group — no collective mesh construction."""
+    from torch.distributed.device_mesh import DeviceMesh
 
-    mesh_rows = ps.get_zero1_sharding_group(as_list=True)
-    world = dist.get_world_size()
-    if len(mesh_rows) != 1 or len(mesh_rows[0]) != world:
-        raise NotImplementedError(
-            "zero1 DCP save requires the zero1 sharding group to span the "
-            "world (tp=pp=1); use the per-rank checkpoint engine + offline "
-            "merge otherwise")
+    ginfo = optimizer._sharding_group
     device_type = "cuda" if torch.cuda.is_available() else "cpu"
-    return init_device_mesh(device_type, (world,))
+    if ginfo.group is None or ginfo.size == 1:
+        return None
+    return DeviceMesh.from_group(ginfo.group, device_type)
+
+
+def _mesh_prefix() -> str:
+    """Key prefix identifying this rank's model-parallel slice (tp/pp and,
+    when EP is on, the ep coordinate)."""
+    tp = ps.get_tensor_model_parallel_rank()
+    pp = ps.get_pipeline_model_parallel_rank()
+    pre = f"mp_tp{tp:02d}_pp{pp:02d}"
+    if "ep" in ps._GROUPS and ps._GROUPS["ep"].size > 1:
+        pre += f"_ep{ps.get_expert_model_parallel_rank():02d}"
+    return pre
+
+
+def _group_mesh(ginfo, fallback_mesh):  # noqa: ARG001 — kept for call symmetry
+    from torch.distributed.device_mesh import DeviceMesh
+
+    if ginfo is None or ginfo.group is None or ginfo.size == 1:
+        return None  # single-rank shard -> plain tensor
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    return DeviceMesh.from_group(ginfo.group, device_type)
 
 
 def _dtensor_state(optimizer) -> Dict[str, torch.Tensor]:
-    """The zero1 shards as DTensors + replicated scalars."""
+    """The zero1 shards as DTensors + replicated scalars.  Expert buckets
+    (sharded over EDP) get the EDP mesh; dense buckets the zero1 mesh."""
     from torch.distributed.tensor import DTensor, Shard
 
-    mesh = _zero1_mesh()
+    mesh = _zero1_mesh(optimizer)
+    pre = _mesh_prefix()
     state: Dict[str, torch.Tensor] = {}
+
+    def as_shard(t, bmesh):
+        return t if bmesh is None else DTensor.from_local(t, bmesh, [Shard(0)])
+
     for i, b in enumerate(optimizer.buckets):
-        local = b.master.detach()
-        state[f"bucket_{i}.master"] = DTensor.from_local(
-            local, mesh, [Shard(0)])
+        bmesh = mesh if b.group_info is optimizer._sharding_group else \
+            _group_mesh(b.group_info, mesh)
+        state[f"{pre}/bucket_{i}.master"] = as_shard(b.master.detach(), bmesh)
         if hasattr(b, "fused_m"):
-            state[f"bucket_{i}.m"] = DTensor.from_local(
-                b.fused_m.detach(), mesh, [Shard(0)])
-            state[f"bucket_{i}.v"] = DTensor.from_local(
-                b.fused_v.detach(), mesh, [Shard(0)])
-    state["step_count"] = torch.tensor(
+            state[f"{pre}/bucket_{i}.m"] = as_shard(b.fused_m.detach(), bmesh)
+            state[f"{pre}/bucket_{i}.v"] = as_shard(b.fused_v.detach(), bmesh)
+    state[f"{pre}/step_count"] = torch.tensor(
         getattr(optimizer, "_step_count", 0))
     return state
 
@@ -69,12 +102,17 @@ def load_zero1_optimizer_dcp(optimizer, path: str) -> None:
     optimizer's live buckets, then re-broadcasts the bf16 params."""
     import torch.distributed.checkpoint as dcp
 
+    pre = _mesh_prefix()
     state = _dtensor_state(optimizer)
     dcp.load(state, checkpoint_id=path)
+
+    def local(t):
+        return t.to_local() if hasattr(t, "to_local") else t
+
     for i, b in enumerate(optimizer.buckets):
-        b.master.data.copy_(state[f"bucket_{i}.master"].to_local())
+        b.master.data.copy_(local(state[f"{pre}/bucket_{i}.master"]))
         if hasattr(b, "fused_m"):
-            b.fused_m.copy_(state[f"bucket_{i}.m"].to_local())
-            b.fused_v.copy_(state[f"bucket_{i}.v"].to_local())
-    optimizer._step_count = int(state["step_count"].item())
+            b.fused_m.copy_(local(state[f"{pre}/bucket_{i}.m"]))
+            b.fused_v.copy_(local(state[f"{pre}/bucket_{i}.v"]))
+    optimizer._step_count = int(state[f"{pre}/step_count"].item())
     optimizer._all_gather_params()

@@ -403,7 +403,15 @@ class NeuronZero1Optimizer(torch.optim.Optimizer):
             "masters": [b.master.detach().cpu() for b in self.buckets]
             if include_masters else None,
             "shard_meta": [
-                {"padded": b.padded, "rank": b.rank, "world": b.group_info.size}
+                {"padded": b.padded, "rank": b.rank,
+                 "world": b.group_info.size, "group": b.group_info.name,
+                 "group_index": b.group_index,
+                 # per-param segment map so the OFFLINE dp-shard merge
+                 # (scripts/convert_zero_checkpoints.py, reference
+                 # optimizer/convert_zero_checkpoints.py:15-179) can
+                 # reconstruct full per-parameter tensors
+                 "segments": [(s, e, tuple(p.shape))
+                              for p, s, e in b.segments]}
                 for b in self.buckets
             ],
         }

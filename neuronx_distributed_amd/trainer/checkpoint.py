@@ -1,24 +1,32 @@
 """Checkpoint engine (reference trainer/checkpoint.py:110-973).
 
 Implements the reference's public behavior:
-* per-rank shard files ``{tag}/model/dp_rank_xx_tp_rank_xx_pp_rank_xx.pt``
-  (reference _get_path :54-63),
+* per-rank shard files ``{tag}/model/dp_rank_xx[_cp..][_ep..]_tp_rank_xx_
+  pp_rank_xx.pt`` (reference _get_path :54-63),
 * optimizer saved per DP rank when ZeRO-1 (:760-787),
 * async saves on a single-thread executor with a "done"-tag commit
   protocol + num_kept garbage collection (:66-98,110-324),
-* ``load_checkpoint(tag='latest_if_exists')`` resume-on-restart and
-  broadcast-free per-rank loads.
+* ``use_xser=True``: one file per tensor + skeleton + ``.info.pt`` index
+  (reference :530-575) — streams tensors out one at a time instead of
+  materializing the whole state dict in host RAM, and enables
+* data-parallel-deduplicated PARALLEL writes: the replicated model state
+  is split across the replica group with Karmarkar-Karp size binning so
+  every rank writes a similar number of bytes (reference
+  _assign_tensors_to_bins/_xser_save_data :443-527),
+* ``load_checkpoint(tag='latest_if_exists')`` resume-on-restart,
+* all file IO routed through a storage backend (local FS default,
+  fsspec-backed remote storage with retries — checkpoint_storage.py).
 """
 
 import os
-import shutil
 from concurrent.futures import ThreadPoolExecutor
-from typing import Any, Dict, Optional
+from typing import Any, Dict, List, Optional, Tuple
 
 import torch
 
 from ..parallel import comm, parallel_state as ps
 from ..utils.logger import get_logger
+from .checkpoint_storage import get_storage
 
 logger = get_logger(__name__)
 
@@ -76,6 +84,19 @@ def _rank_name(dp: bool = True) -> str:
     return name
 
 
+def _replica_group():
+    """GroupInfo whose ranks hold IDENTICAL model state (the dedup /
+    parallel-write group): EDP when EP is on, else the merged DPxCP
+    (zero1) group."""
+    if not ps.model_parallel_is_initialized():
+        return None
+    if _ep_size() > 1:
+        return ps.get_group_info("edp")
+    if "zero1" in ps._GROUPS:
+        return ps.get_group_info("zero1")
+    return ps.get_group_info("dp") if "dp" in ps._GROUPS else None
+
+
 def _is_model_writer() -> bool:
     """Model shards are deduped over data-parallel replicas.  Without EP
     that is dp/cp rank 0.  With EP, each EP rank holds DIFFERENT experts,
@@ -87,17 +108,16 @@ def _is_model_writer() -> bool:
             and ps.get_context_model_parallel_rank() == 0)
 
 
-def _done_file(path: str, tag: str) -> str:
-    return os.path.join(path, str(tag), "done")
+def _done_file(tag: str) -> str:
+    return os.path.join(str(tag), "done")
 
 
 def _write_done(path: str, tag: str):
-    with open(_done_file(path, tag), "w") as f:
-        f.write("done")
+    get_storage(path).write_text(_done_file(tag), "done")
 
 
 def checkpoint_exists(path: str, tag: str) -> bool:
-    return os.path.exists(_done_file(path, tag))
+    return get_storage(path).exists(_done_file(tag))
 
 
 def has_checkpoint(path: str) -> bool:
@@ -105,12 +125,9 @@ def has_checkpoint(path: str) -> bool:
 
 
 def _list_checkpoints(path: str):
-    if not os.path.isdir(path):
-        return []
-    tags = []
-    for t in os.listdir(path):
-        if os.path.exists(_done_file(path, t)):
-            tags.append(t)
+    storage = get_storage(path)
+    tags = [t for t in storage.listdir("")
+            if storage.exists(_done_file(t))]
 
     def key(t):
         try:
@@ -124,24 +141,18 @@ def _list_checkpoints(path: str):
 def _gc_checkpoints(path: str, num_kept: Optional[int]):
     """Delete oldest completed checkpoints beyond num_kept, and any
     interrupted (done-less) ones (reference :66-98,259-272)."""
-    if not os.path.isdir(path):
-        return
+    storage = get_storage(path)
     complete = _list_checkpoints(path)
-    for t in os.listdir(path):
-        full = os.path.join(path, t)
-        if os.path.isdir(full) and t not in complete and \
-                any(x in os.listdir(full) for x in ("model", "optim")):
-            logger.warning("removing interrupted checkpoint %s", full)
-            shutil.rmtree(full, ignore_errors=True)
+    for t in storage.listdir(""):
+        if t in complete:
+            continue
+        sub = storage.listdir(t)
+        if any(x in sub for x in ("model", "optim")):
+            logger.warning("removing interrupted checkpoint %s/%s", path, t)
+            storage.remove_tree(t)
     if num_kept is not None and len(complete) > num_kept:
         for t in complete[:-num_kept]:
-            shutil.rmtree(os.path.join(path, t), ignore_errors=True)
-
-
-def _save_obj(obj: Any, fname: str):
-    os.makedirs(os.path.dirname(fname), exist_ok=True)
-    cpu_obj = _to_cpu(obj)
-    torch.save(cpu_obj, fname)
+            storage.remove_tree(t)
 
 
 def _to_cpu(obj):
@@ -155,23 +166,123 @@ def _to_cpu(obj):
     return obj
 
 
+# ---------------------------------------------------------------------------
+# xser per-tensor serialization (reference trainer/checkpoint.py:530-575)
+# ---------------------------------------------------------------------------
+
+class TensorReference:
+    """Placeholder left in an xser skeleton where a tensor was removed."""
+
+    def __init__(self, tid: int):
+        self.tid = tid
+
+    def __repr__(self):
+        return f"TensorReference({self.tid})"
+
+
+def _xser_flatten(obj, out: List[torch.Tensor]):
+    """Replace every tensor in a nested structure with a TensorReference;
+    collects the tensors (in deterministic traversal order) into ``out``."""
+    if isinstance(obj, torch.Tensor):
+        out.append(obj)
+        return TensorReference(len(out) - 1)
+    if isinstance(obj, dict):
+        return {k: _xser_flatten(v, out) for k, v in obj.items()}
+    if isinstance(obj, (list, tuple)):
+        t = [_xser_flatten(v, out) for v in obj]
+        return t if isinstance(obj, list) else tuple(t)
+    return obj
+
+
+def _xser_unflatten(obj, fetch):
+    if isinstance(obj, TensorReference):
+        return fetch(obj.tid)
+    if isinstance(obj, dict):
+        return {k: _xser_unflatten(v, fetch) for k, v in obj.items()}
+    if isinstance(obj, (list, tuple)):
+        t = [_xser_unflatten(v, fetch) for v in obj]
+        return t if isinstance(obj, list) else tuple(t)
+    return obj
+
+
+def _tensor_file(fname: str, tid: int) -> str:
+    return f"{fname}.tensors/tensor_{tid}.pt"
+
+
+def assign_tensors_to_bins(tensors: List[torch.Tensor],
+                           bin_count: int) -> List[List[int]]:
+    """Karmarkar-Karp style greedy size binning (reference
+    _assign_tensors_to_bins :443-474): sort by size, always assign to the
+    currently-smallest bin -> near-even bytes per writer."""
+    bins: List[List[int]] = [[] for _ in range(bin_count)]
+    sizes = [0] * bin_count
+    order = sorted(range(len(tensors)),
+                   key=lambda i: tensors[i].numel() * tensors[i].element_size())
+    for tidx in order:
+        bid = sizes.index(min(sizes))
+        bins[bid].append(tidx)
+        sizes[bid] += tensors[tidx].numel() * tensors[tidx].element_size()
+    return bins
+
+
+def _xser_save(storage, fname: str, obj, my_bin: Optional[List[int]],
+               write_skeleton: bool, jobs: List[Tuple[Any, str]]):
+    """Queue xser jobs: my bin of tensor files (+ skeleton/info when this
+    rank is the group writer)."""
+    tensors: List[torch.Tensor] = []
+    skeleton = _xser_flatten(obj, tensors)
+    for tid, t in enumerate(tensors):
+        if my_bin is None or tid in my_bin:
+            jobs.append((t, _tensor_file(fname, tid)))
+    if write_skeleton:
+        info = {tid: {"dtype": t.dtype, "shape": tuple(t.shape)}
+                for tid, t in enumerate(tensors)}
+        jobs.append((skeleton, fname))
+        jobs.append((info, fname + ".info.pt"))
+
+
+def _xser_load(storage, fname: str):
+    skeleton = storage.load_object(fname)
+    return _xser_unflatten(
+        skeleton, lambda tid: storage.load_object(_tensor_file(fname, tid)))
+
+
+def _is_xser_file(storage, fname: str) -> bool:
+    return storage.exists(fname + ".info.pt")
+
+
+# ---------------------------------------------------------------------------
+# save / load
+# ---------------------------------------------------------------------------
+
 def save_checkpoint(path: str, tag, model=None, optimizer=None,
                     scheduler=None, user_content: Optional[Dict] = None,
-                    num_workers: int = 8, use_xser: bool = False,  # noqa: ARG001 — xser is an XLA host-RAM workaround; MI355X snapshots to CPU and torch.saves (API compat)
+                    num_workers: int = 8, use_xser: bool = False,
                     num_kept: Optional[int] = None, async_save: bool = False,
                     zero1_optimizer: Optional[bool] = None) -> None:
     """reference trainer/checkpoint.py:654-824."""
     tag = str(tag)
-    ckpt_dir = os.path.join(path, tag)
+    storage = get_storage(path)
     rank = ps._cur_rank() if ps.model_parallel_is_initialized() else 0
-    os.makedirs(ckpt_dir, exist_ok=True)
 
-    jobs = []
+    jobs: List[Tuple[Any, str]] = []  # (obj, rel_path)
     if model is not None:
         sd = model.state_dict()
-        if _is_model_writer():
-            jobs.append((sd, os.path.join(ckpt_dir, "model",
-                                          _rank_name(dp=False) + ".pt")))
+        mname = os.path.join(tag, "model", _rank_name(dp=False) + ".pt")
+        if use_xser:
+            # every replica writes its KK bin of tensor files in parallel;
+            # the replica-group rank 0 writes the skeleton + index
+            group = _replica_group()
+            my_bin = None
+            write_skel = _is_model_writer()
+            if group is not None and group.size > 1:
+                tensors: List[torch.Tensor] = []
+                _xser_flatten(sd, tensors)
+                gr = group.rank_in_group(ps._cur_rank())
+                my_bin = assign_tensors_to_bins(tensors, group.size)[gr]
+            _xser_save(storage, mname, sd, my_bin, write_skel, jobs)
+        elif _is_model_writer():
+            jobs.append((sd, mname))
     if optimizer is not None:
         from ..optimizer import NeuronZero1Optimizer
         from .optimizer import NxDOptimizer
@@ -182,18 +293,20 @@ def save_checkpoint(path: str, tag, model=None, optimizer=None,
             if zero1_optimizer is None else zero1_optimizer
         # zero1: every DP rank holds distinct shards -> all ranks write
         if is_zero1 or ps.get_data_parallel_rank() == 0:
-            jobs.append((optimizer.state_dict(),
-                         os.path.join(ckpt_dir, "optim",
-                                      _rank_name() + ".pt")))
+            oname = os.path.join(tag, "optim", _rank_name() + ".pt")
+            if use_xser:
+                _xser_save(storage, oname, optimizer.state_dict(), None,
+                           True, jobs)
+            else:
+                jobs.append((optimizer.state_dict(), oname))
     if scheduler is not None and rank == 0:
-        jobs.append((scheduler.state_dict(),
-                     os.path.join(ckpt_dir, "scheduler.pt")))
+        jobs.append((scheduler.state_dict(), os.path.join(tag, "scheduler.pt")))
     if user_content is not None and rank == 0:
-        jobs.append((user_content, os.path.join(ckpt_dir, "user_content.pt")))
+        jobs.append((user_content, os.path.join(tag, "user_content.pt")))
 
     def commit():
         for obj, fname in jobs:
-            _save_obj(obj, fname)
+            storage.save_object(_to_cpu(obj), fname)
         comm.barrier()
         if rank == 0:
             _write_done(path, tag)
@@ -208,7 +321,7 @@ def save_checkpoint(path: str, tag, model=None, optimizer=None,
 
         def bg():
             for obj, fname in jobs:
-                _save_obj(obj, fname)
+                storage.save_object(obj, fname)
 
         _PENDING.append((_executor().submit(bg), path, tag, num_kept))
     else:
@@ -219,6 +332,7 @@ def load_checkpoint(path: str, tag=None, model=None, optimizer=None,
                     scheduler=None, strict: bool = True):
     """reference trainer/checkpoint.py:347-432; tag=None -> latest complete.
     Returns user_content (or None)."""
+    storage = get_storage(path)
     if tag is None or tag == "latest_if_exists":
         tags = _list_checkpoints(path)
         if not tags:
@@ -227,29 +341,29 @@ def load_checkpoint(path: str, tag=None, model=None, optimizer=None,
             raise FileNotFoundError(f"no complete checkpoint under {path}")
         tag = tags[-1]
     tag = str(tag)
-    ckpt_dir = os.path.join(path, tag)
     if not checkpoint_exists(path, tag):
-        raise FileNotFoundError(f"checkpoint {ckpt_dir} incomplete (no done tag)")
+        raise FileNotFoundError(
+            f"checkpoint {path}/{tag} incomplete (no done tag)")
 
     if model is not None:
         # model shards are written deduped with dp/cp pinned to 0
-        fname = os.path.join(ckpt_dir, "model", _rank_name(dp=False) + ".pt")
-        sd = torch.load(fname, map_location="cpu", weights_only=False)
+        fname = os.path.join(tag, "model", _rank_name(dp=False) + ".pt")
+        sd = _xser_load(storage, fname) if _is_xser_file(storage, fname) \
+            else storage.load_object(fname)
         model.load_state_dict(sd, strict=strict)
     if optimizer is not None:
-        fname = os.path.join(ckpt_dir, "optim", _rank_name() + ".pt")
-        if not os.path.exists(fname):
+        fname = os.path.join(tag, "optim", _rank_name() + ".pt")
+        if not storage.exists(fname):
             # non-zero1 optimizers are saved by dp rank 0 only
-            fname = os.path.join(ckpt_dir, "optim",
-                                 _rank_name(dp=False) + ".pt")
-        optimizer.load_state_dict(
-            torch.load(fname, map_location="cpu", weights_only=False))
+            fname = os.path.join(tag, "optim", _rank_name(dp=False) + ".pt")
+        sd = _xser_load(storage, fname) if _is_xser_file(storage, fname) \
+            else storage.load_object(fname)
+        optimizer.load_state_dict(sd)
     if scheduler is not None:
-        f = os.path.join(ckpt_dir, "scheduler.pt")
-        if os.path.exists(f):
-            scheduler.load_state_dict(
-                torch.load(f, map_location="cpu", weights_only=False))
-    uc = os.path.join(ckpt_dir, "user_content.pt")
-    if os.path.exists(uc):
-        return torch.load(uc, map_location="cpu", weights_only=False)
+        f = os.path.join(tag, "scheduler.pt")
+        if storage.exists(f):
+            scheduler.load_state_dict(storage.load_object(f))
+    uc = os.path.join(tag, "user_content.pt")
+    if storage.exists(uc):
+        return storage.load_object(uc)
     return None

@@ -1,0 +1,253 @@
+"""Round-2 checkpoint engine depth: xser per-tensor format, KK-binned
+parallel dedup saves, fsspec remote storage with retries, zero-DCP at
+tp>1, and the offline zero-shard merge CLI (reference
+trainer/checkpoint.py:443-575, optimizer/convert_zero_checkpoints.py)."""
+
+import os
+import tempfile
+
+import pytest
+import torch
+
+from dist_utils import run_distributed
+
+
+# ---------------------------------------------------------------------------
+# KK binning
+# ---------------------------------------------------------------------------
+
+def test_kk_binning_balances_bytes():
+    from neuronx_distributed_amd.trainer.checkpoint import \
+        assign_tensors_to_bins
+
+    torch.manual_seed(0)
+    tensors = [torch.empty(n) for n in (1000, 10, 990, 500, 505, 30, 2000)]
+    bins = assign_tensors_to_bins(tensors, 3)
+    assert sorted(sum(bins, [])) == list(range(len(tensors)))  # partition
+    sizes = [sum(tensors[i].numel() * 4 for i in b) for b in bins]
+    assert max(sizes) - min(sizes) <= 2000 * 4  # bounded by largest tensor
+    # determinism (all replicas must compute identical bins)
+    assert bins == assign_tensors_to_bins(tensors, 3)
+
+
+# ---------------------------------------------------------------------------
+# xser per-tensor format
+# ---------------------------------------------------------------------------
+
+def _xser_worker(rank, world, tmpdir):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.trainer import checkpoint as ckpt
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)  # same model on every dp rank (replicas)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 32),
+                                torch.nn.Linear(32, 8))
+    orig = {k: v.clone() for k, v in model.state_dict().items()}
+
+    ckpt.save_checkpoint(tmpdir, "s1", model=model, use_xser=True)
+
+    # per-tensor layout exists: skeleton + info + 4 tensor files
+    mdir = os.path.join(tmpdir, "s1", "model")
+    base = [f for f in os.listdir(mdir) if f.endswith(".pt")]
+    assert len(base) == 2, base  # skeleton + info
+    tdir = [f for f in os.listdir(mdir) if f.endswith(".tensors")]
+    assert len(tdir) == 1
+    tfiles = os.listdir(os.path.join(mdir, tdir[0]))
+    assert len(tfiles) == 4  # 2 weights + 2 biases
+
+    # KK-binned parallel write: with 2 replicas each rank wrote ~half
+    if world > 1:
+        mine = [f for f in tfiles
+                if os.path.getsize(os.path.join(mdir, tdir[0], f)) > 0]
+        assert len(mine) == 4  # all present in the union
+
+    with torch.no_grad():
+        for p in model.parameters():
+            p.zero_()
+    ckpt.load_checkpoint(tmpdir, "s1", model=model)
+    for k, v in model.state_dict().items():
+        assert torch.equal(v, orig[k]), k
+    return 0.0
+
+
+def test_xser_roundtrip_single():
+    with tempfile.TemporaryDirectory() as d:
+        run_distributed(_xser_worker, world_size=1, args=(d,))
+
+
+def test_xser_kk_parallel_write_dp2():
+    with tempfile.TemporaryDirectory() as d:
+        run_distributed(_xser_worker, world_size=2, args=(d,))
+
+
+def _xser_optimizer_worker(rank, world, tmpdir):
+    """xser also applies to (zero1) optimizer shards, per rank."""
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.trainer import checkpoint as ckpt
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    model = torch.nn.Linear(64, 64)
+    opt = NeuronZero1Optimizer(model.parameters(), torch.optim.AdamW,
+                               grad_clipping=False, lr=1e-2,
+                               use_fused_kernel=False)
+    model(torch.randn(4, 64)).sum().backward()
+    opt.step()
+    sd_before = opt.state_dict()
+
+    ckpt.save_checkpoint(tmpdir, "s1", optimizer=opt, use_xser=True)
+    opt2 = NeuronZero1Optimizer(model.parameters(), torch.optim.AdamW,
+                                grad_clipping=False, lr=1e-2,
+                                use_fused_kernel=False)
+    ckpt.load_checkpoint(tmpdir, "s1", optimizer=opt2)
+    for b1, m in zip(opt2.buckets, sd_before["masters"]):
+        assert torch.allclose(b1.master.detach().cpu(), m)
+    return 0.0
+
+
+def test_xser_optimizer_roundtrip():
+    with tempfile.TemporaryDirectory() as d:
+        run_distributed(_xser_optimizer_worker, world_size=2, args=(d,))
+
+
+# ---------------------------------------------------------------------------
+# fsspec storage (memory://) + retry
+# ---------------------------------------------------------------------------
+
+def _memfs_worker(rank, world):
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.trainer import checkpoint as ckpt
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)
+    model = torch.nn.Linear(8, 8)
+    orig = {k: v.clone() for k, v in model.state_dict().items()}
+    uri = "memory://nxda_ckpt_test"
+    ckpt.save_checkpoint(uri, "t1", model=model,
+                         user_content={"step": 7})
+    assert ckpt.checkpoint_exists(uri, "t1")
+    with torch.no_grad():
+        model.weight.zero_()
+    uc = ckpt.load_checkpoint(uri, "t1", model=model)
+    assert uc == {"step": 7}
+    for k, v in model.state_dict().items():
+        assert torch.equal(v, orig[k]), k
+    # latest_if_exists + GC through the storage layer
+    ckpt.save_checkpoint(uri, "t2", model=model, num_kept=1)
+    tags = ckpt._list_checkpoints(uri)
+    assert tags == ["t2"]
+    return 0.0
+
+
+def test_fsspec_memory_storage_roundtrip():
+    run_distributed(_memfs_worker, world_size=1)
+
+
+def test_retry_transient_decrementing():
+    from neuronx_distributed_amd.trainer.checkpoint_storage import \
+        retry_transient
+
+    calls = []
+
+    def flaky():
+        calls.append(1)
+        if len(calls) < 3:
+            raise OSError("slow down")
+        return 42
+
+    assert retry_transient(flaky, attempts=5, first_wait=0.01) == 42
+    assert len(calls) == 3
+
+    with pytest.raises(OSError):
+        retry_transient(lambda: (_ for _ in ()).throw(OSError("x")),
+                        attempts=2, first_wait=0.01)
+
+
+# ---------------------------------------------------------------------------
+# zero-DCP with tp > 1
+# ---------------------------------------------------------------------------
+
+def _dcp_tp2_worker(rank, world, tmpdir):
+    """tp=2 (zero1 group size 1 per slice): per-slice keys keep the DCP
+    save/load valid; each tp rank round-trips its own distinct masters."""
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+    from neuronx_distributed_amd.optimizer.zero_dcp_utils import (
+        load_zero1_optimizer_dcp, save_zero1_optimizer_dcp)
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.layers import ColumnParallelLinear
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    layer = ColumnParallelLinear(16, 32, bias=False, gather_output=False)
+    opt = NeuronZero1Optimizer(layer.parameters(), torch.optim.AdamW,
+                               grad_clipping=False, lr=1e-2,
+                               use_fused_kernel=False)
+    x = torch.randn(4, 16)
+    layer(x).pow(2).sum().backward()
+    opt.step()
+    masters = [b.master.detach().clone() for b in opt.buckets]
+
+    save_zero1_optimizer_dcp(opt, tmpdir)
+
+    for b in opt.buckets:
+        b.master.data.zero_()
+    load_zero1_optimizer_dcp(opt, tmpdir)
+    for b, m in zip(opt.buckets, masters):
+        assert torch.allclose(b.master.detach(), m)
+        # tp slices hold DIFFERENT weights: make sure we didn't cross-load
+    return float(masters[0].sum())
+
+
+def test_zero_dcp_tp2():
+    with tempfile.TemporaryDirectory() as d:
+        res = run_distributed(_dcp_tp2_worker, world_size=2, args=(d,))
+    assert abs(res[0] - res[1]) > 1e-6  # distinct slices really differed
+
+
+# ---------------------------------------------------------------------------
+# offline zero-shard merge CLI
+# ---------------------------------------------------------------------------
+
+def _merge_worker(rank, world, tmpdir):
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.trainer import checkpoint as ckpt
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(0)  # replicas identical
+    model = torch.nn.Sequential(torch.nn.Linear(24, 48, bias=False),
+                                torch.nn.Linear(48, 8, bias=False))
+    opt = NeuronZero1Optimizer(model.parameters(), torch.optim.AdamW,
+                               grad_clipping=False, lr=1e-2,
+                               use_fused_kernel=False)
+    torch.manual_seed(10 + rank)
+    model(torch.randn(4, 24)).pow(2).sum().backward()
+    opt.step()
+    ckpt.save_checkpoint(tmpdir, "step5", optimizer=opt)
+    # post-step parameters (identical on every dp rank after all-gather);
+    # returned as plain lists — tensors over the mp queue race with the
+    # daemon process teardown
+    return [(tuple(p.shape), p.detach().reshape(-1).tolist())
+            for p in model.parameters()]
+
+
+def test_convert_zero_checkpoints_cli():
+    from neuronx_distributed_amd.scripts.convert_zero_checkpoints import \
+        convert
+
+    with tempfile.TemporaryDirectory() as d:
+        params = run_distributed(_merge_worker, world_size=2, args=(d,))
+        out = convert(os.path.join(d, "step5"), os.path.join(d, "merged"))
+        assert len(out) == 1
+        merged = torch.load(out[0], weights_only=False)
+        state = merged["state"]
+        assert len(state) == 2  # two weight matrices
+        # merged fp32 masters reshape to the param shapes and match the
+        # post-step parameters (rank 0's view == rank 1's, all-gathered)
+        for i, (shape, vals) in enumerate(params[0]):
+            p = torch.tensor(vals).reshape(shape)
+            assert state[i]["master"].shape == p.shape
+            assert torch.allclose(state[i]["master"].to(p.dtype), p,
+                                  atol=1e-6), i
+            assert "exp_avg" in state[i]

@@ -727,7 +727,9 @@ def test_checkpoint_storage_backend(tmp_path):
     assert "tag" in st.listdir()
     st.remove_tree("tag")
     assert not st.exists("tag/done")
-    with _pytest.raises(NotImplementedError):
+    # s3:// now routes through FsspecStorage; without s3fs installed the
+    # protocol resolution fails loudly (with s3fs on a cluster it works)
+    with _pytest.raises((ImportError, ValueError)):
         get_storage("s3://bucket/prefix")
 
 
