@@ -28,6 +28,7 @@ from .scheduler import (
     RecvForward,
     ReduceGrads,
     SendBackward,
+    SendBackwardRecvForward,
     SendForward,
     SendForwardRecvBackward,
     Train1F1BSchedule,
@@ -51,6 +52,7 @@ class NxDPPModel(nn.Module):
                  manual_pp_loss_fn=None,
                  broadcast_and_average_loss: bool = True,
                  auto_partition: bool = True,
+                 deallocate_pipeline_outputs: bool = True,
                  _debug_pre_partitioned=False):
         super().__init__()
         self.original_torch_module = module
@@ -63,6 +65,7 @@ class NxDPPModel(nn.Module):
         self.leaf_module_cls = leaf_module_cls
         self.broadcast_and_average_loss = broadcast_and_average_loss
         self.manual_pp_loss_fn = manual_pp_loss_fn
+        self.deallocate_pipeline_outputs = deallocate_pipeline_outputs
 
         self.pp_rank = ps.get_pipeline_model_parallel_rank()
         self.pp_size = ps.get_pipeline_model_parallel_size()
@@ -168,13 +171,21 @@ class NxDPPModel(nn.Module):
             dist.all_reduce(p.grad, group=group)
 
     def _analyze_stage_io(self, split):
-        """From the split top-level graph, derive for every stage: which
+        """From the split top-level graph, derive for every stage which
         placeholders come from original inputs and which from the previous
-        stage (reference partition.py:132-223 stage IO analysis)."""
-        specs = []
+        stage — INCLUDING skip connections whose producer is an earlier
+        (non-adjacent) stage: those values are PASSED THROUGH the
+        intervening stages' P2P wires (reference partition.py:132-223
+        stage-IO analysis with pass-through objects).
+
+        Produces:
+        * ``_stage_specs[s]``: arg descriptors ("input", name) /
+          ("attr", target) / ("prev", recv_idx) per stage,
+        * ``_send_plan[s]``: the wire each stage sends across boundary s —
+          entries ("own", out_idx) or ("fwd", recv_idx)."""
+        raw_args: List[List] = []       # per stage: descriptors w/ ("val",(p,oi))
         producer: Dict[Any, Tuple[int, int]] = {}  # node -> (stage, out_idx)
-        stage_idx = -1
-        stage_nouts: Dict[int, int] = {}
+        max_consumer: Dict[Tuple[int, int], int] = {}
         for node in split.graph.nodes:
             if node.op == "call_module" and node.target.startswith("submod_"):
                 stage_idx = int(node.target.split("_")[1])
@@ -184,25 +195,51 @@ class NxDPPModel(nn.Module):
                         spec.append(("input", arg.target))
                     elif arg.op == "call_module":
                         st, _ = producer[arg]
-                        if st != stage_idx - 1:
-                            raise NotImplementedError(
-                                "cross-stage skip connections not supported")
-                        spec.append(("prev", 0))
+                        spec.append(("val", (st, 0)))
+                        max_consumer[(st, 0)] = max(
+                            max_consumer.get((st, 0), st), stage_idx)
                     elif arg.op == "call_function":  # getitem
                         src = arg.args[0]
                         idx = arg.args[1]
                         st, _ = producer[src]
-                        if st != stage_idx - 1:
-                            raise NotImplementedError(
-                                "cross-stage skip connections not supported")
-                        spec.append(("prev", idx))
+                        spec.append(("val", (st, idx)))
+                        max_consumer[(st, idx)] = max(
+                            max_consumer.get((st, idx), st), stage_idx)
                     elif arg.op == "get_attr":
                         spec.append(("attr", arg.target))
                     else:
                         raise NotImplementedError(f"stage arg {arg.op}")
                 producer[node] = (stage_idx, 0)
-                specs.append(spec)
+                raw_args.append(spec)
+
+        n_stages = len(raw_args)
+        # boundary b (between stage b and b+1) carries every value produced
+        # at stage p <= b still needed by a stage > b, in deterministic order
+        carried: List[List[Tuple[int, int]]] = []
+        for b in range(n_stages - 1):
+            carried.append(sorted(
+                v for v, last in max_consumer.items() if v[0] <= b < last))
+
+        specs = []
+        send_plan: List[List[Tuple[str, int]]] = []
+        for s, spec in enumerate(raw_args):
+            out = []
+            for kind, key in spec:
+                if kind == "val":
+                    out.append(("prev", carried[s - 1].index(key)))
+                else:
+                    out.append((kind, key))
+            specs.append(out)
+            plan = []
+            if s < n_stages - 1:
+                for (p, oi) in carried[s]:
+                    if p == s:
+                        plan.append(("own", oi))
+                    else:
+                        plan.append(("fwd", carried[s - 1].index((p, oi))))
+            send_plan.append(plan)
         self._stage_specs = specs
+        self._send_plan = send_plan
 
     def local_module(self):
         self._maybe_partition()
@@ -284,28 +321,116 @@ class NxDPPModel(nn.Module):
             return list(out)
         return [out]
 
+    @staticmethod
+    def _custom_backward(outputs, grads):
+        """Direct autograd-engine backward that skips the grad/output shape
+        check — required because deallocated outputs have had their .data
+        replaced by a 1-element stub (reference pipeline/model.py:1163-1215
+        deallocate_output + custom_backward)."""
+        from torch.autograd import Variable
+
+        Variable._execution_engine.run_backward(
+            tuple(outputs), tuple(grads), False, False, tuple(), True, True)
+
+    def _deallocate_outputs(self, out_list):
+        """After the forward send is posted, the full activation data of a
+        stage output is only needed DOWNSTREAM: replace .data with a stub
+        so the memory frees as soon as the in-flight send copy drains
+        (peak-memory control during 1F1B warmup; reference model.py:
+        1163-1215)."""
+        for t in out_list:
+            if isinstance(t, torch.Tensor) and t.requires_grad \
+                    and t.is_floating_point():
+                t.data = torch.empty(1, dtype=t.dtype, device=t.device)
+
+    def _wire_send_fwd(self, outputs, recvd_inputs, key):
+        """The forward wire this stage sends: own outputs + pass-through
+        values forwarded from the previous stage, in boundary order."""
+        if self._stage_specs is None:  # manual partition: raw outputs
+            return outputs[key]
+        gstage = key[1] * self.pp_size + self.pp_rank
+        wire = []
+        for kind, idx in self._send_plan[gstage]:
+            wire.append(outputs[key][idx] if kind == "own"
+                        else recvd_inputs[key][idx])
+        return wire
+
     def _run_schedule(self, schedule, kwargs, train: bool):
         self._maybe_partition()
         mbs = self._split_microbatches(kwargs)
         # state keyed by (mb, chunk); chunk is 0 throughout for C == 1
-        recvd_inputs: Dict[Tuple[int, int], List[torch.Tensor]] = {}
-        outputs: Dict[Tuple[int, int], List[torch.Tensor]] = {}
+        recvd_inputs: Dict[Tuple[int, int], List] = {}
+        outputs: Dict[Tuple[int, int], List] = {}
         losses: List[torch.Tensor] = []
         self._pending_sends = []
+        self._pt_grads: Dict[Tuple[int, int], Dict[int, torch.Tensor]] = {}
         C = self.virtual_pipeline_size
+        dealloc = train and getattr(self, "deallocate_pipeline_outputs", True)
 
         def is_loss_stage(chunk):
             return self.pp_rank == self.pp_size - 1 and chunk == C - 1 and \
                 self.output_loss_value_spec
 
+        def attach_recvd(key, items):
+            for t in items:
+                if isinstance(t, torch.Tensor) and t.is_floating_point():
+                    t.requires_grad_(True)
+            recvd_inputs[key] = items
+
+        def send_fwd_wire(key):
+            wire = self._wire_send_fwd(outputs, recvd_inputs, key)
+            return wire
+
+        def backward_grads(key):
+            """Consume the oldest received grad message for this stage's
+            send wire: own-output grads drive backward; grads of forwarded
+            (pass-through) values are stashed for SendBackward."""
+            grads = self._pending_grads.pop(0)
+            gstage = key[1] * self.pp_size + self.pp_rank
+            plan = self._send_plan[gstage] if self._stage_specs is not None \
+                else [("own", i) for i in range(len(outputs[key]))]
+            pairs = []
+            pt: Dict[int, torch.Tensor] = {}
+            gi = 0
+            for kind, idx in plan:
+                t = outputs[key][idx] if kind == "own" \
+                    else recvd_inputs[key][idx]
+                if isinstance(t, torch.Tensor) and t.is_floating_point():
+                    g = grads[gi]
+                    gi += 1
+                    if kind == "own":
+                        if t.requires_grad:
+                            pairs.append((t, g))
+                    else:
+                        pt[idx] = pt[idx] + g if idx in pt else g
+            if pt:
+                self._pt_grads[key] = pt
+            if pairs:
+                if dealloc:
+                    self._custom_backward([p[0] for p in pairs],
+                                          [p[1] for p in pairs])
+                else:
+                    torch.autograd.backward([p[0] for p in pairs],
+                                            [p[1] for p in pairs])
+
+        def grads_to_send(key):
+            """One grad per float tensor of the RECEIVED wire, combining the
+            local autograd .grad with any pass-through grad from
+            downstream."""
+            pt = self._pt_grads.pop(key, {})
+            grads = []
+            for i, t in enumerate(recvd_inputs[key]):
+                if isinstance(t, torch.Tensor) and t.is_floating_point():
+                    g = t.grad
+                    if i in pt:
+                        g = pt[i] if g is None else g + pt[i]
+                    grads.append(g if g is not None else torch.zeros_like(t))
+            return grads
+
         for task in schedule.steps():
             key = (task.mb, task.chunk)
             if isinstance(task, RecvForward):
-                tensors = ppcomm.recv_from(self.prev_rank)
-                for t in tensors:
-                    if t.is_floating_point():
-                        t.requires_grad_(True)
-                recvd_inputs[key] = tensors
+                attach_recvd(key, ppcomm.recv_from(self.prev_rank))
             elif isinstance(task, ForwardStep):
                 with torch.enable_grad() if train else torch.no_grad():
                     out = self._stage_forward(mbs[task.mb],
@@ -318,13 +443,35 @@ class NxDPPModel(nn.Module):
             elif isinstance(task, SendForward):
                 self._purge_pending_sends()
                 self._pending_sends.append(
-                    ppcomm.send_async(outputs[key], self.next_rank))
+                    ppcomm.send_async(send_fwd_wire(key), self.next_rank))
+                if dealloc:
+                    self._deallocate_outputs(outputs[key])
             elif isinstance(task, SendForwardRecvBackward):
                 # both directions batched; grads arrive in mb order and
-                # attach to the oldest un-backwarded microbatch (FIFO)
-                grads = ppcomm.send_recv(outputs[key], self.next_rank,
-                                         self.next_rank)
+                # attach to the oldest un-backwarded microbatch (FIFO).
+                # Send completion is DEFERRED: in the interleaved ring the
+                # isend completes only after the peer's next recv.
+                grads, works, refs = ppcomm.send_recv(
+                    send_fwd_wire(key), self.next_rank, self.next_rank,
+                    defer_sends=True)
+                self._purge_pending_sends()
+                if works:
+                    self._pending_sends.append((works, refs))
+                if dealloc:
+                    self._deallocate_outputs(outputs[key])
                 self._pending_grads.append(grads)
+            elif isinstance(task, SendBackwardRecvForward):
+                # fused steady-state exchange with the PREVIOUS rank
+                # (interleaved schedule): send grads for (mb, chunk), recv
+                # the next forward for (mb2, chunk2)
+                items, works, refs = ppcomm.send_recv(
+                    grads_to_send(key), self.prev_rank, self.prev_rank,
+                    defer_sends=True)
+                self._purge_pending_sends()
+                if works:
+                    self._pending_sends.append((works, refs))
+                del recvd_inputs[key]
+                attach_recvd((task.mb2, task.chunk2), items)
             elif isinstance(task, RecvBackward):
                 self._pending_grads.append(ppcomm.recv_from(self.next_rank))
             elif isinstance(task, BackwardStep):
@@ -332,26 +479,14 @@ class NxDPPModel(nn.Module):
                     loss = outputs[key][0]
                     (loss / self.num_microbatches).backward()
                 else:
-                    grads = self._pending_grads.pop(0)
-                    # downstream sent one grad per FLOAT output, in order
-                    pairs = []
-                    gi = 0
-                    for t in outputs[key]:
-                        if t.is_floating_point():
-                            if t.requires_grad:
-                                pairs.append((t, grads[gi]))
-                            gi += 1
-                    torch.autograd.backward([p[0] for p in pairs],
-                                            [p[1] for p in pairs])
+                    backward_grads(key)
                 # free the graph/output refs
-                outputs[key] = [t.detach() for t in outputs[key]]
+                outputs[key] = [t.detach() if isinstance(t, torch.Tensor)
+                                else t for t in outputs[key]]
             elif isinstance(task, SendBackward):
-                grads = [t.grad if t.grad is not None
-                         else torch.zeros_like(t)
-                         for t in recvd_inputs[key] if t.is_floating_point()]
                 self._purge_pending_sends()
                 self._pending_sends.append(
-                    ppcomm.send_async(grads, self.prev_rank))
+                    ppcomm.send_async(grads_to_send(key), self.prev_rank))
                 del recvd_inputs[key]
             elif isinstance(task, ReduceGrads):
                 # DP grad sync happens in the optimizer step; here only the

@@ -22,6 +22,11 @@ from typing import Iterator
 class PipelineTask:
     mb: int  # microbatch index
     chunk: int = 0  # model chunk (interleaved schedule)
+    # second (mb, chunk) pair for FUSED bidirectional tasks: the RECV side
+    # of SendBackwardRecvForward (reference scheduler.py:281-293 fused
+    # send-recv in the interleaved schedule)
+    mb2: int = -1
+    chunk2: int = -1
 
 
 class RecvForward(PipelineTask):
@@ -177,23 +182,31 @@ class TrainInterleavedSchedule(BaseSchedule):
             if not (self.is_last and c == C - 1):
                 yield SendForward(mb, c)
             fwd_i += 1
+        # steady state with FUSED bidirectional exchanges (reference
+        # scheduler.py:281-293): send-fwd + recv-bwd batched toward the
+        # next rank, send-bwd + recv-next-fwd batched toward the previous
+        # rank — each pair flies in one batched isend/irecv.
         for _ in range(total - warmup):
             c = chunk_of(fwd_i, True)
             mb = mb_of(fwd_i)
             if not (self.is_first and c == 0):
                 yield RecvForward(mb, c)
             yield ForwardStep(mb, c)
-            if not (self.is_last and c == C - 1):
-                yield SendForward(mb, c)
-            fwd_i += 1
             cb = chunk_of(bwd_i, False)
             mbb = mb_of(bwd_i)
-            if not (self.is_last and cb == C - 1):
+            send_f = not (self.is_last and c == C - 1)
+            recv_b = not (self.is_last and cb == C - 1)
+            if send_f and recv_b:
+                yield SendForwardRecvBackward(mb, c)
+            elif send_f:
+                yield SendForward(mb, c)
+            elif recv_b:
                 yield RecvBackward(mbb, cb)
             yield BackwardStep(mbb, cb)
             if not (self.is_first and cb == 0):
                 yield SendBackward(mbb, cb)
             bwd_i += 1
+            fwd_i += 1
         for _ in range(total - bwd_i):
             cb = chunk_of(bwd_i, False)
             mbb = mb_of(bwd_i)
