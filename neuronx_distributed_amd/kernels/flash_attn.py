@@ -45,10 +45,11 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
         from .. import ops
 
         if q.shape[2] != k.shape[2] or q.shape[3] != 128 or \
-                window is not None:
-            # rectangular attention (KV-cache decode), head_dim != 128, or
-            # sliding-window masking: composed path — the HIP MFMA kernel
-            # is D=128 full-causal (the production training shapes)
+                window is not None or q.dtype != torch.bfloat16:
+            # rectangular attention (KV-cache decode), head_dim != 128,
+            # sliding-window masking, or non-bf16 dtype (fp32 golden runs):
+            # composed path — the HIP MFMA kernel is bf16 D=128
+            # full-causal (the production training shapes)
             return _torch_reference(q, k, v, causal, softmax_scale, window)
         if hasattr(ops, "flash_attn") and ops.flash_attn_available():
             return ops.flash_attn(q, k, v, causal=causal,
