@@ -8,3 +8,4 @@ from .moe_parallel_layers import (
 from .shared_experts import SharedExperts
 from .loss_function import load_balancing_loss_func
 from .token_shuffling import token_shuffle, token_unshuffle
+from .moe_fused_tkg import MoEFusedTKG

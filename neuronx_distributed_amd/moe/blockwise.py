@@ -22,10 +22,17 @@ DEFAULT_BLOCK_SIZE = 512
 
 def compute_block_indices(expert_index: torch.Tensor, num_experts: int,
                           block_size: int = DEFAULT_BLOCK_SIZE,
-                          num_blocks: int = None):
+                          num_blocks: int = None,
+                          max_blocks_per_expert: int = None):
     """expert_index (T, k) -> (token_position_to_id (N*B,), block_to_expert
     (N,), n_used_blocks).  Assignment order is arrival order per expert
-    (stable sort), padding slots are -1."""
+    (stable sort), padding slots are -1.
+
+    ``max_blocks_per_expert`` enables the DROPPING variant (reference K5
+    blockwise_mm_shard_intermediate_dropping + SkipMode): each expert gets
+    a static block budget; assignments beyond ``budget * block_size`` are
+    dropped (their slots never enter the scatter, so they contribute zero
+    like the capacity-factor strategy's over-capacity tokens)."""
     T, k = expert_index.shape
     flat_expert = expert_index.reshape(-1)
     token_of = (torch.arange(T * k, device=expert_index.device) // k)
@@ -35,6 +42,8 @@ def compute_block_indices(expert_index: torch.Tensor, num_experts: int,
 
     counts = torch.bincount(flat_expert, minlength=num_experts)
     blocks_per_e = torch.ceil(counts.float() / block_size).long()
+    if max_blocks_per_expert is not None:
+        blocks_per_e = blocks_per_e.clamp(max=max_blocks_per_expert)
     n_blocks = int(blocks_per_e.sum().item())
     if num_blocks is not None:
         assert n_blocks <= num_blocks, "static block budget exceeded"
@@ -55,10 +64,11 @@ def compute_block_indices(expert_index: torch.Tensor, num_experts: int,
         torch.nn.functional.pad(blocks_per_e, (1, 0)), 0)[sorted_expert]
     slot = block_start * block_size + within
 
+    keep = within < blocks_per_e[sorted_expert] * block_size
     token_position_to_id = torch.full((num_blocks * block_size,), -1,
                                       dtype=torch.long,
                                       device=expert_index.device)
-    token_position_to_id[slot] = sorted_token
+    token_position_to_id[slot[keep]] = sorted_token[keep]
     return token_position_to_id, block_to_expert, n_blocks
 
 
@@ -82,11 +92,14 @@ def blockwise_mm(hidden: torch.Tensor, expert_affinities: torch.Tensor,
 
     nb = block_to_expert.numel()
     xb = x.reshape(nb, block_size, H)
-    out_b = torch.empty_like(xb)
+    out_b = torch.zeros_like(xb)
     from .. import ops as _ops
 
-    for e in torch.unique(block_to_expert).tolist():
-        sel = (block_to_expert == e).nonzero().reshape(-1)
+    # SkipMode: blocks with no valid token (padding-only) never touch the
+    # GEMMs (reference K5 SkipMode DMA/compute skip)
+    nonempty = (token_position_to_id.reshape(nb, block_size) >= 0).any(1)
+    for e in torch.unique(block_to_expert[nonempty]).tolist():
+        sel = ((block_to_expert == e) & nonempty).nonzero().reshape(-1)
         xe = xb[sel].reshape(-1, H)
         gu = xe @ gate_up_w[e]
         if glu:

@@ -160,6 +160,33 @@ class ExpertMLPs(nn.Module):
                        gathered * aff[keep].unsqueeze(-1).float())
         return out.to(hidden.dtype)
 
+    def forward_selective(self, hidden, expert_affinities, expert_index,
+                          normalize_top_k_affinities: bool = False):
+        """Selective loading (reference expert_mlps_v2.py:595-689): at
+        token generation with few tokens, gather ONLY the top-k experts'
+        weight slices per token (T*k slices instead of all E experts) and
+        run them as one batched GEMM — vectorized over (token, k) pairs
+        instead of the reference's per-token python loop."""
+        T = hidden.shape[0]
+        k = self.top_k
+        idx = expert_index.reshape(-1)  # (T*k,) local expert ids
+        aff = expert_affinities.gather(-1, expert_index)  # (T,k)
+        if normalize_top_k_affinities:
+            aff = torch.nn.functional.normalize(aff, p=1.0, dim=1)
+
+        x = hidden.unsqueeze(1).expand(T, k, self.hidden_size)
+        x = x.reshape(T * k, 1, self.hidden_size)
+        gu = self.gate_up_proj(x, expert_indices=idx)  # (T*k, 1, 2I/tp)
+        if self.glu_mlp:
+            I = gu.shape[-1] // 2
+            act = ops.swiglu(gu) if gu.dtype == torch.bfloat16 and gu.is_cuda \
+                else torch.nn.functional.silu(gu[..., :I]) * gu[..., I:]
+        else:
+            act = torch.nn.functional.gelu(gu)
+        out = self.down_proj(act, expert_indices=idx)  # (T*k, 1, H)
+        out = out.reshape(T, k, self.hidden_size).float()
+        return (out * aff.unsqueeze(-1).float()).sum(dim=1).to(hidden.dtype)
+
     def forward_blockwise(self, hidden, expert_affinities, expert_index):
         """No-drop blockwise strategy (reference expert_mlps_v2.py:691 +
         blockwise.py K4): fixed-size expert blocks, grouped GEMMs."""
@@ -169,8 +196,10 @@ class ExpertMLPs(nn.Module):
         assert self.ep_size == 1, "blockwise with EP: use capacity_factor"
         block_size = min(DEFAULT_BLOCK_SIZE,
                          max(32, hidden.shape[0] // 4))
-        tpi, b2e, _ = compute_block_indices(expert_index, self.num_experts,
-                                            block_size)
+        tpi, b2e, _ = compute_block_indices(
+            expert_index, self.num_experts, block_size,
+            max_blocks_per_expert=getattr(self, "blockwise_dropping_blocks",
+                                          None))
         return blockwise_mm(hidden, expert_affinities,
                             self.gate_up_proj.weight.to(hidden.dtype),
                             self.down_proj.weight.to(hidden.dtype),
@@ -183,7 +212,8 @@ class ExpertMLPs(nn.Module):
         batches take the fused HIP decode path (K9): per-expert slot
         blocks, weights streamed once, gather/SwiGLU/scatter fused."""
         hidden = copy_to_tensor_model_parallel_region(hidden)
-        if (not isinstance(hidden, torch.fx.Proxy) and not self.training
+        traced = isinstance(hidden, torch.fx.Proxy)
+        if (not traced and not self.training
                 and self.glu_mlp and self.ep_size == 1 and hidden.is_cuda
                 and hidden.dtype == torch.bfloat16
                 and self.gate_up_proj.weight.dtype == torch.bfloat16
@@ -191,6 +221,13 @@ class ExpertMLPs(nn.Module):
             return ops.moe_decode_glu(hidden, self.gate_up_proj.weight,
                                       self.down_proj.weight,
                                       expert_affinities, expert_index)
+        if (not traced and not self.training and self.ep_size == 1
+                and hidden.shape[0] * self.top_k
+                < self.SELECTIVE_LOADING_THRESHOLD * self.num_experts):
+            # token generation with few expert hits: load only the top-k
+            # experts' weights (reference dispatch :1407-1499)
+            return self.forward_selective(hidden, expert_affinities,
+                                          expert_index)
         if self.capacity_factor is not None and self.capacity_factor > 0:
             return self.forward_capacity_factor(hidden, expert_affinities,
                                                 expert_index)

@@ -79,10 +79,15 @@ class ExpertFusedColumnParallelLinear(_ExpertFusedLinearBase):
         self._init_expertwise(self.weight, output_size, input_size, 2,
                               init_method, dtype, stride=stride)
 
-    def forward(self, x):
+    def forward(self, x, expert_indices=None):
         # x (E_local, C, H) -> (E_local, C, I/tp); grad_input all-reduced
-        # over TP in backward by the caller's input copy (delayed reduce)
-        return torch.bmm(x, self.weight.to(x.dtype))
+        # over TP in backward by the caller's input copy (delayed reduce).
+        # expert_indices (N,): compute only those experts' slices, x (N,C,H)
+        # (reference moe_parallel_layers.py:263-276 selective loading)
+        w = self.weight
+        if expert_indices is not None:
+            w = w.index_select(0, expert_indices.reshape(-1))
+        return torch.bmm(x, w.to(x.dtype))
 
 
 class ExpertFusedRowParallelLinear(_ExpertFusedLinearBase):
@@ -104,8 +109,11 @@ class ExpertFusedRowParallelLinear(_ExpertFusedLinearBase):
         self._init_expertwise(self.weight, output_size, input_size, 1,
                               init_method, dtype)
 
-    def forward(self, x):
-        out = torch.bmm(x, self.weight.to(x.dtype))
+    def forward(self, x, expert_indices=None):
+        w = self.weight
+        if expert_indices is not None:
+            w = w.index_select(0, expert_indices.reshape(-1))
+        out = torch.bmm(x, w.to(x.dtype))
         if self.reduce_output:
             from ..parallel.mappings import reduce_from_tensor_model_parallel_region
 
