@@ -92,26 +92,43 @@ def blockwise_mm(hidden: torch.Tensor, expert_affinities: torch.Tensor,
 
     nb = block_to_expert.numel()
     xb = x.reshape(nb, block_size, H)
-    out_b = torch.zeros_like(xb)
     from .. import ops as _ops
 
-    # SkipMode: blocks with no valid token (padding-only) never touch the
-    # GEMMs (reference K5 SkipMode DMA/compute skip)
-    nonempty = (token_position_to_id.reshape(nb, block_size) >= 0).any(1)
-    for e in torch.unique(block_to_expert[nonempty]).tolist():
-        sel = ((block_to_expert == e) & nonempty).nonzero().reshape(-1)
-        xe = xb[sel].reshape(-1, H)
-        gu = xe @ gate_up_w[e]
+    def _act_fn(gu):
         if glu:
             if gu.is_cuda and gu.dtype == torch.bfloat16 and \
                     _ops.is_available():
-                act = _ops.swiglu(gu)  # fused HIP silu(gate)*up
-            else:
-                I = gu.shape[-1] // 2
-                act = torch.nn.functional.silu(gu[..., :I]) * gu[..., I:]
-        else:
-            act = torch.nn.functional.gelu(gu)
-        out_b[sel] = (act @ down_w[e]).reshape(-1, block_size, H)
+                return _ops.swiglu(gu)  # fused HIP silu(gate)*up
+            I = gu.shape[-1] // 2
+            return torch.nn.functional.silu(gu[..., :I]) * gu[..., I:]
+        return torch.nn.functional.gelu(gu)
+
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and hasattr(torch, "_grouped_mm")
+            and gate_up_w.dtype == torch.bfloat16
+            # grouped-GEMM needs expert-contiguous rows; a static
+            # num_blocks budget appends padding blocks of expert 0 at the
+            # END, breaking monotonicity -> use the per-expert loop there
+            and bool((block_to_expert.diff() >= 0).all())):
+        # ONE grouped GEMM per projection over all experts (hipBLASLt
+        # grouped kernels; the blockwise layout is already expert-sorted
+        # and block-padded, so group r = rows [offs[r-1], offs[r]) )
+        E_w = gate_up_w.shape[0]
+        counts = torch.bincount(block_to_expert, minlength=E_w)
+        offs = torch.cumsum(counts * block_size, 0).to(torch.int32)
+        gu = torch._grouped_mm(x, gate_up_w, offs=offs)
+        out_flat2 = torch._grouped_mm(_act_fn(gu), down_w, offs=offs)
+        out_b = out_flat2.reshape(nb, block_size, H)
+    else:
+        out_b = torch.zeros_like(xb)
+        # SkipMode: blocks with no valid token (padding-only) never touch
+        # the GEMMs (reference K5 SkipMode DMA/compute skip)
+        nonempty = (token_position_to_id.reshape(nb, block_size) >= 0).any(1)
+        for e in torch.unique(block_to_expert[nonempty]).tolist():
+            sel = ((block_to_expert == e) & nonempty).nonzero().reshape(-1)
+            xe = xb[sel].reshape(-1, H)
+            act = _act_fn(xe @ gate_up_w[e])
+            out_b[sel] = (act @ down_w[e]).reshape(-1, block_size, H)
 
     # affinity of (token, owning expert of its slot)
     aff = expert_affinities[safe_ids.clamp(max=T - 1),

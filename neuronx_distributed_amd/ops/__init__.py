@@ -155,6 +155,15 @@ def _rope_torch(x, cos, sin, sign=1.0):
 class _RoPEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos, sin, pos_offset):
+        # bounds-check HERE: an out-of-range position would be an
+        # out-of-bounds global read inside the HIP kernel (GPU fault)
+        S = q.shape[1]
+        off = int(pos_offset) if not isinstance(pos_offset, torch.Tensor) \
+            else 0  # device-tensor positions are decode-path (S == 1)
+        if off + S > cos.shape[0]:
+            raise ValueError(
+                f"RoPE table too short: seq {off}+{S} > table "
+                f"{cos.shape[0]} (raise max_position_embeddings)")
         ctx.pos_offset = pos_offset
         ctx.save_for_backward(cos, sin)
         if q.is_cuda:
@@ -488,6 +497,8 @@ def decode_attn_step(q2: torch.Tensor, k2: torch.Tensor, v2: torch.Tensor,
     Smax = kcache.shape[2]
     assert Hq // Hkv in (1, 2, 4, 8), "GQA rep must be 1/2/4/8"
     assert cos.dtype == torch.float32 and pos_t.dtype == torch.int64
+    assert cos.shape[0] >= Smax, (
+        f"RoPE table ({cos.shape[0]}) shorter than KV cache ({Smax})")
     assert q2.stride(1) == 1 and k2.stride(1) == 1 and v2.stride(1) == 1
     assert k2.stride(0) == v2.stride(0)
     out = torch.empty(B, Hq * 128, dtype=torch.bfloat16, device=q2.device)

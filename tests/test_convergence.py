@@ -99,7 +99,8 @@ def test_convergence_gpu_bf16_tracks_fp32():
 
     cfg = get_config("tiny", num_hidden_layers=2, hidden_size=256,
                      intermediate_size=512, num_attention_heads=2,
-                     num_key_value_heads=2, head_dim=128)
+                     num_key_value_heads=2, head_dim_override=128,
+                     max_position_embeddings=512)
     steps = 25
 
     def build(dtype):

@@ -211,6 +211,54 @@ def test_shared_experts_replicated_tp2():
 
 
 @pytest.mark.gpu
+def test_blockwise_grouped_mm_matches_loop(monkeypatch):
+    """The single grouped-GEMM blockwise path (torch._grouped_mm over
+    hipBLASLt grouped kernels) must match the per-expert-loop path in
+    forward AND gradients."""
+    from neuronx_distributed_amd.moe import blockwise as bw
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    import os
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29764")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+
+    T, H, I, E, k, B = 512, 256, 192, 8, 2, 64
+    dev = "cuda"
+    torch.manual_seed(0)
+    x = torch.randn(T, H, device=dev, dtype=torch.bfloat16)
+    gu_w = (torch.randn(E, H, 2 * I, device=dev, dtype=torch.bfloat16)
+            * 0.05).requires_grad_(True)
+    d_w = (torch.randn(E, I, H, device=dev, dtype=torch.bfloat16)
+           * 0.05).requires_grad_(True)
+    vals, idx = torch.topk(torch.softmax(torch.randn(T, E, device=dev), -1),
+                           k, -1)
+    aff = torch.zeros(T, E, device=dev).scatter(-1, idx, vals)
+    tpi, b2e, _ = bw.compute_block_indices(idx, E, B)
+
+    xg = x.clone().requires_grad_(True)
+    out_g = bw.blockwise_mm(xg, aff, gu_w, d_w, tpi, b2e, idx, B)
+    out_g.float().pow(2).sum().backward()
+    grads_g = (xg.grad.clone(), gu_w.grad.clone(), d_w.grad.clone())
+
+    gu_w.grad = None
+    d_w.grad = None
+    monkeypatch.delattr(torch, "_grouped_mm")  # force the loop path
+    xl = x.clone().requires_grad_(True)
+    out_l = bw.blockwise_mm(xl, aff, gu_w, d_w, tpi, b2e, idx, B)
+    out_l.float().pow(2).sum().backward()
+
+    assert (out_g.float() - out_l.float()).abs().max() < 2e-2
+    for a, b in zip(grads_g, (xl.grad, gu_w.grad, d_w.grad)):
+        rel = (a.float() - b.float()).norm() / (b.float().norm() + 1e-9)
+        assert rel < 2e-2, rel
+
+
+@pytest.mark.gpu
 def test_moe_fused_tkg_gpu_fused_matches_unfused():
     from neuronx_distributed_amd.models.llama import RMSNorm
     from neuronx_distributed_amd.moe import ExpertMLPs, RouterTopK
