@@ -67,9 +67,8 @@ class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
         ctx.eps = eps
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0:
             lib = _require_lib()
-            assert x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0
             x2 = x.contiguous()
             rows = x2.numel() // x2.shape[-1]
             out = torch.empty_like(x2)
@@ -90,7 +89,7 @@ class _RMSNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, rstd = ctx.saved_tensors
         H = x.shape[-1]
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16 and H % 8 == 0:
             lib = _require_lib()
             dy2 = dy.contiguous()
             rows = x.numel() // H
@@ -166,7 +165,7 @@ class _RoPEFn(torch.autograd.Function):
                 f"{cos.shape[0]} (raise max_position_embeddings)")
         ctx.pos_offset = pos_offset
         ctx.save_for_backward(cos, sin)
-        if q.is_cuda:
+        if q.is_cuda and q.dtype == torch.bfloat16:
             lib = _require_lib()
             B, S, Hq, D = q.shape
             Hk = k.shape[2]
@@ -188,7 +187,7 @@ class _RoPEFn(torch.autograd.Function):
     def backward(ctx, dq, dk):
         cos, sin = ctx.saved_tensors
         off = ctx.pos_offset
-        if dq.is_cuda:
+        if dq.is_cuda and dq.dtype == torch.bfloat16:
             lib = _require_lib()
             B, S, Hq, D = dq.shape
             Hk = dk.shape[2]
