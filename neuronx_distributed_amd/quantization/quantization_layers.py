@@ -24,7 +24,7 @@ from ..parallel.utils import set_tensor_model_parallel_attributes
 from .quantization_config import (QuantizationConfig, QuantizationType,
                                   QuantizedDtype)
 from .quantization_utils import (dequantize, fp8_scaled_linear,
-                                 quantize_symmetric)
+                                 quantize_symmetric, unpack_x4)
 
 
 class _QuantizedParallelLinearBase(BaseParallelLinear):
@@ -34,10 +34,17 @@ class _QuantizedParallelLinearBase(BaseParallelLinear):
         path (the all-gather is fused into the bf16 linear there)."""
         cfg = self.quantization_config
         return (cfg.quantize_activation
-                and cfg.quantized_dtype == QuantizedDtype.F8E4M3
+                and cfg.quantized_dtype in (QuantizedDtype.F8E4M3,
+                                            QuantizedDtype.F8E4M3FN_X4)
                 and input_.is_cuda
                 and not self.sequence_parallel_enabled
                 and not input_.requires_grad)
+
+    def _fp8_weight(self):
+        """fp8 view of the stored weight; packed X4 unpacks zero-copy."""
+        dt = self.quantization_config.quantized_dtype
+        return self.weight if dt.packed_count == 1 else \
+            unpack_x4(self.weight, dt)
 
     def _make_scale(self, out_rows: int, cfg: QuantizationConfig, shard: bool):
         if cfg.quantization_type == QuantizationType.PER_TENSOR_SYMMETRIC:
@@ -78,14 +85,15 @@ class QuantizedColumnParallel(_QuantizedParallelLinearBase):
 
     def forward(self, input_):
         if self._use_fp8_mm(input_):
-            out = fp8_scaled_linear(input_, self.weight, self.scale,
+            out = fp8_scaled_linear(input_, self._fp8_weight(), self.scale,
                                     self.compute_dtype)
             if self.bias is not None:
                 out = out + self.bias
             if self.gather_output:
                 out = gather_from_tensor_model_parallel_region(out)
             return out
-        w = dequantize(self.weight, self.scale, self.compute_dtype)
+        w = dequantize(self.weight, self.scale, self.compute_dtype,
+                       self.quantization_config.quantized_dtype)
         if not self.sequence_parallel_enabled and \
                 ps.get_tensor_model_parallel_size() > 1:
             input_parallel = input_
@@ -111,10 +119,11 @@ class QuantizedRowParallel(_QuantizedParallelLinearBase):
         if not self.input_is_parallel:
             input_ = scatter_to_tensor_model_parallel_region(input_)
         if self._use_fp8_mm(input_):
-            out = fp8_scaled_linear(input_, self.weight, self.scale,
+            out = fp8_scaled_linear(input_, self._fp8_weight(), self.scale,
                                     self.compute_dtype)
         else:
-            w = dequantize(self.weight, self.scale, self.compute_dtype)
+            w = dequantize(self.weight, self.scale, self.compute_dtype,
+                           self.quantization_config.quantized_dtype)
             out = linear_with_async_allreduce(
                 input_, w, None, async_grad_allreduce=False,
                 sequence_parallel_enabled=False)
@@ -128,3 +137,54 @@ class QuantizedRowParallel(_QuantizedParallelLinearBase):
         if self.bias is not None:
             out = out + self.bias
         return out
+
+
+class QuantizedParallelLinearLayerStateDictAdaptor:
+    """Adapts externally-quantized (HF-style) checkpoints to the layout the
+    quantized parallel layers expect (reference quantization_layers.py:356
+    QuantizedParallelLinearLayerStateDictAdaptor): handles both plain
+    ``{prefix}weight`` entries and torch.ao ``_packed_params`` entries
+    produced by dynamic int8 quantization."""
+
+    @staticmethod
+    def get_weight_from_state_dict(prefix: str, state_dict: dict):
+        if prefix + "weight" in state_dict:
+            return state_dict[prefix + "weight"]
+        if prefix + "_packed_params.dtype" in state_dict:
+            qw = state_dict[prefix + "_packed_params._packed_params"][0]
+            return torch.int_repr(qw)
+        raise RuntimeError(f"Cannot find {prefix}weight in the state_dict")
+
+    @staticmethod
+    def set_weight_to_state_dict(prefix: str, tensor, state_dict: dict):
+        if prefix + "weight" in state_dict:
+            state_dict[prefix + "weight"] = tensor
+        elif prefix + "_packed_params.dtype" in state_dict:
+            packed = list(state_dict[prefix + "_packed_params._packed_params"])
+            packed[0] = tensor
+            state_dict[prefix + "_packed_params._packed_params"] = \
+                tuple(packed)
+        else:
+            raise RuntimeError(
+                f"Cannot find {prefix}weight in the state_dict")
+
+    @staticmethod
+    def get_scale_from_state_dict(prefix: str, state_dict: dict):
+        for key in ("weight_scale", "scale"):
+            if prefix + key in state_dict:
+                return state_dict[prefix + key]
+        if prefix + "_packed_params.dtype" in state_dict:
+            qw = state_dict[prefix + "_packed_params._packed_params"][0]
+            if qw.qscheme() in (torch.per_tensor_affine,
+                                torch.per_tensor_symmetric):
+                return torch.tensor([qw.q_scale()], dtype=torch.float32)
+            return qw.q_per_channel_scales().float()
+        raise RuntimeError(f"Cannot find {prefix}weight_scale in state_dict")
+
+    @staticmethod
+    def get_bias_from_state_dict(prefix: str, state_dict: dict):
+        if prefix + "bias" in state_dict:
+            return state_dict[prefix + "bias"]
+        if prefix + "_packed_params.dtype" in state_dict:
+            return state_dict[prefix + "_packed_params._packed_params"][1]
+        return None

@@ -7,9 +7,56 @@ from .quantization_config import (QuantizationConfig, QuantizationType,
                                   QuantizedDtype)
 
 
+# fp4 e2m1 value grid (sign x {0, .5, 1, 1.5, 2, 3, 4, 6}); nibble =
+# sign<<3 | code
+_FP4_GRID = torch.tensor([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0])
+
+
+def _fp4_encode(x: torch.Tensor) -> torch.Tensor:
+    """fp32 -> nearest e2m1 nibble code (uint8 in [0, 15])."""
+    grid = _FP4_GRID.to(x.device)
+    sign = (x < 0).to(torch.uint8)
+    mag = x.abs().clamp(max=6.0)
+    code = (mag.unsqueeze(-1) - grid).abs().argmin(dim=-1).to(torch.uint8)
+    return (sign << 3) | code
+
+
+def _fp4_decode(nib: torch.Tensor) -> torch.Tensor:
+    grid = _FP4_GRID.to(nib.device)
+    mag = grid[(nib & 0x7).long()]
+    return torch.where((nib >> 3) > 0, -mag, mag)
+
+
+def pack_x4(q: torch.Tensor, dtype: QuantizedDtype) -> torch.Tensor:
+    """Pack 4 quantized lanes along the LAST dim (reference *_X4 storage
+    formats, quantization_config.py:100-125): fp8 -> uint32 (byte
+    reinterpret), fp4 nibbles -> uint16."""
+    assert q.shape[-1] % 4 == 0, "last dim must be divisible by 4"
+    if dtype == QuantizedDtype.F4E2M1FN_X4:
+        nib = q.to(torch.int32).reshape(*q.shape[:-1], -1, 4)
+        packed = (nib[..., 0] | (nib[..., 1] << 4) | (nib[..., 2] << 8)
+                  | (nib[..., 3] << 12))
+        return packed.to(torch.int32).to(torch.uint16)
+    return q.contiguous().view(torch.uint8).reshape(
+        *q.shape[:-1], -1, 4).contiguous().view(torch.uint32).squeeze(-1)
+
+
+def unpack_x4(packed: torch.Tensor, dtype: QuantizedDtype) -> torch.Tensor:
+    """Inverse of pack_x4; fp8 unpack is a zero-copy byte reinterpret."""
+    if dtype == QuantizedDtype.F4E2M1FN_X4:
+        p = packed.to(torch.int32)
+        nibs = torch.stack([(p >> s) & 0xF for s in (0, 4, 8, 12)], dim=-1)
+        return nibs.reshape(*packed.shape[:-1], -1).to(torch.uint8)
+    out = packed.unsqueeze(-1).contiguous().view(torch.uint8)
+    return out.reshape(*packed.shape[:-1], -1).view(
+        dtype.unpacked.torch_dtype)
+
+
 def quantize_symmetric(weight: torch.Tensor, cfg: QuantizationConfig):
-    """Returns (q_weight, scale) with weight ~= q_weight * scale."""
-    qmax = cfg.quantized_dtype.max_value
+    """Returns (q_weight, scale) with weight ~= q_weight * scale.  Packed
+    *_X4 dtypes return the packed storage tensor."""
+    dt = cfg.quantized_dtype
+    qmax = dt.max_value
     if cfg.quantization_type == QuantizationType.PER_TENSOR_SYMMETRIC:
         amax = weight.abs().max().clamp(min=1e-8)
         scale = (amax / qmax).float()
@@ -19,15 +66,26 @@ def quantize_symmetric(weight: torch.Tensor, cfg: QuantizationConfig):
         amax = weight.abs().amax(dim=dims, keepdim=True).clamp(min=1e-8)
         scale = (amax / qmax).float()
     q = (weight.float() / scale)
-    if cfg.quantized_dtype == QuantizedDtype.INT8:
+    if dt == QuantizedDtype.INT8:
         q = q.round().clamp(-qmax, qmax).to(torch.int8)
+    elif dt == QuantizedDtype.F4E2M1FN_X4:
+        q = pack_x4(_fp4_encode(q.clamp(-qmax, qmax)), dt)
+    elif dt.packed_count == 4:
+        q = pack_x4(q.clamp(-qmax, qmax).to(dt.unpacked.torch_dtype), dt)
     else:
-        q = q.clamp(-qmax, qmax).to(cfg.quantized_dtype.torch_dtype)
+        q = q.clamp(-qmax, qmax).to(dt.torch_dtype)
     return q, scale
 
 
 def dequantize(q_weight: torch.Tensor, scale: torch.Tensor,
-               dtype=torch.bfloat16) -> torch.Tensor:
+               dtype=torch.bfloat16,
+               quantized_dtype: QuantizedDtype = None) -> torch.Tensor:
+    if quantized_dtype is not None and quantized_dtype.packed_count == 4:
+        if quantized_dtype == QuantizedDtype.F4E2M1FN_X4:
+            vals = _fp4_decode(unpack_x4(q_weight, quantized_dtype))
+        else:
+            vals = unpack_x4(q_weight, quantized_dtype).float()
+        return (vals.float() * scale.float()).to(dtype)
     return (q_weight.float() * scale.float()).to(dtype)
 
 
