@@ -1,10 +1,43 @@
 """Token shuffling for MoE load balance (reference
 modules/moe/token_shuffling.py:64-117): random permutation + all-to-all
-over the token-shuffle group before routing, inverted after."""
+over the token-shuffle group before routing, inverted after (BASE Layers,
+arXiv:2103.16716 — tokens within one worker's sequence are correlated, so
+shuffling spreads them across workers for better expert balance).
+
+The all-to-all is AUTOGRAD-AWARE (reference _AllToAllForTokenShuffle):
+token shuffling runs in training, so gradients must ride the inverse
+all-to-all back to the producing rank.
+"""
 
 import torch
 
-from ..parallel import comm, parallel_state as ps
+from ..parallel import parallel_state as ps
+from ..parallel.mappings import all_to_all as _autograd_all_to_all
+from ..parallel.mappings import (
+    gather_from_sequence_parallel_region,
+    scatter_to_sequence_parallel_region,
+)
+
+
+def _shuffle_all_to_all(hidden: torch.Tensor) -> torch.Tensor:
+    if "token_shuffle" not in ps._GROUPS or \
+            ps.get_token_shuffle_group_size() == 1:
+        return hidden
+    return _autograd_all_to_all(hidden, 0, 0, group_name="token_shuffle")
+
+
+def all_to_all_for_shuffle(hidden: torch.Tensor,
+                           input_is_sequence_parallel: bool = True):
+    """reference token_shuffling.py all_to_all_for_shuffle: non-SP inputs
+    are scattered to sequence-parallel form around the exchange so each
+    rank trades equal slices."""
+    if not input_is_sequence_parallel:
+        hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
+    hidden = _shuffle_all_to_all(hidden)
+    if not input_is_sequence_parallel:
+        hidden = gather_from_sequence_parallel_region(
+            hidden, seq_dim=0, to_model_parallel=False)
+    return hidden
 
 
 def token_shuffle(hidden: torch.Tensor, seed: int = None):
@@ -17,15 +50,12 @@ def token_shuffle(hidden: torch.Tensor, seed: int = None):
     else:
         perm = torch.randperm(T, device=hidden.device)
     h = hidden[perm]
-    if "token_shuffle" in ps._GROUPS:
-        h = comm.all_to_all(h, 0, 0, group=ps.get_group_info("token_shuffle"))
-    return h, perm
+    return _shuffle_all_to_all(h), perm
 
 
 def token_unshuffle(hidden: torch.Tensor, perm: torch.Tensor):
-    if "token_shuffle" in ps._GROUPS:
-        hidden = comm.all_to_all(hidden, 0, 0,
-                                 group=ps.get_group_info("token_shuffle"))
+    # the equal-split all-to-all is self-inverse
+    hidden = _shuffle_all_to_all(hidden)
     inv = torch.empty_like(perm)
     inv[perm] = torch.arange(perm.numel(), device=perm.device)
     return hidden[inv]

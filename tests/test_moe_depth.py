@@ -290,3 +290,61 @@ def test_moe_fused_tkg_gpu_fused_matches_unfused():
     ref = tkg._unfused(x)
     err = (out.float() - ref.float()).abs().max()
     assert err < 3e-2, err
+
+
+def _shuffle_grad_worker(rank, world):
+    """Token shuffling must be autograd-aware: gradients ride the inverse
+    all-to-all back, so shuffle -> f -> unshuffle has exact identity-like
+    gradients (reference _AllToAllForTokenShuffle)."""
+    from neuronx_distributed_amd.moe.token_shuffling import (token_shuffle,
+                                                             token_unshuffle)
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 token_shuffle_group_size=world)
+    torch.manual_seed(10 + rank)
+    x = torch.randn(8, 4, requires_grad=True)
+    h, perm = token_shuffle(x, seed=42)
+    out = token_unshuffle(h * 2.0, perm)
+    torch.manual_seed(20 + rank)
+    g = torch.randn_like(out)
+    out.backward(g)
+    assert torch.allclose(x.grad, 2.0 * g, atol=1e-6), \
+        (x.grad - 2.0 * g).abs().max()
+    return 0.0
+
+
+def test_token_shuffle_gradients_world2():
+    run_distributed(_shuffle_grad_worker, world_size=2)
+
+
+def test_sampler_cdf_multinomial_and_medusa():
+    from neuronx_distributed_amd.utils.sampling import Sampler
+
+    torch.manual_seed(0)
+    # a concentrated distribution: CDF sampling must pick inside top mass
+    probs = torch.tensor([[0.9, 0.05, 0.03, 0.02]])
+    picks = [int(Sampler._multinomial_cdf(probs)) for _ in range(200)]
+    assert picks.count(0) > 140  # ~90%
+    assert max(picks) <= 3
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    if not ps.model_parallel_is_initialized():
+        import torch.distributed as dist
+        import os
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29767")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        if not dist.is_initialized():
+            dist.init_process_group("gloo", rank=0, world_size=1)
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    logits = torch.randn(3, 64)
+    s = Sampler(do_sample=True, top_k=8, on_device_multinomial=True)
+    ids = s(logits)
+    assert ids.shape == (3,)
+    m = Sampler(do_sample=True, top_k=5, return_topk_indices=True)
+    topk_idx = m(logits)
+    assert topk_idx.shape == (3, 5)
+    ref = logits.topk(5, dim=-1).indices
+    assert torch.equal(topk_idx, ref)
