@@ -133,7 +133,8 @@ def main():
             "decode_tokens_per_s": round(args.batch * steps / decode_s, 1),
             "dtype": "bf16", "data": "synthetic",
         }), flush=True)
-    if on_gpu and os.environ.get("NXDA_TUNE", "0") == "1":
+    if on_gpu and os.environ.get("NXDA_TUNE", "0") == "1" and \
+            hasattr(torch.cuda.tunable, "write_file"):
         torch.cuda.tunable.write_file()
     dist.destroy_process_group()
 
