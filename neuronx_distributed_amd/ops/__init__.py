@@ -500,9 +500,18 @@ def decode_attn_step(q2: torch.Tensor, k2: torch.Tensor, v2: torch.Tensor,
         f"RoPE table ({cos.shape[0]}) shorter than KV cache ({Smax})")
     assert q2.stride(1) == 1 and k2.stride(1) == 1 and v2.stride(1) == 1
     assert k2.stride(0) == v2.stride(0)
+    rep = Hq // Hkv
+    # split-KV workspace (4 splits; see decode_attn.hip) — cached-alloc
+    # tensors with static shapes, so the pair of launches is
+    # hipGraph-capturable
+    part_o = torch.empty(B * Hkv * 4, rep, 128, dtype=torch.float32,
+                         device=q2.device)
+    part_ml = torch.empty(B * Hkv * 4, rep, 2, dtype=torch.float32,
+                          device=q2.device)
     out = torch.empty(B, Hq * 128, dtype=torch.bfloat16, device=q2.device)
     lib.decode_attn(_ptr(q2), _ptr(k2), _ptr(v2), _ptr(kcache), _ptr(vcache),
-                    _ptr(cos), _ptr(sin), _ptr(pos_t), _ptr(out), B, Hq, Hkv,
+                    _ptr(cos), _ptr(sin), _ptr(pos_t), _ptr(part_o),
+                    _ptr(part_ml), _ptr(out), B, Hq, Hkv,
                     Smax, ctypes.c_float(scale), int(q2.stride(0)),
                     int(k2.stride(0)), _stream())
     return out
