@@ -110,30 +110,89 @@ def _initialize_affine_weight(
 # ---------------------------------------------------------------------------
 
 def _sp_overlapped_linear(input_, weight, bias, tp_info):
-    """Chunked SP forward (NXDA_SP_OVERLAP=1): instead of one blocking
-    all-gather followed by one GEMM, post per-source-rank async broadcasts
-    and GEMM each sequence chunk as it lands — the later chunks' transfers
-    overlap the earlier chunks' MFMA work (SURVEY §7 hard-parts: fused SP
-    producer; chunk granularity = 1/tp of the sequence)."""
+    """Ring-pipelined SP forward (NXDA_SP_OVERLAP=1): instead of one
+    blocking all-gather followed by one GEMM, the TP ranks rotate their
+    sequence chunks around the ring with batched isend/irecv while each
+    in-hand chunk is GEMMed — every transfer except the first overlaps
+    MFMA work, and on xGMI each hop is a single point-to-point link
+    (SURVEY §7 hard-parts: fused SP producer; reference
+    layers_utils.py:91-103 async-overlap semantics)."""
     world = tp_info.size
     me = tp_info.rank_in_group(dist.get_rank())
     ranks = tp_info.ranks_of(dist.get_rank())
-    bufs = []
-    works = []
+    nxt = ranks[(me + 1) % world]
+    prv = ranks[(me - 1) % world]
+    S = input_.shape[0]
     src = input_.contiguous()
-    for r in range(world):
-        buf = src if r == me else torch.empty_like(src)
-        bufs.append(buf)
-        works.append(dist.broadcast(buf, src=ranks[r], group=tp_info.group,
-                                    async_op=True))
-    outs = []
-    for r in range(world):
-        works[r].wait()
-        outs.append(F.linear(bufs[r], weight))
-    out = torch.cat(outs, dim=0)
+    cur = src
+    recv_buf = torch.empty_like(src)
+    # src is SAVED for backward — it may be sent from but never received
+    # into, so the rotation uses a second scratch buffer in its place
+    spare = torch.empty_like(src)
+    out = torch.empty((world * S,) + tuple(input_.shape[1:-1])
+                      + (weight.shape[0],),
+                      dtype=input_.dtype, device=input_.device)
+    for t in range(world):
+        owner = (me - t) % world
+        works = []
+        if t < world - 1:
+            ops_ = [dist.P2POp(dist.isend, cur, nxt, group=tp_info.group),
+                    dist.P2POp(dist.irecv, recv_buf, prv,
+                               group=tp_info.group)]
+            works = dist.batch_isend_irecv(ops_)
+        out[owner * S:(owner + 1) * S] = F.linear(cur, weight)
+        for w in works:
+            w.wait()
+        if t < world - 1:
+            nxt_recv = spare if cur is src else cur
+            cur = recv_buf
+            recv_buf = nxt_recv
     if bias is not None:
         out = out + bias
     return out
+
+
+class _RPLOverlappedLinearRS(torch.autograd.Function):
+    """RowParallel forward GEMM with the SP reduce-scatter CHUNKED as a
+    consumer: the partial output for destination rank c is GEMMed then
+    reduced (async) toward c while the next chunk's GEMM runs — the
+    reduce-scatter hides under MFMA instead of trailing it (SURVEY §7
+    hard-parts: fused SP consumer)."""
+
+    @staticmethod
+    def forward(ctx, input_parallel, weight):
+        tp_info = ps.get_group_info("tp")
+        world = tp_info.size
+        me = tp_info.rank_in_group(dist.get_rank())
+        ranks = tp_info.ranks_of(dist.get_rank())
+        ctx.save_for_backward(input_parallel, weight)
+        x = input_parallel.contiguous()
+        assert x.shape[0] % world == 0
+        Sc = x.shape[0] // world
+        handles = []
+        chunks = []
+        mine = None
+        for c in range(world):
+            y = F.linear(x[c * Sc:(c + 1) * Sc], weight)
+            chunks.append(y)  # keep alive until the reduce drains
+            handles.append(dist.reduce(y, dst=ranks[c],
+                                       group=tp_info.group, async_op=True))
+            if c == me:
+                mine = y
+        for h in handles:
+            h.wait()
+        return mine.clone()
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        input_parallel, weight = ctx.saved_tensors
+        tp_info = ps.get_group_info("tp")
+        g = comm.all_gather(grad_output.contiguous(), dim=0, group=tp_info)
+        grad_input = g.matmul(weight)
+        go2d = g.reshape(-1, g.shape[-1])
+        in2d = input_parallel.reshape(-1, input_parallel.shape[-1])
+        grad_weight = go2d.t().matmul(in2d)
+        return grad_input, grad_weight
 
 
 class LinearWithAsyncCommunication(torch.autograd.Function):
@@ -403,6 +462,13 @@ class RowParallelLinear(BaseParallelLinear):
         else:
             input_parallel = scatter_to_tensor_model_parallel_region(input_)
 
+        if (self.sequence_parallel_enabled and self.reduce_output
+                and os.environ.get("NXDA_SP_OVERLAP", "0") == "1"
+                and ps.get_tensor_model_parallel_size() > 1
+                and not ps.is_aot_mode()
+                and not isinstance(input_parallel, torch.fx.Proxy)):
+            return self._forward_sp_overlapped(input_parallel)
+
         output_parallel = linear_with_async_allreduce(
             input_parallel, self.weight, None,
             async_grad_allreduce=False, sequence_parallel_enabled=False)
@@ -414,6 +480,15 @@ class RowParallelLinear(BaseParallelLinear):
                                                                 seq_dim=0)
         else:
             output = reduce_from_tensor_model_parallel_region(output_parallel)
+
+        if self.skip_bias_add:
+            return output, self.bias
+        if self.bias is not None:
+            output = output + self.bias
+        return output
+
+    def _forward_sp_overlapped(self, input_parallel):
+        output = _RPLOverlappedLinearRS.apply(input_parallel, self.weight)
 
         if self.skip_bias_add:
             return output, self.bias
