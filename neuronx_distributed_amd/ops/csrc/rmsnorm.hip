@@ -133,7 +133,13 @@ add_rmsnorm_fwd_kernel(const short* __restrict__ res_in,
 
   int nvec = H >> 3;
   float ss = 0.f;
-  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+  // register-cache the summed row (up to 4 s8v per thread = H <= 8192)
+  // so the second pass never re-reads it — decode rows are tiny and the
+  // kernel is latency-bound, not bandwidth-bound
+  s8v cache[4];
+  const bool cached = nvec <= (int)blockDim.x * 4;
+  int ci = 0;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x, ++ci) {
     s8v rv = *(const s8v*)(rr + i * 8);
     s8v dv = *(const s8v*)(dr + i * 8);
     s8v o;
@@ -145,13 +151,15 @@ add_rmsnorm_fwd_kernel(const short* __restrict__ res_in,
                          // result matches an unfused add -> rmsnorm chain
       ss += f * f;
     }
+    if (cached && ci < 4) cache[ci] = o;
     *(s8v*)(ror + i * 8) = o;
   }
   ss = block_reduce_sum(ss, scratch);
   float rstd = rsqrtf(ss / (float)H + eps);
 
-  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
-    s8v v = *(const s8v*)(ror + i * 8);
+  ci = 0;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x, ++ci) {
+    s8v v = (cached && ci < 4) ? cache[ci] : *(const s8v*)(ror + i * 8);
     s8v wv = *(const s8v*)(w + i * 8);
     s8v o;
 #pragma unroll

@@ -22,6 +22,11 @@ $TR --nproc-per-node 8 bench.py --gpus 8 --steps 4 --warmup 2 \
 $TR --nproc-per-node 8 bench_infer.py \
     | tee results/llama3_8b_decode_tp8.json
 
+# config 2 again with the ring-pipelined SP comm overlap (round-2
+# machinery, opt-in until xGMI-measured — compare against the first run)
+NXDA_SP_OVERLAP=1 $TR --nproc-per-node 8 bench.py --gpus 8 --steps 8 \
+    --warmup 3 | tee results/llama2_7b_tp8_spoverlap.json
+
 # scaling curve (weak): N = 1, 2, 4
 for N in 1 2 4; do
   $TR --nproc-per-node $N bench.py --gpus $N --steps 6 --warmup 2 \
