@@ -44,6 +44,14 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
     if q.is_cuda:
         from .. import ops
 
+        if (window is not None and causal and not torch.is_grad_enabled()
+                and q.shape[2] == k.shape[2] and q.shape[3] == 128
+                and q.dtype == torch.bfloat16
+                and ops.flash_attn_window_available()):
+            # sliding-window MFMA kernel (inference; the backward kernel
+            # is full-causal, so training windows use the composed path)
+            return ops.flash_attn_windowed(q, k, v, int(window),
+                                           softmax_scale)
         if q.shape[2] != k.shape[2] or q.shape[3] != 128 or \
                 window is not None or q.dtype != torch.bfloat16:
             # rectangular attention (KV-cache decode), head_dim != 128,

@@ -337,6 +337,31 @@ def flash_attn(q, k, v, causal=True, softmax_scale=None):
     return _FlashAttnFn.apply(q, k, v, causal, scale)
 
 
+def flash_attn_window_available() -> bool:
+    lib = _load()
+    return lib is not None and hasattr(lib, "flash_attn_fwd_window")
+
+
+def flash_attn_windowed(q, k, v, window: int, softmax_scale=None):
+    """Sliding-window causal flash attention (inference/no-grad): q row i
+    attends kv rows [i-window+1, i].  q (B,Hq,S,128), k/v (B,Hkv,S,128)
+    bf16; out-of-band tiles are skipped entirely (Mistral-style)."""
+    lib = _require_lib()
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    assert D == 128 and q.dtype == torch.bfloat16
+    scale = softmax_scale or 1.0 / math.sqrt(D)
+    q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+    out = torch.empty_like(q)
+    lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
+    lib.flash_attn_fwd_window(_ptr(q), _ptr(k), _ptr(v), _ptr(out),
+                              _ptr(lse), ctypes.c_int(B), ctypes.c_int(Hq),
+                              ctypes.c_int(Hkv), ctypes.c_int(S),
+                              ctypes.c_float(scale), ctypes.c_int(window),
+                              _stream())
+    return out
+
+
 # ---------------------------------------------------------------------------
 # Fused AdamW (ZeRO-1 master-shard update; csrc/adamw.hip)
 # ---------------------------------------------------------------------------

@@ -41,7 +41,10 @@ extern "C" __global__ void __launch_bounds__(512, 2)
 flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                  const short* __restrict__ vp, short* __restrict__ op,
                  float* __restrict__ lsep, int B, int Hq, int Hkv, int S,
-                 float scale, int causal) {
+                 float scale, int causal, int window) {
+  // window > 0: Mistral-style sliding window — q row i attends kv rows
+  // [i - window + 1, i] (causal implied); whole tiles outside the band
+  // are skipped, boundary tiles masked per element.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // THREE rotating K/Vt buffers: tile t is read from buf t%3 while t+1 is
   // written into (t+1)%3, whose previous readers (tile t-2) finished two
@@ -94,6 +97,10 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int ntiles = (kv_end + FA_KV - 1) / FA_KV;
   // this wave can skip tiles fully above its causal row range
   const int my_kv_end = causal ? min(S, qw0 + FA_QW) : S;
+  // sliding window: tiles fully below ANY of this wave's rows' windows
+  // are skipped; the whole WG starts at the block's earliest window tile
+  const int my_kv_begin = window > 0 ? max(0, qw0 - window + 1) : 0;
+  const int t_begin = window > 0 ? max(0, q0 - window + 1) / FA_KV : 0;
 
   const float s2 = scale * LOG2E;
 
@@ -134,16 +141,17 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     }
   };
 
-  issue_loads(0);
-  write_tile(0);
+  issue_loads(t_begin);
+  write_tile(t_begin % 3);
   __syncthreads();
 
-  for (int t = 0; t < ntiles; ++t) {
+  for (int t = t_begin; t < ntiles; ++t) {
     const int kv0 = t * FA_KV;
     const int cur = t % 3;
     if (t + 1 < ntiles) issue_loads(t + 1);
 
-    const bool wave_active = kv0 < my_kv_end;
+    const bool wave_active = kv0 < my_kv_end &&
+        (window <= 0 || kv0 + FA_KV > my_kv_begin);
     if (wave_active) {
       // ---- QK^T: S^T[k][q] = sum_d K[k][d] Q^T[d][q] ------------------
       // All 16 K fragments are prefetched into registers BEFORE the mfma
@@ -172,7 +180,8 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       // ---- online softmax (exp2 domain), lane-local per q row ---------
       float sc[2][16];
       const bool need_mask =
-          (causal && kv0 + FA_KV > qw0) || (kv0 + FA_KV > S);
+          (causal && kv0 + FA_KV > qw0) || (kv0 + FA_KV > S) ||
+          (window > 0 && kv0 < qw0 + FA_QW - 1 - window + 1 + FA_KV);
 #pragma unroll
       for (int kb = 0; kb < 2; ++kb)
 #pragma unroll
@@ -180,7 +189,8 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           float v = acc[kb][r] * s2;
           if (need_mask) {
             int kg = kv0 + 32 * kb + acc_row(r, hi);
-            if ((causal && kg > my_q) || kg >= S) v = NEG_INF;
+            if ((causal && kg > my_q) || kg >= S ||
+                (window > 0 && kg <= my_q - window)) v = NEG_INF;
           }
           sc[kb][r] = v;
         }
@@ -298,5 +308,17 @@ extern "C" void flash_attn_fwd(const void* q, const void* k, const void* v,
   size_t lds = 3 * (K_TILE_B + VT_TILE_B);
   flash_fwd_kernel<<<grid, 512, lds, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out,
-      (float*)lse, B, Hq, Hkv, S, scale, causal);
+      (float*)lse, B, Hq, Hkv, S, scale, causal, 0);
+}
+
+extern "C" void flash_attn_fwd_window(const void* q, const void* k,
+                                      const void* v, void* out, void* lse,
+                                      int B, int Hq, int Hkv, int S,
+                                      float scale, int window,
+                                      hipStream_t stream) {
+  dim3 grid(Hq, (S + FA_QBLK - 1) / FA_QBLK, B);
+  size_t lds = 3 * (K_TILE_B + VT_TILE_B);
+  flash_fwd_kernel<<<grid, 512, lds, stream>>>(
+      (const short*)q, (const short*)k, (const short*)v, (short*)out,
+      (float*)lse, B, Hq, Hkv, S, scale, 1, window);
 }

@@ -468,3 +468,23 @@ def test_fused_ce_ignore_index():
     assert g[2].abs().max().item() == 0.0
     assert g[4].abs().max().item() == 0.0
     assert torch.isfinite(loss).all()
+
+
+@pytest.mark.parametrize("B,Hq,Hkv,S,W", [
+    (1, 4, 4, 1024, 256),
+    (2, 8, 2, 2048, 512),   # GQA + window smaller than seq
+    (1, 2, 2, 512, 4096),   # window larger than seq == plain causal
+])
+def test_flash_attn_sliding_window(B, Hq, Hkv, S, W):
+    """Sliding-window MFMA kernel vs the fp32 composed reference."""
+    if not ops.flash_attn_window_available():
+        pytest.skip("window kernel not built")
+    torch.manual_seed(9)
+    D = 128
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device="cuda") * 0.5
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device="cuda") * 0.5
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device="cuda") * 0.5
+    out = ops.flash_attn_windowed(q, k, v, W)
+    ref = _torch_reference(q.float(), k.float(), v.float(), causal=True,
+                           window=W)
+    _cmp(out, ref, atol=3e-2, rtol=3e-2, name=f"window S={S} W={W}")
