@@ -127,22 +127,33 @@ def main():
                                lr=1.5e-4, betas=(0.9, 0.95), weight_decay=0.1,
                                grad_clipping=True, max_norm=1.0)
 
+    # FRESH synthetic batch every step (pre-generated outside the timed
+    # region): no batch is ever repeated, so the loss reflects actual
+    # optimization rather than memorizing one batch
     torch.manual_seed(4321)
+    n_unique = args.warmup + args.steps
     if args.pp > 1:
-        full = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+        fulls = [torch.randint(0, cfg.vocab_size, (B, S), device=device)
+                 for _ in range(n_unique)]
+        counter = [0]
 
         def step():
+            full = fulls[counter[0] % n_unique]
+            counter[0] += 1
             opt.zero_grad()
             loss = model.run_train(input_ids=full, labels=full)
             opt.step()
             return loss
     else:
         data = [torch.randint(0, cfg.vocab_size, (mbs, S), device=device)
-                for _ in range(n_micro)]
+                for _ in range(n_micro * n_unique)]
+        counter = [0]
 
         def step():
+            base = (counter[0] % n_unique) * n_micro
+            counter[0] += 1
             opt.zero_grad()
-            for x in data:
+            for x in data[base:base + n_micro]:
                 loss = model(x, labels=x)
                 (loss / n_micro).backward()
             opt.step()
