@@ -262,3 +262,63 @@ def _ep2_grad_norm_worker(rank, world):
 def test_ep2_grad_norm_matches_dense():
     res = run_distributed(_ep2_grad_norm_worker, world_size=2)
     assert abs(res[0] - res[1]) < 1e-5  # same norm on every rank
+
+
+def _tp2_ep2_worker(rank, world):
+    """3-D-ish mesh: tp=2 x ep=2 on 4 ranks — the full MoE layer (router +
+    EP all-experts + delayed TP reduce) matches the dense single-process
+    golden for this dp-rank's tokens (dp=1 here; tokens replicated)."""
+    from neuronx_distributed_amd.moe import ExpertMLPs, MoE, RouterTopK
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=2,
+                                 expert_model_parallel_size=2)
+    E, H, I, k, T = 4, 8, 16, 2, 6
+    tp_rank = ps.get_tensor_model_parallel_rank()
+    ep_rank = ps.get_expert_model_parallel_rank()
+
+    mlps = ExpertMLPs(E, H, I, k, capacity_factor=None, dtype=torch.float32)
+    # deterministic weights per (global expert, tp shard): full master
+    # (H, 2I) per expert, column-sharded [gate|up] per tp rank
+    e0 = ep_rank * (E // 2)
+    with torch.no_grad():
+        for j in range(E // 2):
+            g = e0 + j
+            torch.manual_seed(1000 + g)
+            gu = 0.1 * torch.randn(H, 2 * I)
+            torch.manual_seed(2000 + g)
+            dn = 0.1 * torch.randn(I, H)
+            gate, up = gu[:, :I], gu[:, I:]
+            gs = gate.chunk(2, dim=1)[tp_rank]
+            us = up.chunk(2, dim=1)[tp_rank]
+            mlps.gate_up_proj.weight.data[j] = torch.cat([gs, us], dim=1)
+            mlps.down_proj.weight.data[j] = dn.chunk(2, dim=0)[tp_rank]
+    torch.manual_seed(7)
+    router = RouterTopK(E, k, H, dtype=torch.float32)
+    moe = MoE(router, mlps, return_router_logits=False)
+    moe.train()
+
+    torch.manual_seed(50)  # same tokens on every rank (dp=1)
+    x = torch.randn(1, T, H)
+    out = moe(x).reshape(T, H)
+
+    # dense golden
+    ref = torch.zeros(T, H)
+    torch.manual_seed(7)
+    g_router = RouterTopK(E, k, H, dtype=torch.float32)
+    _, aff, idx = g_router(x.reshape(T, H))
+    for g in range(E):
+        torch.manual_seed(1000 + g)
+        gu_w = 0.1 * torch.randn(H, 2 * I)
+        torch.manual_seed(2000 + g)
+        dn_w = 0.1 * torch.randn(I, H)
+        h = x.reshape(T, H) @ gu_w
+        act = torch.nn.functional.silu(h[:, :I]) * h[:, I:]
+        ref += (act @ dn_w) * aff[:, g:g + 1]
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+    return float(out.sum())
+
+
+def test_moe_tp2_ep2_matches_dense():
+    res = run_distributed(_tp2_ep2_worker, world_size=4)
+    assert max(res) - min(res) < 1e-4  # identical output on all ranks
