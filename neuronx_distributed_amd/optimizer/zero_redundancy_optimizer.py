@@ -141,9 +141,13 @@ class _Bucket:
                 p.grad = view
 
     def local_sq_norm(self) -> torch.Tensor:
-        sq = self.grad_shard.float().pow(2).sum()
+        # vector_norm with fp32 accumulation: ONE fused reduction, no
+        # materialized fp32 copy of the (multi-GB) grad buffer
+        sq = torch.linalg.vector_norm(self.grad_shard,
+                                      dtype=torch.float32).pow(2)
         for s, e in self.norm_exclude_ranges:
-            sq -= self.grad_shard[s:e].float().pow(2).sum()
+            sq -= torch.linalg.vector_norm(self.grad_shard[s:e],
+                                           dtype=torch.float32).pow(2)
         return sq
 
 

@@ -153,6 +153,9 @@ def get_grad_norm(parameters, norm_type: float = 2.0,
         g = p.grad.detach()
         if use_inf:
             torch.maximum(acc, g.abs().max().reshape(1).float(), out=acc)
+        elif norm_type == 2.0:
+            # fused single-pass reduction with fp32 accumulation
+            acc += torch.linalg.vector_norm(g, dtype=torch.float32).pow(2)
         else:
             acc += g.detach().abs().float().pow(norm_type).sum()
 
