@@ -142,3 +142,23 @@ def test_memory_stats_cpu():
     s = memory_stats()
     assert set(s) >= {"allocated_gib", "reserved_gib", "peak_allocated_gib"}
     log_memory_stats("test")
+
+
+def test_yaml_converter(tmp_path):
+    """reference scripts/yaml_converter.py parity: YAML training config ->
+    converter JSON (geometry keys + MoE expert count)."""
+    import json
+
+    from neuronx_distributed_amd.scripts.yaml_converter import \
+        convert_yaml_to_json
+
+    y = tmp_path / "train.yaml"
+    y.write_text(
+        "model:\n  num_layers: 12\n  num_attention_heads: 16\n"
+        "  hidden_size: 1024\n  num_kv_heads: 4\n"
+        "  moe:\n    num_experts: 8\n")
+    out = convert_yaml_to_json(str(y), str(tmp_path / "cfg.json"))
+    cfg = json.load(open(out))
+    assert cfg == {"num_hidden_layers": 12, "num_attention_heads": 16,
+                   "hidden_size": 1024, "num_key_value_heads": 4,
+                   "num_local_experts": 8}
