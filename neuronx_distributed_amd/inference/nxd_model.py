@@ -38,6 +38,49 @@ class NxDModel(nn.Module):
     def route(self, kwargs) -> Optional[str]:
         return self._buckets.get(self._shape_key(kwargs))
 
+    # -- deployment artifact (reference nxd_model.py:709,924 torchscript
+    # export; MI355X: weights + bucket manifest, graphs re-capture lazily) --
+
+    def save(self, path: str) -> None:
+        """Write a deployment artifact: ``weights.safetensors`` (deduped,
+        shared tensors cloned) + ``manifest.pt`` with the bucket routing
+        table.  ``NxDModel.load(path, model_fn)`` reconstructs it; the
+        hipGraphs re-capture on first use (they are device-state, not
+        serializable)."""
+        import os
+
+        from ..utils.safetensors_utils import save_safetensors
+
+        os.makedirs(path, exist_ok=True)
+        save_safetensors(self.module.state_dict(),
+                         os.path.join(path, "weights.safetensors"))
+        torch.save({
+            "buckets": self._buckets,
+            "use_hip_graphs": self.use_hip_graphs,
+        }, os.path.join(path, "manifest.pt"))
+
+    @classmethod
+    def load(cls, path: str, model_fn, strict: bool = True,
+             **model_kwargs) -> "NxDModel":
+        """Rebuild from a ``save()`` artifact: ``model_fn(**model_kwargs)``
+        constructs the architecture; weights and the bucket table load
+        from the artifact."""
+        import os
+
+        from ..utils.safetensors_utils import load_safetensors
+
+        module = model_fn(**model_kwargs)
+        sd = load_safetensors(os.path.join(path, "weights.safetensors"))
+        module.load_state_dict(sd, strict=strict)
+        if torch.cuda.is_available():
+            module = module.cuda()
+        module.eval()
+        manifest = torch.load(os.path.join(path, "manifest.pt"),
+                              map_location="cpu", weights_only=False)
+        model = cls(module, use_hip_graphs=manifest["use_hip_graphs"])
+        model._buckets = manifest["buckets"]
+        return model
+
     @torch.no_grad()
     def _capture(self, tag: str, kwargs):
         """Capture one hipGraph for this bucket: static input buffers are

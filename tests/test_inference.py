@@ -240,3 +240,44 @@ def _sampler_topp_worker(rank, world):
 
 def test_sampler_top_p():
     run_distributed(_sampler_topp_worker, world_size=1)
+
+
+def _artifact_worker(rank, world, tmpdir):
+    """Deployment artifact: ModelBuilder -> NxDModel.save -> NxDModel.load
+    with a fresh model_fn reproduces outputs and the bucket routing."""
+    from neuronx_distributed_amd.inference.model_builder import ModelBuilder
+    from neuronx_distributed_amd.inference.nxd_model import NxDModel
+    from neuronx_distributed_amd.models import LlamaForCausalLM, get_config
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("tiny")
+    torch.manual_seed(0)
+    ref = LlamaForCausalLM(cfg).eval()
+    sd = ref.state_dict()
+    builder = ModelBuilder(model_fn=lambda: LlamaForCausalLM(cfg),
+                           checkpoint_loader=lambda: sd)
+    builder.trace({"input_ids": torch.zeros(1, 8, dtype=torch.long)},
+                  tag="short")
+    builder.trace({"input_ids": torch.zeros(1, 16, dtype=torch.long)},
+                  tag="long")
+    nxd = builder.compile(use_hip_graphs=False)
+    nxd.save(tmpdir)
+
+    loaded = NxDModel.load(tmpdir, lambda: LlamaForCausalLM(cfg))
+    assert loaded.route({"input_ids": torch.zeros(1, 16,
+                                                  dtype=torch.long)}) == "long"
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (1, 8))
+    with torch.no_grad():
+        a = nxd(input_ids=x).cpu()
+        b = loaded(input_ids=x).cpu()
+    assert torch.allclose(a, b, atol=1e-5)
+    return 0.0
+
+
+def test_nxd_model_artifact_roundtrip():
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as d:
+        run_distributed(_artifact_worker, world_size=1, args=(d,))
