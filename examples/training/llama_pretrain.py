@@ -108,7 +108,15 @@ def main():
     mb_per_rank = max(1, args.global_batch // dp // args.microbatch)
     torch.manual_seed(42 + nxd.parallel.get_data_parallel_rank())
 
-    t0 = time.time()
+    # results.json metrics + moving-average seq/s throughput (reference
+    # examples/.../training_utils.py:343-369 Throughput +
+    # tp_zero1_llama_hf_pretrain.py:470-515 TrainingMetrics)
+    from neuronx_distributed_amd.utils.training_metrics import (
+        Metric, Throughput, TrainingMetrics)
+
+    metrics = TrainingMetrics("results.json") if rank == 0 else None
+    thr = Throughput(batch_size=args.global_batch // dp, world_size=dp,
+                     grad_accum_usteps=1, moving_avg_window_size=10)
     for step in range(start_step, args.steps):
         opt.zero_grad()
         if args.pp > 1:
@@ -123,15 +131,23 @@ def main():
                 loss = model(x, labels=x)
                 (loss / mb_per_rank).backward()
         opt.step()
+        seq_per_s = thr.get_throughput()
         if rank == 0 and step % 10 == 0:
-            tok_s = args.global_batch * args.seq * (step - start_step + 1) \
-                / (time.time() - t0)
             print(f"step {step}: loss {loss.item():.4f} "
-                  f"({tok_s:,.0f} tokens/s)", flush=True)
+                  f"({seq_per_s * args.seq:,.0f} tokens/s)", flush=True)
         if args.ckpt_dir and (step + 1) % args.ckpt_interval == 0:
             nxd.save_checkpoint(args.ckpt_dir, tag=str(step + 1), model=model,
                                 optimizer=opt, user_content={"step": step + 1},
                                 num_kept=2, async_save=True)
+    if metrics is not None:
+        metrics.store_parameters({"Model": args.model, "Steps": args.steps,
+                                  "TP": args.tp, "PP": args.pp,
+                                  "SeqLen": args.seq,
+                                  "GlobalBatch": args.global_batch})
+        metrics.store_metrics([
+            Metric("Throughput", round(seq_per_s, 2), "seq/s"),
+            Metric("Final loss", round(float(loss.detach()), 4)),
+        ])
     dist.destroy_process_group()
 
 
