@@ -72,6 +72,29 @@ def compute_block_indices(expert_index: torch.Tensor, num_experts: int,
     return token_position_to_id, block_to_expert, n_blocks
 
 
+def blockwise_mm_mx(hidden: torch.Tensor, expert_affinities: torch.Tensor,
+                    gate_up_q: torch.Tensor, gate_up_scales: torch.Tensor,
+                    down_q: torch.Tensor, down_scales: torch.Tensor,
+                    token_position_to_id: torch.Tensor,
+                    block_to_expert: torch.Tensor,
+                    expert_index: torch.Tensor,
+                    block_size: int = DEFAULT_BLOCK_SIZE,
+                    glu: bool = True, fmt: str = "fp4_e2m1") -> torch.Tensor:
+    """Blockwise MoE with OCP-MX-quantized expert weights (reference K5
+    ``bwmm_shard_on_block_mx``, blockwise.py:50-61): per-expert weights
+    stored as MX fp4/fp8 blocks + shared scales, dequantized on the fly
+    into the grouped-GEMM path.  CDNA4 has native MX MFMA; this emulation
+    establishes the format/accuracy contract (quantization/microscaling)."""
+    from ..quantization.microscaling import dequantize_mx
+
+    gu_w = dequantize_mx(gate_up_q, gate_up_scales, axis=1,
+                         dtype=hidden.dtype)
+    d_w = dequantize_mx(down_q, down_scales, axis=1, dtype=hidden.dtype)
+    return blockwise_mm(hidden, expert_affinities, gu_w, d_w,
+                        token_position_to_id, block_to_expert, expert_index,
+                        block_size, glu=glu)
+
+
 def blockwise_mm(hidden: torch.Tensor, expert_affinities: torch.Tensor,
                  gate_up_w: torch.Tensor, down_w: torch.Tensor,
                  token_position_to_id: torch.Tensor,

@@ -348,3 +348,42 @@ def test_sampler_cdf_multinomial_and_medusa():
     assert topk_idx.shape == (3, 5)
     ref = logits.topk(5, dim=-1).indices
     assert torch.equal(topk_idx, ref)
+
+
+def _mx_blockwise_worker(rank, world):
+    """K5 MX variant: blockwise_mm_mx with fp4-e2m1 MX weights tracks the
+    full-precision blockwise output within fp4 quantization error."""
+    from neuronx_distributed_amd.moe.blockwise import (blockwise_mm,
+                                                       blockwise_mm_mx,
+                                                       compute_block_indices)
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.quantization.microscaling import quantize_mx
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    E, H, I, k, T, B = 4, 64, 32, 2, 24, 8
+    torch.manual_seed(0)
+    hidden = torch.randn(T, H)
+    gu_w = torch.randn(E, H, 2 * I) * 0.1
+    d_w = torch.randn(E, I, H) * 0.1
+    vals, idx = torch.topk(torch.softmax(torch.randn(T, E), -1), k, -1)
+    aff = torch.zeros(T, E).scatter(-1, idx, vals)
+    tpi, b2e, _ = compute_block_indices(idx, E, B)
+
+    ref = blockwise_mm(hidden, aff, gu_w, d_w, tpi, b2e, idx, B)
+
+    gq, gs = quantize_mx(gu_w, fmt="fp4_e2m1", axis=1)
+    dq, ds = quantize_mx(d_w, fmt="fp4_e2m1", axis=1)
+    out = blockwise_mm_mx(hidden, aff, gq, gs, dq, ds, tpi, b2e, idx, B)
+    rel = (out - ref).norm() / (ref.norm() + 1e-9)
+    assert rel < 0.35, rel  # fp4 resolution
+    # fp8 variant is much tighter
+    gq8, gs8 = quantize_mx(gu_w, fmt="fp8_e4m3", axis=1)
+    dq8, ds8 = quantize_mx(d_w, fmt="fp8_e4m3", axis=1)
+    out8 = blockwise_mm_mx(hidden, aff, gq8, gs8, dq8, ds8, tpi, b2e, idx, B)
+    rel8 = (out8 - ref).norm() / (ref.norm() + 1e-9)
+    assert rel8 < 0.08, rel8
+    return 0.0
+
+
+def test_blockwise_mx_variants():
+    run_distributed(_mx_blockwise_worker, world_size=1)
