@@ -387,3 +387,77 @@ def _mx_blockwise_worker(rank, world):
 
 def test_blockwise_mx_variants():
     run_distributed(_mx_blockwise_worker, world_size=1)
+
+
+def test_moe_configs_and_validator(tmp_path):
+    """moe_configs dataclasses + MoeConfigValidator rules (reference
+    moe_configs.py / moe_config_validator.py)."""
+    import json
+    import types
+
+    import pytest as _pytest
+
+    from neuronx_distributed_amd.moe import (BlockwiseMatmulConfig,
+                                             GLUType, MoeConfigValidator,
+                                             RoutedExpertsMLPOpsConfig)
+
+    c = RoutedExpertsMLPOpsConfig(num_experts=8, top_k=2, hidden_size=64,
+                                  intermediate_size=128)
+    assert c.glu_type == GLUType.SWIGLU
+    with _pytest.raises(ValueError):
+        RoutedExpertsMLPOpsConfig(num_experts=2, top_k=4, hidden_size=8,
+                                  intermediate_size=8)
+    b = BlockwiseMatmulConfig.from_kwargs(block_size=128, bogus_key=1)
+    assert b.block_size == 128
+
+    hf = tmp_path / "config.json"
+    hf.write_text(json.dumps({"hidden_act": "silu"}))
+    moe = types.SimpleNamespace(dropless=True, capacity_factor=1.5,
+                                glu_mlp=True)
+    cfg = types.SimpleNamespace(
+        model_source="hf",
+        model=types.SimpleNamespace(moe=moe, model_config=str(hf)))
+    MoeConfigValidator(cfg).validate_moe_config()
+    assert cfg.model.moe.capacity_factor == 0.0  # dropless forces 0
+
+    hf.write_text(json.dumps({"hidden_act": "gelu"}))
+    with _pytest.raises(ValueError):
+        MoeConfigValidator(cfg).validate_moe_config()
+
+    moe2 = types.SimpleNamespace(dropless=False, capacity_factor=-1.0,
+                                 glu_mlp=True)
+    cfg2 = types.SimpleNamespace(model_source="megatron",
+                                 model=types.SimpleNamespace(moe=moe2))
+    with _pytest.raises(ValueError):
+        MoeConfigValidator(cfg2).validate_moe_config()
+
+
+def _hybrid_group_worker(rank, world):
+    """Hybrid CTE/TKG sharding groups: prefill pair (tp=world) vs decode
+    pair (ep=world) expose different meshes by phase."""
+    from neuronx_distributed_amd.moe import (
+        destroy_moe_model_parallel, get_moe_ep_group, get_moe_tp_ep_group,
+        init_tensor_expert_parallel_moe_process_groups)
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    init_tensor_expert_parallel_moe_process_groups(
+        tkg_tp_degree=1, tkg_ep_degree=world,
+        cte_tp_degree=world, cte_ep_degree=1)
+    cte_tp = get_moe_tp_ep_group(prefill=True)
+    tkg_tp = get_moe_tp_ep_group(prefill=False)
+    tkg_ep = get_moe_ep_group(prefill=False)
+    assert cte_tp.size == world
+    assert tkg_tp.size == 1
+    assert tkg_ep.size == world
+    # the groups are usable communicators
+    import torch.distributed as dist
+    t = torch.ones(1)
+    dist.all_reduce(t, group=cte_tp.group)
+    assert float(t) == world
+    destroy_moe_model_parallel()
+    return 0.0
+
+
+def test_moe_hybrid_process_groups():
+    run_distributed(_hybrid_group_worker, world_size=2)
