@@ -92,3 +92,23 @@ def model_parallel_manual_seed(seed: int):
 
 
 model_parallel_xla_manual_seed = model_parallel_manual_seed
+
+
+def set_random_seed(seed: int):
+    """Seed python/torch RNGs identically on every rank (reference
+    utils/random.py:8); the model-parallel dual-domain seeding is
+    ``model_parallel_manual_seed``."""
+    import random as _random
+
+    _random.seed(seed)
+    try:
+        import numpy as _np
+
+        _np.random.seed(seed)
+    except ImportError:
+        pass
+    import torch as _torch
+
+    _torch.manual_seed(seed)
+    if _torch.cuda.is_available():
+        _torch.cuda.manual_seed_all(seed)
