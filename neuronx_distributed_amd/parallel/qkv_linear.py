@@ -27,14 +27,70 @@ from .layers import BaseParallelLinear, default_init_method
 from .utils import divide, set_tensor_model_parallel_attributes
 
 
+def _sp_overlapped_qkv(input_, wq, wk, wv, bq, bk, bv, tp_info):
+    """Ring-pipelined SP all-gather fused with the three QKV GEMMs
+    (NXDA_SP_OVERLAP=1): sequence chunks rotate around the TP ring while
+    the in-hand chunk runs its q/k/v GEMMs — mirrors
+    layers._sp_overlapped_linear for the GQA QKV case."""
+    import torch.distributed as dist
+
+    world = tp_info.size
+    me = tp_info.rank_in_group(dist.get_rank())
+    ranks = tp_info.ranks_of(dist.get_rank())
+    nxt = ranks[(me + 1) % world]
+    prv = ranks[(me - 1) % world]
+    S = input_.shape[0]
+    src = input_.contiguous()
+    cur = src
+    recv_buf = torch.empty_like(src)
+    spare = torch.empty_like(src)  # src is saved for backward: never
+    # receive into its storage (see layers._sp_overlapped_linear)
+    lead = tuple(input_.shape[1:-1])
+    q = torch.empty((world * S,) + lead + (wq.shape[0],),
+                    dtype=input_.dtype, device=input_.device)
+    k = torch.empty((world * S,) + lead + (wk.shape[0],),
+                    dtype=input_.dtype, device=input_.device)
+    v = torch.empty((world * S,) + lead + (wv.shape[0],),
+                    dtype=input_.dtype, device=input_.device)
+    for t in range(world):
+        owner = (me - t) % world
+        works = []
+        if t < world - 1:
+            ops_ = [dist.P2POp(dist.isend, cur, nxt, group=tp_info.group),
+                    dist.P2POp(dist.irecv, recv_buf, prv,
+                               group=tp_info.group)]
+            works = dist.batch_isend_irecv(ops_)
+        sl = slice(owner * S, (owner + 1) * S)
+        q[sl] = F.linear(cur, wq, bq)
+        k[sl] = F.linear(cur, wk, bk)
+        v[sl] = F.linear(cur, wv, bv)
+        for w in works:
+            w.wait()
+        if t < world - 1:
+            nxt_recv = spare if cur is src else cur
+            cur = recv_buf
+            recv_buf = nxt_recv
+    return q, k, v
+
+
 class _QKVLinearWithAsyncCommunication(torch.autograd.Function):
     @staticmethod
     def forward(ctx, input_, wq, wk, wv, bq, bk, bv, sequence_parallel, kv_mult):
+        import os
+
         ctx.sequence_parallel = sequence_parallel
         ctx.kv_mult = kv_mult
         ctx.use_bias = bq is not None
+        tp_info = ps.get_group_info("tp")
+        if sequence_parallel and tp_info.size > 1 and \
+                os.environ.get("NXDA_SP_OVERLAP", "0") == "1" and \
+                not ps.is_aot_mode():
+            q, k, v = _sp_overlapped_qkv(input_, wq, wk, wv, bq, bk, bv,
+                                         tp_info)
+            ctx.save_for_backward(input_, wq, wk, wv)
+            return q, k, v
         if sequence_parallel:
-            total_input = comm.all_gather(input_, dim=0, group=ps.get_group_info("tp"))
+            total_input = comm.all_gather(input_, dim=0, group=tp_info)
         else:
             total_input = input_
         q = F.linear(total_input, wq, bq)

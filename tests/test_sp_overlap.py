@@ -61,3 +61,44 @@ def test_sp_overlap_matches_unfused_tp2():
 
 def test_sp_overlap_matches_unfused_tp4():
     run_distributed(_overlap_worker, world_size=4)
+
+
+def _qkv_overlap_worker(rank, world):
+    """Ring-overlapped SP QKV == plain all-gather + 3 GEMMs, fwd + bwd."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.parallel.qkv_linear import (
+        GQAQKVColumnParallelLinear)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    S, B, H = 8, 2, 16
+    torch.manual_seed(0)
+    qkv = GQAQKVColumnParallelLinear(
+        H, [32, 16], head_dim=4, num_attention_heads=8,
+        num_key_value_heads=4, sequence_parallel_enabled=True,
+        dtype=torch.float32)
+    qkv.train()
+    torch.manual_seed(100 + rank)
+    x = torch.randn(S, B, H)
+
+    results = {}
+    for mode in ("0", "1"):
+        os.environ["NXDA_SP_OVERLAP"] = mode
+        xg = x.clone().requires_grad_(True)
+        with torch.enable_grad():
+            q, k, v = qkv(xg)
+            loss = (q.float().pow(2).sum() + k.float().pow(2).sum()
+                    + v.float().pow(2).sum())
+        loss.backward()
+        results[mode] = (q.detach().clone(), k.detach().clone(),
+                         v.detach().clone(), xg.grad.clone(),
+                         qkv.weight_q.grad.clone(),
+                         qkv.weight_k.grad.clone())
+        for w in (qkv.weight_q, qkv.weight_k, qkv.weight_v):
+            w.grad = None
+    for a, b in zip(results["0"], results["1"]):
+        assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
+    return 0.0
+
+
+def test_sp_overlap_qkv_tp4():
+    run_distributed(_qkv_overlap_worker, world_size=4)
