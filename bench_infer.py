@@ -24,6 +24,9 @@ def main():
     p.add_argument("--batch", type=int, default=32)
     p.add_argument("--prompt", type=int, default=1024)
     p.add_argument("--new", type=int, default=64)
+    p.add_argument("--block", type=int, default=4,
+                   help="decode steps captured per hipGraph (greedy "
+                        "feedback inside the graph); 1 = per-step")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -105,18 +108,31 @@ def main():
     prefill_s = time.perf_counter() - t0
 
     next_tok = sampler(logits[:, -1, :])
+    blk = max(1, args.block) if on_gpu else 1
     dec = GraphDecoder(model, caches, start_pos=args.prompt,
-                       batch=args.batch, device=x.device) if on_gpu else None
+                       batch=args.batch, device=x.device,
+                       steps_per_capture=blk) if on_gpu else None
     if dec is not None:
-        dec.capture()  # untimed (one-off per shape bucket)
-        # one replayed step to settle clocks
-        dec.step(next_tok)
+        dec.capture_block()  # untimed (one-off per shape bucket)
+        # one settle replay, then rewind the position for the timed region
+        if blk > 1:
+            toks = dec.step_block(next_tok)
+            dec.pos_t -= blk
+        else:
+            dec.step(next_tok)
 
     t0 = time.perf_counter()
-    steps = max(1, args.new - 1)
-    for _ in range(steps):
-        logits = dec.step(next_tok) if dec is not None else None
-        next_tok = sampler(logits[:, -1, :])
+    if blk > 1:
+        n_calls = max(1, (args.new - 1) // blk)
+        for _ in range(n_calls):
+            toks = dec.step_block(next_tok)
+            next_tok = toks[:, -1]
+        steps = n_calls * blk
+    else:
+        steps = max(1, args.new - 1)
+        for _ in range(steps):
+            logits = dec.step(next_tok) if dec is not None else None
+            next_tok = sampler(logits[:, -1, :])
     if on_gpu:
         torch.cuda.synchronize()
     decode_s = time.perf_counter() - t0

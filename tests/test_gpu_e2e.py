@@ -171,3 +171,67 @@ def test_mixtral_gpu_train_step():
         logits = m(x[:, :8])
     assert logits.shape == (2, 8, cfg.vocab_size)
     assert torch.isfinite(logits.float()).all()
+
+
+def test_block_decode_matches_per_step():
+    """Multi-step graph capture (greedy feedback INSIDE the graph) emits
+    exactly the per-step greedy tokens."""
+    import torch.distributed as dist
+    from neuronx_distributed_amd.inference.decode_graph import GraphDecoder
+    from neuronx_distributed_amd.inference.kv_cache import build_kv_caches
+    from neuronx_distributed_amd.models import get_config, LlamaForCausalLM
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.utils.sampling import Sampler
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29552")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    cfg = get_config("test-d128", max_position_embeddings=128)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.bfloat16)
+    torch.manual_seed(0)
+    with torch.device("cuda"):
+        model = LlamaForCausalLM(cfg)
+    torch.set_default_dtype(prev)
+    model.eval()
+    B, P, BLK, CALLS = 2, 17, 4, 3
+
+    def prefill():
+        caches = build_kv_caches(cfg.num_hidden_layers, B,
+                                 cfg.num_key_value_heads, 64,
+                                 cfg.head_dim, device="cuda")
+        torch.manual_seed(3)
+        x = torch.randint(0, cfg.vocab_size, (B, P), device="cuda")
+        with torch.no_grad():
+            logits = model(x, kv_caches=caches, pos_offset=0)
+        return caches, Sampler(do_sample=False)(logits[:, -1, :])
+
+    # per-step reference
+    caches, tok = prefill()
+    dec1 = GraphDecoder(model, caches, start_pos=P, batch=B, device="cuda")
+    dec1.capture()
+    ref = []
+    t = tok
+    for _ in range(BLK * CALLS):
+        logits = dec1.step(t)
+        t = Sampler(do_sample=False)(logits[:, -1, :])
+        ref.append(t)
+    ref = torch.stack(ref, dim=1)
+
+    # block decode
+    caches, tok2 = prefill()
+    assert torch.equal(tok, tok2)
+    dec4 = GraphDecoder(model, caches, start_pos=P, batch=B, device="cuda",
+                        steps_per_capture=BLK)
+    assert dec4.capture_block()
+    outs = []
+    t = tok2
+    for _ in range(CALLS):
+        toks = dec4.step_block(t)
+        outs.append(toks)
+        t = toks[:, -1]
+    out = torch.cat(outs, dim=1)
+    assert torch.equal(out, ref), (out, ref)
