@@ -195,6 +195,61 @@ class _RPLOverlappedLinearRS(torch.autograd.Function):
         return grad_input, grad_weight
 
 
+def _fp8_forward_enabled() -> bool:
+    return os.environ.get("NXDA_FP8_LINEAR", "0") == "1"
+
+
+def _fp8_eligible(x: torch.Tensor, w: torch.Tensor) -> bool:
+    """fp8 pays ONLY on wide column GEMMs on this hipBLASLt build
+    (measured at the training shapes, M=32k: gate_up N=22016/K=4096 runs
+    2.2 PF fp8 vs 1.5 bf16; square/row shapes LOSE to the rowwise-scale
+    epilogue — profiles/README.md round-2 fp8 notes)."""
+    return (x.is_cuda and x.dtype == torch.bfloat16
+            and w.dtype == torch.bfloat16
+            and x.shape[-1] % 16 == 0 and w.shape[0] % 16 == 0
+            and w.shape[0] >= 2 * w.shape[1]
+            and x.numel() // x.shape[-1] >= 2048)
+
+
+def _fp8_quantized_weight(weight: torch.Tensor):
+    """Per-channel e4m3 weight quantization, cached per weight VERSION so
+    the n_micro forward calls of one optimizer step quantize once."""
+    from ..quantization.quantization_config import (QuantizationConfig,
+                                                    QuantizedDtype)
+    from ..quantization.quantization_utils import quantize_symmetric
+
+    key = (weight.data_ptr(), weight._version)
+    cache = getattr(weight, "_fp8_cache", None)
+    if cache is None or cache[0] != key:
+        cfg = QuantizationConfig(quantized_dtype=QuantizedDtype.F8E4M3,
+                                 quantize_activation=True)
+        with torch.no_grad():
+            qw, ws = quantize_symmetric(weight.detach(), cfg)
+        cache = (key, qw, ws)
+        weight._fp8_cache = cache
+    return cache[1], cache[2]
+
+
+def _forward_gemm(total_input: torch.Tensor, weight: torch.Tensor, bias):
+    """The forward GEMM of the parallel linears.  NXDA_FP8_LINEAR=1 runs
+    the wide (gate_up-class) forward GEMMs on the gfx950 fp8 MFMA pipe:
+    per-channel e4m3 weight scales (quantized once per optimizer step via
+    the version cache) + dynamic per-token activation scales through
+    torch._scaled_mm.  The BACKWARD stays bf16 on the saved tensors — only
+    the forward activations carry fp8 quantization error, like
+    transformer-engine's default training recipe."""
+    if _fp8_forward_enabled() and _fp8_eligible(total_input, weight):
+        from ..quantization.quantization_utils import fp8_scaled_linear
+
+        qw, ws = _fp8_quantized_weight(weight)
+        with torch.no_grad():
+            out = fp8_scaled_linear(total_input, qw, ws, total_input.dtype)
+        if bias is not None:
+            out = out + bias
+        return out
+    return F.linear(total_input, weight, bias)
+
+
 class LinearWithAsyncCommunication(torch.autograd.Function):
     """F.linear with TP/SP collectives placed for overlap.
 
@@ -226,7 +281,7 @@ class LinearWithAsyncCommunication(torch.autograd.Function):
         else:
             total_input = input_
 
-        output = F.linear(total_input, weight, bias)
+        output = _forward_gemm(total_input, weight, bias)
 
         if save_for_backward:
             if ctx.compute_weight_gradient:

@@ -146,3 +146,61 @@ def test_convergence_gpu_bf16_tracks_fp32():
     for i, (a, b) in enumerate(zip(curve_bf, curve_fp)):
         assert abs(a - b) < 0.05 + 0.05 * abs(b), (i, a, b, curve_bf,
                                                    curve_fp)
+
+
+@pytest.mark.gpu
+def test_convergence_gpu_fp8_forward_tracks_bf16():
+    """NXDA_FP8_LINEAR=1 (selective fp8 forward GEMMs on the wide
+    gate_up-class shapes) must track the bf16 loss curve."""
+    import os
+
+    import torch.distributed as dist
+
+    from neuronx_distributed_amd.models import LlamaForCausalLM, get_config
+    from neuronx_distributed_amd.optimizer import NeuronZero1Optimizer
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29762")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    if not ps.model_parallel_is_initialized():
+        ps.initialize_model_parallel(tensor_model_parallel_size=1)
+
+    cfg = get_config("tiny", num_hidden_layers=2, hidden_size=256,
+                     intermediate_size=512, num_attention_heads=2,
+                     num_key_value_heads=2, head_dim_override=128,
+                     max_position_embeddings=512)
+    steps = 15
+
+    def run(fp8: bool):
+        os.environ["NXDA_FP8_LINEAR"] = "1" if fp8 else "0"
+        prev = torch.get_default_dtype()
+        torch.set_default_dtype(torch.bfloat16)
+        torch.manual_seed(0)
+        with torch.device("cuda"):
+            m = LlamaForCausalLM(cfg)
+        torch.set_default_dtype(prev)
+        opt = NeuronZero1Optimizer(m.parameters(), torch.optim.AdamW,
+                                   grad_clipping=True, max_norm=1.0, lr=LR)
+        curve = []
+        for x in _make_batches(cfg.vocab_size, 8, 256, steps):
+            x = x.cuda()
+            loss = m(x, labels=x)
+            loss.backward()
+            opt.step()
+            opt.zero_grad()
+            curve.append(float(loss))
+        return curve
+
+    try:
+        c8 = run(True)
+        cb = run(False)
+    finally:
+        os.environ["NXDA_FP8_LINEAR"] = "0"
+    assert cb[-1] < cb[0] - 0.3
+    assert c8[-1] < c8[0] - 0.3
+    for i, (a, b) in enumerate(zip(c8, cb)):
+        assert abs(a - b) < 0.08 + 0.05 * abs(b), (i, a, b)
