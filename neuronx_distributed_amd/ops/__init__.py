@@ -279,6 +279,30 @@ def flash_attn_available() -> bool:
     return lib is not None and hasattr(lib, "flash_attn_fwd")
 
 
+def _fa_ok_layout(t) -> bool:
+    """The flash kernels take any (batch, head, seq) strides with a DENSE
+    last dim — transpose views of BSHD activations go in with no copy."""
+    return t.stride(3) == 1
+
+
+def _fa_strides(*tensors):
+    vals = []
+    for t in tensors:
+        vals += [t.stride(0), t.stride(1), t.stride(2)]
+    arr = (ctypes.c_long * len(vals))(*vals)
+    return arr
+
+
+def _fa_alloc_like(t):
+    """Allocate an empty tensor with t's LAYOUT (so a BSHD-view input
+    produces a BSHD-view output and the downstream reshape stays a view)."""
+    B, H, S, D = t.shape
+    if t.stride(1) == D and t.stride(2) == H * D:  # BSHD transpose view
+        return torch.empty(B, S, H, D, dtype=t.dtype,
+                           device=t.device).permute(0, 2, 1, 3)
+    return torch.empty_like(t, memory_format=torch.contiguous_format)
+
+
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
@@ -287,14 +311,20 @@ class _FlashAttnFn(torch.autograd.Function):
         Hkv = k.shape[1]
         assert D == 128, "flash kernel supports D=128"
         assert q.dtype == torch.bfloat16
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        out = torch.empty_like(q)
+        if not _fa_ok_layout(q):
+            q = q.contiguous()
+        if not _fa_ok_layout(k):
+            k = k.contiguous()
+        if not _fa_ok_layout(v):
+            v = v.contiguous()
+        out = _fa_alloc_like(q)
         lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
-        lib.flash_attn_fwd(_ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(lse),
-                           ctypes.c_int(B), ctypes.c_int(Hq),
-                           ctypes.c_int(Hkv), ctypes.c_int(S),
-                           ctypes.c_float(scale), ctypes.c_int(1 if causal else 0),
-                           _stream())
+        lib.flash_attn_fwd_strided(
+            _ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(lse),
+            ctypes.c_int(B), ctypes.c_int(Hq), ctypes.c_int(Hkv),
+            ctypes.c_int(S), ctypes.c_float(scale),
+            ctypes.c_int(1 if causal else 0), _fa_strides(q, k, v, out),
+            _stream())
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.causal = causal
         ctx.scale = scale
@@ -308,19 +338,23 @@ class _FlashAttnFn(torch.autograd.Function):
             raise RuntimeError("flash_attn_bwd kernel not built")
         B, Hq, S, D = q.shape
         Hkv = k.shape[1]
-        dout = dout.contiguous()
+        if not _fa_ok_layout(dout):
+            dout = dout.contiguous()
         delta = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
-        dq = torch.empty_like(q)
+        # grads are allocated in their producer's LAYOUT so autograd's
+        # accumulation into BSHD views stays copy-free
+        dq = _fa_alloc_like(q)
         # dk/dv are written bf16 PER Q-HEAD (B,Hq,S,D); GQA replicas are
         # reduced here (each replica covers distinct Q heads -> SUM)
         dk_pq = torch.empty(B, Hq, S, D, dtype=q.dtype, device=q.device)
         dv_pq = torch.empty(B, Hq, S, D, dtype=q.dtype, device=q.device)
-        lib.flash_attn_bwd(_ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(dout),
-                           _ptr(lse), _ptr(delta), _ptr(dq), _ptr(dk_pq),
-                           _ptr(dv_pq), ctypes.c_int(B), ctypes.c_int(Hq),
-                           ctypes.c_int(Hkv), ctypes.c_int(S),
-                           ctypes.c_float(ctx.scale),
-                           ctypes.c_int(1 if ctx.causal else 0), _stream())
+        lib.flash_attn_bwd_strided(
+            _ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(dout),
+            _ptr(lse), _ptr(delta), _ptr(dq), _ptr(dk_pq), _ptr(dv_pq),
+            ctypes.c_int(B), ctypes.c_int(Hq), ctypes.c_int(Hkv),
+            ctypes.c_int(S), ctypes.c_float(ctx.scale),
+            ctypes.c_int(1 if ctx.causal else 0),
+            _fa_strides(q, k, v, out, dout, dq, dk_pq, dv_pq), _stream())
         rep = Hq // Hkv
         if rep > 1:
             dk = dk_pq.view(B, Hkv, rep, S, D).float().sum(2).to(q.dtype)

@@ -41,7 +41,14 @@ extern "C" __global__ void __launch_bounds__(512, 2)
 flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                  const short* __restrict__ vp, short* __restrict__ op,
                  float* __restrict__ lsep, int B, int Hq, int Hkv, int S,
-                 float scale, int causal, int window) {
+                 float scale, int causal, int window,
+                 long q_bs, long q_hs, long q_ss,
+                 long k_bs, long k_hs, long k_ss,
+                 long v_bs, long v_hs, long v_ss,
+                 long o_bs, long o_hs, long o_ss) {
+  // (*_bs, *_hs, *_ss) = element strides of (batch, head, seq); the last
+  // dim (d) is always dense.  BHSD contiguous: (H*S*128, S*128, 128).
+  // BSHD transpose views (no-copy model layout): (S*H*128, 128, H*128).
   // window > 0: Mistral-style sliding window — q row i attends kv rows
   // [i - window + 1, i] (causal implied); whole tiles outside the band
   // are skipped, boundary tiles masked per element.
@@ -72,8 +79,9 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
-  const long q_base = ((long)(b * Hq + h) * S) * FA_D;
-  const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
+  const long q_base = (long)b * q_bs + (long)h * q_hs;
+  const long kv_base_k = (long)b * k_bs + (long)hkv * k_hs;
+  const long kv_base_v = (long)b * v_bs + (long)hkv * v_hs;
 
   const int q0 = qblk * FA_QBLK;
   const int qw0 = q0 + wid * FA_QW;      // this wave's first q row
@@ -84,7 +92,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   frag_u qf[8];
 #pragma unroll
   for (int c = 0; c < 8; ++c) {
-    const short* src = qp + q_base + (long)q_row_ld * FA_D + c * 16 + hi * 8;
+    const short* src = qp + q_base + (long)q_row_ld * q_ss + c * 16 + hi * 8;
     qf[c].u4 = *(const uint4v*)src;
   }
 
@@ -118,10 +126,10 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     int kv0 = t * FA_KV;
     int rr0 = kv0 + st_r0 < S ? kv0 + st_r0 : S - 1;
     int rr1 = kv0 + st_r1 < S ? kv0 + st_r1 : S - 1;
-    kreg[0] = *(const uint4v*)(kp + kv_base + (long)rr0 * FA_D + st_c16 * 8);
-    kreg[1] = *(const uint4v*)(kp + kv_base + (long)rr1 * FA_D + st_c16 * 8);
-    vreg[0] = *(const uint4v*)(vp + kv_base + (long)rr0 * FA_D + st_c16 * 8);
-    vreg[1] = *(const uint4v*)(vp + kv_base + (long)rr1 * FA_D + st_c16 * 8);
+    kreg[0] = *(const uint4v*)(kp + kv_base_k + (long)rr0 * k_ss + st_c16 * 8);
+    kreg[1] = *(const uint4v*)(kp + kv_base_k + (long)rr1 * k_ss + st_c16 * 8);
+    vreg[0] = *(const uint4v*)(vp + kv_base_v + (long)rr0 * v_ss + st_c16 * 8);
+    vreg[1] = *(const uint4v*)(vp + kv_base_v + (long)rr1 * v_ss + st_c16 * 8);
   };
 
   auto write_tile = [&](int buf) {
@@ -279,7 +287,7 @@ flash_fwd_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   // ---- epilogue: normalize, store O (transposed back) and LSE ---------
   if (my_q >= S) return;
   float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
-  long o_row = q_base + (long)my_q * FA_D;
+  long o_row = (long)b * o_bs + (long)h * o_hs + (long)my_q * o_ss;
 #pragma unroll
   for (int nb = 0; nb < 4; ++nb) {
 #pragma unroll
@@ -308,7 +316,25 @@ extern "C" void flash_attn_fwd(const void* q, const void* k, const void* v,
   size_t lds = 3 * (K_TILE_B + VT_TILE_B);
   flash_fwd_kernel<<<grid, 512, lds, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out,
-      (float*)lse, B, Hq, Hkv, S, scale, causal, 0);
+      (float*)lse, B, Hq, Hkv, S, scale, causal, 0,
+      (long)Hq * S * FA_D, (long)S * FA_D, FA_D,
+      (long)Hkv * S * FA_D, (long)S * FA_D, FA_D,
+      (long)Hkv * S * FA_D, (long)S * FA_D, FA_D,
+      (long)Hq * S * FA_D, (long)S * FA_D, FA_D);
+}
+
+extern "C" void flash_attn_fwd_strided(
+    const void* q, const void* k, const void* v, void* out, void* lse,
+    int B, int Hq, int Hkv, int S, float scale, int causal,
+    const long* st, hipStream_t stream) {
+  // st = 12 longs: (bs, hs, ss) x (q, k, v, o)
+  dim3 grid(Hq, (S + FA_QBLK - 1) / FA_QBLK, B);
+  size_t lds = 3 * (K_TILE_B + VT_TILE_B);
+  flash_fwd_kernel<<<grid, 512, lds, stream>>>(
+      (const short*)q, (const short*)k, (const short*)v, (short*)out,
+      (float*)lse, B, Hq, Hkv, S, scale, causal, 0,
+      st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7], st[8],
+      st[9], st[10], st[11]);
 }
 
 extern "C" void flash_attn_fwd_window(const void* q, const void* k,
@@ -320,5 +346,9 @@ extern "C" void flash_attn_fwd_window(const void* q, const void* k,
   size_t lds = 3 * (K_TILE_B + VT_TILE_B);
   flash_fwd_kernel<<<grid, 512, lds, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out,
-      (float*)lse, B, Hq, Hkv, S, scale, 1, window);
+      (float*)lse, B, Hq, Hkv, S, scale, 1, window,
+      (long)Hq * S * FA_D, (long)S * FA_D, FA_D,
+      (long)Hkv * S * FA_D, (long)S * FA_D, FA_D,
+      (long)Hkv * S * FA_D, (long)S * FA_D, FA_D,
+      (long)Hq * S * FA_D, (long)S * FA_D, FA_D);
 }

@@ -35,11 +35,17 @@
 extern "C" __global__ void __launch_bounds__(256)
 fa_bwd_delta_kernel(const short* __restrict__ dout,
                     const short* __restrict__ out, float* __restrict__ delta,
-                    long rows) {
+                    long rows, int Hq, int S,
+                    long do_bs, long do_hs, long do_ss,
+                    long o_bs, long o_hs, long o_ss) {
+  // row enumerates (b, h, s); per-tensor (batch, head, seq) strides
   __shared__ float scratch[16];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
-    const short* d = dout + row * FA_D;
-    const short* o = out + row * FA_D;
+    const long b = row / ((long)Hq * S);
+    const long h = (row / S) % Hq;
+    const long sq = row % S;
+    const short* d = dout + b * do_bs + h * do_hs + sq * do_ss;
+    const short* o = out + b * o_bs + h * o_hs + sq * o_ss;
     float s = 0.f;
     for (int i = threadIdx.x; i < FA_D / 8; i += blockDim.x) {
       s8v dv = *(const s8v*)(d + i * 8);
@@ -202,7 +208,13 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const float* __restrict__ lsep,
                    const float* __restrict__ deltap,
                    short* __restrict__ dkp, short* __restrict__ dvp,
-                   int B, int Hq, int Hkv, int S, float scale, int causal) {
+                   int B, int Hq, int Hkv, int S, float scale, int causal,
+                   long q_bs, long q_hs, long q_ss,
+                   long k_bs, long k_hs, long k_ss,
+                   long v_bs, long v_hs, long v_ss,
+                   long do_bs, long do_hs, long do_ss,
+                   long dk_bs, long dk_hs, long dk_ss,
+                   long dv_bs, long dv_hs, long dv_ss) {
   // 8 waves / 128-row kv block: wave w -> kv rows (w>>1)*32, d-half w&1.
   // Q-TILE = 64 rows per iteration, processed as two 32-q halves that
   // REUSE the st/dpt accumulators: one barrier pair and one staging round
@@ -231,8 +243,10 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int h = blockIdx.x;
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
-  const long q_base = ((long)(b * Hq + h) * S) * FA_D;
-  const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
+  const long q_base = (long)b * q_bs + (long)h * q_hs;
+  const long do_base = (long)b * do_bs + (long)h * do_hs;
+  const long k_base = (long)b * k_bs + (long)hkv * k_hs;
+  const long v_base = (long)b * v_bs + (long)hkv * v_hs;
   const long lse_base = (long)(b * Hq + h) * S;
 
   const int kv0 = kvblk * 128 + kvg * 32;   // this wave's kv rows
@@ -243,8 +257,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     int row = my_k < S ? my_k : S - 1;
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
-      kf[c].u4 = *(const uint4v*)(kp + kv_base + (long)row * FA_D + c * 16 + hi * 8);
-      vf[c].u4 = *(const uint4v*)(vp + kv_base + (long)row * FA_D + c * 16 + hi * 8);
+      kf[c].u4 = *(const uint4v*)(kp + k_base + (long)row * k_ss + c * 16 + hi * 8);
+      vf[c].u4 = *(const uint4v*)(vp + v_base + (long)row * v_ss + c * 16 + hi * 8);
     }
   }
 
@@ -267,9 +281,9 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const float s2 = scale * LOG2E;
   int q_start = causal ? kvblk * 128 : 0;
 
-  stage_tile64(qp + q_base, q_start, FA_D, S - q_start, smem + BW64_LDS_Q,
+  stage_tile64(qp + q_base, q_start, q_ss, S - q_start, smem + BW64_LDS_Q,
                smem + BW64_LDS_QT);
-  stage_tile64(dop + q_base, q_start, FA_D, S - q_start, smem + BW64_LDS_DO,
+  stage_tile64(dop + do_base, q_start, do_ss, S - q_start, smem + BW64_LDS_DO,
                smem + BW64_LDS_DOT);
   write_ld(load_ld(q_start));
 
@@ -278,8 +292,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   for (int q0 = q_start; q0 < S; q0 += 64) {
     __syncthreads();
     if (q0 + 64 < S) {
-      nq = load_tile64(qp + q_base, q0 + 64, FA_D, S - q0 - 64);
-      ndo = load_tile64(dop + q_base, q0 + 64, FA_D, S - q0 - 64);
+      nq = load_tile64(qp + q_base, q0 + 64, q_ss, S - q0 - 64);
+      ndo = load_tile64(dop + do_base, q0 + 64, do_ss, S - q0 - 64);
       nld = load_ld(q0 + 64);
     }
     const bool any_active = !causal || (q0 + 63 >= kv0);
@@ -424,7 +438,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     int kg = kv0 + row;
     if (kg < S) {
       uint4v vv = *(const uint4v*)(otile + (row * 64 + c16 * 8) * 2);
-      *(uint4v*)(dvp + ((long)(b * Hq + h) * S + kg) * FA_D + dhalf * 64
+      *(uint4v*)(dvp + (long)b * dv_bs + (long)h * dv_hs
+                 + (long)kg * dv_ss + dhalf * 64
                  + c16 * 8) = vv;
     }
   }
@@ -445,7 +460,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     int kg = kv0 + row;
     if (kg < S) {
       uint4v vv = *(const uint4v*)(otile + (row * 64 + c16 * 8) * 2);
-      *(uint4v*)(dkp + ((long)(b * Hq + h) * S + kg) * FA_D + dhalf * 64
+      *(uint4v*)(dkp + (long)b * dk_bs + (long)h * dk_hs
+                 + (long)kg * dk_ss + dhalf * 64
                  + c16 * 8) = vv;
     }
   }
@@ -464,7 +480,12 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                  const short* __restrict__ vp, const short* __restrict__ dop,
                  const float* __restrict__ lsep,
                  const float* __restrict__ deltap, short* __restrict__ dqp,
-                 int B, int Hq, int Hkv, int S, float scale, int causal) {
+                 int B, int Hq, int Hkv, int S, float scale, int causal,
+                 long q_bs, long q_hs, long q_ss,
+                 long k_bs, long k_hs, long k_ss,
+                 long v_bs, long v_hs, long v_ss,
+                 long do_bs, long do_hs, long do_ss,
+                 long dq_bs, long dq_hs, long dq_ss) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
   // readfirstlane: provably wave-uniform -> scalar branches for the
@@ -479,8 +500,11 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int h = blockIdx.x;
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
-  const long q_base = ((long)(b * Hq + h) * S) * FA_D;
-  const long kv_base = ((long)(b * Hkv + hkv) * S) * FA_D;
+  const long q_base = (long)b * q_bs + (long)h * q_hs;
+  const long do_base = (long)b * do_bs + (long)h * do_hs;
+  const long dq_base = (long)b * dq_bs + (long)h * dq_hs;
+  const long k_base = (long)b * k_bs + (long)hkv * k_hs;
+  const long v_base = (long)b * v_bs + (long)hkv * v_hs;
   const long lse_base = (long)(b * Hq + h) * S;
 
   const int q0 = qblk * 128 + wid * 32;
@@ -491,8 +515,8 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   frag_u qf[8], dof[8];
 #pragma unroll
   for (int c = 0; c < 8; ++c) {
-    qf[c].u4 = *(const uint4v*)(qp + q_base + (long)q_ld * FA_D + c * 16 + hi * 8);
-    dof[c].u4 = *(const uint4v*)(dop + q_base + (long)q_ld * FA_D + c * 16 + hi * 8);
+    qf[c].u4 = *(const uint4v*)(qp + q_base + (long)q_ld * q_ss + c * 16 + hi * 8);
+    dof[c].u4 = *(const uint4v*)(dop + do_base + (long)q_ld * do_ss + c * 16 + hi * 8);
   }
   const float lse2 = lsep[lse_base + q_ld] * LOG2E;
   const float dlt = deltap[lse_base + q_ld];
@@ -504,15 +528,15 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   const int kv_end = causal ? min(S, qblk * 128 + 128) : S;
   const float s2 = scale * LOG2E;
 
-  stage_tile32(kp + kv_base, 0, FA_D, S, smem + DQ_LDS_K, smem + DQ_LDS_KT);
-  write_tile32_rm(load_tile32(vp + kv_base, 0, FA_D, S), smem + DQ_LDS_V);
+  stage_tile32(kp + k_base, 0, k_ss, S, smem + DQ_LDS_K, smem + DQ_LDS_KT);
+  write_tile32_rm(load_tile32(vp + v_base, 0, v_ss, S), smem + DQ_LDS_V);
 
   StageRegs nk, nv;
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     __syncthreads();
     if (kv0 + 32 < kv_end) {
-      nk = load_tile32(kp + kv_base, kv0 + 32, FA_D, S - kv0 - 32);
-      nv = load_tile32(vp + kv_base, kv0 + 32, FA_D, S - kv0 - 32);
+      nk = load_tile32(kp + k_base, kv0 + 32, k_ss, S - kv0 - 32);
+      nv = load_tile32(vp + v_base, kv0 + 32, v_ss, S - kv0 - 32);
     }
     const bool wave_active = !causal || (kv0 <= q0 + 31);
 
@@ -615,29 +639,31 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     int qg = q0 + row;
     if (qg < S) {
       uint4v vv = *(const uint4v*)(otile + (row * FA_D + c16 * 8) * 2);
-      *(uint4v*)(dqp + q_base + (long)qg * FA_D + c16 * 8) = vv;
+      *(uint4v*)(dqp + dq_base + (long)qg * dq_ss + c16 * 8) = vv;
     }
   }
 }
 
 // ---------------------------------------------------------------------------
-extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
-                               const void* out, const void* dout,
-                               const void* lse, void* delta, void* dq,
-                               void* dk, void* dv, int B, int Hq, int Hkv,
-                               int S, float scale, int causal,
-                               hipStream_t stream) {
+extern "C" void flash_attn_bwd_strided(
+    const void* q, const void* k, const void* v, const void* out,
+    const void* dout, const void* lse, void* delta, void* dq, void* dk,
+    void* dv, int B, int Hq, int Hkv, int S, float scale, int causal,
+    const long* st, hipStream_t stream) {
+  // st = 24 longs: (bs, hs, ss) x (q, k, v, o, dout, dq, dk, dv)
   long rows = (long)B * Hq * S;
   int nb = rows < 2048 ? (int)rows : 2048;
-  fa_bwd_delta_kernel<<<nb, 256, 0, stream>>>((const short*)dout,
-                                              (const short*)out,
-                                              (float*)delta, rows);
+  fa_bwd_delta_kernel<<<nb, 256, 0, stream>>>(
+      (const short*)dout, (const short*)out, (float*)delta, rows, Hq, S,
+      st[12], st[13], st[14], st[9], st[10], st[11]);
   dim3 gkv(Hq, (S + 127) / 128, B);
   size_t lds1 = 2 * 64 * FA_D * 2 + 2 * TR64_TILE_B + 128 * 4;
   fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
-      Hkv, S, scale, causal);
+      Hkv, S, scale, causal, st[0], st[1], st[2], st[3], st[4], st[5],
+      st[6], st[7], st[8], st[12], st[13], st[14], st[18], st[19], st[20],
+      st[21], st[22], st[23]);
   dim3 gq(Hq, (S + 127) / 128, B);
   // staging layout needs 26.6 KB; the epilogue reuses LDS as 4 per-wave
   // 32x128 transpose tiles = 32 KB, which dominates
@@ -645,5 +671,29 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
   fa_bwd_dq_kernel<<<gq, 256, lds2, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dq, B, Hq, Hkv, S,
-      scale, causal);
+      scale, causal, st[0], st[1], st[2], st[3], st[4], st[5], st[6],
+      st[7], st[8], st[12], st[13], st[14], st[15], st[16], st[17]);
+}
+
+extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
+                               const void* out, const void* dout,
+                               const void* lse, void* delta, void* dq,
+                               void* dk, void* dv, int B, int Hq, int Hkv,
+                               int S, float scale, int causal,
+                               hipStream_t stream) {
+  const long qd[3] = {(long)Hq * S * FA_D, (long)S * FA_D, FA_D};
+  const long kd[3] = {(long)Hkv * S * FA_D, (long)S * FA_D, FA_D};
+  long st[24];
+  for (int i = 0; i < 3; ++i) {
+    st[0 + i] = qd[i];            // q
+    st[3 + i] = kd[i];            // k
+    st[6 + i] = kd[i];            // v
+    st[9 + i] = qd[i];            // out
+    st[12 + i] = qd[i];           // dout
+    st[15 + i] = qd[i];           // dq
+    st[18 + i] = qd[i];           // dk (per-Q-head buffers)
+    st[21 + i] = qd[i];           // dv
+  }
+  flash_attn_bwd_strided(q, k, v, out, dout, lse, delta, dq, dk, dv, B, Hq,
+                         Hkv, S, scale, causal, st, stream);
 }
