@@ -162,3 +162,31 @@ def test_yaml_converter(tmp_path):
     assert cfg == {"num_hidden_layers": 12, "num_attention_heads": 16,
                    "hidden_size": 1024, "num_key_value_heads": 4,
                    "num_local_experts": 8}
+
+
+def test_flash_stride_layout_helpers():
+    """CPU coverage of the strided-flash layout plumbing: transpose views
+    are accepted without copies, outputs are allocated in the producer's
+    layout, and the ctypes stride array matches tensor.stride()."""
+    import torch
+
+    from neuronx_distributed_amd.ops import (_fa_alloc_like, _fa_ok_layout,
+                                             _fa_strides)
+
+    B, H, S, D = 2, 4, 16, 128
+    bhsd = torch.zeros(B, H, S, D)
+    bshd_view = torch.zeros(B, S, H, D).permute(0, 2, 1, 3)
+    assert _fa_ok_layout(bhsd) and _fa_ok_layout(bshd_view)
+    assert not _fa_ok_layout(torch.zeros(B, H, S, 2 * D)[..., ::2])
+
+    # layout preservation: BSHD view in -> BSHD storage out (the model's
+    # downstream .transpose(1,2).reshape stays a view)
+    o1 = _fa_alloc_like(bshd_view)
+    assert o1.shape == (B, H, S, D)
+    assert o1.transpose(1, 2).is_contiguous()
+    o2 = _fa_alloc_like(bhsd)
+    assert o2.is_contiguous()
+
+    arr = _fa_strides(bshd_view, bhsd)
+    assert list(arr) == list(bshd_view.stride()[:3]) + \
+        list(bhsd.stride()[:3])
