@@ -38,24 +38,27 @@ fa_bwd_delta_kernel(const short* __restrict__ dout,
                     long rows, int Hq, int S,
                     long do_bs, long do_hs, long do_ss,
                     long o_bs, long o_hs, long o_ss) {
-  // row enumerates (b, h, s); per-tensor (batch, head, seq) strides
-  __shared__ float scratch[16];
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  // row enumerates (b, h, s).  16 lanes own one 128-elem row (8 bf16 per
+  // lane via b128 loads, coalesced 256 B per row segment); 4 shfl_xor
+  // steps fold the 16 partials — no LDS, no block barrier.  This is a
+  // pure streaming op (read 2 x B*Hq*S*128 bf16) and runs at HBM rate;
+  // the previous one-block-per-row form was ~14x off the read floor.
+  const int sub = threadIdx.x & 15;
+  const long row0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 4;
+  const long row_step = ((long)gridDim.x * blockDim.x) >> 4;
+  for (long row = row0; row < rows; row += row_step) {
     const long b = row / ((long)Hq * S);
     const long h = (row / S) % Hq;
     const long sq = row % S;
-    const short* d = dout + b * do_bs + h * do_hs + sq * do_ss;
-    const short* o = out + b * o_bs + h * o_hs + sq * o_ss;
+    s8v dv = *(const s8v*)(dout + b * do_bs + h * do_hs + sq * do_ss
+                           + sub * 8);
+    s8v ov = *(const s8v*)(out + b * o_bs + h * o_hs + sq * o_ss + sub * 8);
     float s = 0.f;
-    for (int i = threadIdx.x; i < FA_D / 8; i += blockDim.x) {
-      s8v dv = *(const s8v*)(d + i * 8);
-      s8v ov = *(const s8v*)(o + i * 8);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) s += bits2f(dv[j]) * bits2f(ov[j]);
-    }
-    s = block_reduce_sum(s, scratch);
-    if (threadIdx.x == 0) delta[row] = s;
-    __syncthreads();
+    for (int j = 0; j < 8; ++j) s += bits2f(dv[j]) * bits2f(ov[j]);
+#pragma unroll
+    for (int off = 8; off >= 1; off >>= 1) s += __shfl_xor(s, off, 64);
+    if (sub == 0) delta[row] = s;
   }
 }
 
@@ -652,7 +655,10 @@ extern "C" void flash_attn_bwd_strided(
     const long* st, hipStream_t stream) {
   // st = 24 longs: (bs, hs, ss) x (q, k, v, o, dout, dq, dk, dv)
   long rows = (long)B * Hq * S;
-  int nb = rows < 2048 ? (int)rows : 2048;
+  // 16 lanes per row -> 16 rows per 256-thread block; cap well above the
+  // 256-CU fill point and grid-stride the rest
+  long blocks = (rows + 15) / 16;
+  int nb = blocks < 8192 ? (int)blocks : 8192;
   fa_bwd_delta_kernel<<<nb, 256, 0, stream>>>(
       (const short*)dout, (const short*)out, (float*)delta, rows, Hq, S,
       st[12], st[13], st[14], st[9], st[10], st[11]);
