@@ -161,6 +161,12 @@ __device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
 #define BW64_LDS_DOT (2 * 64 * FA_D * 2 + TR64_TILE_B)
 // 128 floats: lse*log2(e) for the 64 q rows at [0..64), delta at [64..128)
 #define BW64_LDS_LD (2 * 64 * FA_D * 2 + 2 * TR64_TILE_B)
+// cross-wave A-fragment exchange: 8 slots (kvg x q-half) x 4 frags x
+// 64 lanes x 16 B = 32 KB.  The kernel is VGPR-bound at 1 WG/CU (254
+// VGPRs -> 2 waves/SIMD), so this LDS is free occupancy-wise.
+#define BW64_LDS_X (BW64_LDS_LD + 128 * 4)
+#define BW64_X_SLOT(kvg, half, f) \
+  ((((kvg) * 2 + (half)) * 4 + (f)) * 64 * 16)
 
 struct Stage64Regs { uint4v v0, v1; };
 
@@ -299,15 +305,19 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       ndo = load_tile64(dop + do_base, q0 + 64, do_ss, S - q0 - 64);
       nld = load_ld(q0 + 64);
     }
-    const bool any_active = !causal || (q0 + 63 >= kv0);
-
-    if (any_active) {
-#pragma unroll
-      for (int half = 0; half < 2; ++half) {
-        const int qh0 = q0 + half * 32;
-        const bool half_active = !causal || (qh0 + 31 >= kv0);
-        if (!half_active) continue;
-
+    // ---- PRODUCE: this wave computes S^T/dP^T for q-half = dhalf ONLY.
+    // Its dhalf-partner wave (same kvg) computes the other half; the
+    // packed A fragments are exchanged through LDS, halving the score
+    // mfma work per wave (the old form had both waves of a kvg pair
+    // redo the identical full-d S^T/dP^T for both halves: 48 mfmas per
+    // wave-iteration where 32 are algorithmically needed — measured as
+    // dkdv running at 363 TF vs the forward's 533).
+    {
+      const int half = dhalf;
+      const int qh0 = q0 + half * 32;
+      const bool produce =
+          (!causal || (qh0 + 31 >= kv0)) && qh0 < S;
+      if (produce) {
         const int rm_row = half * 32 + col;  // row inside the 64-row tile
         const int my_kv = kv0 + col;         // this lane's kv column
 
@@ -349,22 +359,6 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           dpt[r] = p * (dpt[r] - dlt_v[r]);
         }
 
-        // prefetch the dV/dK B fragments NOW — they are independent of the
-        // pack below, so the 8 b128 LDS reads land under the pack VALU
-        frag_u dofr[2][2], qfr2[2][2];
-#pragma unroll
-        for (int cq = 0; cq < 2; ++cq)
-#pragma unroll
-          for (int nb = 0; nb < 2; ++nb) {
-            int d = dhalf * 64 + nb * 32 + col;
-            dofr[cq][nb].u4 = *(const uint4v*)(smem + BW64_LDS_DOT
-                                               + d * (TR64_PITCH * 2)
-                                               + (half * 32 + cq * 16 + hi * 8) * 2);
-            qfr2[cq][nb].u4 = *(const uint4v*)(smem + BW64_LDS_QT
-                                               + d * (TR64_PITCH * 2)
-                                               + (half * 32 + cq * 16 + hi * 8) * 2);
-          }
-
         // repack accumulator rows (q) into A-fragment k-dim in REGISTERS
         // (same pack pairs + permlane32_swap as the forward kernel's P
         // repack; lane dim = kv is already in place)
@@ -401,17 +395,55 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
           dA[cc].u[2] = c2; dA[cc].u[3] = c3;
         }
 
-        // dV/dK over this wave's 64-column d-half, q chunk = this half
-        __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-        for (int cq = 0; cq < 2; ++cq)
-#pragma unroll
-          for (int nb = 0; nb < 2; ++nb) {
-            dv_acc[nb] = mfma_bf16(pA[cq].bf, dofr[cq][nb].bf, dv_acc[nb]);
-            dk_acc[nb] = mfma_bf16(dA[cq].bf, qfr2[cq][nb].bf, dk_acc[nb]);
-          }
-        __builtin_amdgcn_s_setprio(0);
+        // publish this half's A fragments for both d-half waves of the
+        // kvg pair (frag-major layout: consecutive lanes x 16 B ->
+        // conflict-free ds_write_b128)
+        char* xb = smem + BW64_LDS_X;
+        *(uint4v*)(xb + BW64_X_SLOT(kvg, half, 0) + lane * 16) = pA[0].u4;
+        *(uint4v*)(xb + BW64_X_SLOT(kvg, half, 1) + lane * 16) = pA[1].u4;
+        *(uint4v*)(xb + BW64_X_SLOT(kvg, half, 2) + lane * 16) = dA[0].u4;
+        *(uint4v*)(xb + BW64_X_SLOT(kvg, half, 3) + lane * 16) = dA[1].u4;
       }
+    }
+    __syncthreads();  // exchange visible
+
+    // ---- CONSUME: both q halves on this wave's 64-column d-half -------
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int qh0 = q0 + half * 32;
+      if ((causal && qh0 + 31 < kv0) || qh0 >= S) continue;
+
+      // prefetch the dV/dK B fragments first — independent of the slot
+      // reads, so all 12 b128 LDS loads pipeline into one counted wait
+      frag_u dofr[2][2], qfr2[2][2];
+#pragma unroll
+      for (int cq = 0; cq < 2; ++cq)
+#pragma unroll
+        for (int nb = 0; nb < 2; ++nb) {
+          int d = dhalf * 64 + nb * 32 + col;
+          dofr[cq][nb].u4 = *(const uint4v*)(smem + BW64_LDS_DOT
+                                             + d * (TR64_PITCH * 2)
+                                             + (half * 32 + cq * 16 + hi * 8) * 2);
+          qfr2[cq][nb].u4 = *(const uint4v*)(smem + BW64_LDS_QT
+                                             + d * (TR64_PITCH * 2)
+                                             + (half * 32 + cq * 16 + hi * 8) * 2);
+        }
+      frag_u pA[2], dA[2];
+      const char* xb = smem + BW64_LDS_X;
+      pA[0].u4 = *(const uint4v*)(xb + BW64_X_SLOT(kvg, half, 0) + lane * 16);
+      pA[1].u4 = *(const uint4v*)(xb + BW64_X_SLOT(kvg, half, 1) + lane * 16);
+      dA[0].u4 = *(const uint4v*)(xb + BW64_X_SLOT(kvg, half, 2) + lane * 16);
+      dA[1].u4 = *(const uint4v*)(xb + BW64_X_SLOT(kvg, half, 3) + lane * 16);
+
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int cq = 0; cq < 2; ++cq)
+#pragma unroll
+        for (int nb = 0; nb < 2; ++nb) {
+          dv_acc[nb] = mfma_bf16(pA[cq].bf, dofr[cq][nb].bf, dv_acc[nb]);
+          dk_acc[nb] = mfma_bf16(dA[cq].bf, qfr2[cq][nb].bf, dk_acc[nb]);
+        }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     __syncthreads();
@@ -663,7 +695,9 @@ extern "C" void flash_attn_bwd_strided(
       (const short*)dout, (const short*)out, (float*)delta, rows, Hq, S,
       st[12], st[13], st[14], st[9], st[10], st[11]);
   dim3 gkv(Hq, (S + 127) / 128, B);
-  size_t lds1 = 2 * 64 * FA_D * 2 + 2 * TR64_TILE_B + 128 * 4;
+  // staging (70,144 B) + A-fragment exchange (32,768 B) = 102,912 B; the
+  // kernel is VGPR-bound at 1 WG/CU so the extra LDS costs no occupancy
+  size_t lds1 = BW64_LDS_X + 8 * 4 * 64 * 16;
   fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
