@@ -60,8 +60,12 @@ extern "C" __global__ void __launch_bounds__(256)
 rmsnorm_bwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
                    const short* __restrict__ dy,
                    const float* __restrict__ rstd_in,
+                   const short* __restrict__ dpass,
                    short* __restrict__ dx, float* __restrict__ dw_f32,
                    int H, int rows) {
+  // dpass != nullptr: dx += dpass — the residual fork's pass-through
+  // gradient is folded into this pass (training fused add+RMSNorm site),
+  // replacing a separate 3-pass eager add kernel.
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* dw_part = smem;           // [H]
   float* scratch = smem + H;       // [16]
@@ -88,16 +92,21 @@ rmsnorm_bwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
     dot = block_reduce_sum(dot, scratch);
     float k = rstd * rstd * rstd * dot / (float)H;
 
+    const short* pr = dpass ? dpass + (long)row * H : nullptr;
     for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
       s8v xv = *(const s8v*)(xr + i * 8);
       s8v dv = *(const s8v*)(dyr + i * 8);
       s8v wv = *(const s8v*)(w + i * 8);
+      s8v pv;
+      if (pr) pv = *(const s8v*)(pr + i * 8);
       s8v o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xf = bits2f(xv[j]);
         float df = bits2f(dv[j]);
-        o[j] = f2bits(rstd * bits2f(wv[j]) * df - k * xf);
+        float g = rstd * bits2f(wv[j]) * df - k * xf;
+        if (pr) g += bits2f(pv[j]);
+        o[j] = f2bits(g);
         dw_part[i * 8 + j] += df * xf * rstd;  // thread-exclusive slot per i
       }
       *(s8v*)(dxr + i * 8) = o;
@@ -121,7 +130,8 @@ add_rmsnorm_fwd_kernel(const short* __restrict__ res_in,
                        const short* __restrict__ delta,
                        const short* __restrict__ w,
                        short* __restrict__ res_out,
-                       short* __restrict__ normed, int H, float eps,
+                       short* __restrict__ normed,
+                       float* __restrict__ rstd_out, int H, float eps,
                        int rows) {
   __shared__ float scratch[16];
   int row = blockIdx.x;
@@ -156,6 +166,7 @@ add_rmsnorm_fwd_kernel(const short* __restrict__ res_in,
   }
   ss = block_reduce_sum(ss, scratch);
   float rstd = rsqrtf(ss / (float)H + eps);
+  if (rstd_out && threadIdx.x == 0) rstd_out[row] = rstd;
 
   ci = 0;
   for (int i = threadIdx.x; i < nvec; i += blockDim.x, ++ci) {
@@ -175,7 +186,17 @@ extern "C" void add_rmsnorm_fwd(const void* res_in, const void* delta,
                                 hipStream_t stream) {
   add_rmsnorm_fwd_kernel<<<rows, 256, 0, stream>>>(
       (const short*)res_in, (const short*)delta, (const short*)w,
-      (short*)res_out, (short*)normed, H, eps, rows);
+      (short*)res_out, (short*)normed, nullptr, H, eps, rows);
+}
+
+// training variant: also emits per-row rstd for the backward
+extern "C" void add_rmsnorm_fwd_train(const void* res_in, const void* delta,
+                                      const void* w, void* res_out,
+                                      void* normed, void* rstd, int rows,
+                                      int H, float eps, hipStream_t stream) {
+  add_rmsnorm_fwd_kernel<<<rows, 256, 0, stream>>>(
+      (const short*)res_in, (const short*)delta, (const short*)w,
+      (short*)res_out, (short*)normed, (float*)rstd, H, eps, rows);
 }
 
 extern "C" void rmsnorm_fwd(const void* x, const void* w, void* out,
@@ -192,5 +213,17 @@ extern "C" void rmsnorm_bwd(const void* x, const void* w, const void* dy,
   size_t lds = (size_t)(H + 16) * sizeof(float);
   rmsnorm_bwd_kernel<<<blocks, 256, lds, stream>>>(
       (const short*)x, (const short*)w, (const short*)dy, (const float*)rstd,
-      (short*)dx, (float*)dw_f32, H, rows);
+      nullptr, (short*)dx, (float*)dw_f32, H, rows);
+}
+
+// fused add+RMSNorm backward: dx = rmsnorm_bwd(dy) + dpass in one pass
+extern "C" void rmsnorm_bwd_add(const void* x, const void* w, const void* dy,
+                                const void* rstd, const void* dpass, void* dx,
+                                void* dw_f32, int rows, int H,
+                                hipStream_t stream) {
+  int blocks = rows < 2048 ? rows : 2048;
+  size_t lds = (size_t)(H + 16) * sizeof(float);
+  rmsnorm_bwd_kernel<<<blocks, 256, lds, stream>>>(
+      (const short*)x, (const short*)w, (const short*)dy, (const float*)rstd,
+      (const short*)dpass, (short*)dx, (float*)dw_f32, H, rows);
 }

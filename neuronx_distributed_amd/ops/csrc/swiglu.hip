@@ -9,11 +9,17 @@
 extern "C" __global__ void __launch_bounds__(256)
 swiglu_fwd_kernel(const short* __restrict__ x, short* __restrict__ out,
                   long N, int I) {
+  // ONE int64 div/mod at entry, then carry-advance (row, i) each
+  // iteration; v_rcp_f32 for the sigmoid (1-ulp fp32 — far below bf16
+  // output rounding).
   int nvec = I >> 3;
-  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < N * nvec; idx += (long)gridDim.x * blockDim.x) {
-    long row = idx / nvec;
-    int i = (int)(idx % nvec);
+  const long stride = (long)gridDim.x * blockDim.x;
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long row = idx / nvec;
+  int i = (int)(idx - row * nvec);
+  const int di = (int)(stride % nvec);
+  const long drow = stride / nvec;
+  for (; idx < N * nvec; idx += stride) {
     const short* g = x + row * 2 * I + i * 8;
     const short* u = x + row * 2 * I + I + i * 8;
     s8v gv = *(const s8v*)g;
@@ -22,21 +28,29 @@ swiglu_fwd_kernel(const short* __restrict__ x, short* __restrict__ out,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float gf = bits2f(gv[j]);
-      float s = gf / (1.f + __builtin_amdgcn_exp2f(-gf * 1.4426950408889634f));
+      float s = gf * __builtin_amdgcn_rcpf(
+          1.f + __builtin_amdgcn_exp2f(-gf * 1.4426950408889634f));
       o[j] = f2bits(s * bits2f(uv[j]));
     }
     *(s8v*)(out + row * I + i * 8) = o;
+    row += drow;
+    i += di;
+    if (i >= nvec) { i -= nvec; ++row; }
   }
 }
 
 extern "C" __global__ void __launch_bounds__(256)
 swiglu_bwd_kernel(const short* __restrict__ x, const short* __restrict__ dy,
                   short* __restrict__ dx, long N, int I) {
+  // same div-free iteration + fast-rcp structure as the forward
   int nvec = I >> 3;
-  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < N * nvec; idx += (long)gridDim.x * blockDim.x) {
-    long row = idx / nvec;
-    int i = (int)(idx % nvec);
+  const long stride = (long)gridDim.x * blockDim.x;
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long row = idx / nvec;
+  int i = (int)(idx - row * nvec);
+  const int di = (int)(stride % nvec);
+  const long drow = stride / nvec;
+  for (; idx < N * nvec; idx += stride) {
     const short* g = x + row * 2 * I + i * 8;
     const short* u = x + row * 2 * I + I + i * 8;
     const short* d = dy + row * I + i * 8;
@@ -48,13 +62,17 @@ swiglu_bwd_kernel(const short* __restrict__ x, const short* __restrict__ dy,
     for (int j = 0; j < 8; ++j) {
       float gf = bits2f(gv[j]);
       float df = bits2f(dv[j]);
-      float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-gf * 1.4426950408889634f));
+      float sig = __builtin_amdgcn_rcpf(
+          1.f + __builtin_amdgcn_exp2f(-gf * 1.4426950408889634f));
       float s = gf * sig;
       dg[j] = f2bits(df * bits2f(uv[j]) * (sig + s * (1.f - sig)));
       du[j] = f2bits(df * s);
     }
     *(s8v*)(dx + row * 2 * I + i * 8) = dg;
     *(s8v*)(dx + row * 2 * I + I + i * 8) = du;
+    row += drow;
+    i += di;
+    if (i >= nvec) { i -= nvec; ++row; }
   }
 }
 
