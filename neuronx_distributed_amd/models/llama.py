@@ -14,6 +14,7 @@ MI355X-first:
 """
 
 import math
+import os
 from dataclasses import dataclass, field
 from typing import Optional
 
@@ -329,6 +330,24 @@ class LlamaDecoderLayer(nn.Module):
             self.post_attention_layernorm.variance_epsilon)
         return residual, self.mlp(normed2)
 
+    def forward_fused_train(self, residual, delta, cos, sin, pos_offset=0):
+        """Training twin of ``forward_fused``: the residual add fuses into
+        the next RMSNorm through the autograd op (ops.add_rmsnorm_train),
+        whose backward folds the residual fork's pass-through gradient
+        into the RMSNorm backward kernel — each residual site costs one
+        kernel per direction instead of norm + eager add."""
+        if delta is None:
+            normed = self.input_layernorm(residual)
+        else:
+            residual, normed = ops.add_rmsnorm_train(
+                delta, residual, self.input_layernorm.weight,
+                self.input_layernorm.variance_epsilon)
+        attn_out = self.self_attn(normed, cos, sin, pos_offset, None)
+        residual, normed2 = ops.add_rmsnorm_train(
+            attn_out, residual, self.post_attention_layernorm.weight,
+            self.post_attention_layernorm.variance_epsilon)
+        return residual, self.mlp(normed2)
+
 
 class LlamaModel(nn.Module):
     def __init__(self, config: LlamaConfig):
@@ -377,6 +396,25 @@ class LlamaModel(nn.Module):
                     pos_offset, kc)
             _, hidden = ops.add_rmsnorm(residual, delta, self.norm.weight,
                                         self.norm.variance_epsilon)
+            return hidden
+        if (not isinstance(hidden, torch.fx.Proxy)
+                and self.training and torch.is_grad_enabled()
+                and hidden.is_cuda and hidden.dtype == torch.bfloat16
+                and not self.config.sequence_parallel_enabled
+                and self.config.hidden_size % 8 == 0
+                and kv_caches is None
+                and ops.add_rmsnorm_train_available()
+                and os.environ.get("NXDA_FUSED_NORM", "1") == "1"):
+            # fused residual+norm TRAINING path (autograd twin of the
+            # inference path above; NXDA_FUSED_NORM=0 reverts)
+            residual, delta = hidden, None
+            for layer in self.layers:
+                residual, delta = layer.forward_fused_train(
+                    residual, delta, self.rope_cos, self.rope_sin,
+                    pos_offset)
+            _, hidden = ops.add_rmsnorm_train(delta, residual,
+                                              self.norm.weight,
+                                              self.norm.variance_epsilon)
             return hidden
         for i, layer in enumerate(self.layers):
             kc = kv_caches[i] if kv_caches is not None else None

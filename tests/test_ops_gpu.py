@@ -531,3 +531,61 @@ def test_flash_attn_sliding_window(B, Hq, Hkv, S, W):
     ref = _torch_reference(q.float(), k.float(), v.float(), causal=True,
                            window=W)
     _cmp(out, ref, atol=3e-2, rtol=3e-2, name=f"window S={S} W={W}")
+
+
+def test_add_rmsnorm_train_matches_unfused_autograd():
+    """Training fused add+RMSNorm (fwd + the dpass-folding backward) vs
+    the unfused bf16 add -> ops.rmsnorm chain: forward outputs equal and
+    gradients equal (the kernel computes norm stats on the rounded sum,
+    so the chains are arithmetically identical)."""
+    torch.manual_seed(11)
+    rows, H = 512, 1024
+    delta = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(H, dtype=torch.bfloat16, device="cuda")
+
+    d1 = delta.clone().requires_grad_(True)
+    r1 = res.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    h1, n1 = ops.add_rmsnorm_train(d1, r1, w1, 1e-5)
+    # both outputs used downstream (residual carried + normed consumed),
+    # like a decoder layer
+    loss1 = (n1.float().pow(2).sum() + 0.5 * h1.float().sum())
+    loss1.backward()
+
+    d2 = delta.clone().requires_grad_(True)
+    r2 = res.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    h2 = d2 + r2
+    n2 = ops.rmsnorm(h2, w2, 1e-5)
+    loss2 = (n2.float().pow(2).sum() + 0.5 * h2.float().sum())
+    loss2.backward()
+    torch.cuda.synchronize()
+
+    assert torch.equal(h1, h2)
+    _cmp(n1, n2, atol=1e-3, name="normed")
+    _cmp(d1.grad, d2.grad, atol=2e-2, rtol=2e-2, name="ddelta")
+    _cmp(r1.grad, r2.grad, atol=2e-2, rtol=2e-2, name="dres")
+    _cmp(w1.grad, w2.grad, atol=2e-1, rtol=2e-2, name="dw")
+
+
+def test_add_rmsnorm_train_final_norm_no_dh():
+    """Final-norm use: the carried h output is dropped (dh is None) —
+    the backward must fall back to the plain rmsnorm_bwd kernel."""
+    torch.manual_seed(12)
+    rows, H = 64, 512
+    delta = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda",
+                        requires_grad=True)
+    res = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(H, dtype=torch.bfloat16, device="cuda",
+                    requires_grad=True)
+    _, normed = ops.add_rmsnorm_train(delta, res, w, 1e-5)
+    normed.float().pow(2).sum().backward()
+
+    d2 = delta.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    n2 = ops.rmsnorm(d2 + res, w2, 1e-5)
+    n2.float().pow(2).sum().backward()
+    torch.cuda.synchronize()
+    _cmp(delta.grad, d2.grad, atol=2e-2, rtol=2e-2, name="ddelta")
+    _cmp(w.grad, w2.grad, atol=2e-1, rtol=2e-2, name="dw")

@@ -190,3 +190,30 @@ def test_flash_stride_layout_helpers():
     arr = _fa_strides(bshd_view, bhsd)
     assert list(arr) == list(bshd_view.stride()[:3]) + \
         list(bhsd.stride()[:3])
+
+
+def test_add_rmsnorm_train_cpu_fallback():
+    """Composed CPU path of the fused training add+RMSNorm autograd op
+    (the GPU kernel path is covered by test_ops_gpu)."""
+    import torch
+
+    from neuronx_distributed_amd import ops
+
+    torch.manual_seed(3)
+    d = torch.randn(8, 64, requires_grad=True)
+    r = torch.randn(8, 64, requires_grad=True)
+    w = torch.randn(64, requires_grad=True)
+    h, n = ops.add_rmsnorm_train(d, r, w, 1e-5)
+    (n.pow(2).sum() + 0.5 * h.sum()).backward()
+
+    d2 = d.detach().clone().requires_grad_(True)
+    r2 = r.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    h2 = d2 + r2
+    n2 = h2 * torch.rsqrt(h2.pow(2).mean(-1, keepdim=True) + 1e-5) * w2
+    (n2.pow(2).sum() + 0.5 * h2.sum()).backward()
+    assert torch.allclose(h, h2, atol=1e-6)
+    assert torch.allclose(n, n2, atol=1e-5)
+    assert torch.allclose(d.grad, d2.grad, atol=1e-4)
+    assert torch.allclose(r.grad, r2.grad, atol=1e-4)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-3)
