@@ -56,9 +56,19 @@ def main():
         try:
             # pre-tuned hipBLASLt/rocBLAS GEMM selections (profiles/, gfx950);
             # tuning itself stays off - unknown shapes use defaults
-            tun = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                               "profiles", "tunableop_gfx950.csv")
-            if os.path.exists(tun) and \
+            tun = os.environ.get(
+                "NXDA_TUNABLEOP_FILE",
+                os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                             "profiles", "tunableop_gfx950.csv"))
+            if os.environ.get("NXDA_TUNE", "0") == "1":
+                # tuning run: search hipBLASLt algos for this run's GEMM
+                # shapes and write the selections (NXDA_TUNE_OUT)
+                torch.cuda.tunable.set_filename(
+                    os.environ.get("NXDA_TUNE_OUT", "tunableop_out.csv"),
+                    insert_device_ordinal=False)
+                torch.cuda.tunable.tuning_enable(True)
+                torch.cuda.tunable.enable(True)
+            elif os.path.exists(tun) and \
                     os.environ.get("NXDA_TUNABLEOP", "1") == "1":
                 torch.cuda.tunable.set_filename(tun, insert_device_ordinal=False)
                 torch.cuda.tunable.tuning_enable(False)
@@ -208,6 +218,9 @@ def main():
             },
         }
         print(json.dumps(result), flush=True)
+    if on_gpu and os.environ.get("NXDA_TUNE", "0") == "1" and \
+            hasattr(torch.cuda.tunable, "write_file"):
+        torch.cuda.tunable.write_file()
     dist.destroy_process_group()
 
 
