@@ -19,6 +19,8 @@ return plain tensors.
 from dataclasses import dataclass
 from typing import Optional
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -150,6 +152,29 @@ class MixtralDecoderLayer(nn.Module):
             moe_out, router_logits = out, None
         return h + moe_out, router_logits
 
+    def forward_fused_train(self, residual, delta, cos, sin, pos_offset=0):
+        """Training path carrying (residual, delta): every residual add
+        fuses into the next RMSNorm via ops.add_rmsnorm_train (same
+        machinery as LlamaDecoderLayer.forward_fused_train)."""
+        if delta is None:
+            normed = self.input_layernorm(residual)
+        else:
+            residual, normed = ops.add_rmsnorm_train(
+                delta, residual, self.input_layernorm.weight,
+                self.input_layernorm.variance_epsilon)
+        attn_out = self.self_attn(normed, cos, sin, pos_offset, None)
+        residual, normed2 = ops.add_rmsnorm_train(
+            attn_out, residual, self.post_attention_layernorm.weight,
+            self.post_attention_layernorm.variance_epsilon)
+        if not self.is_moe:
+            return residual, self.mlp(normed2), None
+        out = self.block_sparse_moe(normed2)
+        if isinstance(out, tuple):
+            moe_out, router_logits = out
+        else:
+            moe_out, router_logits = out, None
+        return residual, moe_out, router_logits
+
 
 class MixtralModel(nn.Module):
     def __init__(self, config: MixtralConfig):
@@ -186,6 +211,26 @@ class MixtralModel(nn.Module):
             hidden = hidden.transpose(0, 1).contiguous()
             hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
         all_router_logits = []
+        if (not isinstance(hidden, torch.fx.Proxy)
+                and self.training and torch.is_grad_enabled()
+                and hidden.is_cuda and hidden.dtype == torch.bfloat16
+                and not self.config.sequence_parallel_enabled
+                and self.config.hidden_size % 8 == 0
+                and kv_caches is None
+                and ops.add_rmsnorm_train_available()
+                and os.environ.get("NXDA_FUSED_NORM", "1") == "1"):
+            # fused residual+norm training path (see models/llama.py)
+            residual, delta = hidden, None
+            for layer in self.layers:
+                residual, delta, router_logits = layer.forward_fused_train(
+                    residual, delta, self.rope_cos, self.rope_sin,
+                    pos_offset)
+                if router_logits is not None:
+                    all_router_logits.append(router_logits)
+            _, hidden = ops.add_rmsnorm_train(delta, residual,
+                                              self.norm.weight,
+                                              self.norm.variance_epsilon)
+            return hidden, all_router_logits
         for i, layer in enumerate(self.layers):
             kc = kv_caches[i] if kv_caches is not None else None
             hidden, router_logits = layer(hidden, self.rope_cos,
