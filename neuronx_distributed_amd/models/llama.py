@@ -297,6 +297,26 @@ class LlamaMLP(nn.Module):
         return self.down_proj(ops.swiglu(self.gate_up_proj(x)))
 
 
+def fused_norm_train_ok(config, hidden, training=True) -> bool:
+    """Gate for the fused residual+RMSNorm TRAINING path.  True when the
+    fused autograd op applies: training under grad, bf16 CUDA rows with
+    H % 8 == 0 and the kernel built.  NXDA_FUSED_NORM=0 disables;
+    NXDA_FUSED_NORM_FORCE=1 bypasses the cuda/bf16/kernel checks so CPU
+    tests can exercise the fused wiring via the op's composed fallback."""
+    if isinstance(hidden, torch.fx.Proxy):
+        return False
+    if not training or not torch.is_grad_enabled():
+        return False
+    if os.environ.get("NXDA_FUSED_NORM", "1") != "1":
+        return False
+    if config.hidden_size % 8 != 0:
+        return False
+    if os.environ.get("NXDA_FUSED_NORM_FORCE", "0") == "1":
+        return True
+    return (hidden.is_cuda and hidden.dtype == torch.bfloat16
+            and ops.add_rmsnorm_train_available())
+
+
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, config: LlamaConfig):
         super().__init__()
@@ -397,16 +417,14 @@ class LlamaModel(nn.Module):
             _, hidden = ops.add_rmsnorm(residual, delta, self.norm.weight,
                                         self.norm.variance_epsilon)
             return hidden
-        if (not isinstance(hidden, torch.fx.Proxy)
-                and self.training and torch.is_grad_enabled()
-                and hidden.is_cuda and hidden.dtype == torch.bfloat16
-                and not self.config.sequence_parallel_enabled
-                and self.config.hidden_size % 8 == 0
-                and kv_caches is None
-                and ops.add_rmsnorm_train_available()
-                and os.environ.get("NXDA_FUSED_NORM", "1") == "1"):
+        if fused_norm_train_ok(self.config, hidden, self.training) \
+                and kv_caches is None:
             # fused residual+norm TRAINING path (autograd twin of the
-            # inference path above; NXDA_FUSED_NORM=0 reverts)
+            # inference path above; NXDA_FUSED_NORM=0 reverts).  Works
+            # under SP too: the adds/norms are per-row ops on the local
+            # seq shard, and the RMSNorm weights keep their
+            # sequence_parallel_enabled PARAM attribute for the SP grad
+            # all-reduce regardless of which op computes the grad.
             residual, delta = hidden, None
             for layer in self.layers:
                 residual, delta = layer.forward_fused_train(
@@ -415,11 +433,12 @@ class LlamaModel(nn.Module):
             _, hidden = ops.add_rmsnorm_train(delta, residual,
                                               self.norm.weight,
                                               self.norm.variance_epsilon)
-            return hidden
-        for i, layer in enumerate(self.layers):
-            kc = kv_caches[i] if kv_caches is not None else None
-            hidden = layer(hidden, self.rope_cos, self.rope_sin, pos_offset, kc)
-        hidden = self.norm(hidden)
+        else:
+            for i, layer in enumerate(self.layers):
+                kc = kv_caches[i] if kv_caches is not None else None
+                hidden = layer(hidden, self.rope_cos, self.rope_sin,
+                               pos_offset, kc)
+            hidden = self.norm(hidden)
         if self.config.sequence_parallel_enabled:
             hidden = gather_from_sequence_parallel_region(
                 hidden, seq_dim=0, to_model_parallel=True)

@@ -36,7 +36,8 @@ from ..parallel.mappings import (
     scatter_to_sequence_parallel_region,
 )
 from .. import ops
-from .llama import LlamaAttention, LlamaMLP, RMSNorm, _init_method
+from .llama import (LlamaAttention, LlamaMLP, RMSNorm, _init_method,
+                    fused_norm_train_ok)
 
 torch.fx.wrap("parallel_cross_entropy")
 
@@ -211,15 +212,11 @@ class MixtralModel(nn.Module):
             hidden = hidden.transpose(0, 1).contiguous()
             hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
         all_router_logits = []
-        if (not isinstance(hidden, torch.fx.Proxy)
-                and self.training and torch.is_grad_enabled()
-                and hidden.is_cuda and hidden.dtype == torch.bfloat16
-                and not self.config.sequence_parallel_enabled
-                and self.config.hidden_size % 8 == 0
-                and kv_caches is None
-                and ops.add_rmsnorm_train_available()
-                and os.environ.get("NXDA_FUSED_NORM", "1") == "1"):
-            # fused residual+norm training path (see models/llama.py)
+        if fused_norm_train_ok(self.config, hidden, self.training) \
+                and kv_caches is None:
+            # fused residual+norm training path (see models/llama.py;
+            # SP-safe — per-row ops on the local seq shard, norm-weight
+            # SP grad marking is a param attribute)
             residual, delta = hidden, None
             for layer in self.layers:
                 residual, delta, router_logits = layer.forward_fused_train(
@@ -230,14 +227,14 @@ class MixtralModel(nn.Module):
             _, hidden = ops.add_rmsnorm_train(delta, residual,
                                               self.norm.weight,
                                               self.norm.variance_epsilon)
-            return hidden, all_router_logits
-        for i, layer in enumerate(self.layers):
-            kc = kv_caches[i] if kv_caches is not None else None
-            hidden, router_logits = layer(hidden, self.rope_cos,
-                                          self.rope_sin, pos_offset, kc)
-            if router_logits is not None:
-                all_router_logits.append(router_logits)
-        hidden = self.norm(hidden)
+        else:
+            for i, layer in enumerate(self.layers):
+                kc = kv_caches[i] if kv_caches is not None else None
+                hidden, router_logits = layer(hidden, self.rope_cos,
+                                              self.rope_sin, pos_offset, kc)
+                if router_logits is not None:
+                    all_router_logits.append(router_logits)
+            hidden = self.norm(hidden)
         if self.config.sequence_parallel_enabled:
             hidden = gather_from_sequence_parallel_region(
                 hidden, seq_dim=0, to_model_parallel=True)

@@ -256,3 +256,47 @@ def _dp2_worker(rank, world):
 
 def test_dp2_grad_parity():
     run_distributed(_dp2_worker, world_size=2)
+
+
+def _fused_norm_sp_worker(rank, world):
+    """tp2 + SP: forced fused residual+RMSNorm training path (composed
+    CPU fallback via NXDA_FUSED_NORM_FORCE=1) vs the plain path — loss
+    and norm-weight/embedding grads must match, and the norm weights
+    must keep their SP grad marking."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(7)
+    x = torch.randint(0, 256, (2, 16))
+
+    results = {}
+    for force in ("0", "1"):
+        os.environ["NXDA_FUSED_NORM_FORCE"] = force
+        try:
+            model = _build_model(sp=True)
+            model.train()
+            loss = model(x, labels=x)
+            loss.backward()
+            lyr = model.model.layers[0]
+            assert lyr.input_layernorm.weight.sequence_parallel_enabled
+            results[force] = (
+                float(loss),
+                lyr.input_layernorm.weight.grad.clone(),
+                lyr.post_attention_layernorm.weight.grad.clone(),
+                model.model.norm.weight.grad.clone(),
+                model.model.embed_tokens.weight.grad.clone(),
+            )
+        finally:
+            os.environ["NXDA_FUSED_NORM_FORCE"] = "0"
+    l0, g0a, g0b, g0n, g0e = results["0"]
+    l1, g1a, g1b, g1n, g1e = results["1"]
+    assert abs(l0 - l1) < 1e-5 * (1 + abs(l0)), (l0, l1)
+    assert torch.allclose(g0a, g1a, atol=1e-5), (g0a - g1a).abs().max()
+    assert torch.allclose(g0b, g1b, atol=1e-5), (g0b - g1b).abs().max()
+    assert torch.allclose(g0n, g1n, atol=1e-5), (g0n - g1n).abs().max()
+    assert torch.allclose(g0e, g1e, atol=1e-4), (g0e - g1e).abs().max()
+    return l1
+
+
+def test_fused_norm_train_sp_matches_plain_tp2():
+    run_distributed(_fused_norm_sp_worker, world_size=2)
