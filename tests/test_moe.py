@@ -546,3 +546,50 @@ def _ep2_parity_worker(rank, world):
 
 def test_moe_ep2_numerical_parity():
     run_distributed(_ep2_parity_worker, world_size=2)
+
+
+def _mixtral_fused_norm_worker(rank, world):
+    """Forced fused residual+RMSNorm path (NXDA_FUSED_NORM_FORCE=1, CPU
+    composed fallback) vs plain path on the MoE model: loss and grads of
+    the norm weights + router must match."""
+    import os
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.models import (MixtralForCausalLM,
+                                                get_moe_config)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (2, 16))
+
+    results = {}
+    for force in ("0", "1"):
+        os.environ["NXDA_FUSED_NORM_FORCE"] = force
+        try:
+            torch.manual_seed(0)
+            m = MixtralForCausalLM(get_moe_config("tiny-moe"))
+            m.train()
+            loss = m(x, labels=x)
+            loss.backward()
+            lyr = m.model.layers[0]
+            results[force] = (
+                float(loss),
+                lyr.input_layernorm.weight.grad.clone(),
+                lyr.post_attention_layernorm.weight.grad.clone(),
+                lyr.block_sparse_moe.router.linear_router.weight.grad.clone()
+                if hasattr(lyr.block_sparse_moe.router, "linear_router")
+                else lyr.block_sparse_moe.router.weight.grad.clone(),
+            )
+        finally:
+            os.environ["NXDA_FUSED_NORM_FORCE"] = "0"
+    l0, a0, b0, r0 = results["0"]
+    l1, a1, b1, r1 = results["1"]
+    assert abs(l0 - l1) < 1e-5 * (1 + abs(l0)), (l0, l1)
+    assert torch.allclose(a0, a1, atol=1e-5), (a0 - a1).abs().max()
+    assert torch.allclose(b0, b1, atol=1e-5), (b0 - b1).abs().max()
+    assert torch.allclose(r0, r1, atol=1e-5), (r0 - r1).abs().max()
+    return l1
+
+
+def test_mixtral_fused_norm_matches_plain():
+    run_distributed(_mixtral_fused_norm_worker, world_size=1)
