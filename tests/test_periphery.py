@@ -217,3 +217,44 @@ def test_add_rmsnorm_train_cpu_fallback():
     assert torch.allclose(d.grad, d2.grad, atol=1e-4)
     assert torch.allclose(r.grad, r2.grad, atol=1e-4)
     assert torch.allclose(w.grad, w2.grad, atol=1e-3)
+
+
+def test_parallel_layer_norm_sp_tagging():
+    """parallel/layer_norm.py: numerics equal nn.LayerNorm and the
+    weight/bias carry the sequence_parallel_enabled attribute grads.py's
+    SP all-reduce looks for (reference parallel_layers/layer_norm.py)."""
+    import torch
+
+    from neuronx_distributed_amd.parallel.layer_norm import LayerNorm
+
+    ln = LayerNorm(32, sequence_parallel_enabled=True)
+    assert ln.weight.sequence_parallel_enabled
+    assert ln.bias.sequence_parallel_enabled
+    ref = torch.nn.LayerNorm(32)
+    with torch.no_grad():
+        ref.weight.copy_(ln.weight)
+        ref.bias.copy_(ln.bias)
+    x = torch.randn(4, 6, 32)
+    assert torch.allclose(ln(x), ref(x), atol=1e-6)
+
+    ln2 = LayerNorm(16)
+    assert not ln2.weight.sequence_parallel_enabled
+
+
+def test_tensor_utils_cumsum():
+    """utils/tensor_utils.py cumsum: fp64-accumulated dim-0 scan with the
+    reference's signature (tril_size accepted, 2-D/dim-0 contract)."""
+    import pytest
+    import torch
+
+    from neuronx_distributed_amd.utils.tensor_utils import cumsum
+
+    x = torch.randn(64, 8, dtype=torch.float32)
+    out = cumsum(x, tril_size=4)
+    ref = torch.cumsum(x.double(), dim=0).float()
+    assert torch.allclose(out, ref, atol=1e-6)
+    assert out.dtype == x.dtype
+    with pytest.raises(ValueError):
+        cumsum(torch.randn(3))
+    with pytest.raises(NotImplementedError):
+        cumsum(x, dim=1)
