@@ -65,6 +65,10 @@ class LlamaConfig:
     sliding_window: Optional[int] = None
     # Qwen-2 style QKV bias
     attention_bias: bool = False
+    # llama-3.1 style long-context RoPE scaling dict (rope_type
+    # "llama3": factor / low_freq_factor / high_freq_factor /
+    # original_max_position_embeddings), or {"rope_type": "linear", ...}
+    rope_scaling: Optional[dict] = None
     # >0: compute lm_head+CE in this many sequence chunks under
     # checkpointing (caps logits memory; utils/chunked_loss.py)
     loss_chunks: int = 0
@@ -99,6 +103,17 @@ CONFIGS = {
                             num_key_value_heads=4, vocab_size=152064,
                             rope_theta=1e6, attention_bias=True,
                             max_position_embeddings=8192),
+    "llama3.1-8b": LlamaConfig(hidden_size=4096, intermediate_size=14336,
+                               num_hidden_layers=32, num_attention_heads=32,
+                               num_key_value_heads=8, vocab_size=128256,
+                               rope_theta=500000.0,
+                               max_position_embeddings=131072,
+                               rope_scaling={
+                                   "rope_type": "llama3", "factor": 8.0,
+                                   "low_freq_factor": 1.0,
+                                   "high_freq_factor": 4.0,
+                                   "original_max_position_embeddings": 8192,
+                               }),
     "llama3-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672,
                               num_hidden_layers=80, num_attention_heads=64,
                               num_key_value_heads=8, vocab_size=128256,
@@ -381,8 +396,10 @@ class LlamaModel(nn.Module):
             LlamaDecoderLayer(config) for _ in range(config.num_hidden_layers))
         self.norm = RMSNorm(config.hidden_size, config.rms_norm_eps,
                             sequence_parallel_enabled=config.sequence_parallel_enabled)
-        cos, sin = ops.precompute_rope_freqs(config.max_position_embeddings,
-                                             config.head_dim, config.rope_theta)
+        cos, sin = ops.precompute_rope_freqs(
+            config.max_position_embeddings, config.head_dim,
+            config.rope_theta,
+            rope_scaling=getattr(config, "rope_scaling", None))
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
@@ -390,7 +407,8 @@ class LlamaModel(nn.Module):
         # meta materialization clobbers non-persistent buffers: recompute
         cos, sin = ops.precompute_rope_freqs(
             self.config.max_position_embeddings, self.config.head_dim,
-            self.config.rope_theta, device=self.rope_cos.device)
+            self.config.rope_theta, device=self.rope_cos.device,
+            rope_scaling=getattr(self.config, "rope_scaling", None))
         with torch.no_grad():
             self.rope_cos.copy_(cos)
             self.rope_sin.copy_(sin)

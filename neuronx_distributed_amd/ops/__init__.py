@@ -292,12 +292,41 @@ def apply_rotary_pos_emb(q, k, cos, sin, pos_offset: int = 0):
     return _RoPEFn.apply(q, k, cos, sin, pos_offset)
 
 
+def _llama3_scale_freqs(inv_freqs: torch.Tensor, scaling: dict):
+    """Llama-3.1 long-context frequency scaling (reference
+    attention/utils.py apply_scaling): high-frequency components (short
+    wavelength vs the original context) pass through, low-frequency ones
+    divide by ``factor``, the band between interpolates smoothly."""
+    import math as _math
+
+    factor = float(scaling["factor"])
+    lo = float(scaling.get("low_freq_factor", 1.0))
+    hi = float(scaling.get("high_freq_factor", 4.0))
+    orig = float(scaling.get("original_max_position_embeddings", 8192))
+    wavelen = 2 * _math.pi / inv_freqs
+    smooth = ((orig / wavelen - lo) / (hi - lo)).clamp(0.0, 1.0)
+    blended = (1.0 - smooth) * inv_freqs / factor + smooth * inv_freqs
+    return torch.where(wavelen < orig / hi, inv_freqs,
+                       torch.where(wavelen > orig / lo, inv_freqs / factor,
+                                   blended))
+
+
 def precompute_rope_freqs(seq_len: int, dim: int, theta: float = 10000.0,
-                          device=None):
+                          device=None, rope_scaling: dict = None):
     """cos/sin tables (S, D/2) fp32 (reference attention/utils.py
-    precompute_freqs_cis)."""
+    precompute_freqs_cis; ``rope_scaling`` with rope_type \"llama3\"
+    applies the 3.1 long-context wavelength interpolation)."""
     inv = 1.0 / (theta ** (torch.arange(0, dim, 2, device=device,
                                         dtype=torch.float32) / dim))
+    if rope_scaling:
+        rt = rope_scaling.get("rope_type", rope_scaling.get("type",
+                                                            "llama3"))
+        if rt == "llama3":
+            inv = _llama3_scale_freqs(inv, rope_scaling)
+        elif rt == "linear":
+            inv = inv / float(rope_scaling["factor"])
+        else:
+            raise NotImplementedError(f"rope_scaling type {rt!r}")
     t = torch.arange(seq_len, device=device, dtype=torch.float32)
     freqs = torch.outer(t, inv)
     return freqs.cos(), freqs.sin()

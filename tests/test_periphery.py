@@ -258,3 +258,49 @@ def test_tensor_utils_cumsum():
         cumsum(torch.randn(3))
     with pytest.raises(NotImplementedError):
         cumsum(x, dim=1)
+
+
+def test_llama3_rope_scaling_matches_transformers():
+    """ops.precompute_rope_freqs(rope_scaling=...) vs the HF transformers
+    llama3 rope init — an INDEPENDENT implementation of the same
+    published scheme (reference attention/utils.py apply_scaling)."""
+    import pytest
+    import torch
+
+    transformers = pytest.importorskip("transformers")
+    from transformers import LlamaConfig
+    from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+
+    from neuronx_distributed_amd import ops
+
+    scaling = {"rope_type": "llama3", "factor": 8.0,
+               "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+               "original_max_position_embeddings": 64}
+    dim, theta, S = 128, 500000.0, 256
+    cfg = LlamaConfig(hidden_size=dim, num_attention_heads=1,
+                      rope_theta=theta, max_position_embeddings=S,
+                      rope_scaling=dict(scaling))
+    hf_inv, att = ROPE_INIT_FUNCTIONS["llama3"](cfg, device="cpu")
+    t = torch.arange(S, dtype=torch.float32)
+    hf_freqs = torch.outer(t, hf_inv.float() * att)
+
+    cos, sin = ops.precompute_rope_freqs(S, dim, theta,
+                                         rope_scaling=scaling)
+    assert torch.allclose(cos, hf_freqs.cos(), atol=1e-5)
+    assert torch.allclose(sin, hf_freqs.sin(), atol=1e-5)
+
+    # linear scaling + unknown type
+    c2, _ = ops.precompute_rope_freqs(
+        16, 8, rope_scaling={"rope_type": "linear", "factor": 2.0})
+    c_ref, _ = ops.precompute_rope_freqs(16, 8)
+    assert torch.allclose(c2[2], c_ref[1], atol=1e-6)  # half-speed angles
+    with pytest.raises(NotImplementedError):
+        ops.precompute_rope_freqs(8, 8, rope_scaling={"rope_type": "yarn"})
+
+
+def test_llama31_config_builds():
+    from neuronx_distributed_amd.models import get_config
+
+    cfg = get_config("llama3.1-8b")
+    assert cfg.rope_scaling["factor"] == 8.0
+    assert cfg.max_position_embeddings == 131072
