@@ -332,6 +332,25 @@ def precompute_rope_freqs(seq_len: int, dim: int, theta: float = 10000.0,
     return freqs.cos(), freqs.sin()
 
 
+def apply_rotary_polar_compatible(query, key, freqs):
+    """Meta-llama INTERLEAVED rotary (reference attention/utils.py:50
+    apply_rotary_polar_compatible): dims pair as (2i, 2i+1) complex
+    components — used for Meta-format checkpoints, whereas the model
+    stack and HIP kernel use the HF neox rotate-half pairing.
+    query/key (B, S, H, D); freqs (S, D/2) fp32 angles."""
+    if freqs.dtype != torch.float32:
+        raise ValueError("freqs must be fp32 for accuracy")
+    phase = torch.polar(torch.ones_like(freqs), freqs)  # e^{i*theta}
+    phase = phase.view(1, freqs.shape[0], 1, freqs.shape[1])
+
+    def rot(x):
+        xc = torch.view_as_complex(
+            x.float().reshape(*x.shape[:-1], x.shape[-1] // 2, 2))
+        return torch.view_as_real(xc * phase).reshape(x.shape).to(x.dtype)
+
+    return rot(query), rot(key)
+
+
 # ---------------------------------------------------------------------------
 # SwiGLU
 # ---------------------------------------------------------------------------

@@ -304,3 +304,34 @@ def test_llama31_config_builds():
     cfg = get_config("llama3.1-8b")
     assert cfg.rope_scaling["factor"] == 8.0
     assert cfg.max_position_embeddings == 131072
+
+
+def test_apply_rotary_polar_compatible():
+    """Interleaved (Meta 'polar') rotary vs an independent construction:
+    de-interleave -> neox rotate-half (_rope_torch) -> re-interleave."""
+    import torch
+
+    from neuronx_distributed_amd.ops import (_rope_torch,
+                                             apply_rotary_polar_compatible,
+                                             precompute_rope_freqs)
+
+    torch.manual_seed(5)
+    B, S, H, D = 2, 16, 3, 32
+    q = torch.randn(B, S, H, D)
+    k = torch.randn(B, S, H, D)
+    cos, sin = precompute_rope_freqs(S, D)
+    freqs = torch.atan2(sin, cos)  # recover angles
+
+    qo, ko = apply_rotary_polar_compatible(q, k, freqs)
+
+    # independent golden: permute even dims first, neox-rotate, invert
+    perm = torch.cat([torch.arange(0, D, 2), torch.arange(1, D, 2)])
+    inv_perm = torch.argsort(perm)
+    q_ref = _rope_torch(q[..., perm], cos, sin)[..., inv_perm]
+    k_ref = _rope_torch(k[..., perm], cos, sin)[..., inv_perm]
+    assert torch.allclose(qo, q_ref, atol=1e-5), (qo - q_ref).abs().max()
+    assert torch.allclose(ko, k_ref, atol=1e-5)
+
+    import pytest
+    with pytest.raises(ValueError):
+        apply_rotary_polar_compatible(q, k, freqs.double())
