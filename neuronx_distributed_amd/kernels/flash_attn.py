@@ -41,6 +41,14 @@ def _torch_reference(q, k, v, causal=True, scale=None, window=None):
 
 
 def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
+    if window is not None and causal and window >= k.shape[2]:
+        # a window covering the whole KV length masks nothing beyond the
+        # causal triangle (row i attends [i-window+1, i]) — run the plain
+        # causal MFMA kernel, fwd AND bwd.  Makes Mistral-style models
+        # train on the fused path whenever seq_len <= sliding_window
+        # (the composed fallback materializes S x S scores and OOMs at
+        # production batch sizes).
+        window = None
     if q.is_cuda:
         from .. import ops
 

@@ -335,3 +335,25 @@ def test_apply_rotary_polar_compatible():
     import pytest
     with pytest.raises(ValueError):
         apply_rotary_polar_compatible(q, k, freqs.double())
+
+
+def test_flash_window_covering_seq_is_causal():
+    """window >= kv len masks nothing beyond causal: flash_attn_func
+    must short-circuit to the plain causal path (CPU reference here;
+    keeps Mistral seq<=window training off the O(S^2) composed path)."""
+    import torch
+
+    from neuronx_distributed_amd.kernels.flash_attn import flash_attn_func
+
+    torch.manual_seed(9)
+    q = torch.randn(1, 2, 32, 128)
+    k = torch.randn(1, 2, 32, 128)
+    v = torch.randn(1, 2, 32, 128)
+    w_eq = flash_attn_func(q, k, v, causal=True, window=32)
+    w_big = flash_attn_func(q, k, v, causal=True, window=1000)
+    plain = flash_attn_func(q, k, v, causal=True)
+    assert torch.allclose(w_eq, plain, atol=1e-6)
+    assert torch.allclose(w_big, plain, atol=1e-6)
+    # a genuinely smaller window must differ
+    w_small = flash_attn_func(q, k, v, causal=True, window=4)
+    assert not torch.allclose(w_small, plain, atol=1e-3)
