@@ -235,3 +235,64 @@ def test_block_decode_matches_per_step():
         t = toks[:, -1]
     out = torch.cat(outs, dim=1)
     assert torch.equal(out, ref), (out, ref)
+
+
+def _family_train_steps(build, vocab):
+    torch.manual_seed(0)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.bfloat16)
+    try:
+        with torch.device("cuda"):
+            m = build()
+    finally:
+        torch.set_default_dtype(prev)
+    opt = torch.optim.AdamW(m.parameters(), lr=1e-3)
+    losses = []
+    for s in range(8):
+        torch.manual_seed(100 + s)
+        x = torch.randint(0, vocab, (2, 64), device="cuda")
+        loss = m(x, labels=x)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    assert all(v == v for v in losses), losses          # finite
+    assert losses[-1] < losses[0] - 0.2, losses         # learns
+    return losses
+
+
+def test_gpt2_gpu_train_step():
+    """GPT-2 family (learned abs pos, LayerNorm, fused c_attn) trains in
+    bf16 on the HIP kernel stack."""
+    from neuronx_distributed_amd.models.gpt2 import (GPT2ForCausalLM,
+                                                     get_gpt2_config)
+
+    _family_train_steps(
+        lambda: GPT2ForCausalLM(get_gpt2_config("gpt2-tiny")), 256)
+
+
+def test_gpt_neox_gpu_train_step():
+    """GPT-NeoX family (parallel residual, partial rotary) trains in
+    bf16 on GPU."""
+    from neuronx_distributed_amd.models.gpt_neox import (GPTNeoXForCausalLM,
+                                                         get_neox_config)
+
+    _family_train_steps(
+        lambda: GPTNeoXForCausalLM(get_neox_config("gpt-neox-tiny")), 256)
+
+
+def test_qwen2_bias_and_rope_scaled_llama_gpu():
+    """qwen2-style attention bias and llama3.1-style scaled rope both
+    train on GPU (tiny shapes, production code paths)."""
+    from neuronx_distributed_amd.models import LlamaForCausalLM, get_config
+
+    cfg_bias = get_config("tiny", attention_bias=True,
+                          max_position_embeddings=512)
+    _family_train_steps(lambda: LlamaForCausalLM(cfg_bias), 256)
+
+    cfg_scaled = get_config(
+        "tiny", max_position_embeddings=512,
+        rope_scaling={"rope_type": "llama3", "factor": 4.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 128})
+    _family_train_steps(lambda: LlamaForCausalLM(cfg_scaled), 256)
