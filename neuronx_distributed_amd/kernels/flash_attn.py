@@ -40,6 +40,51 @@ def _torch_reference(q, k, v, causal=True, scale=None, window=None):
     return torch.matmul(probs, v.float()).to(q.dtype)
 
 
+def _banded_window_attn(q, k, v, scale, window, block=512):
+    """Causal sliding-window attention in O(S*(window+block)) memory:
+    q-row blocks attend only the kv band [q0 - window + 1, q1); each
+    block is gradient-checkpointed so training memory stays per-block
+    (the full composed path materializes S x S scores and OOMs at
+    production shapes).  Numerically identical to _torch_reference with
+    the same window."""
+    import torch.utils.checkpoint as ckpt
+
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    rep = Hq // Hkv
+    scale = scale or 1.0 / math.sqrt(D)
+
+    def run_block(qb, kb, vb, q0, k0, q1):
+        # qb (B,Hq,bq,D); kb/vb (B,Hq,bk,D) already GQA-expanded
+        scores = torch.matmul(qb.float(), kb.float().transpose(-1, -2))
+        scores = scores * scale
+        qi = torch.arange(q0, q1, device=qb.device).unsqueeze(1)
+        kj = torch.arange(k0, k0 + kb.shape[2], device=qb.device)
+        keep = (kj <= qi) & (kj > qi - window)
+        scores = scores.masked_fill(~keep, float("-inf"))
+        probs = torch.softmax(scores, dim=-1)
+        return torch.matmul(probs, vb.float()).to(qb.dtype)
+
+    if rep > 1:
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    outs = []
+    for q0 in range(0, S, block):
+        q1 = min(q0 + block, S)
+        k0 = max(0, q0 - window + 1)
+        qb = q[:, :, q0:q1]
+        kb = k[:, :, k0:q1]
+        vb = v[:, :, k0:q1]
+        if torch.is_grad_enabled() and (qb.requires_grad
+                                        or kb.requires_grad):
+            o = ckpt.checkpoint(run_block, qb, kb, vb, q0, k0, q1,
+                                use_reentrant=False)
+        else:
+            o = run_block(qb, kb, vb, q0, k0, q1)
+        outs.append(o)
+    return torch.cat(outs, dim=2)
+
+
 def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
     if window is not None and causal and window >= k.shape[2]:
         # a window covering the whole KV length masks nothing beyond the
@@ -60,6 +105,12 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
             # is full-causal, so training windows use the composed path)
             return ops.flash_attn_windowed(q, k, v, int(window),
                                            softmax_scale)
+        if window is not None and causal and q.shape[2] == k.shape[2] \
+                and q.shape[2] > 1024:
+            # window < S training/prefill at real seq lengths: banded
+            # blocks in O(S*window) memory (the full composed path
+            # materializes S x S scores and OOMs at production batch)
+            return _banded_window_attn(q, k, v, softmax_scale, int(window))
         if q.shape[2] != k.shape[2] or q.shape[3] != 128 or \
                 window is not None or q.dtype != torch.bfloat16:
             # rectangular attention (KV-cache decode), head_dim != 128,
