@@ -105,11 +105,19 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
             # is full-causal, so training windows use the composed path)
             return ops.flash_attn_windowed(q, k, v, int(window),
                                            softmax_scale)
+        if (window is not None and causal and q.shape[2] == k.shape[2]
+                and q.shape[3] == 128 and q.dtype == torch.bfloat16
+                and ops.flash_attn_available()):
+            # window < S TRAINING on the MFMA kernels: fwd tile-skip +
+            # band masks and the windowed dkdv/dq backward
+            return ops.flash_attn(q, k, v, causal=True,
+                                  softmax_scale=softmax_scale,
+                                  window=int(window))
         if window is not None and causal and q.shape[2] == k.shape[2] \
                 and q.shape[2] > 1024:
-            # window < S training/prefill at real seq lengths: banded
-            # blocks in O(S*window) memory (the full composed path
-            # materializes S x S scores and OOMs at production batch)
+            # window < S training/prefill at real seq lengths without the
+            # kernel: banded blocks in O(S*window) memory (the full
+            # composed path materializes S x S scores and OOMs)
             return _banded_window_attn(q, k, v, softmax_scale, int(window))
         if q.shape[2] != k.shape[2] or q.shape[3] != 128 or \
                 window is not None or q.dtype != torch.bfloat16:

@@ -433,7 +433,7 @@ def _fa_alloc_like(t):
 
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, window=0):
         lib = _require_lib()
         B, Hq, S, D = q.shape
         Hkv = k.shape[1]
@@ -451,11 +451,12 @@ class _FlashAttnFn(torch.autograd.Function):
             _ptr(q), _ptr(k), _ptr(v), _ptr(out), _ptr(lse),
             ctypes.c_int(B), ctypes.c_int(Hq), ctypes.c_int(Hkv),
             ctypes.c_int(S), ctypes.c_float(scale),
-            ctypes.c_int(1 if causal else 0), _fa_strides(q, k, v, out),
-            _stream())
+            ctypes.c_int(1 if causal else 0), ctypes.c_int(window or 0),
+            _fa_strides(q, k, v, out), _stream())
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.window = window or 0
         return out
 
     @staticmethod
@@ -482,6 +483,7 @@ class _FlashAttnFn(torch.autograd.Function):
             ctypes.c_int(B), ctypes.c_int(Hq), ctypes.c_int(Hkv),
             ctypes.c_int(S), ctypes.c_float(ctx.scale),
             ctypes.c_int(1 if ctx.causal else 0),
+            ctypes.c_int(getattr(ctx, "window", 0)),
             _fa_strides(q, k, v, out, dout, dq, dk_pq, dv_pq), _stream())
         rep = Hq // Hkv
         if rep > 1:
@@ -490,13 +492,14 @@ class _FlashAttnFn(torch.autograd.Function):
         else:
             dk = dk_pq
             dv = dv_pq
-        return dq, dk, dv, None, None
+        return dq, dk, dv, None, None, None
 
 
-def flash_attn(q, k, v, causal=True, softmax_scale=None):
-    """q (B,Hq,S,D=128), k/v (B,Hkv,S,D) bf16 on GPU."""
+def flash_attn(q, k, v, causal=True, softmax_scale=None, window=0):
+    """q (B,Hq,S,D=128), k/v (B,Hkv,S,D) bf16 on GPU.  window > 0:
+    sliding-window causal (Mistral), fwd AND bwd on the MFMA kernels."""
     scale = softmax_scale or 1.0 / math.sqrt(q.shape[-1])
-    return _FlashAttnFn.apply(q, k, v, causal, scale)
+    return _FlashAttnFn.apply(q, k, v, causal, scale, window)
 
 
 def flash_attn_window_available() -> bool:

@@ -325,14 +325,14 @@ extern "C" void flash_attn_fwd(const void* q, const void* k, const void* v,
 
 extern "C" void flash_attn_fwd_strided(
     const void* q, const void* k, const void* v, void* out, void* lse,
-    int B, int Hq, int Hkv, int S, float scale, int causal,
+    int B, int Hq, int Hkv, int S, float scale, int causal, int window,
     const long* st, hipStream_t stream) {
   // st = 12 longs: (bs, hs, ss) x (q, k, v, o)
   dim3 grid(Hq, (S + FA_QBLK - 1) / FA_QBLK, B);
   size_t lds = 3 * (K_TILE_B + VT_TILE_B);
   flash_fwd_kernel<<<grid, 512, lds, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out,
-      (float*)lse, B, Hq, Hkv, S, scale, causal, 0,
+      (float*)lse, B, Hq, Hkv, S, scale, causal, window,
       st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7], st[8],
       st[9], st[10], st[11]);
 }

@@ -218,6 +218,7 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                    const float* __restrict__ deltap,
                    short* __restrict__ dkp, short* __restrict__ dvp,
                    int B, int Hq, int Hkv, int S, float scale, int causal,
+                   int window,
                    long q_bs, long q_hs, long q_ss,
                    long k_bs, long k_hs, long k_ss,
                    long v_bs, long v_hs, long v_ss,
@@ -289,6 +290,13 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
   const float s2 = scale * LOG2E;
   int q_start = causal ? kvblk * 128 : 0;
+  // sliding window: q attends k iff k <= q < k + window, so this kv
+  // block's active q rows end at kv_max + window
+  int q_end = S;
+  if (window > 0) {
+    int qe = kvblk * 128 + 127 + window;
+    q_end = qe < S ? qe : S;
+  }
 
   stage_tile64(qp + q_base, q_start, q_ss, S - q_start, smem + BW64_LDS_Q,
                smem + BW64_LDS_QT);
@@ -298,9 +306,9 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 
   Stage64Regs nq, ndo;
   float nld = 0.f;
-  for (int q0 = q_start; q0 < S; q0 += 64) {
+  for (int q0 = q_start; q0 < q_end; q0 += 64) {
     __syncthreads();
-    if (q0 + 64 < S) {
+    if (q0 + 64 < q_end) {
       nq = load_tile64(qp + q_base, q0 + 64, q_ss, S - q0 - 64);
       ndo = load_tile64(dop + do_base, q0 + 64, do_ss, S - q0 - 64);
       nld = load_ld(q0 + 64);
@@ -316,7 +324,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
       const int half = dhalf;
       const int qh0 = q0 + half * 32;
       const bool produce =
-          (!causal || (qh0 + 31 >= kv0)) && qh0 < S;
+          (!causal || (qh0 + 31 >= kv0)) && qh0 < S &&
+          (window <= 0 || qh0 < kv0 + 31 + window);
       if (produce) {
         const int rm_row = half * 32 + col;  // row inside the 64-row tile
         const int my_kv = kv0 + col;         // this lane's kv column
@@ -352,7 +361,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
         for (int r = 0; r < 16; ++r) {
           int ql = half * 32 + acc_row(r, hi);
           int qg = q0 + ql;
-          bool masked = (causal && my_kv > qg) || my_kv >= S || qg >= S;
+          bool masked = (causal && my_kv > qg) || my_kv >= S || qg >= S ||
+              (window > 0 && qg >= my_kv + window);
           float p =
               masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse_v[r]);
           st[r] = p;
@@ -411,7 +421,8 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
       const int qh0 = q0 + half * 32;
-      if ((causal && qh0 + 31 < kv0) || qh0 >= S) continue;
+      if ((causal && qh0 + 31 < kv0) || qh0 >= S ||
+          (window > 0 && qh0 >= kv0 + 31 + window)) continue;
 
       // prefetch the dV/dK B fragments first — independent of the slot
       // reads, so all 12 b128 LDS loads pipeline into one counted wait
@@ -447,7 +458,7 @@ fa_bwd_dkdv_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
     }
 
     __syncthreads();
-    if (q0 + 64 < S) {
+    if (q0 + 64 < q_end) {
       write_tile64(nq, smem + BW64_LDS_Q, smem + BW64_LDS_QT);
       write_tile64(ndo, smem + BW64_LDS_DO, smem + BW64_LDS_DOT);
       write_ld(nld);
@@ -516,6 +527,7 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
                  const float* __restrict__ lsep,
                  const float* __restrict__ deltap, short* __restrict__ dqp,
                  int B, int Hq, int Hkv, int S, float scale, int causal,
+                 int window,
                  long q_bs, long q_hs, long q_ss,
                  long k_bs, long k_hs, long k_ss,
                  long v_bs, long v_hs, long v_ss,
@@ -561,19 +573,26 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
   // WG-uniform loop bound (all waves share barriers); per-wave causal
   // skipping happens via wave_active below
   const int kv_end = causal ? min(S, qblk * 128 + 128) : S;
+  // sliding window: this q block's earliest active kv row (tile-aligned)
+  const int kv_begin = window > 0
+      ? ((qblk * 128 - window + 1 > 0 ? qblk * 128 - window + 1 : 0) & ~31)
+      : 0;
   const float s2 = scale * LOG2E;
 
-  stage_tile32(kp + k_base, 0, k_ss, S, smem + DQ_LDS_K, smem + DQ_LDS_KT);
-  write_tile32_rm(load_tile32(vp + v_base, 0, v_ss, S), smem + DQ_LDS_V);
+  stage_tile32(kp + k_base, kv_begin, k_ss, S - kv_begin, smem + DQ_LDS_K,
+               smem + DQ_LDS_KT);
+  write_tile32_rm(load_tile32(vp + v_base, kv_begin, v_ss, S - kv_begin),
+                  smem + DQ_LDS_V);
 
   StageRegs nk, nv;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+  for (int kv0 = kv_begin; kv0 < kv_end; kv0 += 32) {
     __syncthreads();
     if (kv0 + 32 < kv_end) {
       nk = load_tile32(kp + k_base, kv0 + 32, k_ss, S - kv0 - 32);
       nv = load_tile32(vp + v_base, kv0 + 32, v_ss, S - kv0 - 32);
     }
-    const bool wave_active = !causal || (kv0 <= q0 + 31);
+    const bool wave_active = (!causal || (kv0 <= q0 + 31)) &&
+        (window <= 0 || kv0 + 31 + window > q0);
 
     if (wave_active) {
       // S^T[k][q]: A = K rows (LDS rm, row k=col), B = Q^T (regs)
@@ -595,7 +614,8 @@ fa_bwd_dq_kernel(const short* __restrict__ qp, const short* __restrict__ kp,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int kg = kv0 + acc_row(r, hi);
-        bool masked = (causal && kg > my_q) || kg >= S || my_q >= S;
+        bool masked = (causal && kg > my_q) || kg >= S || my_q >= S ||
+            (window > 0 && kg <= my_q - window);
         float p = masked ? 0.f : __builtin_amdgcn_exp2f(st[r] * s2 - lse2);
         dst[r] = p * (dpt[r] - dlt);
       }
@@ -684,7 +704,7 @@ extern "C" void flash_attn_bwd_strided(
     const void* q, const void* k, const void* v, const void* out,
     const void* dout, const void* lse, void* delta, void* dq, void* dk,
     void* dv, int B, int Hq, int Hkv, int S, float scale, int causal,
-    const long* st, hipStream_t stream) {
+    int window, const long* st, hipStream_t stream) {
   // st = 24 longs: (bs, hs, ss) x (q, k, v, o, dout, dq, dk, dv)
   long rows = (long)B * Hq * S;
   // 16 lanes per row -> 16 rows per 256-thread block; cap well above the
@@ -701,9 +721,9 @@ extern "C" void flash_attn_bwd_strided(
   fa_bwd_dkdv_kernel<<<gkv, 512, lds1, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dk, (short*)dv, B, Hq,
-      Hkv, S, scale, causal, st[0], st[1], st[2], st[3], st[4], st[5],
-      st[6], st[7], st[8], st[12], st[13], st[14], st[18], st[19], st[20],
-      st[21], st[22], st[23]);
+      Hkv, S, scale, causal, window, st[0], st[1], st[2], st[3], st[4],
+      st[5], st[6], st[7], st[8], st[12], st[13], st[14], st[18], st[19],
+      st[20], st[21], st[22], st[23]);
   dim3 gq(Hq, (S + 127) / 128, B);
   // staging layout needs 26.6 KB; the epilogue reuses LDS as 4 per-wave
   // 32x128 transpose tiles = 32 KB, which dominates
@@ -711,8 +731,9 @@ extern "C" void flash_attn_bwd_strided(
   fa_bwd_dq_kernel<<<gq, 256, lds2, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)lse, (const float*)delta, (short*)dq, B, Hq, Hkv, S,
-      scale, causal, st[0], st[1], st[2], st[3], st[4], st[5], st[6],
-      st[7], st[8], st[12], st[13], st[14], st[15], st[16], st[17]);
+      scale, causal, window, st[0], st[1], st[2], st[3], st[4], st[5],
+      st[6], st[7], st[8], st[12], st[13], st[14], st[15], st[16],
+      st[17]);
 }
 
 extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
@@ -735,5 +756,5 @@ extern "C" void flash_attn_bwd(const void* q, const void* k, const void* v,
     st[21 + i] = qd[i];           // dv
   }
   flash_attn_bwd_strided(q, k, v, out, dout, lse, delta, dq, dk, dv, B, Hq,
-                         Hkv, S, scale, causal, st, stream);
+                         Hkv, S, scale, causal, 0, st, stream);
 }
