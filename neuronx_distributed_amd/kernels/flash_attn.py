@@ -101,8 +101,8 @@ def flash_attn_func(q, k, v, causal=True, softmax_scale=None, window=None):
                 and q.shape[2] == k.shape[2] and q.shape[3] == 128
                 and q.dtype == torch.bfloat16
                 and ops.flash_attn_window_available()):
-            # sliding-window MFMA kernel (inference; the backward kernel
-            # is full-causal, so training windows use the composed path)
+            # sliding-window MFMA kernel, no-grad fast path (training
+            # windows route through ops.flash_attn(window=) below)
             return ops.flash_attn_windowed(q, k, v, int(window),
                                            softmax_scale)
         if (window is not None and causal and q.shape[2] == k.shape[2]

@@ -59,9 +59,10 @@ class LlamaConfig:
     # set by pad_llama_config: padded head counts keep the ORIGINAL head
     # size (hidden_size // original_num_heads)
     head_dim_override: Optional[int] = None
-    # Mistral-style sliding-window attention (None = full causal).  Runs
-    # on the composed/batched-GEMM paths; the MFMA flash kernel is
-    # full-causal (window support is round-2 kernel work).
+    # Mistral-style sliding-window attention (None = full causal).
+    # Training AND inference run on the windowed MFMA flash kernels
+    # (band-limited fwd/dkdv/dq); window >= seq short-circuits to plain
+    # causal.
     sliding_window: Optional[int] = None
     # Qwen-2 style QKV bias
     attention_bias: bool = False
