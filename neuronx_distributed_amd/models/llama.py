@@ -222,7 +222,8 @@ class LlamaAttention(nn.Module):
             # slice (utils.batch_utils); K/V ring-rotate over the CP group
             from ..kernels.ring_attn import ring_attn_func
 
-            out = ring_attn_func(q, k, v, causal=True)
+            out = ring_attn_func(q, k, v, causal=True,
+                                 window=self.sliding_window)
         else:
             out = flash_attn_func(q, k, v, causal=True,
                                   window=self.sliding_window)

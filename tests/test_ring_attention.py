@@ -164,3 +164,66 @@ def test_cp_tp_3d():
     out = run_distributed(_cp_tp_worker, world_size=4)
     # all ranks agree after the CP mean (TP pairs compute identical losses)
     assert max(out) - min(out) < 1e-4
+
+
+def _ring_window_worker(rank, world, W):
+    """cp-sharded sliding-window ring attention vs the full windowed
+    reference: fwd and all grads, covering block kinds skip/full/partial
+    /diag across window sizes."""
+    from neuronx_distributed_amd.kernels.flash_attn import _torch_reference
+    from neuronx_distributed_amd.kernels.ring_attn import ring_attn_func
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 context_parallel_size=world)
+    torch.manual_seed(0)
+    B, H, S, D = 1, 2, 64, 128
+    q = torch.randn(B, H, S, D) * 0.5
+    k = torch.randn(B, H, S, D) * 0.5
+    v = torch.randn(B, H, S, D) * 0.5
+    Sl = S // world
+    ql = q[:, :, rank * Sl:(rank + 1) * Sl].clone().requires_grad_(True)
+    kl = k[:, :, rank * Sl:(rank + 1) * Sl].clone().requires_grad_(True)
+    vl = v[:, :, rank * Sl:(rank + 1) * Sl].clone().requires_grad_(True)
+
+    out = ring_attn_func(ql, kl, vl, causal=True, window=W)
+
+    qf = q.clone().requires_grad_(True)
+    kf = k.clone().requires_grad_(True)
+    vf = v.clone().requires_grad_(True)
+    ref = _torch_reference(qf, kf, vf, causal=True, window=W)
+    ref_l = ref[:, :, rank * Sl:(rank + 1) * Sl]
+    assert torch.allclose(out, ref_l, atol=1e-4), \
+        (W, (out - ref_l).abs().max().item())
+
+    torch.manual_seed(7)
+    dy_full = torch.randn_like(ref)
+    out.backward(dy_full[:, :, rank * Sl:(rank + 1) * Sl])
+    ref.backward(dy_full)
+    for g, gf_full, name in ((ql.grad, qf.grad, "dq"),
+                             (kl.grad, kf.grad, "dk"),
+                             (vl.grad, vf.grad, "dv")):
+        gf = gf_full[:, :, rank * Sl:(rank + 1) * Sl]
+        assert torch.allclose(g, gf, atol=1e-4), \
+            f"W={W} {name}: {(g - gf).abs().max().item()}"
+    return True
+
+
+def test_ring_window_small_cp2():
+    # W < C=32: diagonal windowed + all off-diagonal blocks skipped
+    run_distributed(_ring_window_worker, world_size=2, args=(8,))
+
+
+def test_ring_window_partial_cp2():
+    # C < W < 2C: partial boundary block exercised
+    run_distributed(_ring_window_worker, world_size=2, args=(48,))
+
+
+def test_ring_window_cp4_mixed():
+    # C=16, W=24: diag partial-window + partial + skip blocks all appear
+    run_distributed(_ring_window_worker, world_size=4, args=(24,))
+
+
+def test_ring_window_covers_seq_cp2():
+    # W >= S reduces to plain causal
+    run_distributed(_ring_window_worker, world_size=2, args=(64,))
