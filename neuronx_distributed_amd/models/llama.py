@@ -438,7 +438,9 @@ class LlamaModel(nn.Module):
                                         self.norm.variance_epsilon)
             return hidden
         if fused_norm_train_ok(self.config, hidden, self.training) \
-                and kv_caches is None:
+                and kv_caches is None \
+                and all(hasattr(l, "forward_fused_train")
+                        for l in self.layers):
             # fused residual+norm TRAINING path (autograd twin of the
             # inference path above; NXDA_FUSED_NORM=0 reverts).  Works
             # under SP too: the adds/norms are per-row ops on the local

@@ -213,7 +213,9 @@ class MixtralModel(nn.Module):
             hidden = scatter_to_sequence_parallel_region(hidden, seq_dim=0)
         all_router_logits = []
         if fused_norm_train_ok(self.config, hidden, self.training) \
-                and kv_caches is None:
+                and kv_caches is None \
+                and all(hasattr(l, "forward_fused_train")
+                        for l in self.layers):
             # fused residual+norm training path (see models/llama.py;
             # SP-safe — per-row ops on the local seq shard, norm-weight
             # SP grad marking is a param attribute)

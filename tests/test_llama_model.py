@@ -300,3 +300,37 @@ def _fused_norm_sp_worker(rank, world):
 
 def test_fused_norm_train_sp_matches_plain_tp2():
     run_distributed(_fused_norm_sp_worker, world_size=2)
+
+
+def _fused_norm_ckpt_worker(rank, world):
+    """Activation checkpointing wraps the decoder layers in
+    CheckpointWrapper (no forward_fused_train): the fused-norm gate must
+    fall back to the plain path instead of crashing, and training must
+    still work."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.utils.activation_checkpoint import (
+        apply_activation_checkpointing)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    os.environ["NXDA_FUSED_NORM_FORCE"] = "1"
+    try:
+        model = _build_model()
+        from neuronx_distributed_amd.models.llama import LlamaDecoderLayer
+
+        apply_activation_checkpointing(
+            model, check_fn=lambda m: isinstance(m, LlamaDecoderLayer))
+        model.train()
+        torch.manual_seed(7)
+        x = torch.randint(0, 256, (2, 16))
+        loss = model(x, labels=x)
+        loss.backward()
+        assert torch.isfinite(loss)
+        assert model.model.layers[0].module.input_layernorm.weight.grad \
+            is not None
+    finally:
+        os.environ["NXDA_FUSED_NORM_FORCE"] = "0"
+    return float(loss)
+
+
+def test_fused_norm_gate_respects_activation_checkpoint():
+    run_distributed(_fused_norm_ckpt_worker, world_size=1)
