@@ -425,7 +425,9 @@ class LlamaModel(nn.Module):
                 and hidden.is_cuda and hidden.dtype == torch.bfloat16
                 and not self.config.sequence_parallel_enabled
                 and self.config.hidden_size % 8 == 0
-                and ops.add_rmsnorm_available()):
+                and ops.add_rmsnorm_available()
+                and all(hasattr(l, "forward_fused")
+                        for l in self.layers)):
             # fused residual+norm inference path (2 elementwise kernels per
             # layer instead of 4; decode steps are kernel-count-bound)
             residual, delta = hidden, None
