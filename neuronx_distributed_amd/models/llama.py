@@ -442,6 +442,8 @@ class LlamaModel(nn.Module):
         if fused_norm_train_ok(self.config, hidden, self.training) \
                 and kv_caches is None \
                 and all(hasattr(l, "forward_fused_train")
+                        and not l._forward_hooks
+                        and not l._forward_pre_hooks
                         for l in self.layers):
             # fused residual+norm TRAINING path (autograd twin of the
             # inference path above; NXDA_FUSED_NORM=0 reverts).  Works

@@ -215,6 +215,8 @@ class MixtralModel(nn.Module):
         if fused_norm_train_ok(self.config, hidden, self.training) \
                 and kv_caches is None \
                 and all(hasattr(l, "forward_fused_train")
+                        and not l._forward_hooks
+                        and not l._forward_pre_hooks
                         for l in self.layers):
             # fused residual+norm training path (see models/llama.py;
             # SP-safe — per-row ops on the local seq shard, norm-weight

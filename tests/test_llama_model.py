@@ -334,3 +334,33 @@ def _fused_norm_ckpt_worker(rank, world):
 
 def test_fused_norm_gate_respects_activation_checkpoint():
     run_distributed(_fused_norm_ckpt_worker, world_size=1)
+
+
+def _fused_norm_hooks_worker(rank, world):
+    """Module-level forward hooks on decoder layers (tensor capture
+    style) must keep firing: the fused path bypasses __call__, so the
+    gate falls back to the plain path when hooks are registered."""
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    os.environ["NXDA_FUSED_NORM_FORCE"] = "1"
+    try:
+        model = _build_model()
+        fired = []
+        for lyr in model.model.layers:
+            lyr.register_forward_hook(
+                lambda m, i, o: fired.append(type(m).__name__))
+        model.train()
+        torch.manual_seed(7)
+        x = torch.randint(0, 256, (2, 16))
+        loss = model(x, labels=x)
+        loss.backward()
+        assert len(fired) == len(model.model.layers), fired
+        assert torch.isfinite(loss)
+    finally:
+        os.environ["NXDA_FUSED_NORM_FORCE"] = "0"
+    return float(loss)
+
+
+def test_fused_norm_gate_respects_layer_hooks():
+    run_distributed(_fused_norm_hooks_worker, world_size=1)
