@@ -1,0 +1,75 @@
+"""Property-based tests (hypothesis) for serialization-critical paths:
+packed-quantization round-trips, Karmarkar-Karp bin balance, and the
+xser flatten/unflatten structure preservation."""
+
+import hypothesis.strategies as st
+import pytest
+import torch
+from hypothesis import given, settings
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(1, 7), st.integers(4, 64))
+def test_pack_unpack_x4_roundtrip_fp8(rows, cols_quads):
+    from neuronx_distributed_amd.quantization.quantization_utils import (
+        QuantizedDtype, pack_x4, unpack_x4)
+
+    cols = cols_quads  # packed width; unpacked = 4x
+    torch.manual_seed(rows * 100 + cols)
+    bits = torch.randint(0, 256, (rows, cols * 4), dtype=torch.uint8)
+    for dt in (QuantizedDtype.F8E4M3FN_X4, QuantizedDtype.F8E5M2_X4):
+        q = bits.view(dt.unpacked.torch_dtype)
+        packed = pack_x4(q, dt)
+        assert packed.shape == (rows, cols)
+        out = unpack_x4(packed, dt)
+        assert out.dtype == dt.unpacked.torch_dtype
+        assert out.view(torch.uint8).equal(bits)  # exact bit round-trip
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.integers(1, 5000), min_size=1, max_size=40),
+       st.integers(1, 8))
+def test_kk_bins_cover_and_balance(sizes, nbins):
+    from neuronx_distributed_amd.trainer.checkpoint import (
+        assign_tensors_to_bins)
+
+    tensors = [torch.empty(s, dtype=torch.int8) for s in sizes]
+    bins = assign_tensors_to_bins(tensors, nbins)
+    assert len(bins) == nbins
+    seen = sorted(i for b in bins for i in b)
+    assert seen == list(range(len(tensors)))  # exact cover, no dupes
+    if len(sizes) >= nbins:
+        loads = sorted(sum(sizes[i] for i in b) for b in bins)
+        # greedy/KK guarantee: max bin <= mean + max element
+        assert loads[-1] <= sum(sizes) / nbins + max(sizes) + 1
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.recursive(
+    st.one_of(st.integers(-5, 5), st.text(max_size=4),
+              st.booleans(), st.none(),
+              st.integers(1, 4).map(lambda n: torch.arange(n).float())),
+    lambda children: st.one_of(
+        st.lists(children, max_size=3),
+        st.dictionaries(st.text(max_size=3), children, max_size=3)),
+    max_leaves=12))
+def test_xser_flatten_unflatten_roundtrip(obj):
+    from neuronx_distributed_amd.trainer.checkpoint import (
+        _xser_flatten, _xser_unflatten)
+
+    tensors = []
+    skeleton = _xser_flatten(obj, tensors)
+    restored = _xser_unflatten(skeleton, lambda i: tensors[i])
+
+    def eq(a, b):
+        if isinstance(a, torch.Tensor):
+            return isinstance(b, torch.Tensor) and torch.equal(a, b)
+        if isinstance(a, dict):
+            return (isinstance(b, dict) and a.keys() == b.keys()
+                    and all(eq(a[k], b[k]) for k in a))
+        if isinstance(a, list):
+            return (isinstance(b, list) and len(a) == len(b)
+                    and all(eq(x, y) for x, y in zip(a, b)))
+        return a == b
+
+    assert eq(obj, restored)
