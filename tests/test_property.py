@@ -73,3 +73,35 @@ def test_xser_flatten_unflatten_roundtrip(obj):
         return a == b
 
     assert eq(obj, restored)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 4), st.integers(1, 4), st.integers(1, 6))
+def test_hf_state_dict_converters_roundtrip(layers, heads, i_mult):
+    """HF -> native -> HF llama state-dict conversion is lossless for
+    arbitrary layer counts / widths (fused gate_up split included)."""
+    from neuronx_distributed_amd.overrides import (
+        convert_hf_llama_state_dict, convert_to_hf_llama_state_dict)
+
+    H, inter = heads * 8, i_mult * 8
+    hf = {"model.embed_tokens.weight": torch.randn(16, H),
+          "model.norm.weight": torch.randn(H),
+          "lm_head.weight": torch.randn(16, H)}
+    for n in range(layers):
+        p = f"model.layers.{n}."
+        hf[p + "self_attn.q_proj.weight"] = torch.randn(H, H)
+        hf[p + "self_attn.k_proj.weight"] = torch.randn(H // 2, H)
+        hf[p + "self_attn.v_proj.weight"] = torch.randn(H // 2, H)
+        hf[p + "self_attn.o_proj.weight"] = torch.randn(H, H)
+        hf[p + "mlp.gate_proj.weight"] = torch.randn(inter, H)
+        hf[p + "mlp.up_proj.weight"] = torch.randn(inter, H)
+        hf[p + "mlp.down_proj.weight"] = torch.randn(H, inter)
+        hf[p + "input_layernorm.weight"] = torch.randn(H)
+        hf[p + "post_attention_layernorm.weight"] = torch.randn(H)
+
+    native = convert_hf_llama_state_dict(hf)
+    assert not any("gate_proj" in k or "up_proj" in k for k in native)
+    back = convert_to_hf_llama_state_dict(native)
+    assert back.keys() == hf.keys()
+    for k in hf:
+        assert torch.equal(back[k], hf[k]), k
