@@ -100,7 +100,9 @@ def test_hf_state_dict_converters_roundtrip(layers, heads, i_mult):
         hf[p + "post_attention_layernorm.weight"] = torch.randn(H)
 
     native = convert_hf_llama_state_dict(hf)
-    assert not any("gate_proj" in k or "up_proj" in k for k in native)
+    assert not any(k.endswith(("gate_proj.weight", "up_proj.weight"))
+                   and not k.endswith("gate_up_proj.weight")
+                   for k in native)
     back = convert_to_hf_llama_state_dict(native)
     assert back.keys() == hf.keys()
     for k in hf:
