@@ -227,3 +227,43 @@ def test_ring_window_cp4_mixed():
 def test_ring_window_covers_seq_cp2():
     # W >= S reduces to plain causal
     run_distributed(_ring_window_worker, world_size=2, args=(64,))
+
+
+def _ring_window_gqa_worker(rank, world):
+    """GQA (Hq=4, Hkv=2) + sliding window through the ring — the block
+    primitives GQA-expand inside each kind (kernel, masked, full)."""
+    from neuronx_distributed_amd.kernels.flash_attn import _torch_reference
+    from neuronx_distributed_amd.kernels.ring_attn import ring_attn_func
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1,
+                                 context_parallel_size=world)
+    torch.manual_seed(3)
+    B, Hq, Hkv, S, D, W = 1, 4, 2, 64, 128, 40
+    q = torch.randn(B, Hq, S, D) * 0.5
+    k = torch.randn(B, Hkv, S, D) * 0.5
+    v = torch.randn(B, Hkv, S, D) * 0.5
+    Sl = S // world
+    sl = slice(rank * Sl, (rank + 1) * Sl)
+    ql = q[:, :, sl].clone().requires_grad_(True)
+    kl = k[:, :, sl].clone().requires_grad_(True)
+    vl = v[:, :, sl].clone().requires_grad_(True)
+    out = ring_attn_func(ql, kl, vl, causal=True, window=W)
+
+    qf = q.clone().requires_grad_(True)
+    kf = k.clone().requires_grad_(True)
+    vf = v.clone().requires_grad_(True)
+    ref = _torch_reference(qf, kf, vf, causal=True, window=W)
+    assert torch.allclose(out, ref[:, :, sl], atol=1e-4)
+    torch.manual_seed(8)
+    dy = torch.randn_like(ref)
+    out.backward(dy[:, :, sl])
+    ref.backward(dy)
+    assert torch.allclose(ql.grad, qf.grad[:, :, sl], atol=1e-4)
+    assert torch.allclose(kl.grad, kf.grad[:, :, sl], atol=1e-4)
+    assert torch.allclose(vl.grad, vf.grad[:, :, sl], atol=1e-4)
+    return True
+
+
+def test_ring_window_gqa_cp2():
+    run_distributed(_ring_window_gqa_worker, world_size=2)
