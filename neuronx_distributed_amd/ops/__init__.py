@@ -174,6 +174,10 @@ class _AddRMSNormFn(torch.autograd.Function):
     def backward(ctx, dh, dnormed):
         h, weight, rstd = ctx.saved_tensors
         H = h.shape[-1]
+        if dnormed is None:
+            # normed output unused: the op reduces to the residual add —
+            # grads pass straight through, no weight grad
+            return dh, dh, None, None
         if h.is_cuda and h.dtype == torch.bfloat16 and H % 8 == 0:
             lib = _require_lib()
             rows = h.numel() // H

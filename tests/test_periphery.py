@@ -386,3 +386,20 @@ def test_banded_window_attention_matches_reference():
     assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
     for a, b in zip(g, (q.grad, k.grad, v.grad)):
         assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
+
+
+def test_add_rmsnorm_train_unused_normed():
+    """Dropping the normed output must not crash the backward (dnormed
+    is None): the op degrades to a pass-through residual add."""
+    import torch
+
+    from neuronx_distributed_amd import ops
+
+    d = torch.randn(4, 32, requires_grad=True)
+    r = torch.randn(4, 32, requires_grad=True)
+    w = torch.randn(32, requires_grad=True)
+    h, _ = ops.add_rmsnorm_train(d, r, w, 1e-5)
+    h.sum().backward()
+    assert torch.allclose(d.grad, torch.ones_like(d))
+    assert torch.allclose(r.grad, torch.ones_like(r))
+    assert w.grad is None
