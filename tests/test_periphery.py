@@ -402,4 +402,4 @@ def test_add_rmsnorm_train_unused_normed():
     h.sum().backward()
     assert torch.allclose(d.grad, torch.ones_like(d))
     assert torch.allclose(r.grad, torch.ones_like(r))
-    assert w.grad is None
+    assert w.grad is None or torch.all(w.grad == 0)
