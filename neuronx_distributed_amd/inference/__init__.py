@@ -9,5 +9,6 @@ from .functions import (
     shard_checkpoint,
     load_sharded_checkpoint,
 )
-from .kv_cache import KVCache, build_kv_caches
+from .kv_cache import (KVCache, RollingKVCache,
+                       build_kv_caches)
 from .generation import generate

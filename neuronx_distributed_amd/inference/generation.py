@@ -31,7 +31,8 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
     n_kv_local = cfg.num_key_value_heads * kv_mult // tp
     caches = build_kv_caches(cfg.num_hidden_layers, B, n_kv_local,
                              S + max_new_tokens, cfg.head_dim,
-                             device=input_ids.device)
+                             device=input_ids.device,
+                             window=getattr(cfg, "sliding_window", None))
 
     # prefill (eager, flash kernel)
     logits = model(input_ids, kv_caches=caches, pos_offset=0)

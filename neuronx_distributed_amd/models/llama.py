@@ -283,8 +283,15 @@ class LlamaAttention(nn.Module):
         # every step) -> (B,Hkv,rep,Smax)
         qg = q.view(B, self.num_kv_local, rep, self.head_dim)
         scores = (qg @ K.transpose(-1, -2)).float() * scale
-        idx = torch.arange(Smax, device=q.device)
-        invalid = idx > pos_t
+        if hasattr(kv_cache, "position_index"):
+            # rolling window cache: slots carry their GLOBAL position
+            # (-1 = unwritten -> masked by idx > pos_t never holding...
+            # masked via idx < 0 below)
+            idx = kv_cache.position_index()
+            invalid = (idx > pos_t) | (idx < 0)
+        else:
+            idx = torch.arange(Smax, device=q.device)
+            invalid = idx > pos_t
         if self.sliding_window is not None:
             invalid = invalid | (idx <= pos_t - self.sliding_window)
         scores = scores.masked_fill(invalid, float("-inf"))

@@ -281,3 +281,50 @@ def test_nxd_model_artifact_roundtrip():
 
     with tempfile.TemporaryDirectory() as d:
         run_distributed(_artifact_worker, world_size=1, args=(d,))
+
+
+def _rolling_cache_worker(rank, world):
+    """RollingKVCache (O(window) memory) must produce the same decode
+    logits as the full-context KVCache for a sliding-window model,
+    through prefill + enough decode steps to wrap the ring twice."""
+    import torch
+
+    from neuronx_distributed_amd.inference.kv_cache import (KVCache,
+                                                            RollingKVCache)
+    from neuronx_distributed_amd.models import LlamaForCausalLM, get_config
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    W = 8
+    cfg = get_config("tiny", sliding_window=W, max_position_embeddings=64)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(cfg).eval()
+    B, S, steps = 2, 6, 20
+    n_kv = cfg.num_key_value_heads
+    full = [KVCache(B, n_kv, S + steps + 1, cfg.head_dim,
+                    dtype=torch.float32, device="cpu")
+            for _ in range(cfg.num_hidden_layers)]
+    roll = [RollingKVCache(B, n_kv, W, cfg.head_dim, dtype=torch.float32,
+                           device="cpu")
+            for _ in range(cfg.num_hidden_layers)]
+    assert roll[0].k.shape[2] == W  # bounded memory
+
+    torch.manual_seed(1)
+    x = torch.randint(0, 256, (B, S))
+    with torch.no_grad():
+        lf = model(x, kv_caches=full, pos_offset=0)
+        lr = model(x, kv_caches=roll, pos_offset=0)
+        assert torch.allclose(lf, lr, atol=1e-5)
+        tok = lf[:, -1, :].argmax(-1, keepdim=True)
+        for step in range(steps):
+            pos = torch.tensor([S + step])
+            lf = model(tok, kv_caches=full, pos_offset=pos)
+            lr = model(tok, kv_caches=roll, pos_offset=pos)
+            assert torch.allclose(lf, lr, atol=1e-5), \
+                (step, (lf - lr).abs().max())
+            tok = lf[:, -1, :].argmax(-1, keepdim=True)
+    return True
+
+
+def test_rolling_kv_cache_matches_full():
+    run_distributed(_rolling_cache_worker, world_size=1)
