@@ -29,9 +29,10 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
     B, S = input_ids.shape
     kv_mult = max(1, tp // cfg.num_key_value_heads)
     n_kv_local = cfg.num_key_value_heads * kv_mult // tp
+    model_dtype = next(model.parameters()).dtype
     caches = build_kv_caches(cfg.num_hidden_layers, B, n_kv_local,
                              S + max_new_tokens, cfg.head_dim,
-                             device=input_ids.device,
+                             dtype=model_dtype, device=input_ids.device,
                              window=getattr(cfg, "sliding_window", None))
 
     # prefill (eager, flash kernel)
@@ -53,10 +54,17 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int = 32,
                 break
         return torch.cat(out, dim=1)
 
+    # eager decode: tensor positions when supported — the tensor-pos
+    # decode step masks by cache position_index(), which RollingKVCache
+    # (sliding-window models) requires; int positions assume ordered
+    # cache rows
+    tensor_pos = getattr(model, "supports_tensor_position", False)
     pos = S
     for _ in range(max_new_tokens - 1):
         step_in = next_tok.unsqueeze(1)
-        logits = model(step_in, kv_caches=caches, pos_offset=pos)
+        po = torch.tensor([pos], device=input_ids.device) if tensor_pos \
+            else pos
+        logits = model(step_in, kv_caches=caches, pos_offset=po)
         next_tok = sampler(logits[:, -1, :])
         out.append(next_tok.unsqueeze(1))
         pos += 1

@@ -27,8 +27,8 @@ class KVCache:
         decode): the write becomes index_copy_ and the FULL buffers are
         returned (attention masks by position — no dynamic shapes)."""
         if isinstance(pos, torch.Tensor):
-            self.k.index_copy_(2, pos, k_new)
-            self.v.index_copy_(2, pos, v_new)
+            self.k.index_copy_(2, pos, k_new.to(self.k.dtype))
+            self.v.index_copy_(2, pos, v_new.to(self.v.dtype))
             return self.k, self.v
         S = k_new.shape[2]
         self.k[:, :, pos:pos + S] = k_new
@@ -74,8 +74,8 @@ class RollingKVCache(KVCache):
     def update(self, k_new: torch.Tensor, v_new: torch.Tensor, pos):
         if isinstance(pos, torch.Tensor):  # decode: one token
             slot = pos % self.window
-            self.k.index_copy_(2, slot, k_new)
-            self.v.index_copy_(2, slot, v_new)
+            self.k.index_copy_(2, slot, k_new.to(self.k.dtype))
+            self.v.index_copy_(2, slot, v_new.to(self.v.dtype))
             self.slot_pos.index_copy_(0, slot, pos)
             return self.k, self.v
         S = k_new.shape[2]
@@ -87,8 +87,8 @@ class RollingKVCache(KVCache):
         tail = min(S, self.window)
         idx = torch.arange(S - tail, S, device=self.k.device)
         slot = idx % self.window
-        self.k.index_copy_(2, slot, k_new[:, :, S - tail:])
-        self.v.index_copy_(2, slot, v_new[:, :, S - tail:])
+        self.k.index_copy_(2, slot, k_new[:, :, S - tail:].to(self.k.dtype))
+        self.v.index_copy_(2, slot, v_new[:, :, S - tail:].to(self.v.dtype))
         self.slot_pos.index_copy_(0, slot, idx)
         # prefill attention runs over the CURRENT chunk directly (the
         # cache held nothing before position 0)
