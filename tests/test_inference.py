@@ -328,3 +328,35 @@ def _rolling_cache_worker(rank, world):
 
 def test_rolling_kv_cache_matches_full():
     run_distributed(_rolling_cache_worker, world_size=1)
+
+
+def test_rolling_kv_cache_units():
+    """Unit edges: prefill longer than the window keeps only the
+    attendable tail; chunked prefill (pos>0, S>1) raises; slot_pos
+    tracks global positions through wraps."""
+    import pytest
+    import torch
+
+    from neuronx_distributed_amd.inference.kv_cache import RollingKVCache
+
+    W = 4
+    c = RollingKVCache(1, 1, W, 8, dtype=torch.float32, device="cpu")
+    k = torch.arange(6, dtype=torch.float32).view(1, 1, 6, 1).expand(
+        1, 1, 6, 8).contiguous()
+    ret_k, _ = c.update(k, k.clone(), 0)
+    assert ret_k.shape[2] == 6          # prefill attends its own chunk
+    # tail rows 2..5 stored at slots 2,3,0,1
+    assert c.slot_pos.tolist() == [4, 5, 2, 3]
+    assert c.k[0, 0, 0, 0].item() == 4.0
+    assert c.k[0, 0, 2, 0].item() == 2.0
+
+    with pytest.raises(NotImplementedError):
+        c.update(k[:, :, :2], k[:, :, :2], 3)
+
+    # decode step wraps: position 6 -> slot 2
+    one = torch.full((1, 1, 1, 8), 9.0)
+    K, _ = c.update(one, one.clone(), torch.tensor([6]))
+    assert K.shape[2] == W              # decode returns the ring
+    assert c.slot_pos.tolist() == [4, 5, 6, 3]
+    assert c.k[0, 0, 2, 0].item() == 9.0
+    assert c.position_index() is c.slot_pos
