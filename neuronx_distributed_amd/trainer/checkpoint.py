@@ -157,7 +157,10 @@ def _gc_checkpoints(path: str, num_kept: Optional[int]):
 
 def _to_cpu(obj):
     if isinstance(obj, torch.Tensor):
-        return obj.detach().cpu()
+        t = obj.detach()
+        # .cpu() on a CPU tensor is a NO-COPY alias: clone so the async
+        # writer never races the optimizer mutating the live weights
+        return t.cpu() if t.is_cuda else t.clone()
     if isinstance(obj, dict):
         return {k: _to_cpu(v) for k, v in obj.items()}
     if isinstance(obj, (list, tuple)):

@@ -316,3 +316,35 @@ def test_zero_dcp_ep2():
     with tempfile.TemporaryDirectory() as d:
         res = run_distributed(_dcp_ep2_worker, world_size=2, args=(d,))
     assert abs(res[0] - res[1]) > 1e-6  # distinct expert shards per rank
+
+
+def _async_snapshot_worker(rank, world):
+    """Async save must SNAPSHOT the weights: mutating the model after
+    save_checkpoint() returns (before the background write finishes)
+    must not leak into the checkpoint — .cpu() on CPU tensors is a
+    no-copy alias, so the snapshot needs a clone."""
+    import torch
+
+    from neuronx_distributed_amd.parallel import parallel_state as ps
+    from neuronx_distributed_amd.trainer.checkpoint import (
+        finalize_checkpoints, load_checkpoint, save_checkpoint)
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    import tempfile
+
+    d = tempfile.mkdtemp()
+    m = torch.nn.Linear(4, 4, bias=False)
+    with torch.no_grad():
+        m.weight.fill_(1.0)
+    save_checkpoint(d, tag="t0", model=m, async_save=True)
+    with torch.no_grad():
+        m.weight.fill_(2.0)  # optimizer-step stand-in, racing the writer
+    finalize_checkpoints()
+    m2 = torch.nn.Linear(4, 4, bias=False)
+    load_checkpoint(d, tag="t0", model=m2)
+    assert torch.all(m2.weight == 1.0), m2.weight
+    return True
+
+
+def test_async_save_snapshots_weights():
+    run_distributed(_async_snapshot_worker, world_size=1)
