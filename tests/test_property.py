@@ -107,3 +107,41 @@ def test_hf_state_dict_converters_roundtrip(layers, heads, i_mult):
     assert back.keys() == hf.keys()
     for k in hf:
         assert torch.equal(back[k], hf[k]), k
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.recursive(
+    st.one_of(st.integers(-9, 9), st.text(max_size=3), st.none(),
+              st.tuples(st.integers(1, 3), st.integers(1, 3)).map(
+                  lambda s: torch.randn(*s))),
+    lambda ch: st.one_of(st.lists(ch, max_size=3),
+                         st.tuples(ch, ch),
+                         st.dictionaries(st.text(max_size=2), ch,
+                                         max_size=3)),
+    max_leaves=10))
+def test_pipeline_serialization_roundtrip(obj):
+    """SerializationManager (the PP wire format): arbitrary nested
+    structures split into (skeleton, metas, tensors) and rebuild exactly
+    — tuples stay tuples, tensor identity by index."""
+    from neuronx_distributed_amd.utils.serialization import (
+        SerializationManager)
+
+    m = SerializationManager()
+    skel, metas, tensors = m.serialize(obj)
+    rebuilt = m.deserialize(skel, list(tensors))
+
+    def eq(a, b):
+        if isinstance(a, torch.Tensor):
+            return isinstance(b, torch.Tensor) and torch.equal(a, b)
+        if type(a) is not type(b):
+            return False
+        if isinstance(a, dict):
+            return a.keys() == b.keys() and all(eq(a[k], b[k]) for k in a)
+        if isinstance(a, (list, tuple)):
+            return len(a) == len(b) and all(eq(x, y) for x, y in zip(a, b))
+        return a == b
+
+    assert eq(obj, rebuilt)
+    assert len(metas) == len(tensors)
+    for mt, t in zip(metas, tensors):
+        assert mt.shape == tuple(t.shape) and mt.dtype == t.dtype
