@@ -145,3 +145,30 @@ def test_pipeline_serialization_roundtrip(obj):
     assert len(metas) == len(tensors)
     for mt, t in zip(metas, tensors):
         assert mt.shape == tuple(t.shape) and mt.dtype == t.dtype
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.sampled_from([1, 2, 4, 8]), st.sampled_from([1, 2]),
+       st.integers(1, 4), st.integers(0, 1))
+def test_create_local_weight_shards_reassemble(tp, stride, blocks, dim):
+    """Stride-aware TP sharding (the fused gate-up/QKV layout): slicing
+    every rank's shard and reassembling by stride sub-blocks recovers
+    the full weight exactly, for any tp/stride/width/partition dim."""
+    from neuronx_distributed_amd.parallel.utils import create_local_weight
+
+    full_dim = tp * stride * blocks  # smallest legal multiple times blocks
+    shape = [full_dim, 3] if dim == 0 else [3, full_dim]
+    torch.manual_seed(full_dim + dim)
+    full = torch.randn(*shape)
+    per = full_dim // tp
+    shards = [create_local_weight(full, dim, per, stride, rank=r,
+                                  world_size=tp) for r in range(tp)]
+    assert all(s.shape[dim] == per for s in shards)
+    # reassemble: for each stride sub-block b, ranks contribute their
+    # b-th sub-slice in rank order
+    sub = per // stride
+    parts = []
+    for b in range(stride):
+        for r in range(tp):
+            parts.append(shards[r].narrow(dim, b * sub, sub))
+    assert torch.equal(torch.cat(parts, dim=dim), full)
