@@ -172,3 +172,27 @@ def test_create_local_weight_shards_reassemble(tp, stride, blocks, dim):
         for r in range(tp):
             parts.append(shards[r].narrow(dim, b * sub, sub))
     assert torch.equal(torch.cat(parts, dim=dim), full)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 3), st.integers(1, 16))
+def test_lse_merge_associative(B, S):
+    """The (O, lse) online-softmax merge (ring attention / split-KV
+    decode) must be associative and order-independent — the property the
+    ring schedule relies on when blocks arrive in rotation order."""
+    from neuronx_distributed_amd.kernels.ring_attn import _merge
+
+    torch.manual_seed(B * 100 + S)
+    parts = []
+    for _ in range(3):
+        o = torch.randn(B, 2, S, 8)
+        l = torch.randn(B, 2, S) * 3
+        parts.append((o, l))
+    (o1, l1), (o2, l2), (o3, l3) = parts
+    a_o, a_l = _merge(*_merge(o1, l1, o2, l2), o3, l3)
+    b_o, b_l = _merge(o1, l1, *_merge(o2, l2, o3, l3))
+    c_o, c_l = _merge(o3, l3, *_merge(o2, l2, o1, l1))
+    assert torch.allclose(a_l, b_l, atol=1e-5)
+    assert torch.allclose(a_o, b_o, atol=1e-5)
+    assert torch.allclose(a_l, c_l, atol=1e-5)
+    assert torch.allclose(a_o, c_o, atol=1e-5)
