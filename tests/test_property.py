@@ -196,3 +196,31 @@ def test_lse_merge_associative(B, S):
     assert torch.allclose(a_o, b_o, atol=1e-5)
     assert torch.allclose(a_l, c_l, atol=1e-5)
     assert torch.allclose(a_o, c_o, atol=1e-5)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.integers(1, 6), st.integers(1, 8), st.integers(1, 40))
+def test_ring_window_block_classification_exact(cp, C, W):
+    """_block_kind covers the sliding-window attention EXACTLY: skip
+    blocks contain no in-window (q, k) pair, full blocks are entirely
+    in-window, and the union of non-skip blocks covers every in-window
+    causal pair — for arbitrary cp / block size / window."""
+    from neuronx_distributed_amd.kernels.ring_attn import _block_kind
+
+    S = cp * C
+    for i in range(cp):
+        for blk in range(cp):
+            kind = _block_kind(i, blk, C, W, causal=True)
+            pairs = [(q, k) for q in range(i * C, (i + 1) * C)
+                     for k in range(blk * C, (blk + 1) * C)]
+            in_win = [(q, k) for q, k in pairs if k <= q and k > q - W]
+            if kind == "skip":
+                assert not in_win, (i, blk, C, W)
+            elif kind == "full":
+                causal_pairs = [(q, k) for q, k in pairs if k <= q]
+                assert in_win == causal_pairs and in_win, (i, blk, C, W)
+            elif kind == "diag":
+                assert blk == i
+            else:
+                assert kind == "partial"
+                assert in_win, (i, blk, C, W)  # partial is never empty
